@@ -1,0 +1,169 @@
+"""FCMA compute pipeline: correlation → Fisher-z normalization → Gram.
+
+This is the MI355X re-design of the reference's kernel chain
+(ref src/brainiak/fcma/cython_blas.pyx:20-207 — per-epoch sgemm + per-voxel
+ssyrk — and src/brainiak/fcma/src/fcma_extension.cc:29-86 — OpenMP
+Fisher-z/z-score).  Stages:
+
+  1. ``correlate_chunk``  — corr[c, e, v] = Σ_t A[e, t, s+c]·A2[e, t, v]
+     (a batched GEMM over epochs; epochs are padded to a common length —
+     z-scored columns make the pad zeros inert).
+  2. ``normalize_correlation_`` — in-place Fisher z (with the reference's
+     1±r ≤ 0 → 1e-4 clamp) + per-(voxel, subject, column) z-score over
+     ``epochs_per_subj`` (ddof=0, zero-variance → 0).
+  3. ``gram_matrices`` — per selected voxel the [E, E] linear-SVM Gram
+     matrix over its V-length normalized correlation vectors, plus the
+     reference's magnitude shrink (10**(2-digits) when the leading
+     diagonal entry has >2 integer digits).
+
+On CUDA (= ROCm) tensors stages 1–3 dispatch to the hand-written
+HIP/CDNA4 kernels in ``brainiak_amd.ops`` (MFMA bf16 tiles for the GEMM
+classes, a fused LDS pass for the normalization).  The torch
+implementations below are the CPU path and the numerics oracle the GPU
+kernels are tested against.
+"""
+
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from .. import ops
+
+__all__ = [
+    "CorrelationPipeline",
+    "gram_matrices",
+    "normalize_correlation_",
+    "stack_epochs",
+]
+
+
+def stack_epochs(raw_data: List[np.ndarray], device,
+                 dtype=torch.float32) -> torch.Tensor:
+    """Stack per-epoch [len_e, V] matrices into [E, L, V] (zero-padded).
+
+    The inputs are z-scored and 1/sqrt(len)-scaled, so zero padding does
+    not change any correlation dot product.
+    """
+    E = len(raw_data)
+    L = max(m.shape[0] for m in raw_data)
+    V = raw_data[0].shape[1]
+    out = torch.zeros((E, L, V), dtype=dtype, device=device)
+    for e, m in enumerate(raw_data):
+        out[e, :m.shape[0], :] = torch.as_tensor(
+            np.ascontiguousarray(m), dtype=dtype).to(device)
+    return out
+
+
+def normalize_correlation_(corr: torch.Tensor, epochs_per_subj: int
+                           ) -> torch.Tensor:
+    """In-place Fisher-z + within-subject z-score of corr [C, E, V].
+
+    Matches the reference's native kernel semantics exactly
+    (fcma_extension.cc:54-83): clamp 1±r ≤ 0 to 1e-4, z-score with the
+    biased variance over each subject's ``epochs_per_subj`` epochs, and
+    zero out zero-variance entries.
+    """
+    if corr.is_cuda and ops.has_hip():
+        ops.fcma_normalize_(corr, epochs_per_subj)
+        return corr
+    C, E, V = corr.shape
+    n_subj = E // epochs_per_subj
+    # exact reference semantics: only non-positive arguments are clamped
+    # (a tiny positive 1±r stays, so r≈±1 is intentionally noise-driven —
+    # fcma_extension.cc:68-71 behaves identically)
+    num = 1.0 + corr
+    den = 1.0 - corr
+    num = torch.where(num <= 0, torch.full_like(num, 1e-4), num)
+    den = torch.where(den <= 0, torch.full_like(den, 1e-4), den)
+    z = 0.5 * torch.log(num / den)
+    z = z.view(C, n_subj, epochs_per_subj, V)
+    mean = z.mean(dim=2, keepdim=True)
+    var = (z * z).mean(dim=2, keepdim=True) - mean * mean
+    inv_std = torch.where(var > 0, var.rsqrt(),
+                          torch.zeros_like(var))
+    corr.copy_(((z - mean) * inv_std).view(C, E, V))
+    return corr
+
+
+def _shrink_(gram: torch.Tensor) -> torch.Tensor:
+    """Reference's magnitude shrink (voxelselector.py:407-412): if the
+    first diagonal entry has more than 2 integer digits, scale the whole
+    [E, E] matrix by 10**(2-digits)."""
+    lead = gram[:, 0, 0].abs().clamp_min(1.0)
+    digits = torch.floor(torch.log10(lead)) + 1
+    scale = torch.where(digits > 2, torch.pow(10.0, 2 - digits),
+                        torch.ones_like(lead))
+    gram.mul_(scale[:, None, None])
+    return gram
+
+
+def gram_matrices(corr_norm: torch.Tensor, shrink: bool = True
+                  ) -> torch.Tensor:
+    """Per-voxel [E, E] Gram matrices of corr_norm [C, E, V]."""
+    if corr_norm.is_cuda and ops.has_hip():
+        gram = ops.fcma_gram(corr_norm)
+    else:
+        gram = torch.bmm(corr_norm, corr_norm.transpose(1, 2))
+    if shrink:
+        _shrink_(gram)
+    return gram
+
+
+class CorrelationPipeline:
+    """Holds the stacked epoch tensors and runs the chunk pipeline.
+
+    Parameters
+    ----------
+    raw_data, raw_data2 : list of [len_e, V] z-scored epoch matrices
+        (raw_data2 None → self-correlation).
+    epochs_per_subj : int
+    device : torch device for the compute.
+    use_bf16 : cast the GEMM inputs to bf16 on GPU (the BASELINE dtype);
+        accumulation stays fp32.
+    """
+
+    def __init__(self, raw_data, raw_data2, epochs_per_subj, device=None,
+                 use_bf16: Optional[bool] = None):
+        self.device = torch.device(device) if device is not None else (
+            torch.device("cuda") if torch.cuda.is_available()
+            else torch.device("cpu"))
+        if use_bf16 is None:
+            use_bf16 = self.device.type == "cuda"
+        self.use_bf16 = use_bf16
+        gemm_dtype = torch.bfloat16 if use_bf16 else torch.float32
+        self.epochs_per_subj = epochs_per_subj
+        self.data = stack_epochs(raw_data, self.device, gemm_dtype)
+        self.data2 = (stack_epochs(raw_data2, self.device, gemm_dtype)
+                      if raw_data2 is not None else self.data)
+        self.num_epochs = self.data.shape[0]
+        self.num_voxels = self.data.shape[2]
+        self.num_voxels2 = self.data2.shape[2]
+
+    def correlate_chunk(self, start: int, count: int) -> torch.Tensor:
+        """corr [count, E, V2] fp32 for voxels [start, start+count)."""
+        if self.device.type == "cuda" and ops.has_hip():
+            return ops.fcma_correlate(self.data, self.data2, start, count)
+        a = self.data[:, :, start:start + count].to(torch.float32)
+        b = self.data2.to(torch.float32)
+        # [E, C, V] then→ [C, E, V]
+        out = torch.bmm(a.transpose(1, 2), b)
+        return out.transpose(0, 1).contiguous()
+
+    def chunk_kernel_matrices(self, start: int, count: int,
+                              shrink: bool = True) -> torch.Tensor:
+        """Full fused chunk: correlation → normalize → Gram [count, E, E].
+
+        On gfx950 this is ONE kernel launch (ops.fcma_fused_gram): the
+        [count, E, V] intermediate lives in LDS/registers, never in HBM.
+        """
+        if self.device.type == "cuda" and ops.has_hip() and \
+                ops.has_fused_gram():
+            gram = ops.fcma_fused_gram(
+                self.data, self.data2, start, count, self.epochs_per_subj)
+            if shrink:
+                _shrink_(gram)
+            return gram
+        corr = self.correlate_chunk(start, count)
+        normalize_correlation_(corr, self.epochs_per_subj)
+        return gram_matrices(corr, shrink=shrink)
