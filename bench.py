@@ -1,0 +1,192 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: FCMA voxel selection, whole-brain 34k voxels x 16
+subjects, bf16 compute, on N MI355X GPUs (BASELINE.json config 2; the
+driver's headline metric).
+
+One *step* scores ``--voxels-per-step`` voxels PER GPU against all
+``--num-voxels`` voxels: the full selection pipeline — fused HIP
+correlation + Fisher-z normalization + per-voxel [E,E] SVM Gram
+matrices (MFMA), then batched k-fold SVM cross-validation on device.
+Per-GPU work is fixed as N grows (weak scaling): rank r scores its own
+voxel slice, exactly how `VoxelSelector` shards whole-brain selection.
+
+Metric: voxel-pairs/sec = (voxels_scored_per_step * num_voxels * N)
+/ step_time, aggregated over the whole job.  Data is synthetic
+(random-init, z-scored epochs of the named shape) — there is no network
+for datasets; BASELINE.md documents that the reference publishes no
+in-repo numbers, so `vs_baseline` is null.
+
+Usage (driver contract):
+    python bench.py --gpus N --steps K --warmup W
+    torchrun --nproc-per-node N bench.py --gpus N ...   (N > 1)
+"""
+
+import argparse
+import json
+import math
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--num-voxels", type=int, default=34470)
+    ap.add_argument("--subjects", type=int, default=16)
+    ap.add_argument("--epochs-per-subj", type=int, default=4)
+    ap.add_argument("--epoch-len", type=int, default=12)
+    ap.add_argument("--voxels-per-step", type=int, default=2048)
+    ap.add_argument("--num-folds", type=int, default=4)
+    ap.add_argument("--chunk", type=int, default=512,
+                    help="pipeline chunk (voxels per kernel pass)")
+    ap.add_argument("--no-cv", action="store_true",
+                    help="skip the SVM CV stage (pipeline only)")
+    ap.add_argument("--device", default=None)
+    return ap.parse_args()
+
+
+def make_synthetic_epochs(rng, n_epochs, length, voxels, device):
+    """Z-scored, 1/sqrt(n)-scaled epochs, generated on device."""
+    g = torch.Generator(device="cpu").manual_seed(rng)
+    out = []
+    for _ in range(n_epochs):
+        m = torch.randn((length, voxels), generator=g, dtype=torch.float32)
+        m = m.to(device)
+        m = (m - m.mean(dim=0)) / m.std(dim=0, unbiased=False).clamp_min(
+            1e-12)
+        out.append(m / math.sqrt(length))
+    return out
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    distributed = world > 1
+
+    if torch.cuda.is_available():
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        device = torch.device("cuda", local_rank % torch.cuda.device_count())
+        torch.cuda.set_device(device)
+        backend = "nccl"
+    else:  # CPU smoke path (no GPU in the dev container)
+        device = torch.device("cpu")
+        backend = "gloo"
+    if args.device:
+        device = torch.device(args.device)
+
+    if distributed:
+        import torch.distributed as dist
+        dist.init_process_group(backend=backend)
+
+    from brainiak_amd.fcma.core import CorrelationPipeline
+    from brainiak_amd.fcma.svm import cross_validate_voxels
+
+    E = args.subjects * args.epochs_per_subj
+    V = args.num_voxels
+    # epochs stacked directly on device; bf16 on GPU (BASELINE dtype)
+    raw = make_synthetic_epochs(1234, E, args.epoch_len, V, device)
+    labels = np.array([e % 2 for e in range(E)])
+    pipeline = CorrelationPipeline(
+        raw, None, args.epochs_per_subj, device=device,
+        use_bf16=device.type == "cuda")
+    del raw
+
+    # this rank's voxel range rotates so steps touch different voxels
+    vps = min(args.voxels_per_step, V)
+    chunk = min(args.chunk, vps)
+
+    def one_step(step_idx):
+        start0 = (step_idx * vps) % V
+        grams = []
+        done = 0
+        while done < vps:
+            count = min(chunk, vps - done)
+            start = (start0 + done) % V
+            count = min(count, V - start)
+            grams.append(pipeline.chunk_kernel_matrices(start, count))
+            done += count
+        if not args.no_cv:
+            kernels = torch.cat(grams, dim=0)
+            if device.type == "cuda":
+                accs = cross_validate_voxels(kernels, labels,
+                                             args.num_folds)
+            else:  # CPU smoke: tiny sklearn sample to keep runtime sane
+                accs = cross_validate_voxels(kernels[:8], labels,
+                                             args.num_folds)
+            return float(np.mean(accs))
+        return float(grams[0].float().mean())
+
+    def barrier_sync():
+        if distributed:
+            import torch.distributed as dist
+            dist.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    for w in range(args.warmup):
+        one_step(w)
+    barrier_sync()
+
+    t0 = time.perf_counter()
+    sink = 0.0
+    for k in range(args.steps):
+        sink += one_step(args.warmup + k)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if distributed:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if backend == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t[0])
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    pairs_per_sec = (vps * world * V) / (elapsed / args.steps)
+
+    if rank == 0:
+        result = {
+            "metric": "fcma_voxel_pairs_per_sec",
+            "value": pairs_per_sec,
+            "unit": "voxel-pairs/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if device.type == "cuda" else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "fcma_voxel_selection",
+                "num_voxels": V,
+                "subjects": args.subjects,
+                "epochs_per_subj": args.epochs_per_subj,
+                "epoch_len": args.epoch_len,
+                "voxels_per_step_per_gpu": vps,
+                "num_folds": args.num_folds,
+                "cv": not args.no_cv,
+                "global_batch": vps * world,
+                "seq_len": args.epoch_len,
+                "parallelism": f"voxel-sharded dp{world}",
+            },
+            "_sink": sink,
+        }
+        del result["_sink"]
+        print(json.dumps(result))
+
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -122,6 +122,12 @@ def fcma_fused_gram(data: torch.Tensor, data2: torch.Tensor, start: int,
 # SRM Procrustes
 # ---------------------------------------------------------------------------
 
+def jacobi_eigh(G: torch.Tensor):
+    """Batched symmetric eigensolve of G [B, K, K] (K <= 64) → (evals,
+    evecs), one wavefront per matrix."""
+    return _ext().jacobi_eigh(G)
+
+
 def batched_polar(A: torch.Tensor, perturb: float = 0.001) -> torch.Tensor:
     """Batched orthogonal Procrustes factor of A [B, V, K] → [B, V, K].
 

@@ -1,0 +1,500 @@
+// FCMA HIP/CDNA4 kernels for MI355X (gfx950).
+//
+// MI355X-native re-design of the reference's FCMA native layer:
+//  - ref src/brainiak/fcma/cython_blas.pyx:20-116  (per-epoch sgemm)  -> k_corr_norm
+//  - ref src/brainiak/fcma/src/fcma_extension.cc:29-86 (OpenMP norm)  -> k_normalize / fused
+//  - ref src/brainiak/fcma/cython_blas.pyx:118-207 (per-voxel ssyrk)  -> k_gram_bf16 (MFMA) / k_gram_f32
+//
+// Design (per /opt/skills/guides/cdna_hip_programming.md):
+//  * wave64, 256-thread blocks (4 waves).
+//  * k_corr_norm fuses correlation + Fisher-z + within-subject z-score per
+//    (voxel-tile, subject, column-tile): the fp32 correlation tensor stays
+//    in LDS; only bf16 Z (half bytes) or the requested fp32 view reaches HBM.
+//  * k_gram_bf16: G_c = Z_c Z_c^T via v_mfma_f32_16x16x32_bf16, one block
+//    per voxel x 64x64 band pair, double-buffered LDS K-tiles, padded rows.
+//  * epochs_per_subj is a template parameter for the common values so the
+//    per-column z[] arrays stay in registers (runtime-indexed ext arrays
+//    drop to scratch - guide rule 20); a generic two-pass path covers the
+//    rest.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+typedef __hip_bfloat16 bf16_t;
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef short bf16x8 __attribute__((ext_vector_type(8)));
+typedef long long ll;
+
+// ---------------------------------------------------------------------------
+// MFMA fragment maps, v_mfma_f32_16x16x32_bf16 (gfx950):
+//   A (16x32): lane l, j in 0..7 -> A[l%16][frag_k(l,j)]
+//   B (32x16): lane l, j in 0..7 -> B[frag_k(l,j)][l%16]
+//   C/D (16x16, f32x4): col = lane&15, row = (lane>>4)*4 + reg
+// frag_k isolated so a layout flip is one line, guarded by GPU numerics
+// tests (tests/ops/test_hip_ops.py::test_gram_identity_asymmetric).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int frag_k(int lane, int j) {
+    return 8 * (lane >> 4) + j;
+}
+
+__device__ __forceinline__ float fisher_z(float r) {
+    float num = 1.0f + r;
+    float den = 1.0f - r;
+    num = (num <= 0.0f) ? 1e-4f : num;   // reference clamp semantics
+    den = (den <= 0.0f) ? 1e-4f : den;
+    return 0.5f * __logf(num / den);
+}
+
+// ===========================================================================
+// k_normalize: standalone in-place Fisher-z + within-subject z-score of
+// corr [C, E, V] fp32.  One thread per (c, subject, v) column.
+// ===========================================================================
+template <int TP>
+__global__ __launch_bounds__(256) void k_normalize(
+    float* __restrict__ corr, ll C, ll E, ll V, int Prt) {
+    const int P = TP > 0 ? TP : Prt;
+    const ll nSubj = E / P;
+    const ll vBlocks = (V + 255) / 256;
+    ll b = blockIdx.x;
+    const ll vb = b % vBlocks; b /= vBlocks;
+    const ll s = b % nSubj;    b /= nSubj;
+    const ll c = b;
+    const ll v = vb * 256 + threadIdx.x;
+    if (c >= C || v >= V) return;
+
+    float* col = corr + (c * E + s * (ll)P) * V + v;
+    if (TP > 0) {
+        float z[TP > 0 ? TP : 1];
+        float mean = 0.f, sq = 0.f;
+        #pragma unroll
+        for (int p = 0; p < P; ++p) {
+            z[p] = fisher_z(col[(ll)p * V]);
+            mean += z[p]; sq += z[p] * z[p];
+        }
+        mean /= (float)P;
+        float var = sq / (float)P - mean * mean;
+        float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+        #pragma unroll
+        for (int p = 0; p < P; ++p)
+            col[(ll)p * V] = (z[p] - mean) * inv;
+    } else {
+        float mean = 0.f, sq = 0.f;
+        for (int p = 0; p < P; ++p) {
+            float zv = fisher_z(col[(ll)p * V]);
+            mean += zv; sq += zv * zv;
+        }
+        mean /= (float)P;
+        float var = sq / (float)P - mean * mean;
+        float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+        for (int p = 0; p < P; ++p)
+            col[(ll)p * V] = (fisher_z(col[(ll)p * V]) - mean) * inv;
+    }
+}
+
+// ===========================================================================
+// k_corr_norm: fused correlation + normalization for one
+// (voxel-tile CT=16, subject, column-tile VT=64):
+//   corr[c,p,v] = sum_k A[s*P+p, k, s0+c0+c] * B[s*P+p, k, v0+v]
+//   out        = zscore_p(fisher_z(corr))        (modes 0/1)
+// A: [E, L, VA] bf16, B: [E, L, VB] bf16 (stacked z-scored epochs; zero
+// padding rows are inert).  mode 0: bf16 Z out; 1: fp32 normalized out;
+// 2: fp32 RAW correlation out (no normalization).
+// B-tile elements are reused across the CT voxels from per-thread
+// registers - LLC traffic scales as 1/CT.
+// ===========================================================================
+#define CN_CT 16
+#define CN_VT 64
+#define CN_MAXL 40
+
+template <int TP>
+__global__ __launch_bounds__(256) void k_corr_norm(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ zOut, float* __restrict__ fOut,
+    ll E, ll L, ll VA, ll VB, ll s0, ll C, int Prt, int mode) {
+    const int P = TP > 0 ? TP : Prt;
+    const ll nSubj = E / P;
+    const ll cTiles = (C + CN_CT - 1) / CN_CT;
+    const ll vTiles = (VB + CN_VT - 1) / CN_VT;
+    ll b = blockIdx.x;
+    const ll vt = b % vTiles; b /= vTiles;
+    const ll s = b % nSubj;   b /= nSubj;
+    const ll ct = b;
+    if (ct >= cTiles) return;
+
+    const ll c0 = ct * CN_CT;
+    const ll v0 = vt * CN_VT;
+    const int CT = (int)min((ll)CN_CT, C - c0);
+    const int VT = (int)min((ll)CN_VT, VB - v0);
+    const int tid = threadIdx.x;
+
+    extern __shared__ char smem[];
+    bf16_t* a_tile = (bf16_t*)smem;                    // [P][L][CN_CT]
+    float* corr = (float*)(smem + (size_t)P * L * CN_CT * sizeof(bf16_t));
+    // corr: [CN_CT][P][CN_VT]
+
+    for (int idx = tid; idx < P * (int)L * CN_CT; idx += 256) {
+        int c = idx % CN_CT;
+        int k = (idx / CN_CT) % (int)L;
+        int p = idx / (CN_CT * (int)L);
+        bf16_t val = (bf16_t)0.0f;
+        if (c < CT)
+            val = A[((s * P + p) * L + k) * VA + (s0 + c0 + c)];
+        a_tile[idx] = val;
+    }
+    __syncthreads();
+
+    for (int base = tid; base < P * CN_VT; base += 256) {
+        int v = base % CN_VT;
+        int p = base / CN_VT;
+        if (v < VT) {
+            float breg[CN_MAXL];
+            const bf16_t* brow = B + ((s * P + p) * L) * VB + (v0 + v);
+            for (int k = 0; k < (int)L; ++k)
+                breg[k] = (float)brow[(ll)k * VB];
+            for (int c = 0; c < CT; ++c) {
+                float acc = 0.f;
+                const bf16_t* arow = a_tile + ((size_t)p * L) * CN_CT + c;
+                for (int k = 0; k < (int)L; ++k)
+                    acc = fmaf((float)arow[(size_t)k * CN_CT], breg[k], acc);
+                corr[((size_t)c * P + p) * CN_VT + v] = acc;
+            }
+        }
+    }
+    __syncthreads();
+
+    if (mode == 2) {
+        for (int base = tid; base < CT * P * CN_VT; base += 256) {
+            int v = base % CN_VT;
+            int p = (base / CN_VT) % P;
+            int c = base / (CN_VT * P);
+            if (v < VT)
+                fOut[((c0 + c) * E + (s * P + p)) * VB + v0 + v] =
+                    corr[((size_t)c * P + p) * CN_VT + v];
+        }
+        return;
+    }
+
+    // normalization: one thread per (c, v) column over the P epochs
+    for (int base = tid; base < CT * CN_VT; base += 256) {
+        int v = base % CN_VT;
+        int c = base / CN_VT;
+        if (v >= VT) continue;
+        float* col = corr + ((size_t)c * P) * CN_VT + v;
+        float mean = 0.f, sq = 0.f;
+        if (TP > 0) {
+            float z[TP > 0 ? TP : 1];
+            #pragma unroll
+            for (int p = 0; p < P; ++p) {
+                z[p] = fisher_z(col[(size_t)p * CN_VT]);
+                mean += z[p]; sq += z[p] * z[p];
+            }
+            mean /= (float)P;
+            float var = sq / (float)P - mean * mean;
+            float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                col[(size_t)p * CN_VT] = (z[p] - mean) * inv;
+        } else {
+            for (int p = 0; p < P; ++p) {
+                float zv = fisher_z(col[(size_t)p * CN_VT]);
+                mean += zv; sq += zv * zv;
+                col[(size_t)p * CN_VT] = zv;   // store transformed
+            }
+            mean /= (float)P;
+            float var = sq / (float)P - mean * mean;
+            float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+            for (int p = 0; p < P; ++p)
+                col[(size_t)p * CN_VT] = (col[(size_t)p * CN_VT] - mean)
+                                         * inv;
+        }
+    }
+    __syncthreads();
+
+    for (int base = tid; base < CT * P * CN_VT; base += 256) {
+        int v = base % CN_VT;
+        int p = (base / CN_VT) % P;
+        int c = base / (CN_VT * P);
+        if (v >= VT) continue;
+        float zv = corr[((size_t)c * P + p) * CN_VT + v];
+        ll off = ((c0 + c) * E + (s * P + p)) * VB + v0 + v;
+        if (mode == 0) zOut[off] = (bf16_t)zv;
+        else fOut[off] = zv;
+    }
+}
+
+// ===========================================================================
+// k_gram_bf16: G_c = Z_c Z_c^T per voxel with bf16 MFMA.
+// Z: [C, E, V] bf16; G: [C, E, E] fp32.  E % 64 == 0 (host pads).
+// Block = (c, band_i, band_j>=band_i); 4 waves own the 32x32 quadrants.
+// ===========================================================================
+#define GR_KT 32
+#define GR_PAD 8
+#define GR_ROW (GR_KT + GR_PAD)
+
+__global__ __launch_bounds__(256) void k_gram_bf16(
+    const bf16_t* __restrict__ Z, float* __restrict__ G,
+    ll C, ll E, ll V) {
+    const ll eb = E / 64;
+    ll b = blockIdx.x;
+    const ll band_j = b % eb; b /= eb;
+    const ll band_i = b % eb; b /= eb;
+    const ll c = b;
+    if (c >= C || band_j < band_i) return;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int w = tid >> 6;
+    const int wr = (w >> 1) * 32;
+    const int wc = (w & 1) * 32;
+
+    __shared__ bf16_t zi[2][64][GR_ROW];
+    __shared__ bf16_t zj[2][64][GR_ROW];
+    const bool diag = (band_i == band_j);
+    const bf16_t* Zc = Z + c * E * V;
+    const ll rows_i = band_i * 64;
+    const ll rows_j = band_j * 64;
+
+    f32x4 acc00 = (f32x4)0.f, acc01 = (f32x4)0.f;
+    f32x4 acc10 = (f32x4)0.f, acc11 = (f32x4)0.f;
+
+    auto load_tile = [&](bf16_t dst[64][GR_ROW], const bf16_t* src,
+                         ll rows0, ll k0) {
+        int row = tid >> 2;
+        int col = (tid & 3) * 8;
+        const bf16_t* s = src + (rows0 + row) * V + k0 + col;
+        bf16_t tmp[8];
+        if (k0 + col + 8 <= V) {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) tmp[j] = s[j];
+        } else {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+                tmp[j] = (k0 + col + j < V) ? s[j] : (bf16_t)0.0f;
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) dst[row][col + j] = tmp[j];
+    };
+
+    const ll kTiles = (V + GR_KT - 1) / GR_KT;
+    int cur = 0;
+    load_tile(zi[0], Zc, rows_i, 0);
+    if (!diag) load_tile(zj[0], Zc, rows_j, 0);
+    __syncthreads();
+
+    for (ll kt = 0; kt < kTiles; ++kt) {
+        if (kt + 1 < kTiles) {
+            load_tile(zi[cur ^ 1], Zc, rows_i, (kt + 1) * GR_KT);
+            if (!diag) load_tile(zj[cur ^ 1], Zc, rows_j, (kt + 1) * GR_KT);
+        }
+        const int frow = lane & 15;
+        const int fk = 8 * (lane >> 4);
+        bf16x8 fi0 = *(const bf16x8*)&zi[cur][wr + frow][fk];
+        bf16x8 fi1 = *(const bf16x8*)&zi[cur][wr + 16 + frow][fk];
+        const bf16_t (*zjs)[GR_ROW] = diag ? zi[cur] : zj[cur];
+        bf16x8 fj0 = *(const bf16x8*)&zjs[wc + frow][fk];
+        bf16x8 fj1 = *(const bf16x8*)&zjs[wc + 16 + frow][fk];
+        acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi0, fj0, acc00,
+                                                        0, 0, 0);
+        acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi0, fj1, acc01,
+                                                        0, 0, 0);
+        acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi1, fj0, acc10,
+                                                        0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi1, fj1, acc11,
+                                                        0, 0, 0);
+        __syncthreads();
+        cur ^= 1;
+    }
+
+    float* Gc = G + c * E * E;
+    const int dcol = lane & 15;
+    const int drow = (lane >> 4) * 4;
+    const f32x4* accs[4] = {&acc00, &acc01, &acc10, &acc11};
+    #pragma unroll
+    for (int q = 0; q < 4; ++q) {
+        int mi = q >> 1, ni = q & 1;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            ll gr = rows_i + wr + 16 * mi + drow + r;
+            ll gc = rows_j + wc + 16 * ni + dcol;
+            float val = (*accs[q])[r];
+            Gc[gr * E + gc] = val;
+            if (!diag || gr != gc) Gc[gc * E + gr] = val;
+        }
+    }
+}
+
+// ===========================================================================
+// k_gram_f32: the same contraction from fp32 via exact-f32 MFMA
+// v_mfma_f32_16x16x4_f32 (A: lane l -> A[l&15][l>>4]).
+// ===========================================================================
+#define GF_KT 32
+#define GF_PAD 4
+#define GF_ROW (GF_KT + GF_PAD)
+
+__global__ __launch_bounds__(256) void k_gram_f32(
+    const float* __restrict__ Zf, float* __restrict__ G,
+    ll C, ll E, ll V) {
+    const ll eb = E / 64;
+    ll b = blockIdx.x;
+    const ll band_j = b % eb; b /= eb;
+    const ll band_i = b % eb; b /= eb;
+    const ll c = b;
+    if (c >= C || band_j < band_i) return;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int w = tid >> 6;
+    const int wr = (w >> 1) * 32;
+    const int wc = (w & 1) * 32;
+
+    __shared__ float zi[2][64][GF_ROW];
+    __shared__ float zj[2][64][GF_ROW];
+    const bool diag = (band_i == band_j);
+    const float* Zc = Zf + c * E * V;
+    const ll rows_i = band_i * 64;
+    const ll rows_j = band_j * 64;
+
+    f32x4 acc00 = (f32x4)0.f, acc01 = (f32x4)0.f;
+    f32x4 acc10 = (f32x4)0.f, acc11 = (f32x4)0.f;
+
+    auto load_tile = [&](float dst[64][GF_ROW], const float* src,
+                         ll rows0, ll k0) {
+        int row = tid >> 2;
+        int col = (tid & 3) * 8;
+        const float* s = src + (rows0 + row) * V + k0 + col;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+            dst[row][col + j] = (k0 + col + j < V) ? s[j] : 0.0f;
+    };
+
+    const ll kTiles = (V + GF_KT - 1) / GF_KT;
+    int cur = 0;
+    load_tile(zi[0], Zc, rows_i, 0);
+    if (!diag) load_tile(zj[0], Zc, rows_j, 0);
+    __syncthreads();
+
+    for (ll kt = 0; kt < kTiles; ++kt) {
+        if (kt + 1 < kTiles) {
+            load_tile(zi[cur ^ 1], Zc, rows_i, (kt + 1) * GF_KT);
+            if (!diag) load_tile(zj[cur ^ 1], Zc, rows_j, (kt + 1) * GF_KT);
+        }
+        const int frow = lane & 15;
+        const int fk0 = lane >> 4;
+        #pragma unroll
+        for (int step = 0; step < GF_KT / 4; ++step) {
+            float ai0 = zi[cur][wr + frow][fk0 + 4 * step];
+            float ai1 = zi[cur][wr + 16 + frow][fk0 + 4 * step];
+            const float (*zjs)[GF_ROW] = diag ? zi[cur] : zj[cur];
+            float aj0 = zjs[wc + frow][fk0 + 4 * step];
+            float aj1 = zjs[wc + 16 + frow][fk0 + 4 * step];
+            acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(ai0, aj0, acc00,
+                                                         0, 0, 0);
+            acc01 = __builtin_amdgcn_mfma_f32_16x16x4f32(ai0, aj1, acc01,
+                                                         0, 0, 0);
+            acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(ai1, aj0, acc10,
+                                                         0, 0, 0);
+            acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(ai1, aj1, acc11,
+                                                         0, 0, 0);
+        }
+        __syncthreads();
+        cur ^= 1;
+    }
+
+    float* Gc = G + c * E * E;
+    const int dcol = lane & 15;
+    const int drow = (lane >> 4) * 4;
+    const f32x4* accs[4] = {&acc00, &acc01, &acc10, &acc11};
+    #pragma unroll
+    for (int q = 0; q < 4; ++q) {
+        int mi = q >> 1, ni = q & 1;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            ll gr = rows_i + wr + 16 * mi + drow + r;
+            ll gc = rows_j + wc + 16 * ni + dcol;
+            float val = (*accs[q])[r];
+            Gc[gr * E + gc] = val;
+            if (!diag || gr != gc) Gc[gc * E + gr] = val;
+        }
+    }
+}
+
+// ===========================================================================
+// host-side launchers (P-templated dispatch)
+// ===========================================================================
+
+static inline ll ceil_div(ll a, ll b) { return (a + b - 1) / b; }
+
+template <int TP>
+static void launch_normalize_t(float* corr, ll C, ll E, ll V, int P,
+                               hipStream_t stream) {
+    ll nSubj = E / P;
+    ll grid = C * nSubj * ceil_div(V, 256);
+    hipLaunchKernelGGL((k_normalize<TP>), dim3(grid), dim3(256), 0, stream,
+                       corr, C, E, V, P);
+}
+
+extern "C" void launch_fcma_normalize(float* corr, ll C, ll E, ll V, int P,
+                                      hipStream_t stream) {
+    switch (P) {
+        case 2:  launch_normalize_t<2>(corr, C, E, V, P, stream); break;
+        case 4:  launch_normalize_t<4>(corr, C, E, V, P, stream); break;
+        case 8:  launch_normalize_t<8>(corr, C, E, V, P, stream); break;
+        case 16: launch_normalize_t<16>(corr, C, E, V, P, stream); break;
+        case 32: launch_normalize_t<32>(corr, C, E, V, P, stream); break;
+        default: launch_normalize_t<0>(corr, C, E, V, P, stream); break;
+    }
+}
+
+template <int TP>
+static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
+                               float* fOut, ll E, ll L, ll VA, ll VB,
+                               ll s0, ll C, int P, int mode, size_t smem,
+                               hipStream_t stream) {
+    ll nSubj = E / P;
+    ll grid = ceil_div(C, CN_CT) * nSubj * ceil_div(VB, CN_VT);
+    hipLaunchKernelGGL((k_corr_norm<TP>), dim3(grid), dim3(256), smem,
+                       stream, (const bf16_t*)A, (const bf16_t*)B,
+                       (bf16_t*)zOut, fOut, E, L, VA, VB, s0, C, P, mode);
+}
+
+extern "C" int fcma_corr_norm_smem(ll L, int P) {
+    size_t smem = (size_t)P * L * CN_CT * sizeof(bf16_t)
+                + (size_t)CN_CT * P * CN_VT * sizeof(float);
+    return (int)smem;
+}
+
+extern "C" void launch_fcma_corr_norm(const void* A, const void* B,
+                                      void* zOut, float* fOut, ll E, ll L,
+                                      ll VA, ll VB, ll s0, ll C, int P,
+                                      int mode, hipStream_t stream) {
+    size_t smem = (size_t)fcma_corr_norm_smem(L, P);
+    switch (P) {
+        case 2:  launch_corr_norm_t<2>(A, B, zOut, fOut, E, L, VA, VB, s0,
+                                       C, P, mode, smem, stream); break;
+        case 4:  launch_corr_norm_t<4>(A, B, zOut, fOut, E, L, VA, VB, s0,
+                                       C, P, mode, smem, stream); break;
+        case 8:  launch_corr_norm_t<8>(A, B, zOut, fOut, E, L, VA, VB, s0,
+                                       C, P, mode, smem, stream); break;
+        case 16: launch_corr_norm_t<16>(A, B, zOut, fOut, E, L, VA, VB, s0,
+                                        C, P, mode, smem, stream); break;
+        default: launch_corr_norm_t<0>(A, B, zOut, fOut, E, L, VA, VB, s0,
+                                       C, P, mode, smem, stream); break;
+    }
+}
+
+extern "C" void launch_fcma_gram_bf16(const void* Z, float* G, ll C, ll E,
+                                      ll V, hipStream_t stream) {
+    ll eb = E / 64;
+    ll grid = C * eb * eb;
+    hipLaunchKernelGGL(k_gram_bf16, dim3(grid), dim3(256), 0, stream,
+                       (const bf16_t*)Z, G, C, E, V);
+}
+
+extern "C" void launch_fcma_gram_f32(const float* Z, float* G, ll C, ll E,
+                                     ll V, hipStream_t stream) {
+    ll eb = E / 64;
+    ll grid = C * eb * eb;
+    hipLaunchKernelGGL(k_gram_f32, dim3(grid), dim3(256), 0, stream,
+                       Z, G, C, E, V);
+}

@@ -1,0 +1,173 @@
+"""GPU numerics tests for the hand-written HIP/CDNA4 kernels.
+
+Every kernel is validated against a plain PyTorch fp32 reference of the
+same op (computed on CPU or via eager torch on device).
+"""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ops():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from brainiak_amd import ops as _ops
+    if _ops.load_extension() is None:
+        pytest.fail("HIP extension not built — GPU box must load it")
+    return _ops
+
+
+def _zscored_epochs(g, E, L, V, device):
+    out = torch.empty((E, L, V), dtype=torch.float32)
+    for e in range(E):
+        m = torch.randn((L, V), generator=g)
+        m = (m - m.mean(0)) / m.std(0, unbiased=False).clamp_min(1e-12)
+        out[e] = m / math.sqrt(L)
+    return out.to(device)
+
+
+def _ref_normalize(corr, P):
+    num = 1.0 + corr
+    den = 1.0 - corr
+    num = torch.where(num <= 0, torch.full_like(num, 1e-4), num)
+    den = torch.where(den <= 0, torch.full_like(den, 1e-4), den)
+    z = 0.5 * torch.log(num / den)
+    C, E, V = corr.shape
+    z = z.view(C, E // P, P, V)
+    mean = z.mean(dim=2, keepdim=True)
+    var = (z * z).mean(dim=2, keepdim=True) - mean * mean
+    inv = torch.where(var > 0, var.rsqrt(), torch.zeros_like(var))
+    return ((z - mean) * inv).view(C, E, V)
+
+
+def test_normalize_matches_reference(ops):
+    g = torch.Generator().manual_seed(0)
+    corr = (torch.rand((3, 16, 277), generator=g) * 2 - 1).float()
+    corr[0, 0, 0] = 1.0
+    corr[0, 1, 1] = -1.0
+    expected = _ref_normalize(corr.clone(), 4)
+    dev = corr.cuda()
+    ops.fcma_normalize_(dev, 4)
+    assert torch.allclose(dev.cpu(), expected, atol=1e-4)
+
+
+def test_normalize_large_p_generic_path(ops):
+    g = torch.Generator().manual_seed(1)
+    corr = (torch.rand((2, 2 * 48, 64), generator=g) * 2 - 1).float()
+    expected = _ref_normalize(corr.clone(), 48)
+    dev = corr.cuda()
+    ops.fcma_normalize_(dev, 48)
+    assert torch.allclose(dev.cpu(), expected, atol=1e-4)
+
+
+def test_correlate_matches_torch(ops):
+    g = torch.Generator().manual_seed(2)
+    A = _zscored_epochs(g, 8, 12, 300, "cuda")
+    B = _zscored_epochs(g, 8, 12, 150, "cuda")
+    out = ops.fcma_correlate(A.to(torch.bfloat16).contiguous(),
+                             B.to(torch.bfloat16).contiguous(), 17, 41)
+    ref = torch.einsum('elc,elv->cev',
+                       A[:, :, 17:58].to(torch.bfloat16).float(),
+                       B.to(torch.bfloat16).float())
+    assert out.shape == (41, 8, 150)
+    assert torch.allclose(out.cpu(), ref.cpu(), atol=2e-2, rtol=2e-2)
+
+
+def test_gram_f32_identity_asymmetric(ops):
+    """Transpose-detecting check of the f32 MFMA fragment maps (guide G9):
+    asymmetric Z, compare G = Z Z^T elementwise."""
+    g = torch.Generator().manual_seed(3)
+    Z = torch.randn((2, 64, 96), generator=g).float()
+    Z[0, 0, :] = torch.arange(96).float() / 96.0  # strongly asymmetric
+    G = ops.fcma_gram(Z.cuda())
+    ref = torch.bmm(Z, Z.transpose(1, 2))
+    assert torch.allclose(G.cpu(), ref, atol=1e-3, rtol=1e-4)
+
+
+def test_gram_f32_nonmultiple_shapes(ops):
+    g = torch.Generator().manual_seed(4)
+    Z = torch.randn((3, 24, 100), generator=g).float()  # E=24 → pad to 64
+    G = ops.fcma_gram(Z.cuda())
+    ref = torch.bmm(Z, Z.transpose(1, 2))
+    assert G.shape == (3, 24, 24)
+    assert torch.allclose(G.cpu(), ref, atol=1e-3, rtol=1e-4)
+
+
+def test_gram_bf16_matches_reference(ops):
+    g = torch.Generator().manual_seed(5)
+    Z = torch.randn((2, 128, 77), generator=g).to(torch.bfloat16)
+    G = ops.fcma_gram_bf16(Z.cuda().contiguous())
+    Zf = Z.float()
+    ref = torch.bmm(Zf, Zf.transpose(1, 2))
+    assert G.shape == (2, 128, 128)
+    assert torch.allclose(G.cpu(), ref, atol=0.5, rtol=2e-2)
+
+
+def test_fused_gram_matches_staged_cpu(ops):
+    g = torch.Generator().manual_seed(6)
+    E, L, V, P = 16, 12, 256, 4
+    A = _zscored_epochs(g, E, L, V, "cpu")
+    B = _zscored_epochs(g, E, L, 192, "cpu")
+    Ab = A.to(torch.bfloat16)
+    Bb = B.to(torch.bfloat16)
+    G = ops.fcma_fused_gram(Ab.cuda().contiguous(), Bb.cuda().contiguous(),
+                            5, 32, P)
+    # CPU reference from the same bf16-rounded inputs
+    corr = torch.einsum('elc,elv->cev', Ab.float()[:, :, 5:37], Bb.float())
+    nc = _ref_normalize(corr, P)
+    ref = torch.bmm(nc, nc.transpose(1, 2))
+    assert G.shape == (32, E, E)
+    assert torch.allclose(G.cpu(), ref, atol=1.0, rtol=3e-2)
+
+
+def test_batched_polar_matches_svd(ops):
+    g = torch.Generator().manual_seed(7)
+    A = torch.randn((5, 200, 50), generator=g).float().cuda()
+    W = ops.batched_polar(A, 0.0)
+    for b in range(5):
+        U, _, Vt = torch.linalg.svd(A[b].cpu().double(),
+                                    full_matrices=False)
+        ref = (U @ Vt).float()
+        assert torch.allclose(W[b].cpu(), ref, atol=5e-3)
+    # orthogonality
+    WtW = torch.bmm(W.transpose(1, 2), W).cpu()
+    eye = torch.eye(50).expand(5, -1, -1)
+    assert torch.allclose(WtW, eye, atol=5e-3)
+
+
+def test_jacobi_eigh(ops):
+    g = torch.Generator().manual_seed(8)
+    M = torch.randn((4, 32, 32), generator=g).float()
+    G = torch.bmm(M, M.transpose(1, 2)).cuda().contiguous()
+    evals, evecs = ops.jacobi_eigh(G)
+    for b in range(4):
+        lam_ref = torch.linalg.eigvalsh(G[b].cpu().double())
+        lam = evals[b].cpu().double().sort().values
+        assert torch.allclose(lam, lam_ref, atol=1e-2, rtol=1e-4)
+        # V diag(lam) V^T == G
+        rec = (evecs[b].cpu() * evals[b].cpu()) @ evecs[b].cpu().T
+        assert torch.allclose(rec, G[b].cpu(), atol=1e-2, rtol=1e-3)
+
+
+def test_tfa_factor_and_recon(ops):
+    g = torch.Generator().manual_seed(9)
+    K, V, T = 20, 1000, 50
+    centers = torch.randn((K, 3), generator=g).float().cuda()
+    widths = (torch.rand((K,), generator=g) * 4 + 1).float().cuda()
+    coords = torch.randn((V, 3), generator=g).float().cuda()
+    F = ops.tfa_factor(centers, widths, coords)
+    d = coords[:, None, :] - centers[None, :, :]
+    ref_F = torch.exp(-(d * d).sum(-1) / widths[None, :])
+    assert torch.allclose(F, ref_F, atol=1e-4, rtol=1e-4)
+
+    X = torch.randn((V, T), generator=g).float().cuda()
+    W = torch.randn((K, T), generator=g).float().cuda()
+    R = ops.tfa_recon(X, W, F, 0.5)
+    ref_R = (0.5 * (X - F @ W)).reshape(-1)
+    assert torch.allclose(R, ref_R, atol=1e-3, rtol=1e-3)
