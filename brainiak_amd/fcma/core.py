@@ -50,8 +50,12 @@ def stack_epochs(raw_data: List[np.ndarray], device,
     V = raw_data[0].shape[1]
     out = torch.zeros((E, L, V), dtype=dtype, device=device)
     for e, m in enumerate(raw_data):
-        out[e, :m.shape[0], :] = torch.as_tensor(
-            np.ascontiguousarray(m), dtype=dtype).to(device)
+        if isinstance(m, torch.Tensor):
+            t = m.to(device=device, dtype=dtype)
+        else:
+            t = torch.as_tensor(np.ascontiguousarray(m),
+                                dtype=torch.float32).to(device).to(dtype)
+        out[e, :m.shape[0], :] = t
     return out
 
 

@@ -110,6 +110,11 @@ def fcma_gram(corr_norm: torch.Tensor) -> torch.Tensor:
     return _ext().fcma_gram(corr_norm)
 
 
+def fcma_gram_bf16(Z: torch.Tensor) -> torch.Tensor:
+    """Per-voxel Gram [C, E, E] fp32 of Z [C, E, V] bf16 (E % 64 == 0)."""
+    return _ext().fcma_gram_bf16(Z)
+
+
 def fcma_fused_gram(data: torch.Tensor, data2: torch.Tensor, start: int,
                     count: int, epochs_per_subj: int) -> torch.Tensor:
     """Fused correlate→normalize→Gram for one voxel chunk; the [C, E, V]
