@@ -47,6 +47,13 @@ def stack_epochs(raw_data: List[np.ndarray], device,
     """
     E = len(raw_data)
     L = max(m.shape[0] for m in raw_data)
+    if device is not None and torch.device(device).type == "cuda":
+        # pad to the HIP kernel's supported epoch lengths; zero rows are
+        # inert for z-scored data
+        for opt in (8, 16, 24, 32, 40):
+            if L <= opt:
+                L = opt
+                break
     V = raw_data[0].shape[1]
     out = torch.zeros((E, L, V), dtype=dtype, device=device)
     for e, m in enumerate(raw_data):

@@ -106,6 +106,39 @@ def smo_batch_train(K: torch.Tensor, y: torch.Tensor, C: float = 1.0,
     return alpha, b
 
 
+def _accuracy_gpu_hip(kernels: torch.Tensor, labels: np.ndarray,
+                      num_folds: int, C: float, tol: float) -> np.ndarray:
+    """Whole-CV-in-one-launch path: one wavefront per (voxel, fold) QP
+    (ops.svm_cv HIP kernel)."""
+    from .. import ops
+    device = kernels.device
+    n_vox, E, _ = kernels.shape
+    classes = np.unique(labels)
+    y_np = np.where(labels == classes[1], 1.0, -1.0).astype(np.float32)
+    folds = stratified_folds(labels, num_folds)
+    F = len(folds)
+    train_idx = np.zeros((F, E), dtype=np.int32)
+    test_idx = np.zeros((F, E), dtype=np.int32)
+    n_train = np.zeros(F, dtype=np.int32)
+    n_test = np.zeros(F, dtype=np.int32)
+    for f, (tr, te) in enumerate(folds):
+        train_idx[f, :len(tr)] = tr
+        test_idx[f, :len(te)] = te
+        n_train[f] = len(tr)
+        n_test[f] = len(te)
+    correct = ops.svm_cv(
+        kernels.to(torch.float32).contiguous(),
+        torch.as_tensor(y_np, device=device),
+        torch.as_tensor(train_idx, device=device),
+        torch.as_tensor(test_idx, device=device),
+        torch.as_tensor(n_train, device=device),
+        torch.as_tensor(n_test, device=device),
+        C=C, tol=tol)
+    acc_per_fold = correct.to(torch.float32) / \
+        torch.as_tensor(n_test, device=device, dtype=torch.float32)
+    return acc_per_fold.mean(dim=1).cpu().numpy()
+
+
 def _accuracy_gpu(kernels: torch.Tensor, labels: np.ndarray, num_folds: int,
                   C: float, tol: float) -> np.ndarray:
     """Batched k-fold CV accuracy for kernels [Cvox, E, E]."""
@@ -116,6 +149,12 @@ def _accuracy_gpu(kernels: torch.Tensor, labels: np.ndarray, num_folds: int,
     if len(classes) != 2:
         raise ValueError("GPU batched SVM supports binary labels; got "
                          f"{len(classes)} classes")
+    from .. import ops
+    if ops.has_hip():
+        folds = stratified_folds(labels, num_folds)
+        if max(len(tr) for tr, _ in folds) <= 64 and \
+                max(len(te) for _, te in folds) <= 64:
+            return _accuracy_gpu_hip(kernels, labels, num_folds, C, tol)
     y_np = np.where(labels == classes[1], 1.0, -1.0)
     folds = stratified_folds(labels, num_folds)
 

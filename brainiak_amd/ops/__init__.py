@@ -127,6 +127,18 @@ def fcma_fused_gram(data: torch.Tensor, data2: torch.Tensor, start: int,
 # SRM Procrustes
 # ---------------------------------------------------------------------------
 
+def svm_cv(kernels: torch.Tensor, y: torch.Tensor, train_idx, test_idx,
+           n_train, n_test, C: float = 1.0, tol: float = 1e-3,
+           max_iter: int = 10000) -> torch.Tensor:
+    """Batched precomputed-kernel SVC k-fold CV fully on device.
+
+    One wavefront per (voxel, fold) dual QP (SMO, WSS1); returns
+    correct-prediction counts [C, F] int32.  n_train/n_test <= 64.
+    """
+    return _ext().svm_cv(kernels, y, train_idx, test_idx, n_train, n_test,
+                         float(C), float(tol), int(max_iter))
+
+
 def jacobi_eigh(G: torch.Tensor):
     """Batched symmetric eigensolve of G [B, K, K] (K <= 64) → (evals,
     evecs), one wavefront per matrix."""

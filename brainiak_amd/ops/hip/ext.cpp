@@ -22,6 +22,10 @@ void launch_tfa_factor(const float*, const float*, const float*, float*,
                        ll, int, void*);
 void launch_tfa_recon(const float*, const float*, const float*, float*,
                       ll, ll, int, float, void*);
+void launch_svm_cv(const float*, const float*, const int*, const int*,
+                   const int*, const int*, int*, ll, int, int, float,
+                   float, int, void*);
+ll fcma_supported_L(ll);
 }
 
 static void* cur_stream() {
@@ -59,7 +63,8 @@ torch::Tensor fcma_correlate(torch::Tensor A, torch::Tensor B,
     ll E = A.size(0), L = A.size(1), VA = A.size(2), VB = B.size(2);
     TORCH_CHECK(B.size(0) == E && B.size(1) == L, "A/B epoch shapes differ");
     TORCH_CHECK(start >= 0 && start + count <= VA, "voxel range");
-    TORCH_CHECK(L <= 40, "epoch length > 40 unsupported by fused kernel");
+    TORCH_CHECK(fcma_supported_L(L) == L,
+                "epoch length must be padded to one of {8,16,24,32,40}");
     int P = pick_p_for_raw(E);
     auto out = torch::empty({count, E, VB},
                             A.options().dtype(torch::kFloat32));
@@ -79,7 +84,8 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
     TORCH_CHECK(start >= 0 && start + count <= VA, "voxel range");
     TORCH_CHECK(P > 0 && E % P == 0, "epochs_per_subj must divide E");
     TORCH_CHECK(P <= 32, "epochs_per_subj > 32: use the staged path");
-    TORCH_CHECK(L <= 40, "epoch length > 40 unsupported by fused kernel");
+    TORCH_CHECK(fcma_supported_L(L) == L,
+                "epoch length must be padded to one of {8,16,24,32,40}");
     ll Eout = std::max((ll)padE, E);
     auto Z = (Eout == E)
         ? torch::empty({count, E, VB}, A.options())
@@ -219,7 +225,36 @@ torch::Tensor tfa_recon(torch::Tensor X, torch::Tensor W, torch::Tensor F,
     return R;
 }
 
+torch::Tensor svm_cv(torch::Tensor kernels, torch::Tensor y,
+                     torch::Tensor train_idx, torch::Tensor test_idx,
+                     torch::Tensor n_train, torch::Tensor n_test,
+                     double Creg, double tol, int64_t max_iter) {
+    check_3d(kernels, torch::kFloat32, "kernels");
+    ll C = kernels.size(0);
+    int E = (int)kernels.size(1);
+    TORCH_CHECK(kernels.size(2) == E, "kernels must be [C,E,E]");
+    int F = (int)train_idx.size(0);
+    TORCH_CHECK(y.is_cuda() && y.scalar_type() == torch::kFloat32 &&
+                y.numel() == E, "y must be fp32 [E] on GPU");
+    TORCH_CHECK(train_idx.scalar_type() == torch::kInt32 &&
+                test_idx.scalar_type() == torch::kInt32, "idx int32");
+    TORCH_CHECK(train_idx.size(1) == E && test_idx.size(1) == E,
+                "idx must be [F,E]");
+    auto nt = n_train.to(torch::kInt32).contiguous();
+    auto ns = n_test.to(torch::kInt32).contiguous();
+    auto correct = torch::zeros({C, F}, kernels.options()
+                                            .dtype(torch::kInt32));
+    launch_svm_cv(kernels.data_ptr<float>(), y.data_ptr<float>(),
+                  train_idx.data_ptr<int>(), test_idx.data_ptr<int>(),
+                  nt.data_ptr<int>(), ns.data_ptr<int>(),
+                  correct.data_ptr<int>(), C, E, F, (float)Creg,
+                  (float)tol, (int)max_iter, cur_stream());
+    return correct;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("svm_cv", &svm_cv,
+          "batched precomputed-kernel SVC cross-validation");
     m.def("fcma_normalize_", &fcma_normalize_,
           "in-place Fisher-z + within-subject z-score [C,E,V]");
     m.def("fcma_correlate", &fcma_correlate,

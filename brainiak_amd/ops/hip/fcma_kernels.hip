@@ -107,12 +107,17 @@ __global__ __launch_bounds__(256) void k_normalize(
 #define CN_VT 64
 #define CN_MAXL 40
 
-template <int TP>
+// TL = compile-time epoch length (host pads epochs to {8,16,24,32,40}
+// with zero rows, which are inert for z-scored data) so the per-thread
+// B column stays in registers with fully unrolled FMA chains.
+template <int TP, int TL>
 __global__ __launch_bounds__(256) void k_corr_norm(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ zOut, float* __restrict__ fOut,
-    ll E, ll L, ll VA, ll VB, ll s0, ll C, int Prt, int mode) {
+    ll E, ll Lrt, ll VA, ll VB, ll s0, ll C, int Prt, int mode) {
     const int P = TP > 0 ? TP : Prt;
+    const ll L = TL;
+    (void)Lrt;
     const ll nSubj = E / P;
     const ll cTiles = (C + CN_CT - 1) / CN_CT;
     const ll vTiles = (VB + CN_VT - 1) / CN_VT;
@@ -148,14 +153,16 @@ __global__ __launch_bounds__(256) void k_corr_norm(
         int v = base % CN_VT;
         int p = base / CN_VT;
         if (v < VT) {
-            float breg[CN_MAXL];
+            float breg[TL];
             const bf16_t* brow = B + ((s * P + p) * L) * VB + (v0 + v);
-            for (int k = 0; k < (int)L; ++k)
+            #pragma unroll
+            for (int k = 0; k < TL; ++k)
                 breg[k] = (float)brow[(ll)k * VB];
             for (int c = 0; c < CT; ++c) {
                 float acc = 0.f;
                 const bf16_t* arow = a_tile + ((size_t)p * L) * CN_CT + c;
-                for (int k = 0; k < (int)L; ++k)
+                #pragma unroll
+                for (int k = 0; k < TL; ++k)
                     acc = fmaf((float)arow[(size_t)k * CN_CT], breg[k], acc);
                 corr[((size_t)c * P + p) * CN_VT + v] = acc;
             }
@@ -446,14 +453,14 @@ extern "C" void launch_fcma_normalize(float* corr, ll C, ll E, ll V, int P,
     }
 }
 
-template <int TP>
+template <int TP, int TL>
 static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
                                float* fOut, ll E, ll L, ll VA, ll VB,
                                ll s0, ll C, int P, int mode, size_t smem,
                                hipStream_t stream) {
     ll nSubj = E / P;
     ll grid = ceil_div(C, CN_CT) * nSubj * ceil_div(VB, CN_VT);
-    hipLaunchKernelGGL((k_corr_norm<TP>), dim3(grid), dim3(256), smem,
+    hipLaunchKernelGGL((k_corr_norm<TP, TL>), dim3(grid), dim3(256), smem,
                        stream, (const bf16_t*)A, (const bf16_t*)B,
                        (bf16_t*)zOut, fOut, E, L, VA, VB, s0, C, P, mode);
 }
@@ -464,22 +471,54 @@ extern "C" int fcma_corr_norm_smem(ll L, int P) {
     return (int)smem;
 }
 
+template <int TL>
+static void dispatch_p(const void* A, const void* B, void* zOut,
+                       float* fOut, ll E, ll L, ll VA, ll VB, ll s0, ll C,
+                       int P, int mode, size_t smem, hipStream_t stream) {
+    switch (P) {
+        case 2:  launch_corr_norm_t<2, TL>(A, B, zOut, fOut, E, L, VA, VB,
+                                           s0, C, P, mode, smem, stream);
+                 break;
+        case 4:  launch_corr_norm_t<4, TL>(A, B, zOut, fOut, E, L, VA, VB,
+                                           s0, C, P, mode, smem, stream);
+                 break;
+        case 8:  launch_corr_norm_t<8, TL>(A, B, zOut, fOut, E, L, VA, VB,
+                                           s0, C, P, mode, smem, stream);
+                 break;
+        case 16: launch_corr_norm_t<16, TL>(A, B, zOut, fOut, E, L, VA, VB,
+                                            s0, C, P, mode, smem, stream);
+                 break;
+        default: launch_corr_norm_t<0, TL>(A, B, zOut, fOut, E, L, VA, VB,
+                                           s0, C, P, mode, smem, stream);
+                 break;
+    }
+}
+
+// host pads L to one of these (zero rows are inert for z-scored epochs)
+extern "C" ll fcma_supported_L(ll L) {
+    const ll opts[5] = {8, 16, 24, 32, 40};
+    for (int i = 0; i < 5; ++i)
+        if (L <= opts[i]) return opts[i];
+    return -1;
+}
+
 extern "C" void launch_fcma_corr_norm(const void* A, const void* B,
                                       void* zOut, float* fOut, ll E, ll L,
                                       ll VA, ll VB, ll s0, ll C, int P,
                                       int mode, hipStream_t stream) {
     size_t smem = (size_t)fcma_corr_norm_smem(L, P);
-    switch (P) {
-        case 2:  launch_corr_norm_t<2>(A, B, zOut, fOut, E, L, VA, VB, s0,
-                                       C, P, mode, smem, stream); break;
-        case 4:  launch_corr_norm_t<4>(A, B, zOut, fOut, E, L, VA, VB, s0,
-                                       C, P, mode, smem, stream); break;
-        case 8:  launch_corr_norm_t<8>(A, B, zOut, fOut, E, L, VA, VB, s0,
-                                       C, P, mode, smem, stream); break;
-        case 16: launch_corr_norm_t<16>(A, B, zOut, fOut, E, L, VA, VB, s0,
-                                        C, P, mode, smem, stream); break;
-        default: launch_corr_norm_t<0>(A, B, zOut, fOut, E, L, VA, VB, s0,
-                                       C, P, mode, smem, stream); break;
+    switch (L) {
+        case 8:  dispatch_p<8>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
+                               mode, smem, stream); break;
+        case 16: dispatch_p<16>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
+                                mode, smem, stream); break;
+        case 24: dispatch_p<24>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
+                                mode, smem, stream); break;
+        case 32: dispatch_p<32>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
+                                mode, smem, stream); break;
+        case 40: dispatch_p<40>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
+                                mode, smem, stream); break;
+        default: break;  // host guarantees L in the supported set
     }
 }
 

@@ -126,6 +126,33 @@ def test_fused_gram_matches_staged_cpu(ops):
     assert torch.allclose(G.cpu(), ref, atol=1.0, rtol=3e-2)
 
 
+def test_svm_cv_kernel_matches_sklearn(ops):
+    from sklearn import model_selection, svm as sksvm
+
+    from brainiak_amd.fcma.svm import _accuracy_gpu_hip
+    g = torch.Generator().manual_seed(11)
+    n, E = 12, 32
+    y = np.array([0, 1] * (E // 2))
+    kernels = []
+    for i in range(n):
+        sep = 2.0 * i / n
+        X = torch.randn((E, 10), generator=g).numpy() + sep * y[:, None]
+        kernels.append((X @ X.T).astype(np.float32))
+    kt = torch.tensor(np.stack(kernels)).cuda()
+    accs = _accuracy_gpu_hip(kt, y, num_folds=4, C=1.0, tol=1e-3)
+    skf = model_selection.StratifiedKFold(n_splits=4, shuffle=False)
+    for i in range(n):
+        ref = model_selection.cross_val_score(
+            sksvm.SVC(kernel='precomputed', C=1.0), kernels[i], y=y,
+            cv=skf).mean()
+        assert abs(accs[i] - ref) <= 0.15, (i, accs[i], ref)
+    # average agreement should be tight
+    refs = [model_selection.cross_val_score(
+        sksvm.SVC(kernel='precomputed', C=1.0), kernels[i], y=y,
+        cv=skf).mean() for i in range(n)]
+    assert abs(np.mean(accs) - np.mean(refs)) < 0.05
+
+
 def test_batched_polar_matches_svd(ops):
     g = torch.Generator().manual_seed(7)
     A = torch.randn((5, 200, 50), generator=g).float().cuda()
