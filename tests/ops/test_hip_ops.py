@@ -24,11 +24,13 @@ def ops():
 
 
 def _zscored_epochs(g, E, L, V, device):
-    out = torch.empty((E, L, V), dtype=torch.float32)
+    # pad L to the kernel-supported set (zero rows are inert)
+    Lpad = next(opt for opt in (8, 16, 24, 32, 40) if L <= opt)
+    out = torch.zeros((E, Lpad, V), dtype=torch.float32)
     for e in range(E):
         m = torch.randn((L, V), generator=g)
         m = (m - m.mean(0)) / m.std(0, unbiased=False).clamp_min(1e-12)
-        out[e] = m / math.sqrt(L)
+        out[e, :L] = m / math.sqrt(L)
     return out.to(device)
 
 
