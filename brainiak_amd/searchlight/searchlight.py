@@ -1,0 +1,250 @@
+"""Distributed Searchlight (API parity: ref src/brainiak/searchlight/
+searchlight.py:34-578).
+
+Same contract as the reference: halo-padded spatial blocks, cyclic block
+layout across ranks, per-voxel user function applied inside each block,
+borders trimmed on stitch.  The communication layer is the RCCL/gloo
+DistContext instead of mpi4py: ownership collection is one
+``all_gather_object``, block/mask/subject distribution uses rounds of
+``scatter_object`` (keeping the reference's cyclic layout contract,
+searchlight.py:284-325), and results return via ``gather_object``.
+
+The per-voxel function is user code (numpy views in, anything out), so
+the GPU plays through the *user's* voxel_fn — `bcast_var` commonly holds
+device state.  Block-level CPU parallelism via multiprocessing.Pool is
+kept (``pool_size``).
+"""
+
+from multiprocessing import Pool
+
+import numpy as np
+from scipy.spatial.distance import cityblock, euclidean
+
+from ..parallel import DistContext
+from ..utils.utils import usable_cpu_count
+
+__all__ = ["Ball", "Cube", "Diamond", "Searchlight", "Shape"]
+
+
+class Shape:
+    """Searchlight shape inside a (2*rad+1)^3 cube; mask_ marks it."""
+
+    def __init__(self, rad):
+        self.rad = rad
+
+
+class Cube(Shape):
+    """Full (2*rad+1)^3 cube."""
+
+    def __init__(self, rad):
+        super().__init__(rad)
+        self.mask_ = np.ones((2 * rad + 1,) * 3, dtype=bool)
+
+
+class Diamond(Shape):
+    """Manhattan ball of radius rad."""
+
+    def __init__(self, rad):
+        super().__init__(rad)
+        self.mask_ = np.zeros((2 * rad + 1,) * 3, dtype=bool)
+        for r1 in range(2 * rad + 1):
+            for r2 in range(2 * rad + 1):
+                for r3 in range(2 * rad + 1):
+                    if cityblock((r1, r2, r3), (rad, rad, rad)) <= rad:
+                        self.mask_[r1, r2, r3] = True
+
+
+class Ball(Shape):
+    """Euclidean ball of radius rad."""
+
+    def __init__(self, rad):
+        super().__init__(rad)
+        self.mask_ = np.zeros((2 * rad + 1,) * 3, dtype=bool)
+        for r1 in range(2 * rad + 1):
+            for r2 in range(2 * rad + 1):
+                for r3 in range(2 * rad + 1):
+                    if euclidean((r1, r2, r3), (rad, rad, rad)) <= rad:
+                        self.mask_[r1, r2, r3] = True
+
+
+class Searchlight:
+    """Distributed searchlight over halo-padded blocks.
+
+    Parameters match the reference: ``sl_rad`` (halo radius),
+    ``max_blk_edge`` (inner block edge), ``shape`` (a Shape subclass),
+    ``min_active_voxels_proportion``; plus ``comm`` (a DistContext) and
+    ``pool_size``.
+    """
+
+    def __init__(self, sl_rad=1, max_blk_edge=10, shape=Cube,
+                 min_active_voxels_proportion=0, comm=None, pool_size=None):
+        self.sl_rad = sl_rad
+        self.max_blk_edge = max_blk_edge
+        self.min_active_voxels_proportion = min_active_voxels_proportion
+        self.comm = comm if isinstance(comm, DistContext) else DistContext()
+        self.shape = shape(sl_rad).mask_
+        self.bcast_var = None
+        self.pool_size = pool_size
+
+    # -- decomposition -----------------------------------------------------
+
+    def _get_ownership(self, data):
+        """Rank owning each subject (entries are non-None on one rank)."""
+        mine = [(self.comm.rank, idx) for idx, c in enumerate(data)
+                if c is not None]
+        everyone = self.comm.all_gather_object(mine)
+        ownership = [None] * len(data)
+        for part in everyone:
+            for rank, idx in part:
+                ownership[idx] = rank
+        return ownership
+
+    def _get_blocks(self, mask):
+        """Halo-padded block decomposition; blocks with no active voxels
+        in their interior are dropped."""
+        blocks = []
+        outer = self.max_blk_edge + 2 * self.sl_rad
+        for i in range(0, mask.shape[0], self.max_blk_edge):
+            for j in range(0, mask.shape[1], self.max_blk_edge):
+                for k in range(0, mask.shape[2], self.max_blk_edge):
+                    shape = mask[i:i + outer, j:j + outer,
+                                 k:k + outer].shape
+                    if np.any(mask[i + self.sl_rad:i + shape[0]
+                                   - self.sl_rad,
+                                   j + self.sl_rad:j + shape[1]
+                                   - self.sl_rad,
+                                   k + self.sl_rad:k + shape[2]
+                                   - self.sl_rad]):
+                        blocks.append(((i, j, k), shape))
+        return blocks
+
+    @staticmethod
+    def _get_block_data(mat, block):
+        (pt, sz) = block
+        sl = np.s_[pt[0]:pt[0] + sz[0], pt[1]:pt[1] + sz[1],
+                   pt[2]:pt[2] + sz[2]]
+        return mat[sl].copy()
+
+    def _split_volume(self, mat, blocks):
+        return [self._get_block_data(mat, b) for b in blocks]
+
+    def _scatter_list(self, data, owner):
+        """Cyclic scatter of a list from ``owner`` (rounds of world_size)."""
+        size = self.comm.world_size
+        out = []
+        nblocks = self.comm.broadcast_object(
+            len(data) if self.comm.rank == owner else None, src=owner)
+        for idx in range(0, nblocks, size):
+            padded = None
+            if self.comm.rank == owner:
+                padded = list(data[idx:idx + size])
+                padded += [None] * (size - len(padded))
+            mine = self.comm.scatter_object(padded, src=owner)
+            if mine is not None:
+                out.append(mine)
+        return out
+
+    # -- public API --------------------------------------------------------
+
+    def distribute(self, subjects, mask):
+        """Decompose the volume and distribute blocks/masks/subject data."""
+        if mask.ndim != 3:
+            raise ValueError('mask should be a 3D array')
+        for idx, subj in enumerate(subjects):
+            if subj is not None and subj.ndim != 4:
+                raise ValueError('subjects[{}] must be 4D'.format(idx))
+
+        self.mask = mask
+        ownership = self._get_ownership(subjects)
+        all_blocks = self._get_blocks(mask) if self.comm.is_root else None
+        all_blocks = self.comm.broadcast_object(all_blocks)
+
+        splitsubj = [self._split_volume(s, all_blocks)
+                     if s is not None else None for s in subjects]
+        submasks = self._split_volume(mask, all_blocks)
+
+        self.blocks = self._scatter_list(all_blocks, 0)
+        self.submasks = self._scatter_list(submasks, 0)
+        self.subproblems = [self._scatter_list(s, ownership[i])
+                            for i, s in enumerate(splitsubj)]
+
+    def broadcast(self, bcast_var):
+        """Share ``bcast_var`` with every rank (available as the
+        voxel_fn's bcast_var argument)."""
+        self.bcast_var = self.comm.broadcast_object(bcast_var)
+
+    def run_block_function(self, block_fn, extra_block_fn_params=None,
+                           pool_size=None):
+        """Apply ``block_fn`` to every local block, gather and stitch."""
+        if pool_size is None and self.pool_size is not None:
+            pool_size = self.pool_size
+        processes = usable_cpu_count() if pool_size is None else \
+            min(pool_size, usable_cpu_count())
+
+        results = []
+        if processes > 1:
+            with Pool(processes) as pool:
+                handles = []
+                for idx, block in enumerate(self.blocks):
+                    handles.append((block[0], pool.apply_async(
+                        block_fn,
+                        ([sub[idx] for sub in self.subproblems],
+                         self.submasks[idx], self.sl_rad, self.bcast_var,
+                         extra_block_fn_params))))
+                results = [(pt, h.get()) for pt, h in handles]
+        else:
+            for idx, block in enumerate(self.blocks):
+                out = block_fn([sub[idx] for sub in self.subproblems],
+                               self.submasks[idx], self.sl_rad,
+                               self.bcast_var, extra_block_fn_params)
+                results.append((block[0], out))
+
+        global_outputs = self.comm.gather_object(results)
+
+        outmat = np.empty(self.mask.shape, dtype=object)
+        if self.comm.is_root:
+            for rank_out in global_outputs:
+                for pt, mat in rank_out:
+                    coords = np.s_[
+                        pt[0] + self.sl_rad:pt[0] + self.sl_rad
+                        + mat.shape[0],
+                        pt[1] + self.sl_rad:pt[1] + self.sl_rad
+                        + mat.shape[1],
+                        pt[2] + self.sl_rad:pt[2] + self.sl_rad
+                        + mat.shape[2]]
+                    outmat[coords] = mat
+        return outmat
+
+    def run_searchlight(self, voxel_fn, pool_size=None):
+        """Apply ``voxel_fn`` at every active voxel; returns an object
+        volume (None where inactive / in the trimmed border)."""
+        if pool_size is None and self.pool_size is not None:
+            pool_size = self.pool_size
+        extra = (voxel_fn, self.shape, self.min_active_voxels_proportion)
+        return self.run_block_function(_singlenode_searchlight, extra,
+                                       pool_size)
+
+
+def _singlenode_searchlight(data, msk, mysl_rad, bcast_var, extra_params):
+    """Serial per-block kernel: apply voxel_fn to each active center."""
+    voxel_fn, shape_mask, min_active_voxels_proportion = extra_params
+    outmat = np.empty(msk.shape, dtype=object)
+    if mysl_rad > 0:
+        outmat = outmat[mysl_rad:-mysl_rad, mysl_rad:-mysl_rad,
+                        mysl_rad:-mysl_rad]
+    for i in range(outmat.shape[0]):
+        for j in range(outmat.shape[1]):
+            for k in range(outmat.shape[2]):
+                if msk[i + mysl_rad, j + mysl_rad, k + mysl_rad]:
+                    sl = np.s_[i:i + 2 * mysl_rad + 1,
+                               j:j + 2 * mysl_rad + 1,
+                               k:k + 2 * mysl_rad + 1]
+                    voxel_fn_mask = msk[sl] * shape_mask
+                    if (min_active_voxels_proportion == 0
+                            or np.count_nonzero(voxel_fn_mask)
+                            / voxel_fn_mask.size
+                            > min_active_voxels_proportion):
+                        outmat[i, j, k] = voxel_fn(
+                            [subject[sl] for subject in data],
+                            voxel_fn_mask, mysl_rad, bcast_var)
+    return outmat

@@ -1,0 +1,127 @@
+import numpy as np
+import pytest
+
+from brainiak_amd.parallel import spawn_ranks
+from brainiak_amd.searchlight import Ball, Cube, Diamond, Searchlight
+
+
+def test_shapes():
+    assert Cube(1).mask_.sum() == 27
+    assert Diamond(1).mask_.sum() == 7   # center + 6 neighbours
+    assert Ball(1).mask_.sum() == 7
+    assert Ball(2).mask_.shape == (5, 5, 5)
+    d2 = Diamond(2).mask_
+    assert d2[2, 2, 2] and d2[0, 2, 2] and not d2[0, 0, 0]
+
+
+def _sum_fn(subjects, mask, rad, bcast):
+    # sum of masked voxel values across the searchlight + bcast offset
+    return float(sum(s[mask].sum() for s in subjects)) + bcast
+
+
+def _run_serial(data, mask, rad=1, blk=3, bcast=0.0):
+    sl = Searchlight(sl_rad=rad, max_blk_edge=blk)
+    sl.distribute([data], mask)
+    sl.broadcast(bcast)
+    return sl.run_searchlight(_sum_fn, pool_size=1)
+
+
+def test_searchlight_serial_correctness(seeded_rng):
+    dim = (7, 8, 9)
+    data = seeded_rng.rand(*dim, 4).astype(np.float32)
+    mask = np.ones(dim, dtype=bool)
+    out = _run_serial(data, mask, rad=1, blk=3, bcast=10.0)
+    assert out.shape == dim
+    # border trimmed
+    assert out[0, 0, 0] is None
+    # interior voxel: cube sum of 3^3 neighbourhood over all TRs + 10
+    i, j, k = 3, 4, 5
+    expected = data[i - 1:i + 2, j - 1:j + 2, k - 1:k + 2, :].sum() + 10.0
+    assert np.isclose(out[i, j, k], expected, rtol=1e-5)
+
+
+def test_searchlight_masked_voxels_skipped(seeded_rng):
+    dim = (6, 6, 6)
+    data = seeded_rng.rand(*dim, 3).astype(np.float32)
+    mask = np.zeros(dim, dtype=bool)
+    mask[2:4, 2:4, 2:4] = True
+    out = _run_serial(data, mask, rad=1, blk=4)
+    assert out[2, 2, 2] is not None
+    assert out[1, 1, 1] is None
+    # voxel_fn's mask argument excludes inactive voxels: compare against
+    # explicitly masked sum
+    i, j, k = 2, 3, 3
+    region = data[i - 1:i + 2, j - 1:j + 2, k - 1:k + 2, :]
+    m = mask[i - 1:i + 2, j - 1:j + 2, k - 1:k + 2]
+    assert np.isclose(out[i, j, k], region[m].sum(), rtol=1e-5)
+
+
+def test_searchlight_ball_shape_masks_voxel_fn(seeded_rng):
+    dim = (5, 5, 5)
+    data = seeded_rng.rand(*dim, 2).astype(np.float32)
+    mask = np.ones(dim, dtype=bool)
+    sl = Searchlight(sl_rad=1, max_blk_edge=5, shape=Diamond)
+    sl.distribute([data], mask)
+    sl.broadcast(0.0)
+    out = sl.run_searchlight(_sum_fn, pool_size=1)
+    i, j, k = 2, 2, 2
+    region = data[i - 1:i + 2, j - 1:j + 2, k - 1:k + 2, :]
+    expected = region[Diamond(1).mask_].sum()
+    assert np.isclose(out[i, j, k], expected, rtol=1e-5)
+
+
+def _dist_searchlight(ctx, outfile):
+    rng = np.random.RandomState(5)
+    dim = (9, 9, 9)
+    data = rng.rand(*dim, 4).astype(np.float32)
+    mask = np.ones(dim, dtype=bool)
+    # subject owned by rank 0 only
+    subjects = [data if ctx.rank == 0 else None]
+    sl = Searchlight(sl_rad=1, max_blk_edge=3, comm=ctx)
+    sl.distribute(subjects, mask)
+    sl.broadcast(1.5)
+    out = sl.run_searchlight(_sum_fn, pool_size=1)
+    if ctx.rank == 0:
+        vol = np.full(dim, np.nan)
+        for idx in np.ndindex(dim):
+            if out[idx] is not None:
+                vol[idx] = out[idx]
+        np.save(outfile, vol)
+
+
+@pytest.mark.slow
+def test_searchlight_distributed_matches_serial(tmp_path):
+    out = str(tmp_path / "vol.npy")
+    spawn_ranks(_dist_searchlight, world_size=2, args=(out,))
+    vol_dist = np.load(out)
+
+    rng = np.random.RandomState(5)
+    dim = (9, 9, 9)
+    data = rng.rand(*dim, 4).astype(np.float32)
+    mask = np.ones(dim, dtype=bool)
+    serial = _run_serial(data, mask, rad=1, blk=3, bcast=1.5)
+    vol_serial = np.full(dim, np.nan)
+    for idx in np.ndindex(dim):
+        if serial[idx] is not None:
+            vol_serial[idx] = serial[idx]
+    assert np.allclose(vol_dist, vol_serial, equal_nan=True, rtol=1e-6)
+
+
+def test_mvpa_voxelselector(seeded_rng):
+    from sklearn import svm
+
+    from brainiak_amd.fcma.mvpa_voxelselector import MVPAVoxelSelector
+    dim = (6, 6, 6)
+    n_epochs = 20
+    labels = np.array([e % 2 for e in range(n_epochs)])
+    data = seeded_rng.randn(*dim, n_epochs).astype(np.float32)
+    # inject discriminative activity at one location
+    data[3, 3, 3, :] += labels * 5.0
+    mask = np.ones(dim, dtype=bool)
+    sl = Searchlight(sl_rad=1, max_blk_edge=4, pool_size=1)
+    mvs = MVPAVoxelSelector(data, mask, labels, 4, sl)
+    vol, results = mvs.run(svm.SVC(kernel='linear'))
+    assert len(results) == mask.sum()
+    scores = [s for _, s in results]
+    assert scores == sorted(scores, reverse=True)
+    assert scores[0] > 0.8
