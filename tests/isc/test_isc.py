@@ -1,0 +1,180 @@
+import numpy as np
+import pytest
+
+from brainiak_amd.isc import (
+    bootstrap_isc,
+    compute_summary_statistic,
+    isc,
+    isfc,
+    isfc_distributed,
+    permutation_isc,
+    phaseshift_isc,
+    squareform_isfc,
+    timeshift_isc,
+)
+from brainiak_amd.parallel import spawn_ranks
+
+
+def _correlated_data(rng, n_TRs=60, n_voxels=10, n_subjects=5, strength=0.7):
+    signal = rng.randn(n_TRs, n_voxels)
+    data = strength * signal[:, :, np.newaxis] + \
+        (1 - strength) * rng.randn(n_TRs, n_voxels, n_subjects)
+    return data
+
+
+def test_isc_loo_basic(seeded_rng):
+    data = _correlated_data(seeded_rng)
+    iscs = isc(data)
+    assert iscs.shape == (5, 10)
+    assert np.all(iscs > 0.3)  # strong shared signal
+    # reference check for one subject/voxel
+    s, v = 2, 3
+    others = np.mean(np.delete(data, s, axis=2), axis=2)
+    expected = np.corrcoef(data[:, v, s], others[:, v])[0, 1]
+    assert np.isclose(iscs[s, v], expected, atol=1e-10)
+
+
+def test_isc_pairwise(seeded_rng):
+    data = _correlated_data(seeded_rng, n_subjects=4)
+    iscs = isc(data, pairwise=True)
+    assert iscs.shape == (6, 10)  # 4 choose 2
+    v = 0
+    expected = np.corrcoef(data[:, v, 0], data[:, v, 1])[0, 1]
+    assert np.isclose(iscs[0, v], expected, atol=1e-10)
+
+
+def test_isc_two_subjects_and_summary(seeded_rng):
+    data = _correlated_data(seeded_rng, n_subjects=2)
+    iscs = isc(data)
+    assert iscs.shape == (10,)
+    data5 = _correlated_data(seeded_rng)
+    m = isc(data5, summary_statistic='mean')
+    md = isc(data5, summary_statistic='median')
+    assert m.shape == (10,) and md.shape == (10,)
+    raw = isc(data5)
+    assert np.allclose(m, np.tanh(np.mean(np.arctanh(raw), axis=0)))
+
+
+def test_isc_list_input_and_nans(seeded_rng):
+    data = _correlated_data(seeded_rng)
+    as_list = [data[..., s] for s in range(5)]
+    assert np.allclose(isc(as_list), isc(data), equal_nan=True)
+    # NaN voxel
+    data_nan = data.copy()
+    data_nan[:, 2, :] = np.nan
+    iscs = isc(data_nan, tolerate_nans=True)
+    assert np.all(np.isnan(iscs[:, 2]))
+    assert np.all(~np.isnan(iscs[:, [0, 1, 3]]))
+
+
+def test_isfc_matches_manual(seeded_rng):
+    data = _correlated_data(seeded_rng, n_voxels=6, n_subjects=4)
+    isfcs, iscs = isfc(data)
+    assert isfcs.shape == (4, 15)  # condensed 6*5/2
+    assert iscs.shape == (4, 6)
+    # diagonal of ISFC == ISC values (leave-one-out, symmetrized)
+    sq = squareform_isfc(isfcs, iscs)
+    assert sq.shape == (4, 6, 6)
+    # manual check: subject 0 voxel (1,2) symmetrized cross-correlation
+    others = np.mean(data[..., 1:], axis=2)
+    c12 = np.corrcoef(data[:, 1, 0], others[:, 2])[0, 1]
+    c21 = np.corrcoef(data[:, 2, 0], others[:, 1])[0, 1]
+    assert np.isclose(sq[0, 1, 2], (c12 + c21) / 2, atol=1e-5)
+
+
+def test_isfc_targets(seeded_rng):
+    data = _correlated_data(seeded_rng, n_voxels=6, n_subjects=3)
+    targets = _correlated_data(seeded_rng, n_voxels=4, n_subjects=3)
+    out = isfc(data, targets=targets)
+    assert out.shape == (3, 6, 4)  # asymmetric, not vectorized
+
+
+def test_squareform_isfc_roundtrip(seeded_rng):
+    mats = seeded_rng.rand(3, 5, 5)
+    mats = (mats + mats.transpose(0, 2, 1)) / 2
+    condensed, diag = squareform_isfc(mats)
+    back = squareform_isfc(condensed, diag)
+    assert np.allclose(back, mats)
+
+
+def test_bootstrap_isc(seeded_rng):
+    data = _correlated_data(seeded_rng, n_subjects=8)
+    iscs = isc(data)
+    observed, ci, p, dist = bootstrap_isc(iscs, n_bootstraps=50,
+                                          random_state=0)
+    assert observed.shape == (10,)
+    assert dist.shape == (50, 10)
+    assert np.all(p < 0.5)  # real signal → small p
+    assert np.all(ci[0] <= ci[1])
+    # deterministic under the same seed
+    o2, _, p2, _ = bootstrap_isc(iscs, n_bootstraps=50, random_state=0)
+    assert np.allclose(p, p2)
+
+
+def test_permutation_isc_one_sample(seeded_rng):
+    data = _correlated_data(seeded_rng, n_subjects=8)  # 2^8 > 100 → MC
+    iscs = isc(data)
+    observed, p, dist = permutation_isc(iscs, n_permutations=100,
+                                        random_state=0)
+    assert dist.shape == (100, 10)
+    assert np.all(p < 0.5)
+    # exact test triggers when 2**n <= n_permutations
+    iscs_small = isc(_correlated_data(seeded_rng, n_subjects=4))
+    _, p_exact, dist_exact = permutation_isc(iscs_small,
+                                             n_permutations=16)
+    assert dist_exact.shape[0] == 16
+
+
+def test_permutation_isc_two_sample(seeded_rng):
+    d1 = _correlated_data(seeded_rng, n_subjects=5, strength=0.8)
+    d2 = _correlated_data(seeded_rng, n_subjects=5, strength=0.1)
+    iscs = np.vstack((isc(d1), isc(d2)))
+    groups = [0] * 5 + [1] * 5
+    observed, p, dist = permutation_isc(iscs, group_assignment=groups,
+                                        n_permutations=100, random_state=0)
+    assert observed.shape == (10,)
+    assert np.all(observed > 0)  # group 1 stronger
+    assert dist.shape == (100, 10)
+
+
+def test_timeshift_and_phaseshift(seeded_rng):
+    data = _correlated_data(seeded_rng, n_TRs=40, n_voxels=3, n_subjects=4)
+    obs_t, p_t, dist_t = timeshift_isc(data, n_shifts=20, random_state=0)
+    assert dist_t.shape == (20, 3)
+    assert np.all(p_t <= 1.0)
+    obs_p, p_p, dist_p = phaseshift_isc(data, n_shifts=20, random_state=0)
+    assert dist_p.shape == (20, 3)
+    # null distributions centered near zero, observed well above
+    assert np.abs(np.mean(dist_t)) < 0.2
+    assert np.all(obs_t > np.mean(dist_t, axis=0))
+
+
+def test_compute_summary_statistic_errors():
+    with pytest.raises(ValueError):
+        compute_summary_statistic(np.ones(3), 'mode')
+
+
+def _dist_isfc(ctx, outfile):
+    rng = np.random.RandomState(3)
+    data = _correlated_data(rng, n_TRs=30, n_voxels=8, n_subjects=4)
+    mine = [data[..., s] for s in range(4)
+            if s % ctx.world_size == ctx.rank]
+    result = isfc_distributed(mine, ctx, summary_statistic='mean')
+    if ctx.rank == 0:
+        np.save(outfile, result)
+
+
+@pytest.mark.slow
+def test_isfc_distributed_matches_serial(tmp_path):
+    out = str(tmp_path / "isfc.npy")
+    spawn_ranks(_dist_isfc, world_size=2, args=(out,))
+    dist_result = np.load(out)
+
+    rng = np.random.RandomState(3)
+    data = _correlated_data(rng, n_TRs=30, n_voxels=8, n_subjects=4)
+    serial = isfc(data, summary_statistic='mean', vectorize_isfcs=False)
+    assert dist_result.shape == (8, 8)
+    # off-diagonal entries must agree (diagonal: distributed keeps the
+    # self-ISC values where serial's squareform drops/refills them)
+    off = ~np.eye(8, dtype=bool)
+    assert np.allclose(dist_result[off], serial[off], atol=1e-4)
