@@ -1,0 +1,109 @@
+import numpy as np
+import pytest
+
+from brainiak_amd.eventseg.event import EventSegment, NotFittedError
+
+
+def _event_data(rng, n_events=4, event_len=15, n_voxels=20, noise=0.5):
+    """Piecewise-constant event patterns + noise."""
+    patterns = rng.randn(n_events, n_voxels) * 2
+    rows = []
+    bounds = []
+    for e in range(n_events):
+        rows.append(np.tile(patterns[e], (event_len, 1)))
+        bounds.append(e * event_len)
+    data = np.vstack(rows) + noise * rng.randn(n_events * event_len,
+                                               n_voxels)
+    return data, np.array(bounds[1:]), patterns
+
+
+def test_fit_recovers_event_boundaries(seeded_rng):
+    data, bounds, _ = _event_data(seeded_rng)
+    es = EventSegment(n_events=4, n_iter=100)
+    es.fit(data)
+    assert es.event_pat_.shape == (20, 4)
+    assert len(es.segments_) == 1
+    seg = es.segments_[0]
+    assert seg.shape == (60, 4)
+    # each timepoint's soft assignment sums to 1
+    assert np.allclose(seg.sum(axis=1), 1.0, atol=1e-6)
+    # hard segmentation should match the true boundaries closely
+    labels = np.argmax(seg, axis=1)
+    true_labels = np.repeat(np.arange(4), 15)
+    assert np.mean(labels == true_labels) > 0.85
+    # log-likelihood increased during fitting
+    assert es.ll_.shape[0] >= 2
+    assert np.mean(es.ll_[-1]) >= np.mean(es.ll_[0])
+
+
+def test_find_events_on_new_data(seeded_rng):
+    data, _, patterns = _event_data(seeded_rng)
+    es = EventSegment(n_events=4, n_iter=60).fit(data)
+    data2, _, _ = _event_data(np.random.RandomState(1))
+    # transfer the learned patterns via set_event_patterns instead
+    es2 = EventSegment(n_events=4)
+    es2.set_event_patterns(es.event_pat_)
+    seg, ll = es2.find_events(data, var=1.0)
+    assert seg.shape == (60, 4)
+    assert np.isfinite(ll)
+    labels = np.argmax(seg, axis=1)
+    true_labels = np.repeat(np.arange(4), 15)
+    assert np.mean(labels == true_labels) > 0.8
+
+
+def test_predict_and_notfitted(seeded_rng):
+    data, _, _ = _event_data(seeded_rng)
+    es = EventSegment(n_events=4, n_iter=60)
+    with pytest.raises(NotFittedError):
+        es.predict(data)
+    es.fit(data)
+    labels = es.predict(data)
+    assert labels.shape == (60,)
+    assert set(labels) <= set(range(4))
+    # monotonically nondecreasing labels (left-to-right chain)
+    assert np.all(np.diff(labels) >= 0)
+
+
+def test_multiple_datasets(seeded_rng):
+    d1, _, _ = _event_data(seeded_rng)
+    d2, _, _ = _event_data(seeded_rng)
+    es = EventSegment(n_events=4, n_iter=60).fit([d1, d2])
+    assert len(es.segments_) == 2
+    assert es.ll_.shape[1] == 2
+
+
+def test_model_prior():
+    es = EventSegment(n_events=3)
+    seg, ll = es.model_prior(30)
+    assert seg.shape == (30, 3)
+    assert np.allclose(seg.sum(axis=1), 1.0)
+    # prior probability of event 0 decreasing over time, event K-1
+    # increasing
+    assert seg[0, 0] > seg[-1, 0]
+    assert seg[-1, 2] > seg[0, 2]
+
+
+def test_split_merge_runs(seeded_rng):
+    data, _, _ = _event_data(seeded_rng, n_events=3, event_len=10)
+    es = EventSegment(n_events=3, n_iter=30, split_merge=True)
+    es.fit(data)
+    assert hasattr(es, "event_pat_")
+
+
+def test_event_chains():
+    es = EventSegment(n_events=4, event_chains=np.array([0, 0, 1, 1]))
+    with pytest.raises(RuntimeError):
+        es.fit(np.random.randn(40, 10))
+    # with patterns set, find_events works on chains
+    es.set_event_patterns(np.random.randn(10, 4))
+    seg, ll = es.find_events(np.random.randn(40, 10), var=1.0)
+    assert seg.shape == (40, 4)
+
+
+def test_calc_weighted_event_var(seeded_rng):
+    data, _, patterns = _event_data(seeded_rng, noise=0.1)
+    es = EventSegment(n_events=4, n_iter=80).fit(data)
+    ev_var = es.calc_weighted_event_var(data, es.segments_[0],
+                                        es.event_pat_)
+    assert ev_var.shape == (4,)
+    assert np.all(ev_var >= 0)
