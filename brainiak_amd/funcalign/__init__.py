@@ -6,6 +6,7 @@ Citation context (same methods as the reference implements):
 Datasets", IEEE Big Data 2016.
 """
 
+from .rsrm import RSRM  # noqa: F401
 from .srm import SRM, DetSRM, load  # noqa: F401
 
-__all__ = ["SRM", "DetSRM", "load"]
+__all__ = ["RSRM", "SRM", "DetSRM", "load"]

@@ -56,7 +56,15 @@ def _polar_orthogonal(A, perturb=0.001):
     through an eigendecomposition of the K×K Gram matrix — K is small
     (≈50), so device work is V-independent.  The ε diagonal perturbation
     matches the reference's conditioning trick (srm.py:598-599).
+
+    On gfx950 the eigensolve runs in the hand-written batched Jacobi
+    kernel (ops.batched_polar, one wavefront per Gram matrix).
     """
+    if A.is_cuda and A.shape[1] <= 64:
+        from .. import ops
+        if ops.has_hip():
+            return ops.batched_polar(A[None].contiguous(),
+                                     perturb=perturb)[0].to(A.dtype)
     if perturb:
         A = A.clone()
         d = min(A.shape)
