@@ -149,22 +149,55 @@ __global__ __launch_bounds__(256) void k_corr_norm(
     }
     __syncthreads();
 
-    for (int base = tid; base < P * CN_VT; base += 256) {
-        int v = base % CN_VT;
-        int p = base / CN_VT;
-        if (v < VT) {
-            float breg[TL];
+    if ((VB & 1) == 0 && VT == CN_VT) {
+        // fast path: each thread owns a PAIR of adjacent columns and
+        // loads both with one u32 (2×bf16) — halves the global-load
+        // instruction count and amortizes the a_tile LDS reads
+        for (int base = tid; base < P * (CN_VT / 2); base += 256) {
+            int v = 2 * (base % (CN_VT / 2));
+            int p = base / (CN_VT / 2);
+            float b0[TL], b1[TL];
             const bf16_t* brow = B + ((s * P + p) * L) * VB + (v0 + v);
             #pragma unroll
-            for (int k = 0; k < TL; ++k)
-                breg[k] = (float)brow[(ll)k * VB];
+            for (int k = 0; k < TL; ++k) {
+                unsigned u = *(const unsigned*)&brow[(ll)k * VB];
+                b0[k] = (float)(*(const bf16_t*)&u);
+                bf16_t hi = *(((const bf16_t*)&u) + 1);
+                b1[k] = (float)hi;
+            }
             for (int c = 0; c < CT; ++c) {
-                float acc = 0.f;
+                float acc0 = 0.f, acc1 = 0.f;
                 const bf16_t* arow = a_tile + ((size_t)p * L) * CN_CT + c;
                 #pragma unroll
+                for (int k = 0; k < TL; ++k) {
+                    float a = (float)arow[(size_t)k * CN_CT];
+                    acc0 = fmaf(a, b0[k], acc0);
+                    acc1 = fmaf(a, b1[k], acc1);
+                }
+                corr[((size_t)c * P + p) * CN_VT + v] = acc0;
+                corr[((size_t)c * P + p) * CN_VT + v + 1] = acc1;
+            }
+        }
+    } else {
+        for (int base = tid; base < P * CN_VT; base += 256) {
+            int v = base % CN_VT;
+            int p = base / CN_VT;
+            if (v < VT) {
+                float breg[TL];
+                const bf16_t* brow = B + ((s * P + p) * L) * VB + (v0 + v);
+                #pragma unroll
                 for (int k = 0; k < TL; ++k)
-                    acc = fmaf((float)arow[(size_t)k * CN_CT], breg[k], acc);
-                corr[((size_t)c * P + p) * CN_VT + v] = acc;
+                    breg[k] = (float)brow[(ll)k * VB];
+                for (int c = 0; c < CT; ++c) {
+                    float acc = 0.f;
+                    const bf16_t* arow = a_tile
+                        + ((size_t)p * L) * CN_CT + c;
+                    #pragma unroll
+                    for (int k = 0; k < TL; ++k)
+                        acc = fmaf((float)arow[(size_t)k * CN_CT],
+                                   breg[k], acc);
+                    corr[((size_t)c * P + p) * CN_VT + v] = acc;
+                }
             }
         }
     }
@@ -270,17 +303,19 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
         int row = tid >> 2;
         int col = (tid & 3) * 8;
         const bf16_t* s = src + (rows0 + row) * V + k0 + col;
-        bf16_t tmp[8];
-        if (k0 + col + 8 <= V) {
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) tmp[j] = s[j];
+        if (k0 + col + 8 <= V && (((uintptr_t)s) & 15) == 0) {
+            // one 16-B vector load per thread (coalesced 128 B per
+            // quarter-wave) instead of 8 scalar bf16 loads
+            bf16x8 v = *(const bf16x8*)s;
+            *(bf16x8*)&dst[row][col] = v;
         } else {
+            bf16_t tmp[8];
             #pragma unroll
             for (int j = 0; j < 8; ++j)
                 tmp[j] = (k0 + col + j < V) ? s[j] : (bf16_t)0.0f;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) dst[row][col + j] = tmp[j];
         }
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) dst[row][col + j] = tmp[j];
     };
 
     const ll kTiles = (V + GR_KT - 1) / GR_KT;
