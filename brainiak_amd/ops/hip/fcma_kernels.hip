@@ -161,9 +161,10 @@ __global__ __launch_bounds__(256) void k_corr_norm(
             #pragma unroll
             for (int k = 0; k < TL; ++k) {
                 unsigned u = *(const unsigned*)&brow[(ll)k * VB];
-                b0[k] = (float)(*(const bf16_t*)&u);
-                bf16_t hi = *(((const bf16_t*)&u) + 1);
-                b1[k] = (float)hi;
+                // bf16 -> f32 is a plain 16-bit left shift (no locals,
+                // no scratch)
+                b0[k] = __uint_as_float(u << 16);
+                b1[k] = __uint_as_float(u & 0xffff0000u);
             }
             for (int c = 0; c < CT; ++c) {
                 float acc0 = 0.f, acc1 = 0.f;
