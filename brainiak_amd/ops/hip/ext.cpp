@@ -26,6 +26,7 @@ void launch_svm_cv(const float*, const float*, const int*, const int*,
                    const int*, const int*, int*, ll, int, int, float,
                    float, int, void*);
 ll fcma_supported_L(ll);
+int fcma_corr_norm_smem(ll, int);
 }
 
 static void* cur_stream() {
@@ -84,6 +85,8 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
     TORCH_CHECK(start >= 0 && start + count <= VA, "voxel range");
     TORCH_CHECK(P > 0 && E % P == 0, "epochs_per_subj must divide E");
     TORCH_CHECK(P <= 32, "epochs_per_subj > 32: use the staged path");
+    TORCH_CHECK(fcma_corr_norm_smem(L, (int)P) <= 160 * 1024,
+                "corr tile exceeds gfx950 LDS; use the staged path");
     TORCH_CHECK(fcma_supported_L(L) == L,
                 "epoch length must be padded to one of {8,16,24,32,40}");
     ll Eout = std::max((ll)padE, E);

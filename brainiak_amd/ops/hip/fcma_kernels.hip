@@ -24,6 +24,9 @@
 typedef __hip_bfloat16 bf16_t;
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 typedef short bf16x8 __attribute__((ext_vector_type(8)));
+// 4-byte-aligned variant for global loads: gfx950 dwordx4 loads need
+// only dword alignment, and bf16 rows are 4B- but not 16B-aligned
+typedef short bf16x8_u __attribute__((ext_vector_type(8), aligned(4)));
 typedef long long ll;
 
 // ---------------------------------------------------------------------------
@@ -134,72 +137,54 @@ __global__ __launch_bounds__(256) void k_corr_norm(
     const int tid = threadIdx.x;
 
     extern __shared__ char smem[];
-    bf16_t* a_tile = (bf16_t*)smem;                    // [P][L][CN_CT]
-    float* corr = (float*)(smem + (size_t)P * L * CN_CT * sizeof(bf16_t));
+    float* a_tile = (float*)smem;                      // [P][L][CN_CT] f32
+    float* corr = a_tile + (size_t)P * L * CN_CT;
     // corr: [CN_CT][P][CN_VT]
 
+    // stage A columns converted to fp32 ONCE (per-FMA bf16 converts in
+    // the hot loop were ~1/3 of the kernel's VALU work)
     for (int idx = tid; idx < P * (int)L * CN_CT; idx += 256) {
         int c = idx % CN_CT;
         int k = (idx / CN_CT) % (int)L;
         int p = idx / (CN_CT * (int)L);
-        bf16_t val = (bf16_t)0.0f;
+        float val = 0.0f;
         if (c < CT)
-            val = A[((s * P + p) * L + k) * VA + (s0 + c0 + c)];
+            val = (float)A[((s * P + p) * L + k) * VA + (s0 + c0 + c)];
         a_tile[idx] = val;
     }
     __syncthreads();
 
-    if ((VB & 1) == 0 && VT == CN_VT) {
-        // fast path: each thread owns a PAIR of adjacent columns and
-        // loads both with one u32 (2×bf16) — halves the global-load
-        // instruction count and amortizes the a_tile LDS reads
-        for (int base = tid; base < P * (CN_VT / 2); base += 256) {
-            int v = 2 * (base % (CN_VT / 2));
-            int p = base / (CN_VT / 2);
-            float b0[TL], b1[TL];
+    for (int base = tid; base < P * CN_VT; base += 256) {
+        int v = base % CN_VT;
+        int p = base / CN_VT;
+        if (v < VT) {
+            float breg[TL];
             const bf16_t* brow = B + ((s * P + p) * L) * VB + (v0 + v);
             #pragma unroll
+            for (int k = 0; k < TL; ++k)
+                breg[k] = (float)brow[(ll)k * VB];
+            // k-outer / c-inner with a 16-wide register accumulator:
+            // 4x f32x4 LDS broadcast reads + 16 FMA per k-step
+            f32x4 acc[CN_CT / 4];
+            #pragma unroll
+            for (int q = 0; q < CN_CT / 4; ++q) acc[q] = (f32x4)0.f;
+            const f32x4* arow4 =
+                (const f32x4*)(a_tile + ((size_t)p * L) * CN_CT);
+            #pragma unroll
             for (int k = 0; k < TL; ++k) {
-                unsigned u = *(const unsigned*)&brow[(ll)k * VB];
-                // bf16 -> f32 is a plain 16-bit left shift (no locals,
-                // no scratch)
-                b0[k] = __uint_as_float(u << 16);
-                b1[k] = __uint_as_float(u & 0xffff0000u);
-            }
-            for (int c = 0; c < CT; ++c) {
-                float acc0 = 0.f, acc1 = 0.f;
-                const bf16_t* arow = a_tile + ((size_t)p * L) * CN_CT + c;
+                float bk = breg[k];
                 #pragma unroll
-                for (int k = 0; k < TL; ++k) {
-                    float a = (float)arow[(size_t)k * CN_CT];
-                    acc0 = fmaf(a, b0[k], acc0);
-                    acc1 = fmaf(a, b1[k], acc1);
-                }
-                corr[((size_t)c * P + p) * CN_VT + v] = acc0;
-                corr[((size_t)c * P + p) * CN_VT + v + 1] = acc1;
-            }
-        }
-    } else {
-        for (int base = tid; base < P * CN_VT; base += 256) {
-            int v = base % CN_VT;
-            int p = base / CN_VT;
-            if (v < VT) {
-                float breg[TL];
-                const bf16_t* brow = B + ((s * P + p) * L) * VB + (v0 + v);
-                #pragma unroll
-                for (int k = 0; k < TL; ++k)
-                    breg[k] = (float)brow[(ll)k * VB];
-                for (int c = 0; c < CT; ++c) {
-                    float acc = 0.f;
-                    const bf16_t* arow = a_tile
-                        + ((size_t)p * L) * CN_CT + c;
-                    #pragma unroll
-                    for (int k = 0; k < TL; ++k)
-                        acc = fmaf((float)arow[(size_t)k * CN_CT],
-                                   breg[k], acc);
-                    corr[((size_t)c * P + p) * CN_VT + v] = acc;
+                for (int q = 0; q < CN_CT / 4; ++q) {
+                    f32x4 a4 = arow4[k * (CN_CT / 4) + q];
+                    acc[q] += a4 * bk;
                 }
             }
+            #pragma unroll
+            for (int q = 0; q < CN_CT / 4; ++q)
+                #pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    corr[((size_t)(4 * q + j) * P + p) * CN_VT + v] =
+                        acc[q][j];
         }
     }
     __syncthreads();
@@ -304,10 +289,11 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
         int row = tid >> 2;
         int col = (tid & 3) * 8;
         const bf16_t* s = src + (rows0 + row) * V + k0 + col;
-        if (k0 + col + 8 <= V && (((uintptr_t)s) & 15) == 0) {
+        if (k0 + col + 8 <= V && (((uintptr_t)s) & 3) == 0) {
             // one 16-B vector load per thread (coalesced 128 B per
-            // quarter-wave) instead of 8 scalar bf16 loads
-            bf16x8 v = *(const bf16x8*)s;
+            // quarter-wave) instead of 8 scalar bf16 loads; rows are
+            // only 4B-aligned, hence the aligned(4) vector type
+            bf16x8 v = (bf16x8)(*(const bf16x8_u*)s);
             *(bf16x8*)&dst[row][col] = v;
         } else {
             bf16_t tmp[8];
@@ -502,7 +488,7 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
 }
 
 extern "C" int fcma_corr_norm_smem(ll L, int P) {
-    size_t smem = (size_t)P * L * CN_CT * sizeof(bf16_t)
+    size_t smem = (size_t)P * L * CN_CT * sizeof(float)   // fp32 a_tile
                 + (size_t)CN_CT * P * CN_VT * sizeof(float);
     return (int)smem;
 }
