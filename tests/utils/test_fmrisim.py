@@ -1,0 +1,167 @@
+import numpy as np
+import pytest
+
+from brainiak_amd.utils import fmrisim as sim
+
+
+@pytest.fixture(autouse=True)
+def _seed():
+    np.random.seed(0)
+
+
+def test_generate_signal_shapes():
+    dims = np.array([16, 16, 16])
+    for ftype, check in [('cube', None), ('sphere', None),
+                         ('loop', None), ('cavity', None)]:
+        vol = sim.generate_signal(dimensions=dims,
+                                  feature_coordinates=np.array([[8, 8, 8]]),
+                                  feature_size=[5],
+                                  feature_type=[ftype],
+                                  signal_magnitude=[2.0])
+        assert vol.shape == tuple(dims)
+        assert vol.max() == 2.0
+        assert vol.min() == 0.0
+    # cube of size 3 centered at (8,8,8): 27 voxels
+    vol = sim.generate_signal(dims, np.array([[8, 8, 8]]), [3], ['cube'])
+    assert (vol > 0).sum() == 27
+
+
+def test_generate_stimfunction_and_exports(tmp_path):
+    stim = sim.generate_stimfunction(onsets=[10, 30], event_durations=[5],
+                                     total_time=60)
+    assert stim.shape == (6000, 1)
+    assert stim[1100, 0] == 1 and stim[2000, 0] == 0
+    with pytest.raises(ValueError):
+        sim.generate_stimfunction([100], [5], total_time=60)
+
+    # 3-column export roundtrip
+    f3 = tmp_path / "events.txt"
+    sim.export_3_column(stim, str(f3))
+    back = np.loadtxt(f3)
+    assert back.shape == (2, 3)
+    assert np.allclose(back[:, 0], [10, 30], atol=0.02)
+    assert np.allclose(back[:, 1], [5, 5], atol=0.02)
+
+    # epoch file export
+    fe = tmp_path / "epochs.npy"
+    sim.export_epoch_file([stim], str(fe), tr_duration=2.0)
+    epochs = np.load(fe, allow_pickle=True)
+    assert epochs[0].shape[0] == 1      # one condition
+    assert epochs[0].shape[1] == 2      # two epochs
+    assert epochs[0].dtype == bool
+
+
+def test_hrf_and_convolution():
+    hrf = sim._double_gamma_hrf()
+    hrf = np.asarray(hrf)
+    # peak around 5-7 s at 100 Hz resolution
+    assert 400 < np.argmax(hrf) < 800
+    # undershoot exists
+    assert hrf.min() < 0
+
+    stim = sim.generate_stimfunction(onsets=[10], event_durations=[2],
+                                     total_time=80)
+    signal_func = sim.convolve_hrf(stim, tr_duration=2.0)
+    assert signal_func.shape == (40, 1)
+    assert np.isclose(signal_func.max(), 1.0)  # scaled
+    # response peaks a few TRs after the onset (TR 5)
+    assert 5 <= np.argmax(signal_func[:, 0]) <= 10
+
+
+def test_apply_signal():
+    vol = np.zeros((4, 4, 4))
+    vol[1, 1, 1] = 2.0
+    vol[2, 2, 2] = 1.0
+    sf = np.vstack([np.linspace(0, 1, 10)]).T
+    out = sim.apply_signal(sf, vol)
+    assert out.shape == (4, 4, 4, 10)
+    assert np.isclose(out[1, 1, 1, -1], 2.0)
+    assert np.isclose(out[2, 2, 2, -1], 1.0)
+    assert out[0, 0, 0].sum() == 0
+
+
+def test_generate_noise_and_calc_noise():
+    dims = np.array([16, 16, 16])
+    stim = sim.generate_stimfunction(onsets=[10], event_durations=[4],
+                                     total_time=100)
+    stim_tr = stim[::int(2.0 * 100), 0]
+    # brain in the center, empty border (needed for SNR estimation)
+    template = np.zeros(dims)
+    template[6:10, 6:10, 6:10] = 0.8
+    mask = (template > 0).astype(float)
+    nd = {'sfnr': 60, 'snr': 30, 'max_activity': 500, 'matched': 0}
+    noise = sim.generate_noise(dims, stim_tr, 2.0, template, mask,
+                               noise_dict=dict(nd))
+    assert noise.shape == (16, 16, 16, 50)
+    assert np.all(noise >= 0)
+    # mean brain activity should be near max_activity * template
+    brain_mean = noise[mask > 0].mean()
+    assert 200 < brain_mean < 600
+
+    est = sim.calc_noise(noise, mask, template)
+    assert 'sfnr' in est and 'snr' in est and 'fwhm' in est
+    assert est['sfnr'] > 0 and est['snr'] > 0
+    assert np.isfinite(est['auto_reg_rho'][0])
+
+
+def test_noise_matching_converges():
+    """With matched=1, the fitted noise's SFNR should approach target."""
+    dims = np.array([8, 8, 8])
+    stim_tr = np.zeros(40)
+    template = np.ones(dims) * 0.9
+    mask = np.ones(dims)
+    target = {'sfnr': 50, 'snr': 40, 'max_activity': 800, 'matched': 1,
+              'auto_reg_rho': [0.5], 'ma_rho': [0.0]}
+    noise = sim.generate_noise(dims, stim_tr, 2.0, template, mask,
+                               noise_dict=dict(target),
+                               iterations=[8, 4])
+    sfnr = sim._calc_sfnr(noise, mask)
+    assert abs(sfnr - 50) / 50 < 0.5
+
+
+def test_mask_brain_self():
+    vol = np.zeros((10, 10, 10, 5))
+    vol[3:7, 3:7, 3:7, :] = 100 + np.random.randn(4, 4, 4, 5)
+    mask, template = sim.mask_brain(vol, mask_threshold=0.5)
+    assert mask.shape == (10, 10, 10)
+    assert mask[5, 5, 5] == 1
+    assert mask[0, 0, 0] == 0
+    assert template.max() <= 1.0
+
+
+def test_compute_signal_change():
+    sf = np.vstack([np.sin(np.linspace(0, 6, 50))]).T
+    noise = np.ones((50, 1)) * 200
+    nd = {'sfnr': 50}
+    scaled = sim.compute_signal_change(sf.copy(), noise, nd, [2.0],
+                                       method='PSC')
+    # PSC: peak = mean/100 * magnitude = 4
+    assert np.isclose(np.max(np.abs(scaled)), 4.0, rtol=1e-6)
+    scaled2 = sim.compute_signal_change(sf.copy(), noise, nd, [1.0],
+                                        method='SFNR')
+    assert np.isclose(np.max(np.abs(scaled2)), 200 / 50, rtol=1e-6)
+    with pytest.raises(ValueError):
+        sim.compute_signal_change(sf.copy(), noise, nd, [1.0],
+                                  method='bogus')
+
+
+def test_ar_estimation_recovers_rho():
+    """Yule-Walker AR(1) estimate on a synthetic AR(1) process."""
+    rho = 0.6
+    n = 2000
+    x = np.zeros(n)
+    eps = np.random.randn(n)
+    for t in range(1, n):
+        x[t] = rho * x[t - 1] + eps[t]
+    ar, ma = sim._estimate_ar_ma(x, 1, 1)
+    assert abs(ar[0] - rho) < 0.1
+
+
+def test_1d_rfs():
+    rfs, tuning = sim.generate_1d_gaussian_rfs(
+        20, 180, (0, 179), random_tuning=False)
+    assert rfs.shape == (20, 180)
+    assert np.allclose(rfs.max(axis=1), 1.0)
+    data = sim.generate_1d_rf_responses(rfs, np.array([10, 90, 170]),
+                                        180, (0, 179))
+    assert data.shape == (20, 3)
