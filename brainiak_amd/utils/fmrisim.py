@@ -151,7 +151,7 @@ def generate_stimfunction(onsets, event_durations, total_time, weights=[1],
         event_durations = event_durations * len(onsets)
     if len(weights) == 1:
         weights = weights * len(onsets)
-    if np.max(onsets) > total_time:
+    if len(onsets) and np.max(onsets) > total_time:
         raise ValueError('Onsets outside of range of total time.')
 
     stimfunction = np.zeros(
@@ -275,7 +275,7 @@ def convolve_hrf(stimfunction, tr_duration, hrf_type='double_gamma',
         tmp = np.convolve(stimfunction[:, i], hrf)
         tmp = tmp[:duration * stride]
         vox = tmp[int(stride / 2)::stride]
-        if scale_function:
+        if scale_function and np.max(np.abs(vox)) > 0:
             vox = vox / np.max(vox)
         if signal_function is None:
             signal_function = np.zeros((len(vox), list_num))
