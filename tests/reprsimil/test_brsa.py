@@ -1,0 +1,140 @@
+"""BRSA/GBRSA: planted-covariance recovery + autograd gradient checks
+(the reference validates its hand-written gradients with numdifftools;
+here finite differences check the autograd value-and-grad)."""
+
+import numpy as np
+import pytest
+import torch
+
+from brainiak_amd.reprsimil.brsa import (
+    BRSA,
+    GBRSA,
+    Ncomp_SVHT_MG_DLD_approx,
+    _ar1_quadforms,
+)
+from brainiak_amd.utils.utils import cov2corr
+
+
+def _gen_brsa_data(rng, T=150, V=60, C=5, snr_scale=1.0, rho=0.3):
+    """Data generated exactly from the BRSA model."""
+    U = np.eye(C) * 0.5
+    U[0, 1] = U[1, 0] = 0.4
+    if C > 3:
+        U[2, 3] = U[3, 2] = -0.3
+    design = rng.randn(T, C)
+    # smooth the design a little like an HRF would
+    for c in range(C):
+        design[:, c] = np.convolve(design[:, c], np.ones(5) / 5,
+                                   mode='same')
+    snr = np.exp(rng.randn(V) * 0.3) * snr_scale
+    sigma = 0.5 + rng.rand(V)
+    beta = np.linalg.cholesky(U + 1e-9 * np.eye(C)) @ rng.randn(C, V)
+    beta = beta * (snr * sigma)[None, :]
+    noise = np.zeros((T, V))
+    eps = rng.randn(T, V) * sigma[None, :]
+    noise[0] = eps[0]
+    for t in range(1, T):
+        noise[t] = rho * noise[t - 1] + eps[t]
+    Y = design @ beta + noise + 10.0  # DC offset
+    return Y, design, U
+
+
+def test_brsa_recovers_planted_covariance(seeded_rng):
+    Y, design, U = _gen_brsa_data(seeded_rng)
+    model = BRSA(rank=None, auto_nuisance=False, random_state=0,
+                 minimize_options={'maxiter': 300, 'disp': False})
+    model.fit(X=Y, design=design)
+    assert model.U_.shape == (5, 5)
+    c_est = model.C_
+    c_true = cov2corr(U)
+    off = ~np.eye(5, dtype=bool)
+    r = np.corrcoef(c_est[off], c_true[off])[0, 1]
+    assert r > 0.7, (c_est, c_true)
+    # AR coefficient in the right ballpark
+    assert 0.0 < np.median(model.rho_) < 0.6
+    assert model.beta_.shape == (5, 60)
+    assert np.all(model.sigma_ > 0)
+
+
+def test_brsa_gradient_matches_finite_difference(seeded_rng):
+    Y, design, _ = _gen_brsa_data(seeded_rng, T=60, V=8, C=3)
+    from brainiak_amd.reprsimil.brsa import _project_out
+    X0 = np.ones((60, 1))
+    Xp = torch.as_tensor(_project_out(design, X0))
+    Yp = torch.as_tensor(_project_out(Y, X0))
+    quads = _ar1_quadforms(Xp, Yp)
+    C, V, T, rank = 3, 8, 60, 3
+    n = C * rank + 2 * V
+    rng = np.random.RandomState(1)
+    theta = torch.tensor(rng.randn(n) * 0.3, dtype=torch.float64,
+                         requires_grad=True)
+    loss = BRSA._neg_loglik(theta, *quads, C=C, V=V, T=T, rank=rank)
+    loss.backward()
+    g = theta.grad.numpy()
+    eps = 1e-6
+    for i in [0, 4, C * rank + 1, n - 2]:
+        tp = theta.detach().numpy().copy()
+        tp[i] += eps
+        lp = float(BRSA._neg_loglik(torch.tensor(tp), *quads, C=C, V=V,
+                                    T=T, rank=rank))
+        tm = theta.detach().numpy().copy()
+        tm[i] -= eps
+        lm = float(BRSA._neg_loglik(torch.tensor(tm), *quads, C=C, V=V,
+                                    T=T, rank=rank))
+        num = (lp - lm) / (2 * eps)
+        assert np.isclose(g[i], num, rtol=1e-4, atol=1e-6), (i, g[i], num)
+
+
+def test_brsa_transform_and_score(seeded_rng):
+    Y, design, _ = _gen_brsa_data(seeded_rng, T=120, V=40)
+    model = BRSA(auto_nuisance=False, random_state=0,
+                 minimize_options={'maxiter': 150, 'disp': False})
+    model.fit(X=Y, design=design)
+    ts, ts0 = model.transform(Y)
+    assert ts.shape == (120, 5)
+    # decoded time courses should correlate with the true design
+    rs = [np.corrcoef(ts[:, c], design[:, c])[0, 1] for c in range(5)]
+    assert np.mean(rs) > 0.25
+    # score on training data better than score on pure-noise data
+    s_good = model.score(Y, design)
+    noise_Y = seeded_rng.randn(*Y.shape) * Y.std() + Y.mean()
+    s_null = model.score(noise_Y, design)
+    assert s_good > s_null
+
+
+def test_brsa_auto_nuisance_runs(seeded_rng):
+    Y, design, _ = _gen_brsa_data(seeded_rng, T=80, V=30)
+    # add a shared nuisance time course
+    Y += np.outer(np.sin(np.arange(80) / 5), seeded_rng.randn(30)) * 3
+    model = BRSA(auto_nuisance=True, n_nureg=2, random_state=0,
+                 minimize_options={'maxiter': 100, 'disp': False})
+    model.fit(X=Y, design=design)
+    assert model.X0_.shape[1] == 3  # 2 PCs + DC
+    assert np.all(np.isfinite(model.U_))
+
+
+def test_brsa_requires_design(seeded_rng):
+    with pytest.raises(AssertionError):
+        BRSA().fit(X=np.zeros((10, 5)))
+
+
+def test_gbrsa_recovers_structure(seeded_rng):
+    Y1, design1, U = _gen_brsa_data(seeded_rng, T=120, V=40)
+    Y2, design2, _ = _gen_brsa_data(seeded_rng, T=120, V=40)
+    model = GBRSA(auto_nuisance=False, SNR_bins=8, rho_bins=8,
+                  random_state=0,
+                  minimize_options={'maxiter': 120, 'disp': False})
+    model.fit(X=[Y1, Y2], design=[design1, design1])
+    c_true = cov2corr(U)
+    off = ~np.eye(5, dtype=bool)
+    r = np.corrcoef(model.C_[off], c_true[off])[0, 1]
+    assert r > 0.6
+
+
+def test_ncomp_svht(seeded_rng):
+    # low-rank + noise: estimate close to the true rank
+    U = seeded_rng.randn(200, 3)
+    V = seeded_rng.randn(3, 50)
+    X = U @ V + 0.05 * seeded_rng.randn(200, 50)
+    n = Ncomp_SVHT_MG_DLD_approx(X, zscore=False)
+    assert 1 <= n <= 6
