@@ -1,0 +1,54 @@
+#!/usr/bin/env python3
+"""BASELINE config 1: probabilistic SRM, 5 subjects x 200 voxels x 50
+TRs, CPU, world_size=1 (the plumbing check).  One step = one full
+SRM fit (n_iter EM iterations)."""
+
+import argparse
+import sys
+from pathlib import Path
+
+import numpy as np
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+from benchmarks.common import dist_setup, emit, teardown, timed_steps  # noqa: E402
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--subjects", type=int, default=5)
+    ap.add_argument("--voxels", type=int, default=200)
+    ap.add_argument("--trs", type=int, default=50)
+    ap.add_argument("--features", type=int, default=10)
+    ap.add_argument("--n-iter", type=int, default=10)
+    args = ap.parse_args()
+
+    rank, world, device, _ = dist_setup()
+    from brainiak_amd.funcalign.srm import SRM
+
+    rng = np.random.RandomState(0)
+    S = rng.randn(args.features, args.trs)
+    data = []
+    for _ in range(args.subjects):
+        q, _ = np.linalg.qr(rng.randn(args.voxels, args.features))
+        data.append(q @ S + 0.1 * rng.randn(args.voxels, args.trs))
+
+    def step(i):
+        SRM(n_iter=args.n_iter, features=args.features, rand_seed=i,
+            device="cpu").fit(data)
+
+    elapsed = timed_steps(step, args.steps, args.warmup, world, device)
+    fits_per_sec = args.steps / elapsed
+    emit(rank, "srm_fits_per_sec", fits_per_sec * world, "fits/s", world,
+         args.steps, args.warmup, elapsed, True, "weak", "fp64",
+         {"model": "probabilistic_srm", "subjects": args.subjects,
+          "voxels": args.voxels, "trs": args.trs,
+          "features": args.features, "n_iter": args.n_iter,
+          "global_batch": args.subjects, "seq_len": args.trs,
+          "parallelism": f"subject-sharded dp{world}"})
+    teardown(world)
+
+
+if __name__ == "__main__":
+    main()

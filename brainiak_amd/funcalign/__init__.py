@@ -6,7 +6,9 @@ Citation context (same methods as the reference implements):
 Datasets", IEEE Big Data 2016.
 """
 
+from .fastsrm import FastSRM  # noqa: F401
 from .rsrm import RSRM  # noqa: F401
 from .srm import SRM, DetSRM, load  # noqa: F401
+from .sssrm import SSSRM  # noqa: F401
 
-__all__ = ["RSRM", "SRM", "DetSRM", "load"]
+__all__ = ["FastSRM", "RSRM", "SRM", "SSSRM", "DetSRM", "load"]
