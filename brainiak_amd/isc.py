@@ -213,8 +213,10 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
     import torch
 
     dev = torch.device(device) if device is not None else comm.device
-    local = [torch.as_tensor(np.ascontiguousarray(d),
-                             dtype=torch.float32).to(dev)
+    local = [d.to(device=dev, dtype=torch.float32)
+             if isinstance(d, torch.Tensor)
+             else torch.as_tensor(np.ascontiguousarray(d),
+                                  dtype=torch.float32).to(dev)
              for d in local_data]
     V = local[0].shape[1] if local else 0
     T = local[0].shape[0] if local else 0
