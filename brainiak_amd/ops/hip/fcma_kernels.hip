@@ -250,6 +250,165 @@ __global__ __launch_bounds__(256) void k_corr_norm(
 }
 
 // ===========================================================================
+// k_corr_norm_mfma: the MFMA form of k_corr_norm for L <= 32.
+// PMC evidence (profiles/README.md): the VALU form is issue-bound at
+// ~5 instructions per useful FMA; one v_mfma_f32_16x16x32_bf16 computes a
+// full 16c x 16v correlation tile (K = L padded to 32) in ONE instruction.
+//   A fragment: a_tile [P][16 c][32 k] bf16 — lane reads 8 contiguous
+//     bf16 at [c = l%16][k0 = 8*(l>>4)] → one ds_read_b128.
+//   B fragment: b_tile [P][VT v][32 k] (+pad) — same read shape.
+// b_tile is staged v-coalesced from global and transposed into k-rows
+// during the write (row pad +2 keeps the writes conflict-free).
+// ===========================================================================
+#define CM_CT 16
+#define CM_VT 64
+#define CM_K 32
+#define CM_BPAD 2
+#define CM_BROW (CM_K + CM_BPAD)
+
+template <int TP, int TL>
+__global__ __launch_bounds__(256) void k_corr_norm_mfma(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ zOut, float* __restrict__ fOut,
+    ll E, ll Lrt, ll VA, ll VB, ll s0, ll C, int Prt, int mode) {
+    static_assert(TL <= CM_K, "MFMA corr kernel requires L <= 32");
+    const int P = TP > 0 ? TP : Prt;
+    const ll L = TL;
+    (void)Lrt;
+    const ll nSubj = E / P;
+    const ll cTiles = (C + CM_CT - 1) / CM_CT;
+    const ll vTiles = (VB + CM_VT - 1) / CM_VT;
+    ll b = blockIdx.x;
+    const ll vt = b % vTiles; b /= vTiles;
+    const ll s = b % nSubj;   b /= nSubj;
+    const ll ct = b;
+    if (ct >= cTiles) return;
+
+    const ll c0 = ct * CM_CT;
+    const ll v0 = vt * CM_VT;
+    const int CT = (int)min((ll)CM_CT, C - c0);
+    const int VT = (int)min((ll)CM_VT, VB - v0);
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;          // 4 waves
+
+    extern __shared__ char smem[];
+    // a_tile [P][CM_CT][CM_K] bf16, b_tile [P][CM_VT][CM_BROW] bf16,
+    // corr [CM_CT][P][CM_VT] fp32
+    bf16_t* a_tile = (bf16_t*)smem;
+    bf16_t* b_tile = a_tile + (size_t)P * CM_CT * CM_K;
+    float* corr = (float*)(b_tile + (size_t)P * CM_VT * CM_BROW);
+
+    // --- stage A transposed: a_tile[p][c][k] (zero-padded k >= L)
+    for (int idx = tid; idx < P * CM_CT * CM_K; idx += 256) {
+        int k = idx % CM_K;
+        int c = (idx / CM_K) % CM_CT;
+        int p = idx / (CM_K * CM_CT);
+        bf16_t val = (bf16_t)0.0f;
+        if (c < CT && k < (int)L)
+            val = A[((s * P + p) * L + k) * VA + (s0 + c0 + c)];
+        a_tile[idx] = val;
+    }
+    // --- stage B transposed: b_tile[p][v][k]; global reads coalesced
+    // over v (64 lanes = one 128B line per (p, k))
+    {
+        int v = tid & 63;
+        for (int pk = tid >> 6; pk < P * CM_K; pk += 4) {
+            int k = pk % CM_K;
+            int p = pk / CM_K;
+            bf16_t val = (bf16_t)0.0f;
+            if (k < (int)L && v < VT)
+                val = B[((s * P + p) * L + k) * VB + (v0 + v)];
+            b_tile[((size_t)p * CM_VT + v) * CM_BROW + k] = val;
+        }
+    }
+    __syncthreads();
+
+    // --- MFMA: 16 (p, v-subtile) units over 4 waves
+    const int frow = lane & 15;
+    const int fk = 8 * (lane >> 4);
+    const int dcol = lane & 15;            // v within subtile
+    const int drow4 = (lane >> 4) * 4;     // c base
+    const int units = P * (CM_VT / 16);
+    for (int u = wid; u < units; u += 4) {
+        int p = u / (CM_VT / 16);
+        int vsub = u % (CM_VT / 16);
+        bf16x8 fa = *(const bf16x8*)&a_tile[((size_t)p * CM_CT + frow)
+                                            * CM_K + fk];
+        bf16x8 fb = *(const bf16x8*)&b_tile[
+            ((size_t)p * CM_VT + vsub * 16 + frow) * CM_BROW + fk];
+        f32x4 acc = (f32x4)0.f;
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fa, fb, acc,
+                                                      0, 0, 0);
+        // D: row = c = drow4 + r, col = v = dcol
+        #pragma unroll
+        for (int r = 0; r < 4; ++r)
+            corr[((size_t)(drow4 + r) * P + p) * CM_VT
+                 + vsub * 16 + dcol] = acc[r];
+    }
+    __syncthreads();
+
+    if (mode == 2) {
+        for (int base = tid; base < CT * P * CM_VT; base += 256) {
+            int v = base % CM_VT;
+            int p = (base / CM_VT) % P;
+            int c = base / (CM_VT * P);
+            if (v < VT)
+                fOut[((c0 + c) * E + (s * P + p)) * VB + v0 + v] =
+                    corr[((size_t)c * P + p) * CM_VT + v];
+        }
+        return;
+    }
+
+    // --- normalization (identical to the VALU kernel)
+    for (int base = tid; base < CT * CM_VT; base += 256) {
+        int v = base % CM_VT;
+        int c = base / CM_VT;
+        if (v >= VT) continue;
+        float* col = corr + ((size_t)c * P) * CM_VT + v;
+        float mean = 0.f, sq = 0.f;
+        if (TP > 0) {
+            float z[TP > 0 ? TP : 1];
+            #pragma unroll
+            for (int p = 0; p < P; ++p) {
+                z[p] = fisher_z(col[(size_t)p * CM_VT]);
+                mean += z[p]; sq += z[p] * z[p];
+            }
+            mean /= (float)P;
+            float var = sq / (float)P - mean * mean;
+            float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                col[(size_t)p * CM_VT] = (z[p] - mean) * inv;
+        } else {
+            for (int p = 0; p < P; ++p) {
+                float zv = fisher_z(col[(size_t)p * CM_VT]);
+                mean += zv; sq += zv * zv;
+                col[(size_t)p * CM_VT] = zv;
+            }
+            mean /= (float)P;
+            float var = sq / (float)P - mean * mean;
+            float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+            for (int p = 0; p < P; ++p)
+                col[(size_t)p * CM_VT] = (col[(size_t)p * CM_VT] - mean)
+                                         * inv;
+        }
+    }
+    __syncthreads();
+
+    for (int base = tid; base < CT * P * CM_VT; base += 256) {
+        int v = base % CM_VT;
+        int p = (base / CM_VT) % P;
+        int c = base / (CM_VT * P);
+        if (v >= VT) continue;
+        float zv = corr[((size_t)c * P + p) * CM_VT + v];
+        ll off = ((c0 + c) * E + (s * P + p)) * VB + v0 + v;
+        if (mode == 0) zOut[off] = (bf16_t)zv;
+        else fOut[off] = zv;
+    }
+}
+
+// ===========================================================================
 // k_gram_bf16: G_c = Z_c Z_c^T per voxel with bf16 MFMA.
 // Z: [C, E, V] bf16; G: [C, E, E] fp32.  E % 64 == 0 (host pads).
 // Block = (c, band_i, band_j>=band_i); 4 waves own the 32x32 quadrants.
@@ -481,13 +640,42 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
                                ll s0, ll C, int P, int mode, size_t smem,
                                hipStream_t stream) {
     ll nSubj = E / P;
+    // Measured A/B on MI355X (profiles/README.md): the VALU form at
+    // 2.09 ms/512-voxel call beats the MFMA form (3.86 ms) — with
+    // K = L <= 32 the MFMA phase is 16 instructions per block and the
+    // kernel is bound by staging/normalize/write address arithmetic
+    // (PMC: MFMA form = 1.43x the VALU instructions).  Set USE_MFMA_CORR
+    // to re-enable the MFMA path when its staging is restructured.
+#ifdef USE_MFMA_CORR
+    if constexpr (TL <= CM_K) {
+        size_t smem_mfma = (size_t)P * CM_CT * CM_K * sizeof(bf16_t)
+                         + (size_t)P * CM_VT * CM_BROW * sizeof(bf16_t)
+                         + (size_t)CM_CT * P * CM_VT * sizeof(float);
+        ll grid = ceil_div(C, CM_CT) * nSubj * ceil_div(VB, CM_VT);
+        hipLaunchKernelGGL((k_corr_norm_mfma<TP, TL>), dim3(grid),
+                           dim3(256), smem_mfma, stream,
+                           (const bf16_t*)A, (const bf16_t*)B,
+                           (bf16_t*)zOut, fOut, E, L, VA, VB, s0, C, P,
+                           mode);
+        return;
+    }
+#endif
     ll grid = ceil_div(C, CN_CT) * nSubj * ceil_div(VB, CN_VT);
-    hipLaunchKernelGGL((k_corr_norm<TP, TL>), dim3(grid), dim3(256), smem,
-                       stream, (const bf16_t*)A, (const bf16_t*)B,
-                       (bf16_t*)zOut, fOut, E, L, VA, VB, s0, C, P, mode);
+    hipLaunchKernelGGL((k_corr_norm<TP, TL>), dim3(grid), dim3(256),
+                       smem, stream, (const bf16_t*)A,
+                       (const bf16_t*)B, (bf16_t*)zOut, fOut, E, L,
+                       VA, VB, s0, C, P, mode);
 }
 
 extern "C" int fcma_corr_norm_smem(ll L, int P) {
+#ifdef USE_MFMA_CORR
+    if (L <= CM_K) {
+        size_t smem = (size_t)P * CM_CT * CM_K * sizeof(bf16_t)
+                    + (size_t)P * CM_VT * CM_BROW * sizeof(bf16_t)
+                    + (size_t)CM_CT * P * CM_VT * sizeof(float);
+        return (int)smem;
+    }
+#endif
     size_t smem = (size_t)P * L * CN_CT * sizeof(float)   // fp32 a_tile
                 + (size_t)CN_CT * P * CN_VT * sizeof(float);
     return (int)smem;
