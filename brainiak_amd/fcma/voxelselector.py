@@ -53,7 +53,7 @@ class VoxelSelector:
     """
 
     def __init__(self, labels, epochs_per_subj, num_folds, raw_data,
-                 raw_data2=None, voxel_unit=64, process_num=4,
+                 raw_data2=None, voxel_unit=None, process_num=4,
                  master_rank=0, comm=None, device=None,
                  use_gpu_svm: Optional[bool] = None):
         self.labels = np.asarray(labels)
@@ -61,6 +61,11 @@ class VoxelSelector:
         self.num_folds = num_folds
         self.raw_data = raw_data
         self.raw_data2 = raw_data2
+        if voxel_unit is None:
+            # GPU chunks amortize kernel launches; 64 is the reference's
+            # CPU task size (voxelselector.py:89)
+            import torch as _torch
+            voxel_unit = 1024 if _torch.cuda.is_available() else 64
         self.num_voxels = raw_data[0].shape[1]
         self.num_voxels2 = (raw_data2[0].shape[1] if raw_data2 is not None
                             else self.num_voxels)
