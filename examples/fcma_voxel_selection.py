@@ -8,9 +8,13 @@ Single process uses one GPU (or CPU); scale to 8 GPUs with:
 ref docs/examples/fcma/FCMA_script/fcma_voxel_selection_cv.py).
 """
 
+import sys
+from pathlib import Path
+
 import numpy as np
 from sklearn import svm
 
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 from brainiak_amd.fcma import VoxelSelector
 from brainiak_amd.fcma.preprocessing import _separate_epochs
 from brainiak_amd.parallel import DistContext

@@ -6,8 +6,12 @@ Distributed across GPUs with rank-cyclic subject ownership:
         examples/srm_alignment.py
 """
 
+import sys
+from pathlib import Path
+
 import numpy as np
 
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 from brainiak_amd.funcalign import SRM
 from brainiak_amd.parallel import DistContext
 
