@@ -46,7 +46,9 @@ __device__ __forceinline__ float fisher_z(float r) {
     float den = 1.0f - r;
     num = (num <= 0.0f) ? 1e-4f : num;   // reference clamp semantics
     den = (den <= 0.0f) ? 1e-4f : den;
-    return 0.5f * __logf(num / den);
+    // log(num/den) = log(num) - log(den): two v_log_f32 instead of a
+    // full-precision divide chain (v_div_scale/v_rcp, ~10 VALU)
+    return 0.5f * (__logf(num) - __logf(den));
 }
 
 // ===========================================================================
