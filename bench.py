@@ -103,16 +103,16 @@ def main():
 
     def one_step(step_idx):
         start0 = (step_idx * vps) % V
-        grams = []
+        chunks = []
         done = 0
         while done < vps:
             count = min(chunk, vps - done)
             start = (start0 + done) % V
             count = min(count, V - start)
-            grams.append(pipeline.chunk_kernel_matrices(start, count))
+            chunks.append((start, count))
             done += count
+        kernels = pipeline.pipelined_kernel_matrices(chunks)
         if not args.no_cv:
-            kernels = torch.cat(grams, dim=0)
             if device.type == "cuda":
                 accs = cross_validate_voxels(kernels, labels,
                                              args.num_folds)
@@ -120,7 +120,7 @@ def main():
                 accs = cross_validate_voxels(kernels[:8], labels,
                                              args.num_folds)
             return float(np.mean(accs))
-        return float(grams[0].float().mean())
+        return float(kernels[0].float().mean())
 
     def barrier_sync():
         if distributed:

@@ -108,14 +108,34 @@ class VoxelSelector:
 
         my = self.ctx.shard(self.num_voxels)
         results = []
-        start = my.start
-        while start < my.stop:
-            count = min(self.voxel_unit, my.stop - start)
-            scores = self._score_chunk(pipeline, clf, start, count,
-                                       precomputed, C, tol, use_gpu_svm)
-            results.extend(
-                (start + i, float(scores[i])) for i in range(count))
-            start += count
+        if precomputed and pipeline.device.type == "cuda":
+            # whole-shard path: stream-pipelined corr+norm/Gram kernels,
+            # then ONE batched CV launch over every scored voxel
+            chunks = []
+            start = my.start
+            while start < my.stop:
+                count = min(self.voxel_unit, my.stop - start)
+                chunks.append((start, count))
+                start += count
+            kernels = pipeline.pipelined_kernel_matrices(chunks)
+            if use_gpu_svm:
+                scores = cross_validate_voxels(kernels, self.labels,
+                                               self.num_folds, C=C,
+                                               tol=tol)
+            else:
+                scores = self._sklearn_cv(clf, kernels.cpu().numpy())
+            results.extend((my.start + i, float(scores[i]))
+                           for i in range(my.stop - my.start))
+        else:
+            start = my.start
+            while start < my.stop:
+                count = min(self.voxel_unit, my.stop - start)
+                scores = self._score_chunk(pipeline, clf, start, count,
+                                           precomputed, C, tol,
+                                           use_gpu_svm)
+                results.extend(
+                    (start + i, float(scores[i])) for i in range(count))
+                start += count
         logger.info('rank %d scored voxels [%d, %d)', self.ctx.rank,
                     my.start, my.stop)
 
