@@ -40,9 +40,9 @@ def parse_args():
     ap.add_argument("--subjects", type=int, default=16)
     ap.add_argument("--epochs-per-subj", type=int, default=4)
     ap.add_argument("--epoch-len", type=int, default=12)
-    ap.add_argument("--voxels-per-step", type=int, default=2048)
+    ap.add_argument("--voxels-per-step", type=int, default=4096)
     ap.add_argument("--num-folds", type=int, default=4)
-    ap.add_argument("--chunk", type=int, default=512,
+    ap.add_argument("--chunk", type=int, default=1024,
                     help="pipeline chunk (voxels per kernel pass)")
     ap.add_argument("--no-cv", action="store_true",
                     help="skip the SVM CV stage (pipeline only)")
