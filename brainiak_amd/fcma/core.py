@@ -197,32 +197,49 @@ class CorrelationPipeline:
         E = self.num_epochs
         Epad = ((E + 63) // 64) * 64
         ext = ops.load_extension()
-        corr_stream = torch.cuda.Stream(device=self.device)
-        gram_stream = torch.cuda.Stream(device=self.device)
+        max_count = max(c for _, c in chunks)
+        # persistent double-buffered Z workspace: repeated multi-GB
+        # allocations churn the caching allocator across streams
+        if getattr(self, "_zbuf", None) is None or \
+                self._zbuf[0].shape[0] < max_count or \
+                self._zbuf[0].shape[1] != Epad:
+            self._zbuf = [
+                torch.zeros((max_count, Epad, self.num_voxels2),
+                            dtype=torch.bfloat16, device=self.device)
+                for _ in range(2)]
+        if getattr(self, "_streams", None) is None:
+            self._streams = (torch.cuda.Stream(device=self.device),
+                             torch.cuda.Stream(device=self.device))
+        corr_stream, gram_stream = self._streams
         grams = []
-        pending = None   # (z, event) — at most two Z buffers live
+        pending = None          # (z, ready-event, buffer index)
+        buf_free = [None, None]  # event: gram done reading buffer b
 
-        def _consume(z, ev):
+        def _consume(z, ev, bidx):
             with torch.cuda.stream(gram_stream):
                 gram_stream.wait_event(ev)
                 g = ops.fcma_gram_bf16(z)
                 if Epad != E:
                     g = g[:, :E, :E].contiguous()
-                # z was allocated on corr_stream; tell the caching
-                # allocator it is still in use by gram_stream
-                z.record_stream(gram_stream)
+                done = torch.cuda.Event()
+                done.record(gram_stream)
+                buf_free[bidx] = done
                 grams.append(g)
 
-        for start, count in chunks:
+        for i, (start, count) in enumerate(chunks):
+            bidx = i % 2
             with torch.cuda.stream(corr_stream):
+                # don't overwrite a buffer the gram stream still reads
+                if buf_free[bidx] is not None:
+                    corr_stream.wait_event(buf_free[bidx])
                 z = ext.fcma_corr_norm_z(
                     self.data, self.data2, start, count,
-                    self.epochs_per_subj, Epad)
+                    self.epochs_per_subj, Epad, out=self._zbuf[bidx])
                 ev = torch.cuda.Event()
                 ev.record(corr_stream)
             if pending is not None:
                 _consume(*pending)
-            pending = (z, ev)
+            pending = (z, ev, bidx)
         _consume(*pending)
 
         cur = torch.cuda.current_stream(self.device)

@@ -13,7 +13,7 @@ typedef long long ll;
 extern "C" {
 void launch_fcma_normalize(float*, ll, ll, ll, int, void*);
 void launch_fcma_corr_norm(const void*, const void*, void*, float*, ll, ll,
-                           ll, ll, ll, ll, int, int, void*);
+                           ll, ll, ll, ll, int, int, ll, void*);
 int fcma_corr_norm_smem(ll, int);
 void launch_fcma_gram_bf16(const void*, float*, ll, ll, ll, void*);
 void launch_fcma_gram_f32(const float*, float*, ll, ll, ll, void*);
@@ -71,13 +71,14 @@ torch::Tensor fcma_correlate(torch::Tensor A, torch::Tensor B,
                             A.options().dtype(torch::kFloat32));
     launch_fcma_corr_norm(A.data_ptr(), B.data_ptr(), nullptr,
                           out.data_ptr<float>(), E, L, VA, VB, start,
-                          count, P, /*mode=*/2, cur_stream());
+                          count, P, /*mode=*/2, E, cur_stream());
     return out;
 }
 
 torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
                                int64_t start, int64_t count, int64_t P,
-                               int64_t padE) {
+                               int64_t padE,
+                               c10::optional<torch::Tensor> out_opt) {
     check_3d(A, torch::kBFloat16, "A");
     check_3d(B, torch::kBFloat16, "B");
     ll E = A.size(0), L = A.size(1), VA = A.size(2), VB = B.size(2);
@@ -90,25 +91,26 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
     TORCH_CHECK(fcma_supported_L(L) == L,
                 "epoch length must be padded to one of {8,16,24,32,40}");
     ll Eout = std::max((ll)padE, E);
-    auto Z = (Eout == E)
-        ? torch::empty({count, E, VB}, A.options())
-        : torch::zeros({count, Eout, VB}, A.options());
-    // write into the leading E rows; kernel indexes output by its own E
-    // so for padded output we pass stride via a narrow view being
-    // contiguous only when Eout == E — handle by writing into a separate
-    // tensor then copying.  Simpler: kernel writes [count, E, VB] and we
-    // zero-pad afterwards.
-    if (Eout == E) {
-        launch_fcma_corr_norm(A.data_ptr(), B.data_ptr(), Z.data_ptr(),
-                              nullptr, E, L, VA, VB, start, count, (int)P,
-                              /*mode=*/0, cur_stream());
+    torch::Tensor Z;
+    if (out_opt.has_value()) {
+        // caller-provided persistent buffer (rows >= E; padding rows
+        // must be pre-zeroed once by the caller)
+        Z = out_opt.value();
+        TORCH_CHECK(Z.is_cuda() && Z.is_contiguous()
+                    && Z.scalar_type() == torch::kBFloat16
+                    && Z.size(0) >= count && Z.size(1) >= Eout
+                    && Z.size(2) == VB, "bad out buffer");
+        Eout = Z.size(1);
+        if (Z.size(0) != count)
+            Z = Z.narrow(0, 0, count);
     } else {
-        auto Ztmp = torch::empty({count, E, VB}, A.options());
-        launch_fcma_corr_norm(A.data_ptr(), B.data_ptr(), Ztmp.data_ptr(),
-                              nullptr, E, L, VA, VB, start, count, (int)P,
-                              /*mode=*/0, cur_stream());
-        Z.narrow(1, 0, E).copy_(Ztmp);
+        Z = (Eout == E) ? torch::empty({count, E, VB}, A.options())
+                        : torch::zeros({count, Eout, VB}, A.options());
     }
+    // the kernel writes rows [0, E) with row stride Eout directly
+    launch_fcma_corr_norm(A.data_ptr(), B.data_ptr(), Z.data_ptr(),
+                          nullptr, E, L, VA, VB, start, count, (int)P,
+                          /*mode=*/0, Eout, cur_stream());
     return Z;
 }
 
@@ -144,7 +146,7 @@ torch::Tensor fcma_fused_gram(torch::Tensor A, torch::Tensor B,
                               int64_t start, int64_t count, int64_t P) {
     ll E = A.size(0);
     ll Epad = ((E + 63) / 64) * 64;
-    auto Z = fcma_corr_norm_z(A, B, start, count, P, Epad);
+    auto Z = fcma_corr_norm_z(A, B, start, count, P, Epad, c10::nullopt);
     auto G = fcma_gram_bf16(Z);
     if (Epad != E)
         return G.narrow(1, 0, E).narrow(2, 0, E).contiguous();
@@ -263,7 +265,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fcma_correlate", &fcma_correlate,
           "raw chunk correlation [count,E,VB] fp32");
     m.def("fcma_corr_norm_z", &fcma_corr_norm_z,
-          "fused corr+norm -> bf16 Z");
+          "fused corr+norm -> bf16 Z",
+          pybind11::arg("A"), pybind11::arg("B"), pybind11::arg("start"),
+          pybind11::arg("count"), pybind11::arg("P"),
+          pybind11::arg("padE"),
+          pybind11::arg("out") = pybind11::none());
     m.def("fcma_gram", &fcma_gram, "per-voxel Gram from fp32 [C,E,V]");
     m.def("fcma_gram_bf16", &fcma_gram_bf16,
           "per-voxel Gram from bf16 Z [C,E,V]");

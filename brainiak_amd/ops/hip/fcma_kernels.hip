@@ -119,7 +119,8 @@ template <int TP, int TL>
 __global__ __launch_bounds__(256) void k_corr_norm(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ zOut, float* __restrict__ fOut,
-    ll E, ll Lrt, ll VA, ll VB, ll s0, ll C, int Prt, int mode) {
+    ll E, ll Lrt, ll VA, ll VB, ll s0, ll C, int Prt, int mode,
+    ll zstride) {
     const int P = TP > 0 ? TP : Prt;
     const ll L = TL;
     (void)Lrt;
@@ -245,9 +246,11 @@ __global__ __launch_bounds__(256) void k_corr_norm(
         int c = base / (CN_VT * P);
         if (v >= VT) continue;
         float zv = corr[((size_t)c * P + p) * CN_VT + v];
-        ll off = ((c0 + c) * E + (s * P + p)) * VB + v0 + v;
-        if (mode == 0) zOut[off] = (bf16_t)zv;
-        else fOut[off] = zv;
+        if (mode == 0)
+            zOut[((c0 + c) * zstride + (s * P + p)) * VB + v0 + v] =
+                (bf16_t)zv;
+        else
+            fOut[((c0 + c) * E + (s * P + p)) * VB + v0 + v] = zv;
     }
 }
 
@@ -272,7 +275,8 @@ template <int TP, int TL>
 __global__ __launch_bounds__(256) void k_corr_norm_mfma(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ zOut, float* __restrict__ fOut,
-    ll E, ll Lrt, ll VA, ll VB, ll s0, ll C, int Prt, int mode) {
+    ll E, ll Lrt, ll VA, ll VB, ll s0, ll C, int Prt, int mode,
+    ll zstride) {
     static_assert(TL <= CM_K, "MFMA corr kernel requires L <= 32");
     const int P = TP > 0 ? TP : Prt;
     const ll L = TL;
@@ -404,9 +408,11 @@ __global__ __launch_bounds__(256) void k_corr_norm_mfma(
         int c = base / (CM_VT * P);
         if (v >= VT) continue;
         float zv = corr[((size_t)c * P + p) * CM_VT + v];
-        ll off = ((c0 + c) * E + (s * P + p)) * VB + v0 + v;
-        if (mode == 0) zOut[off] = (bf16_t)zv;
-        else fOut[off] = zv;
+        if (mode == 0)
+            zOut[((c0 + c) * zstride + (s * P + p)) * VB + v0 + v] =
+                (bf16_t)zv;
+        else
+            fOut[((c0 + c) * E + (s * P + p)) * VB + v0 + v] = zv;
     }
 }
 
@@ -415,7 +421,7 @@ __global__ __launch_bounds__(256) void k_corr_norm_mfma(
 // Z: [C, E, V] bf16; G: [C, E, E] fp32.  E % 64 == 0 (host pads).
 // Block = (c, band_i, band_j>=band_i); 4 waves own the 32x32 quadrants.
 // ===========================================================================
-#define GR_KT 32
+#define GR_KT 64
 #define GR_PAD 8
 #define GR_ROW (GR_KT + GR_PAD)
 
@@ -435,8 +441,8 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
     const int wr = (w >> 1) * 32;
     const int wc = (w & 1) * 32;
 
-    __shared__ bf16_t zi[2][64][GR_ROW];
-    __shared__ bf16_t zj[2][64][GR_ROW];
+    __shared__ bf16_t zi[64][GR_ROW];
+    __shared__ bf16_t zj[64][GR_ROW];
     const bool diag = (band_i == band_j);
     const bf16_t* Zc = Z + c * E * V;
     const ll rows_i = band_i * 64;
@@ -445,55 +451,68 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
     f32x4 acc00 = (f32x4)0.f, acc01 = (f32x4)0.f;
     f32x4 acc10 = (f32x4)0.f, acc11 = (f32x4)0.f;
 
-    auto load_tile = [&](bf16_t dst[64][GR_ROW], const bf16_t* src,
-                         ll rows0, ll k0) {
-        int row = tid >> 2;
-        int col = (tid & 3) * 8;
-        const bf16_t* s = src + (rows0 + row) * V + k0 + col;
-        if (k0 + col + 8 <= V && (((uintptr_t)s) & 3) == 0) {
-            // one 16-B vector load per thread (coalesced 128 B per
-            // quarter-wave) instead of 8 scalar bf16 loads; rows are
-            // only 4B-aligned, hence the aligned(4) vector type
-            bf16x8 v = (bf16x8)(*(const bf16x8_u*)s);
-            *(bf16x8*)&dst[row][col] = v;
-        } else {
-            bf16_t tmp[8];
-            #pragma unroll
-            for (int j = 0; j < 8; ++j)
-                tmp[j] = (k0 + col + j < V) ? s[j] : (bf16_t)0.0f;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) dst[row][col + j] = tmp[j];
+    // T14 issue-early/write-late staging: each thread holds the NEXT
+    // K-tile (2x bf16x8 = 32 B) in registers while MFMAs consume the
+    // CURRENT tile from LDS, so the HBM latency hides under compute
+    // instead of draining at the barrier.
+    const int srow = tid >> 2;            // 64 rows, 4 threads each
+    const int scol = (tid & 3) * 16;      // 2 vector loads per thread
+
+    auto issue_loads = [&](const bf16_t* src, ll rows0, ll k0,
+                           bf16x8 regs[2]) {
+        #pragma unroll
+        for (int h = 0; h < 2; ++h) {
+            ll kk = k0 + scol + 8 * h;
+            const bf16_t* s = src + (rows0 + srow) * V + kk;
+            if (kk + 8 <= V && (((uintptr_t)s) & 3) == 0) {
+                regs[h] = (bf16x8)(*(const bf16x8_u*)s);
+            } else {
+                bf16_t tmp[8];
+                #pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    tmp[j] = (kk + j < V) ? s[j] : (bf16_t)0.0f;
+                regs[h] = *(const bf16x8*)tmp;
+            }
         }
+    };
+    auto write_tile = [&](bf16_t dst[64][GR_ROW], bf16x8 regs[2]) {
+        #pragma unroll
+        for (int h = 0; h < 2; ++h)
+            *(bf16x8*)&dst[srow][scol + 8 * h] = regs[h];
     };
 
     const ll kTiles = (V + GR_KT - 1) / GR_KT;
-    int cur = 0;
-    load_tile(zi[0], Zc, rows_i, 0);
-    if (!diag) load_tile(zj[0], Zc, rows_j, 0);
-    __syncthreads();
+    bf16x8 ri[2], rj[2];
+    issue_loads(Zc, rows_i, 0, ri);
+    if (!diag) issue_loads(Zc, rows_j, 0, rj);
 
     for (ll kt = 0; kt < kTiles; ++kt) {
-        if (kt + 1 < kTiles) {
-            load_tile(zi[cur ^ 1], Zc, rows_i, (kt + 1) * GR_KT);
-            if (!diag) load_tile(zj[cur ^ 1], Zc, rows_j, (kt + 1) * GR_KT);
+        __syncthreads();              // previous tile's reads complete
+        write_tile(zi, ri);
+        if (!diag) write_tile(zj, rj);
+        if (kt + 1 < kTiles) {        // issue next tile early
+            issue_loads(Zc, rows_i, (kt + 1) * GR_KT, ri);
+            if (!diag) issue_loads(Zc, rows_j, (kt + 1) * GR_KT, rj);
         }
+        __syncthreads();              // tile visible
         const int frow = lane & 15;
-        const int fk = 8 * (lane >> 4);
-        bf16x8 fi0 = *(const bf16x8*)&zi[cur][wr + frow][fk];
-        bf16x8 fi1 = *(const bf16x8*)&zi[cur][wr + 16 + frow][fk];
-        const bf16_t (*zjs)[GR_ROW] = diag ? zi[cur] : zj[cur];
-        bf16x8 fj0 = *(const bf16x8*)&zjs[wc + frow][fk];
-        bf16x8 fj1 = *(const bf16x8*)&zjs[wc + 16 + frow][fk];
-        acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi0, fj0, acc00,
-                                                        0, 0, 0);
-        acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi0, fj1, acc01,
-                                                        0, 0, 0);
-        acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi1, fj0, acc10,
-                                                        0, 0, 0);
-        acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi1, fj1, acc11,
-                                                        0, 0, 0);
-        __syncthreads();
-        cur ^= 1;
+        #pragma unroll
+        for (int ks = 0; ks < GR_KT / 32; ++ks) {
+            const int fk = 8 * (lane >> 4) + 32 * ks;
+            bf16x8 fi0 = *(const bf16x8*)&zi[wr + frow][fk];
+            bf16x8 fi1 = *(const bf16x8*)&zi[wr + 16 + frow][fk];
+            const bf16_t (*zjs)[GR_ROW] = diag ? zi : zj;
+            bf16x8 fj0 = *(const bf16x8*)&zjs[wc + frow][fk];
+            bf16x8 fj1 = *(const bf16x8*)&zjs[wc + 16 + frow][fk];
+            acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi0, fj0,
+                                                            acc00, 0, 0, 0);
+            acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi0, fj1,
+                                                            acc01, 0, 0, 0);
+            acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi1, fj0,
+                                                            acc10, 0, 0, 0);
+            acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(fi1, fj1,
+                                                            acc11, 0, 0, 0);
+        }
     }
 
     float* Gc = G + c * E * E;
@@ -639,8 +658,8 @@ extern "C" void launch_fcma_normalize(float* corr, ll C, ll E, ll V, int P,
 template <int TP, int TL>
 static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
                                float* fOut, ll E, ll L, ll VA, ll VB,
-                               ll s0, ll C, int P, int mode, size_t smem,
-                               hipStream_t stream) {
+                               ll s0, ll C, int P, int mode, ll zstride,
+                               size_t smem, hipStream_t stream) {
     ll nSubj = E / P;
     // Measured A/B on MI355X (profiles/README.md): the VALU form at
     // 2.09 ms/512-voxel call beats the MFMA form (3.86 ms) — with
@@ -658,7 +677,7 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
                            dim3(256), smem_mfma, stream,
                            (const bf16_t*)A, (const bf16_t*)B,
                            (bf16_t*)zOut, fOut, E, L, VA, VB, s0, C, P,
-                           mode);
+                           mode, zstride);
         return;
     }
 #endif
@@ -666,7 +685,7 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
     hipLaunchKernelGGL((k_corr_norm<TP, TL>), dim3(grid), dim3(256),
                        smem, stream, (const bf16_t*)A,
                        (const bf16_t*)B, (bf16_t*)zOut, fOut, E, L,
-                       VA, VB, s0, C, P, mode);
+                       VA, VB, s0, C, P, mode, zstride);
 }
 
 extern "C" int fcma_corr_norm_smem(ll L, int P) {
@@ -686,22 +705,28 @@ extern "C" int fcma_corr_norm_smem(ll L, int P) {
 template <int TL>
 static void dispatch_p(const void* A, const void* B, void* zOut,
                        float* fOut, ll E, ll L, ll VA, ll VB, ll s0, ll C,
-                       int P, int mode, size_t smem, hipStream_t stream) {
+                       int P, int mode, ll zstride, size_t smem,
+                       hipStream_t stream) {
     switch (P) {
         case 2:  launch_corr_norm_t<2, TL>(A, B, zOut, fOut, E, L, VA, VB,
-                                           s0, C, P, mode, smem, stream);
+                                           s0, C, P, mode, zstride,
+                                           smem, stream);
                  break;
         case 4:  launch_corr_norm_t<4, TL>(A, B, zOut, fOut, E, L, VA, VB,
-                                           s0, C, P, mode, smem, stream);
+                                           s0, C, P, mode, zstride,
+                                           smem, stream);
                  break;
         case 8:  launch_corr_norm_t<8, TL>(A, B, zOut, fOut, E, L, VA, VB,
-                                           s0, C, P, mode, smem, stream);
+                                           s0, C, P, mode, zstride,
+                                           smem, stream);
                  break;
         case 16: launch_corr_norm_t<16, TL>(A, B, zOut, fOut, E, L, VA, VB,
-                                            s0, C, P, mode, smem, stream);
+                                           s0, C, P, mode, zstride,
+                                           smem, stream);
                  break;
         default: launch_corr_norm_t<0, TL>(A, B, zOut, fOut, E, L, VA, VB,
-                                           s0, C, P, mode, smem, stream);
+                                           s0, C, P, mode, zstride,
+                                           smem, stream);
                  break;
     }
 }
@@ -717,19 +742,20 @@ extern "C" ll fcma_supported_L(ll L) {
 extern "C" void launch_fcma_corr_norm(const void* A, const void* B,
                                       void* zOut, float* fOut, ll E, ll L,
                                       ll VA, ll VB, ll s0, ll C, int P,
-                                      int mode, hipStream_t stream) {
+                                      int mode, ll zstride,
+                                      hipStream_t stream) {
     size_t smem = (size_t)fcma_corr_norm_smem(L, P);
     switch (L) {
         case 8:  dispatch_p<8>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                               mode, smem, stream); break;
+                               mode, zstride, smem, stream); break;
         case 16: dispatch_p<16>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                                mode, smem, stream); break;
+                               mode, zstride, smem, stream); break;
         case 24: dispatch_p<24>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                                mode, smem, stream); break;
+                               mode, zstride, smem, stream); break;
         case 32: dispatch_p<32>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                                mode, smem, stream); break;
+                               mode, zstride, smem, stream); break;
         case 40: dispatch_p<40>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                                mode, smem, stream); break;
+                               mode, zstride, smem, stream); break;
         default: break;  // host guarantees L in the supported set
     }
 }
