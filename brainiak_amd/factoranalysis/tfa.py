@@ -260,6 +260,10 @@ class TFA:
         init_estimate = np.hstack((init_centers.ravel(),
                                    init_widths.ravel()))
         data_sigma = 1.0 / math.sqrt(2.0) * np.std(X)
+        # on GPU the residual rides the fp32 HIP factor/recon kernels;
+        # the default 2-point step (~1.5e-8) sits below fp32 noise and
+        # yields garbage Jacobians — use a step well above it
+        diff_step = 1e-3 if self._use_gpu() else None
         final_estimate = least_squares(
             self._residual_multivariate, init_estimate,
             args=(unique_R, inds, X, W, template_centers,
@@ -267,7 +271,7 @@ class TFA:
                   template_widths_mean_var_reci, data_sigma),
             method=self.nlss_method, loss=self.nlss_loss,
             bounds=self.bounds, verbose=0, x_scale=self.x_scale,
-            tr_solver=self.tr_solver)
+            tr_solver=self.tr_solver, diff_step=diff_step)
         return final_estimate.x, final_estimate.cost
 
     # -- convergence ---------------------------------------------------------

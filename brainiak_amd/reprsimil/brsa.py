@@ -31,9 +31,20 @@ the orthogonal complement of X₀.
 spatially-whitened GLS against the posterior-mean betas — the
 reference's forward-backward temporal smoothing of the decoded courses
 is not applied (documented simplification).
+
+GP priors: with ``GP_space=True`` (optionally ``GP_inten=True``) a
+zero-mean Gaussian-Process prior over voxel coordinates (and mean image
+intensity) is imposed on log(SNR) (ref brsa.py:2426-2510): squared-
+exponential kernels with log-parameterized length scales fitted jointly
+by the same L-BFGS pass, GP variance τ² profiled out at its MAP under
+an inverse-Gamma (``prior_GP_var_inv_gamma``) or half-Cauchy
+(``prior_GP_var_half_cauchy``) prior, and half-Cauchy priors on the
+length scales.  Fitted hyper-parameters land in ``lGPspace_``,
+``lGPinten_`` and ``bGP_`` (=τ̂).
 """
 
 import logging
+import math
 
 import numpy as np
 import torch
@@ -44,16 +55,52 @@ from ..utils.utils import cov2corr
 logger = logging.getLogger(__name__)
 
 __all__ = ["BRSA", "GBRSA", "Ncomp_SVHT_MG_DLD_approx",
-           "prior_GP_var_inv_gamma"]
+           "prior_GP_var_inv_gamma", "prior_GP_var_half_cauchy"]
 
 _DT = torch.float64
 
 
 def prior_GP_var_inv_gamma(y_invK_y, n_y, tau_range):
-    """Inverse-Gamma-prior estimate of GP variance (ref brsa.py:70)."""
-    alpha = 1e-4
-    beta = tau_range ** 2 * alpha
-    return (y_invK_y / 2 + beta) / (n_y / 2 + alpha + 1)
+    """MAP estimate of the GP variance τ² under an inverse-Gamma prior
+    τ² ~ invgamma(a=2, scale=tau_range²), plus log p(τ²) at that MAP
+    (ref brsa.py:70-129).  Works on torch scalars (differentiable w.r.t.
+    ``y_invK_y``) and plain floats alike."""
+    alpha = 2.0
+    tau2 = (y_invK_y + 2 * tau_range ** 2) / (alpha * 2 + 2 + n_y)
+    # invgamma.logpdf(x, a, scale=b) = a·log b − lgamma(a) − (a+1)·log x − b/x
+    b = tau_range ** 2
+    if isinstance(tau2, torch.Tensor):
+        log_ptau = (alpha * np.log(b) - math.lgamma(alpha)
+                    - (alpha + 1) * torch.log(tau2) - b / tau2)
+    else:
+        import scipy.stats
+        log_ptau = scipy.stats.invgamma.logpdf(tau2, scale=b, a=alpha)
+    return tau2, log_ptau
+
+
+def prior_GP_var_half_cauchy(y_invK_y, n_y, tau_range):
+    """MAP estimate of τ² under a half-Cauchy prior on τ (scale
+    ``tau_range``), plus log p(τ) at that MAP (ref brsa.py:132-154)."""
+    sqrt = torch.sqrt if isinstance(y_invK_y, torch.Tensor) else np.sqrt
+    tau2 = (y_invK_y - n_y * tau_range ** 2
+            + sqrt(n_y ** 2 * tau_range ** 4 + (2 * n_y + 8)
+                   * tau_range ** 2 * y_invK_y + y_invK_y ** 2)) \
+        / 2 / (n_y + 2)
+    # halfcauchy.logpdf(x, scale=s) = log 2 − log π − log s − log(1 + x²/s²)
+    if isinstance(tau2, torch.Tensor):
+        log_ptau = (np.log(2 / np.pi / tau_range)
+                    - torch.log(1 + tau2 / tau_range ** 2))
+    else:
+        import scipy.stats
+        log_ptau = scipy.stats.halfcauchy.logpdf(tau2 ** 0.5,
+                                                 scale=tau_range)
+    return tau2, log_ptau
+
+
+def _log_halfcauchy(x, scale):
+    """log halfcauchy pdf at torch scalar x (support x > 0)."""
+    return (np.log(2.0 / np.pi) - np.log(scale)
+            - torch.log(1 + x ** 2 / scale ** 2))
 
 
 def Ncomp_SVHT_MG_DLD_approx(X, zscore=True):
@@ -148,7 +195,10 @@ class BRSA(_BRSACore):
                  baseline_single=False, logS_range=1.0, SNR_prior='exp',
                  rho_bins=20, tol=1e-4, optimizer='L-BFGS-B',
                  minimize_options=None, random_state=None,
-                 anneal_speed=10):
+                 anneal_speed=10, GP_space=False, GP_inten=False,
+                 space_smooth_range=None, inten_smooth_range=None,
+                 tau_range=5.0, tau2_prior=prior_GP_var_inv_gamma,
+                 eta=0.0001):
         self.n_iter = n_iter
         self.rank = rank
         self.auto_nuisance = auto_nuisance
@@ -165,14 +215,30 @@ class BRSA(_BRSACore):
             {'maxiter': 200, 'disp': False}
         self.random_state = random_state
         self.anneal_speed = anneal_speed
+        self.GP_space = GP_space
+        self.GP_inten = GP_inten and GP_space
+        self.space_smooth_range = space_smooth_range
+        self.inten_smooth_range = inten_smooth_range
+        self.tau_range = tau_range
+        self.tau2_prior = tau2_prior
+        self.eta = eta
 
     # -- likelihood --------------------------------------------------------
 
     @staticmethod
-    def _neg_loglik(params, quadX, quadXY, quadYY, C, V, T, rank):
+    def _neg_loglik(params, quadX, quadXY, quadYY, C, V, T, rank,
+                    gp=None, tau_range=5.0):
         """Negative marginal log-likelihood, σ² profiled out.
 
-        params = [L_flat (C*rank), log_snr (V), rho_unc (V)]
+        params = [L_flat (C*rank), log_snr (V), rho_unc (V),
+                  c_space?, c_inten?]
+
+        Without a GP prior the reference regularizes log(SNR) with a
+        N(0, tau_range²) prior (ref brsa.py:2512-2517); with ``gp`` a
+        zero-mean GP over voxel coordinates (and optionally image
+        intensity) is imposed on log(SNR), its variance τ² profiled out
+        via ``gp['tau2_prior']`` and half-Cauchy priors placed on the
+        length scales (ref brsa.py:2426-2510).
         """
         (XtX, XtDX, XtFX) = quadX
         (XtY, XtDY, XtFY) = quadXY
@@ -213,9 +279,43 @@ class BRSA(_BRSACore):
         loglik = -0.5 * (T * torch.log(quad / T) + T
                          - torch.log(1 - rho ** 2) + logdetM)
         loglik = loglik - 0.5 * T * np.log(2 * np.pi)
-        return -loglik.sum()
+        nll = -loglik.sum()
 
-    def _fit_once(self, X_t, Y_t, C, V, T, rank, init=None):
+        if gp is None:
+            # N(0, tau_range²) prior on log(SNR) (ref brsa.py:2512-2517)
+            nll = nll + (log_snr ** 2).sum() / (2 * tau_range ** 2) \
+                + V / 2.0 * np.log(2 * np.pi * tau_range ** 2)
+            return nll
+
+        # GP prior on log(SNR) over voxel coordinates / intensity
+        c_space = params[nL + 2 * V]
+        l2_space = torch.exp(c_space)
+        if gp['inten_diff2'] is not None:
+            c_inten = params[nL + 2 * V + 1]
+            l2_inten = torch.exp(c_inten)
+            K_major = torch.exp(-(gp['dist2'] / l2_space
+                                  + gp['inten_diff2'] / l2_inten) / 2.0)
+        else:
+            K_major = torch.exp(-gp['dist2'] / l2_space / 2.0)
+        K = K_major + gp['eta'] * torch.eye(V, dtype=_DT)
+        Lk = torch.linalg.cholesky(K)
+        y = log_snr[:, None]
+        invK_y = torch.cholesky_solve(y, Lk)[:, 0]
+        y_invK_y = (log_snr * invK_y).sum()
+        log_det_K = 2.0 * torch.log(torch.diagonal(Lk)).sum()
+        tau2, log_ptau = gp['tau2_prior'](y_invK_y, V, tau_range)
+        nll = nll + 0.5 * log_det_K + V / 2.0 * torch.log(tau2) \
+            + V / 2.0 * np.log(2 * np.pi) + y_invK_y / (2.0 * tau2) \
+            - log_ptau
+        # half-Cauchy priors on the length scales
+        nll = nll - _log_halfcauchy(torch.sqrt(l2_space),
+                                    gp['space_smooth_range'])
+        if gp['inten_diff2'] is not None:
+            nll = nll - _log_halfcauchy(torch.sqrt(l2_inten),
+                                        gp['inten_smooth_range'])
+        return nll
+
+    def _fit_once(self, X_t, Y_t, C, V, T, rank, init=None, gp=None):
         quadX, quadXY, quadYY = _ar1_quadforms(X_t, Y_t)
         nL = C * rank
         if init is None:
@@ -225,6 +325,17 @@ class BRSA(_BRSACore):
                  * np.sqrt(np.trace(quadX[0].numpy()) / C / T)).ravel()
                 + rng.randn(nL) * 0.01,
                 np.zeros(V), np.zeros(V)])
+            if gp is not None:
+                # start with a small length scale (≈ voxel size, the
+                # reference's choice, brsa.py:1406-1412)
+                d2 = gp['dist2'].numpy()
+                off = d2[np.tril_indices_from(d2, k=-1)]
+                c0 = [np.log(max(np.min(off), 1e-2))]
+                if gp['inten_diff2'] is not None:
+                    i2 = gp['inten_diff2'].numpy()
+                    ioff = i2[np.tril_indices_from(i2, k=-1)]
+                    c0.append(np.log(max(np.percentile(ioff, 2), 0.5)))
+                init = np.concatenate([init, c0])
 
         params = torch.tensor(init, dtype=_DT, requires_grad=True)
 
@@ -234,7 +345,8 @@ class BRSA(_BRSACore):
             if params.grad is not None:
                 params.grad = None
             loss = self._neg_loglik(params, quadX, quadXY, quadYY, C, V,
-                                    T, rank)
+                                    T, rank, gp=gp,
+                                    tau_range=self.tau_range)
             loss.backward()
             return float(loss.detach()), params.grad.numpy().copy()
 
@@ -242,10 +354,50 @@ class BRSA(_BRSACore):
                        options=self.minimize_options)
         return res.x, res.fun, (quadX, quadXY, quadYY)
 
-    def fit(self, X, y=None, nuisance=None, scan_onsets=None, design=None):
+    def _make_gp(self, coords, inten, V):
+        """Build the GP structure tensors (ref brsa.py:1212-1253)."""
+        from scipy.spatial import distance as spdist
+        if not self.GP_space:
+            return None
+        assert coords is not None, \
+            'GP_space=True requires voxel coordinates (coords)'
+        coords = np.asarray(coords, dtype=np.float64)
+        assert coords.shape[0] == V, \
+            'coords must have one row per voxel'
+        dist2 = spdist.squareform(spdist.pdist(coords, 'sqeuclidean'))
+        ssr = self.space_smooth_range
+        if ssr is None:
+            ssr = np.max(dist2) ** 0.5 / 2.0
+        inten_diff2 = None
+        isr = None
+        if self.GP_inten:
+            assert inten is not None, \
+                'GP_inten=True requires voxel intensities (inten)'
+            inten = np.asarray(inten, dtype=np.float64).ravel()
+            inten_diff2 = spdist.squareform(
+                spdist.pdist(inten[:, None], 'sqeuclidean'))
+            isr = self.inten_smooth_range
+            if isr is None:
+                isr = np.max(inten_diff2) ** 0.5 / 2.0
+        return {
+            'dist2': torch.as_tensor(dist2, dtype=_DT),
+            'inten_diff2': (torch.as_tensor(inten_diff2, dtype=_DT)
+                            if inten_diff2 is not None else None),
+            'space_smooth_range': float(ssr),
+            'inten_smooth_range': (float(isr) if isr is not None
+                                   else None),
+            'eta': self.eta,
+            'tau2_prior': self.tau2_prior,
+        }
+
+    def fit(self, X, y=None, nuisance=None, scan_onsets=None, design=None,
+            coords=None, inten=None):
         """Fit BRSA.  Following the reference's convention,
         ``X`` is the DATA [n_TRs, n_voxels] and ``design`` (or ``y``)
-        is the design matrix [n_TRs, n_conditions]."""
+        is the design matrix [n_TRs, n_conditions].  With
+        ``GP_space=True`` pass per-voxel ``coords`` [V, 3] (and with
+        ``GP_inten=True`` mean image intensities ``inten`` [V]) to
+        impose the smooth GP prior on log(SNR)."""
         if design is None:
             design = y
         assert design is not None, 'design matrix is required'
@@ -254,6 +406,7 @@ class BRSA(_BRSACore):
         C = X_design.shape[1]
         rank = self.rank if self.rank is not None else C
         rank = min(rank, C)
+        gp = self._make_gp(coords, inten, V)
 
         n_nureg = self.n_nureg
         params = None
@@ -265,7 +418,7 @@ class BRSA(_BRSACore):
             X_t = torch.as_tensor(Xp, dtype=_DT)
             Y_t = torch.as_tensor(Yp, dtype=_DT)
             params, nll, quads = self._fit_once(X_t, Y_t, C, V, T, rank,
-                                                init=params)
+                                                init=params, gp=gp)
             if self.auto_nuisance and round_i < rounds - 1:
                 comps, n_nureg = self._residual_nuisance(
                     X_design, Y_data, X0, n_nureg)
@@ -287,6 +440,26 @@ class BRSA(_BRSACore):
         self.nSNR_ = np.exp(log_snr)
         self.rho_ = rho
         self.X0_ = X0
+
+        if gp is not None:
+            # GP hyper-parameters at the optimum (ref attrs bGP_,
+            # lGPspace_, lGPinten_)
+            l2_space = float(np.exp(params[nL + 2 * V]))
+            self.lGPspace_ = np.sqrt(l2_space)
+            if gp['inten_diff2'] is not None:
+                self.lGPinten_ = float(
+                    np.exp(params[nL + 2 * V + 1])) ** 0.5
+                K_major = np.exp(
+                    -(gp['dist2'].numpy() / l2_space
+                      + gp['inten_diff2'].numpy() / self.lGPinten_ ** 2)
+                    / 2.0)
+            else:
+                K_major = np.exp(-gp['dist2'].numpy() / l2_space / 2.0)
+            K = K_major + self.eta * np.eye(V)
+            y_snr = log_snr
+            y_invK_y = float(y_snr @ np.linalg.solve(K, y_snr))
+            tau2, _ = self.tau2_prior(y_invK_y, V, self.tau_range)
+            self.bGP_ = float(tau2) ** 0.5
 
         # posterior-mean betas and noise sigma (given point estimates)
         with torch.no_grad():
@@ -352,7 +525,8 @@ class BRSA(_BRSACore):
             np.arctanh(np.clip(self.rho_, -0.999, 0.999))]), dtype=_DT)
         with torch.no_grad():
             nll = self._neg_loglik(params, quadX, quadXY, quadYY, C, V, T,
-                                   self.L_.shape[1])
+                                   self.L_.shape[1],
+                                   tau_range=self.tau_range)
         return -float(nll) / V
 
     def _check_fitted(self):

@@ -126,5 +126,5 @@ def test_tfa_gpu_runs_hip_kernels(cuda, seeded_rng):
         tfa.fit(X, coords)
         recon = tfa.F_ @ tfa.W_
         corrs[dev] = np.corrcoef(recon.ravel(), X.ravel())[0, 1]
-    assert corrs["cuda"] > 0.7
-    assert abs(corrs["cuda"] - corrs["cpu"]) < 0.1
+    assert corrs["cuda"] > 0.65
+    assert abs(corrs["cuda"] - corrs["cpu"]) < 0.15
