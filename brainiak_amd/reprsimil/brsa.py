@@ -27,10 +27,12 @@ are re-estimated between fitting rounds (ref behaviour).  X₀'s betas
 carry an improper flat prior, implemented by projecting X and Y onto
 the orthogonal complement of X₀.
 
-``transform`` decodes per-timepoint condition activity from new data by
-spatially-whitened GLS against the posterior-mean betas — the
-reference's forward-backward temporal smoothing of the decoded courses
-is not applied (documented simplification).
+``transform`` decodes per-timepoint condition and nuisance activity
+from new data by a forward-backward (Kalman/RTS) smoother: AR(1)
+latent time courses observed through the posterior-mean loadings under
+per-voxel AR(1) noise, with the V×V innovation covariance avoided via
+measurement differencing + information-form updates (the reference's
+forward/backward algorithm, brsa.py:1530-1582, re-derived).
 
 GP priors: with ``GP_space=True`` (optionally ``GP_inten=True``) a
 zero-mean Gaussian-Process prior over voxel coordinates (and mean image
@@ -145,6 +147,99 @@ def _project_out(M, X0):
     """Residual-forming projection I − X₀(X₀ᵀX₀)⁻¹X₀ᵀ applied to M."""
     Q, _ = np.linalg.qr(X0)
     return M - Q @ (Q.T @ M)
+
+
+def _ar1_params(M):
+    """Per-column Yule-Walker AR(1) estimates (ρ, innovation σ²).
+
+    Zero-variance columns (e.g. the DC baseline regressor) get a white
+    unit-variance prior (ρ=0, σ²=1) so their latent course stays free.
+    """
+    M = np.asarray(M, dtype=np.float64)
+    Mc = M - M.mean(axis=0)
+    var = (Mc * Mc).mean(axis=0)
+    lag1 = (Mc[1:] * Mc[:-1]).mean(axis=0)
+    ok = var > 1e-12
+    rho = np.where(ok, lag1 / np.maximum(var, 1e-12), 0.0)
+    rho = np.clip(rho, -0.99, 0.99)
+    sigma2 = np.where(ok, np.maximum(var * (1 - rho ** 2), 1e-12), 1.0)
+    return rho, sigma2
+
+
+def _kalman_rts(Y, W, rho_x, sig2_x, rho_e, sig2_e):
+    """Decode latent AR(1) time courses from data with AR(1) noise.
+
+    Model (the reference's transform model, ref brsa.py:1530-1582):
+        z_t = diag(ρ_x) z_{t-1} + w_t,  w_t ~ N(0, diag(σ²_x))
+        y_t = Wᵀ z_t + e_t,             e_t per-voxel AR(1)(ρ_e, σ²_e)
+
+    Our MI355X-native re-derivation: whiten the observation noise by
+    measurement differencing (ỹ_t = y_t − ρ_e ⊙ y_{t-1} observes
+    [z_t, z_{t-1}] with white noise), run a Kalman filter on the
+    augmented state [z_t, z_{t-1}] with information-form measurement
+    updates (the V×V innovation covariance never forms — only K×K
+    normal equations), then RTS-smooth backwards.
+
+    Parameters: Y [T, V]; W [K, V] latent-to-voxel loadings; rho_x,
+    sig2_x [K] latent AR(1); rho_e, sig2_e [V] noise AR(1).
+    Returns the smoothed latent means [T, K].
+    """
+    T_len, V = Y.shape
+    K = W.shape[0]
+    H1 = W.T                                   # [V, K], t = 1
+    A = np.zeros((2 * K, 2 * K))
+    A[:K, :K] = np.diag(rho_x)
+    A[K:, :K] = np.eye(K)
+    Qa = np.zeros((2 * K, 2 * K))
+    Qa[:K, :K] = np.diag(sig2_x)
+    P1z = sig2_x / np.maximum(1 - rho_x ** 2, 1e-6)
+
+    R1 = sig2_e / np.maximum(1 - rho_e ** 2, 1e-6)   # stationary var
+    jitter = 1e-9 * np.eye(2 * K)
+
+    # augmented observation for t >= 2: ỹ_t = [Wᵀ, −ρ_e⊙Wᵀ]·s_t + η
+    H2 = np.concatenate([W.T, -rho_e[:, None] * W.T], axis=1)  # [V, 2K]
+    S2 = (H2 / sig2_e[:, None]).T @ H2                         # [2K, 2K]
+
+    m_pred = np.zeros((T_len, 2 * K))
+    P_pred = np.zeros((T_len, 2 * K, 2 * K))
+    m_filt = np.zeros((T_len, 2 * K))
+    P_filt = np.zeros((T_len, 2 * K, 2 * K))
+
+    # t = 1: prior on s_1 = [z_1, dummy]; observe y_1 with stationary R
+    m = np.zeros(2 * K)
+    P = np.diag(np.concatenate([P1z, P1z]))
+    S1 = np.zeros((2 * K, 2 * K))
+    S1[:K, :K] = (H1 / R1[:, None]).T @ H1
+    c1 = np.zeros(2 * K)
+    c1[:K] = H1.T @ (Y[0] / R1)
+    Pinv = np.linalg.inv(P + jitter)
+    P = np.linalg.inv(Pinv + S1 + jitter)
+    m = P @ (Pinv @ m + c1)
+    m_pred[0], P_pred[0] = 0.0, np.diag(np.concatenate([P1z, P1z]))
+    m_filt[0], P_filt[0] = m, P
+
+    dY = Y[1:] - rho_e[None, :] * Y[:-1]       # whitened observations
+    Hc = H2 / sig2_e[:, None]                  # reused every step
+    for t in range(1, T_len):
+        m_p = A @ m
+        P_p = A @ P @ A.T + Qa
+        c = Hc.T @ dY[t - 1]
+        Pinv = np.linalg.inv(P_p + jitter)
+        P = np.linalg.inv(Pinv + S2 + jitter)
+        m = P @ (Pinv @ m_p + c)
+        m_pred[t], P_pred[t] = m_p, P_p
+        m_filt[t], P_filt[t] = m, P
+
+    # RTS backward pass
+    m_s = m_filt[-1].copy()
+    out = np.zeros((T_len, K))
+    out[-1] = m_s[:K]
+    for t in range(T_len - 2, -1, -1):
+        G = P_filt[t] @ A.T @ np.linalg.inv(P_pred[t + 1] + jitter)
+        m_s = m_filt[t] + G @ (m_s - m_pred[t + 1])
+        out[t] = m_s[:K]
+    return out
 
 
 class _BRSACore:
@@ -482,27 +577,41 @@ class BRSA(_BRSACore):
             beta = (snr2[:, None] * w).numpy().T        # [C, V]
         self.sigma_ = np.sqrt(sigma2)
         self.beta_ = beta
+        # nuisance loadings + AR(1) stats of the training time courses,
+        # used by transform()'s forward-backward smoother
+        resid0 = Y_data - X_design @ beta
+        self.beta0_ = np.linalg.lstsq(X0, resid0, rcond=None)[0]
+        self._rho_design_, self._sigma2_design_ = _ar1_params(X_design)
+        self._rho_X0_, self._sigma2_X0_ = _ar1_params(X0)
         self._fitted_nll = nll
         return self
 
     def transform(self, X, y=None, scan_onsets=None):
-        """Decode per-TR condition activity from new data X [T, V]:
-        spatially-whitened GLS against the posterior-mean betas.
+        """Decode per-TR condition activity from new data X [T, V]
+        by the forward-backward (Kalman/RTS) smoother: condition and
+        nuisance time courses are AR(1) latents (parameters estimated
+        from the training design / X0 courses), observed through the
+        posterior-mean loadings [beta_; beta0_] under per-voxel AR(1)
+        noise (ref brsa.py:793-852, 1530-1582).
         Returns (ts [T, C], ts0 [T, n_X0])."""
         self._check_fitted()
         Y = np.asarray(X, dtype=np.float64)
-        W = 1.0 / (self.sigma_ ** 2)
-        B = self.beta_                      # [C, V]
-        G = (B * W[None, :]) @ B.T + np.eye(B.shape[0])
-        ts = np.linalg.solve(G, (B * W[None, :]) @ Y.T).T
-        # nuisance amplitudes: least squares of the residual on X0's
-        # column space, expressed as per-component time courses
-        X0 = self.X0_ if self.X0_.shape[0] == Y.shape[0] else \
-            np.ones((Y.shape[0], 1))
-        resid = Y - ts @ B
-        beta0 = np.linalg.lstsq(X0, resid, rcond=None)[0]   # [n0, V]
-        ts0 = X0 * np.linalg.norm(beta0, axis=1)[None, :]
-        return ts, ts0
+        C = self.beta_.shape[0]
+        W = np.concatenate([self.beta_, self.beta0_], axis=0)  # [K, V]
+        rho_x = np.concatenate([self._rho_design_, self._rho_X0_])
+        sig2_x = np.concatenate([self._sigma2_design_, self._sigma2_X0_])
+        if scan_onsets is None:
+            onsets = np.array([0], dtype=int)
+        else:
+            onsets = np.unique(np.asarray(scan_onsets, dtype=int))
+            assert onsets[0] == 0, 'scan_onsets must include 0'
+        bounds = list(onsets) + [Y.shape[0]]
+        zs = [
+            _kalman_rts(Y[bounds[i]:bounds[i + 1]], W, rho_x, sig2_x,
+                        self.rho_, self.sigma_ ** 2)
+            for i in range(len(onsets))]
+        z = np.concatenate(zs, axis=0)
+        return z[:, :C], z[:, C:]
 
     def score(self, X, design, scan_onsets=None):
         """Mean per-voxel marginal log-likelihood of new data under the

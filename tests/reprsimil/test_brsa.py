@@ -206,3 +206,44 @@ def test_brsa_gp_inten_runs(seeded_rng):
     assert hasattr(m, 'lGPspace_') and hasattr(m, 'lGPinten_')
     assert np.isfinite(m.lGPinten_) and m.lGPinten_ > 0
     assert np.all(np.isfinite(m.nSNR_))
+
+
+def test_brsa_transform_scan_onsets(seeded_rng):
+    """transform with scan_onsets == concatenated per-scan decodes."""
+    Y, design, _ = _gen_brsa_data(seeded_rng, T=120, V=40)
+    model = BRSA(auto_nuisance=False, random_state=0,
+                 minimize_options={'maxiter': 120, 'disp': False})
+    model.fit(X=Y, design=design)
+    ts_split, ts0_split = model.transform(
+        Y, scan_onsets=np.array([0, 60]))
+    ts_a, _ = model.transform(Y[:60])
+    ts_b, _ = model.transform(Y[60:])
+    assert np.allclose(ts_split, np.vstack([ts_a, ts_b]), atol=1e-8)
+
+
+def test_kalman_rts_recovers_smooth_latent(seeded_rng):
+    """The smoother beats per-TR GLS on a smooth latent course."""
+    from brainiak_amd.reprsimil.brsa import _kalman_rts
+    rng = seeded_rng
+    T, V, K = 200, 30, 2
+    rho_x = np.array([0.9, 0.8])
+    sig2_x = 1 - rho_x ** 2
+    z = np.zeros((T, K))
+    for t in range(1, T):
+        z[t] = rho_x * z[t - 1] + np.sqrt(sig2_x) * rng.randn(K)
+    W = rng.randn(K, V)
+    rho_e = np.full(V, 0.3)
+    sig2_e = np.full(V, 4.0)
+    e = np.zeros((T, V))
+    e[0] = rng.randn(V) * np.sqrt(sig2_e / (1 - rho_e ** 2))
+    for t in range(1, T):
+        e[t] = rho_e * e[t - 1] + np.sqrt(sig2_e) * rng.randn(V)
+    Y = z @ W + e
+    z_hat = _kalman_rts(Y, W, rho_x, sig2_x, rho_e, sig2_e)
+    # per-TR GLS ignoring temporal structure
+    G = (W / sig2_e[None, :]) @ W.T
+    z_gls = np.linalg.solve(G, (W / sig2_e[None, :]) @ Y.T).T
+    err_kalman = np.mean((z_hat - z) ** 2)
+    err_gls = np.mean((z_gls - z) ** 2)
+    assert err_kalman < err_gls
+    assert np.corrcoef(z_hat[:, 0], z[:, 0])[0, 1] > 0.8
