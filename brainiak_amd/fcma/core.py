@@ -215,10 +215,17 @@ class CorrelationPipeline:
         pending = None          # (z, ready-event, buffer index)
         buf_free = [None, None]  # event: gram done reading buffer b
 
+        import os
+        skip_gram = bool(os.environ.get("BRAINIAK_SKIP_GRAM"))  # probe
+
         def _consume(z, ev, bidx):
             with torch.cuda.stream(gram_stream):
                 gram_stream.wait_event(ev)
-                g = ops.fcma_gram_bf16(z)
+                if skip_gram:
+                    g = torch.zeros((z.shape[0], Epad, Epad),
+                                    dtype=torch.float32, device=z.device)
+                else:
+                    g = ops.fcma_gram_bf16(z)
                 if Epad != E:
                     g = g[:, :E, :E].contiguous()
                 done = torch.cuda.Event()

@@ -19,6 +19,8 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <stdlib.h>
+#include <string.h>
 
 #define WAVE 64
 typedef __hip_bfloat16 bf16_t;
@@ -46,9 +48,13 @@ __device__ __forceinline__ float fisher_z(float r) {
     float den = 1.0f - r;
     num = (num <= 0.0f) ? 1e-4f : num;   // reference clamp semantics
     den = (den <= 0.0f) ? 1e-4f : den;
-    // log(num/den) = log(num) - log(den): two v_log_f32 instead of a
-    // full-precision divide chain (v_div_scale/v_rcp, ~10 VALU)
-    return 0.5f * (__logf(num) - __logf(den));
+    // raw v_log_f32 (log2), no denormal fixup: the clamp bounds the
+    // arguments to [1e-4, 2], far from the denormal range — __logf's
+    // ~10-instruction range-check chain is dead weight here (PMC: the
+    // normalize stage was the largest VALU block of the corr kernel)
+    const float half_ln2 = 0.34657359028f;   // 0.5 * ln(2)
+    return half_ln2 * (__builtin_amdgcn_logf(num)
+                       - __builtin_amdgcn_logf(den));
 }
 
 // ===========================================================================
@@ -251,6 +257,181 @@ __global__ __launch_bounds__(256) void k_corr_norm(
                 (bf16_t)zv;
         else
             fOut[((c0 + c) * E + (s * P + p)) * VB + v0 + v] = zv;
+    }
+}
+
+// ===========================================================================
+// k_corr_norm_dot2: the v_dot2c_f32_bf16 form of the corr stage.
+// ISA evidence for the classic kernel (77 s_waitcnt per 64 ds_read_b128 +
+// 128 v_pk_fma_f32 in the unrolled k-loop): the wave stalls on nearly
+// every LDS read and pays 8 b128 broadcasts per 32 MACs.  Here the A tile
+// stays bf16 (its source dtype) in k-pair-interleaved layout
+// [p][k/2][c][2], so one ds_read_b128 feeds FOUR c-columns x k-pair and
+// one v_dot2c_f32_bf16 retires TWO MACs with exact f32 accumulation —
+// half the LDS bytes AND half the VALU issue of the pk_fma form — and
+// the next k-pair's A quads are register-staged while the current pair
+// is dotted (T14), so each k-step carries one batched s_waitcnt.
+// ===========================================================================
+typedef __bf16 bf16x2_t __attribute__((ext_vector_type(2)));
+typedef short short2_t __attribute__((ext_vector_type(2)));
+
+template <int TP, int TL>
+__global__ __launch_bounds__(256) void k_corr_norm_dot2(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ zOut, float* __restrict__ fOut,
+    ll E, ll Lrt, ll VA, ll VB, ll s0, ll C, int Prt, int mode,
+    ll zstride) {
+    static_assert(TL % 2 == 0, "dot2 corr kernel requires even L");
+    const int P = TP > 0 ? TP : Prt;
+    const ll L = TL;
+    (void)Lrt;
+    const ll nSubj = E / P;
+    const ll cTiles = (C + CN_CT - 1) / CN_CT;
+    const ll vTiles = (VB + CN_VT - 1) / CN_VT;
+    ll b = blockIdx.x;
+    const ll vt = b % vTiles; b /= vTiles;
+    const ll s = b % nSubj;   b /= nSubj;
+    const ll ct = b;
+    if (ct >= cTiles) return;
+
+    const ll c0 = ct * CN_CT;
+    const ll v0 = vt * CN_VT;
+    const int CT = (int)min((ll)CN_CT, C - c0);
+    const int VT = (int)min((ll)CN_VT, VB - v0);
+    const int tid = threadIdx.x;
+
+    extern __shared__ char smem[];
+    // a_tile [P][L/2][CN_CT][2] bf16 (k-pair interleaved), corr fp32
+    bf16_t* a_tile = (bf16_t*)smem;
+    float* corr = (float*)(a_tile + (size_t)P * TL * CN_CT);
+    constexpr int KP = TL / 2;
+
+    for (int idx = tid; idx < P * (int)L * CN_CT; idx += 256) {
+        int c = idx % CN_CT;
+        int k = (idx / CN_CT) % (int)L;
+        int p = idx / (CN_CT * (int)L);
+        bf16_t val = (bf16_t)0.0f;
+        if (c < CT)
+            val = A[((s * P + p) * L + k) * VA + (s0 + c0 + c)];
+        a_tile[(((size_t)p * KP + (k >> 1)) * CN_CT + c) * 2 + (k & 1)]
+            = val;
+    }
+    __syncthreads();
+
+    for (int base = tid; base < P * CN_VT; base += 256) {
+        int v = base % CN_VT;
+        int p = base / CN_VT;
+        if (v < VT) {
+            // b column as bf16 k-pairs (kept in source dtype)
+            bf16x2_t bp[KP];
+            const bf16_t* brow = B + ((ll)(s * P + p) * L) * VB + (v0 + v);
+            #pragma unroll
+            for (int kp = 0; kp < KP; ++kp) {
+                bf16x2_t t;
+                t[0] = *(const __bf16*)&brow[(ll)(2 * kp) * VB];
+                t[1] = *(const __bf16*)&brow[(ll)(2 * kp + 1) * VB];
+                bp[kp] = t;
+            }
+            float acc[CN_CT];
+            #pragma unroll
+            for (int c = 0; c < CN_CT; ++c) acc[c] = 0.f;
+            // explicit ds_read_b128 of 4 c-pairs; the scheduler runs the
+            // reads several k-pairs ahead with counted lgkmcnt waits
+            // (verified in the ISA) — no manual pipeline needed
+            const bf16x8* arow8 = (const bf16x8*)
+                (a_tile + ((size_t)p * KP) * CN_CT * 2);
+            #pragma unroll
+            for (int kp = 0; kp < KP; ++kp)
+                #pragma unroll
+                for (int q = 0; q < CN_CT / 4; ++q) {
+                    bf16x8 raw = arow8[kp * (CN_CT / 4) + q];
+                    #pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        short2_t s2 = {raw[2 * j], raw[2 * j + 1]};
+                        acc[4 * q + j] = __builtin_amdgcn_fdot2_f32_bf16(
+                            __builtin_bit_cast(bf16x2_t, s2), bp[kp],
+                            acc[4 * q + j], false);
+                    }
+                }
+            #pragma unroll
+            for (int c = 0; c < CN_CT; ++c)
+                corr[((size_t)c * P + p) * CN_VT + v] = acc[c];
+        }
+    }
+    __syncthreads();
+
+    if (mode == 2) {
+        for (int base = tid; base < CT * P * CN_VT; base += 256) {
+            int v = base % CN_VT;
+            int p = (base / CN_VT) % P;
+            int c = base / (CN_VT * P);
+            if (v < VT)
+                fOut[((c0 + c) * E + (s * P + p)) * VB + v0 + v] =
+                    corr[((size_t)c * P + p) * CN_VT + v];
+        }
+        return;
+    }
+
+    // normalization FUSED with the output store: one thread per (c, v)
+    // column; z stays in registers between fisher_z and the
+    // (lane-coalesced) store — no corr writeback, no re-read pass, one
+    // barrier fewer than the classic kernel
+    for (int base = tid; base < CT * CN_VT; base += 256) {
+        int v = base % CN_VT;
+        int c = base / CN_VT;
+        if (v >= VT) continue;
+        float* col = corr + ((size_t)c * P) * CN_VT + v;
+        float mean = 0.f, sq = 0.f;
+        if (TP > 0) {
+            float z[TP > 0 ? TP : 1];
+            #pragma unroll
+            for (int p = 0; p < P; ++p) {
+                z[p] = fisher_z(col[(size_t)p * CN_VT]);
+                mean += z[p]; sq += z[p] * z[p];
+            }
+            mean /= (float)P;
+            float var = sq / (float)P - mean * mean;
+            float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+            if (mode == 0) {
+                bf16_t* dst = zOut
+                    + ((c0 + c) * zstride + s * (ll)P) * VB + v0 + v;
+                #pragma unroll
+                for (int p = 0; p < P; ++p)
+                    dst[(size_t)p * VB] = (bf16_t)((z[p] - mean) * inv);
+            } else if (mode == 3) {
+                // write-skip PROBE (bounds the fused-kernel gain): the
+                // never-true predicate keeps the compute alive
+                #pragma unroll
+                for (int p = 0; p < P; ++p)
+                    if (zstride < 0)
+                        zOut[(size_t)tid] =
+                            (bf16_t)((z[p] - mean) * inv);
+            } else {
+                float* dst = fOut
+                    + ((c0 + c) * E + s * (ll)P) * VB + v0 + v;
+                #pragma unroll
+                for (int p = 0; p < P; ++p)
+                    dst[(size_t)p * VB] = (z[p] - mean) * inv;
+            }
+        } else {
+            for (int p = 0; p < P; ++p) {
+                float zv = fisher_z(col[(size_t)p * CN_VT]);
+                mean += zv; sq += zv * zv;
+            }
+            mean /= (float)P;
+            float var = sq / (float)P - mean * mean;
+            float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+            for (int p = 0; p < P; ++p) {
+                float zv = (fisher_z(col[(size_t)p * CN_VT]) - mean)
+                           * inv;
+                if (mode == 0)
+                    zOut[((c0 + c) * zstride + (s * P + p)) * VB
+                         + v0 + v] = (bf16_t)zv;
+                else
+                    fOut[((c0 + c) * E + (s * P + p)) * VB + v0 + v] =
+                        zv;
+            }
+        }
     }
 }
 
@@ -655,6 +836,17 @@ extern "C" void launch_fcma_normalize(float* corr, ll C, ll E, ll V, int P,
     }
 }
 
+// runtime corr-kernel selector for within-probe A/B
+// (BRAINIAK_CORR_KERNEL=classic forces the pk_fma form)
+static int corr_variant() {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("BRAINIAK_CORR_KERNEL");
+        v = (e && strcmp(e, "classic") == 0) ? 0 : 1;
+    }
+    return v;
+}
+
 template <int TP, int TL>
 static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
                                float* fOut, ll E, ll L, ll VA, ll VB,
@@ -682,6 +874,13 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
     }
 #endif
     ll grid = ceil_div(C, CN_CT) * nSubj * ceil_div(VB, CN_VT);
+    if (corr_variant() == 1 && (TL % 2) == 0) {
+        hipLaunchKernelGGL((k_corr_norm_dot2<TP, TL>), dim3(grid),
+                           dim3(256), smem, stream, (const bf16_t*)A,
+                           (const bf16_t*)B, (bf16_t*)zOut, fOut, E, L,
+                           VA, VB, s0, C, P, mode, zstride);
+        return;
+    }
     hipLaunchKernelGGL((k_corr_norm<TP, TL>), dim3(grid), dim3(256),
                        smem, stream, (const bf16_t*)A,
                        (const bf16_t*)B, (bf16_t*)zOut, fOut, E, L,
@@ -697,6 +896,11 @@ extern "C" int fcma_corr_norm_smem(ll L, int P) {
         return (int)smem;
     }
 #endif
+    if (corr_variant() == 1 && (L % 2) == 0) {
+        size_t smem = (size_t)P * L * CN_CT * sizeof(bf16_t)  // bf16 a
+                    + (size_t)CN_CT * P * CN_VT * sizeof(float);
+        return (int)smem;
+    }
     size_t smem = (size_t)P * L * CN_CT * sizeof(float)   // fp32 a_tile
                 + (size_t)CN_CT * P * CN_VT * sizeof(float);
     return (int)smem;
@@ -744,6 +948,8 @@ extern "C" void launch_fcma_corr_norm(const void* A, const void* B,
                                       ll VA, ll VB, ll s0, ll C, int P,
                                       int mode, ll zstride,
                                       hipStream_t stream) {
+    static int probe_mode3 = getenv("BRAINIAK_CORR_MODE3") ? 1 : 0;
+    if (probe_mode3 && mode == 0 && corr_variant() == 1) mode = 3;
     size_t smem = (size_t)fcma_corr_norm_smem(L, P);
     switch (L) {
         case 8:  dispatch_p<8>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
