@@ -23,6 +23,7 @@ implementations below are the CPU path and the numerics oracle the GPU
 kernels are tested against.
 """
 
+import os
 from typing import List, Optional
 
 import numpy as np
@@ -197,6 +198,19 @@ class CorrelationPipeline:
         E = self.num_epochs
         Epad = ((E + 63) // 64) * 64
         ext = ops.load_extension()
+        if getattr(ext, "fcma_fused_gram_native", None) is not None and \
+                ext.fcma_fused_gram_native(E, self.epochs_per_subj,
+                                           self.data.shape[1]) and \
+                not os.environ.get("BRAINIAK_NO_FUSED"):
+            # single-kernel corr+gram: Z never touches HBM, so there is
+            # nothing to double-buffer or overlap
+            gram = torch.cat([
+                ops.fcma_fused_gram(self.data, self.data2, s, c,
+                                    self.epochs_per_subj)
+                for s, c in chunks], dim=0)
+            if shrink:
+                _shrink_(gram)
+            return gram
         max_count = max(c for _, c in chunks)
         # persistent double-buffered Z workspace: repeated multi-GB
         # allocations churn the caching allocator across streams
@@ -215,7 +229,6 @@ class CorrelationPipeline:
         pending = None          # (z, ready-event, buffer index)
         buf_free = [None, None]  # event: gram done reading buffer b
 
-        import os
         skip_gram = bool(os.environ.get("BRAINIAK_SKIP_GRAM"))  # probe
 
         def _consume(z, ev, bidx):

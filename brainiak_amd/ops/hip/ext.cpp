@@ -27,6 +27,9 @@ void launch_svm_cv(const float*, const float*, const int*, const int*,
                    float, int, void*);
 ll fcma_supported_L(ll);
 int fcma_corr_norm_smem(ll, int);
+int fcma_fused_gram_supported(ll, int, ll);
+void launch_fcma_fused_corr_gram(const void*, const void*, float*, ll,
+                                 ll, ll, ll, ll, int, void*);
 }
 
 static void* cur_stream() {
@@ -144,13 +147,38 @@ torch::Tensor fcma_gram(torch::Tensor corr) {
 
 torch::Tensor fcma_fused_gram(torch::Tensor A, torch::Tensor B,
                               int64_t start, int64_t count, int64_t P) {
-    ll E = A.size(0);
+    ll E = A.size(0), L = A.size(1), VA = A.size(2), VB = B.size(2);
+    if (fcma_fused_gram_supported(E, (int)P, L)) {
+        // single-kernel path: Z never leaves LDS (see
+        // fcma_kernels.hip::k_fused_corr_gram)
+        check_3d(A, torch::kBFloat16, "A");
+        check_3d(B, torch::kBFloat16, "B");
+        TORCH_CHECK(B.size(0) == E && B.size(1) == L,
+                    "A/B epoch shapes differ");
+        TORCH_CHECK(start >= 0 && start + count <= VA, "voxel range");
+        ll cTiles = (count + 7) / 8;
+        ll vTiles = std::max((ll)1, (VB + 63) / 64);
+        // enough workgroups to fill 256 CUs (1 resident WG per CU),
+        // but never more v-splits than v-windows
+        int nsplit = (int)std::min(
+            vTiles, std::max((ll)1, (511 + cTiles) / cTiles));
+        auto G = torch::empty({nsplit, count, E, E},
+                              A.options().dtype(torch::kFloat32));
+        launch_fcma_fused_corr_gram(A.data_ptr(), B.data_ptr(),
+                                    G.data_ptr<float>(), L, VA, VB,
+                                    start, count, nsplit, cur_stream());
+        return nsplit == 1 ? G.squeeze(0) : G.sum(0);
+    }
     ll Epad = ((E + 63) / 64) * 64;
     auto Z = fcma_corr_norm_z(A, B, start, count, P, Epad, c10::nullopt);
     auto G = fcma_gram_bf16(Z);
     if (Epad != E)
         return G.narrow(1, 0, E).narrow(2, 0, E).contiguous();
     return G;
+}
+
+bool fcma_fused_gram_native(int64_t E, int64_t P, int64_t L) {
+    return fcma_fused_gram_supported(E, (int)P, L) != 0;
 }
 
 // ---------------------------------------------------------------------------
@@ -273,6 +301,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fcma_gram", &fcma_gram, "per-voxel Gram from fp32 [C,E,V]");
     m.def("fcma_gram_bf16", &fcma_gram_bf16,
           "per-voxel Gram from bf16 Z [C,E,V]");
+    m.def("fcma_fused_gram_native", &fcma_fused_gram_native,
+          "true when the single-kernel corr+gram path covers (E, P, L)");
     m.def("fcma_fused_gram", &fcma_fused_gram,
           "corr+norm+Gram for a voxel chunk");
     m.def("jacobi_eigh", &jacobi_eigh, "batched KxK symmetric eigensolve");

@@ -598,6 +598,214 @@ __global__ __launch_bounds__(256) void k_corr_norm_mfma(
 }
 
 // ===========================================================================
+// k_fused_corr_gram: the WHOLE chunk pipeline in one kernel for the
+// headline shape (E = 64 epochs, P = 4 epochs/subject):
+//   corr -> Fisher-z -> z-score -> per-voxel [64,64] Gram (MFMA),
+// with the normalized Z living ONLY in LDS.  Probe evidence
+// (profiles/README.md): the two-kernel pipeline pays ~7.3 ms/step for
+// Z's HBM round trip (bf16 write + Gram re-read, 36 GB/step); here the
+// per-(c-tile, v-window) Z tile is produced in LDS and immediately
+// rank-64-updated into per-wave MFMA accumulators, so HBM traffic per
+// step drops to B reads (MALL-cached) + the 64 MB Gram output.
+//
+// Structure: 512-thread workgroup (8 waves) owns FG_CB=8 selected
+// voxels; grid = c-tiles x nsplit v-ranges; each workgroup loops its
+// v-windows (VT=64):
+//   per subject s: stage A-pairs -> dot2 corr (thread = (chalf,p,v))
+//     -> fused normalize writing bf16 z into z_smem[c][epoch][v]
+//   after 16 subjects: wave w rank-updates G_{c0+w} from z_smem[w]
+//     (16 MFMA tiles x 2 k-steps, same fragment map as k_gram_bf16).
+// Gram accumulators persist in VGPRs across the whole v-range; the
+// [nsplit, C, 64, 64] partials are summed on the host when nsplit > 1.
+// ===========================================================================
+#define FG_CB 8
+#define FG_VT 64
+#define FG_ROW 68   // bf16 z row stride: 136 B spreads the 16-lane
+                    // fragment groups across banks (see LDS table)
+
+template <int TL>
+__global__ __launch_bounds__(512) void k_fused_corr_gram(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    float* __restrict__ Gpart, ll VA, ll VB, ll s0, ll C, int nsplit) {
+    static_assert(TL % 2 == 0, "fused kernel requires even L");
+    constexpr int P = 4, E = 64, NSUBJ = 16, KP = TL / 2;
+    const ll cTiles = (C + FG_CB - 1) / FG_CB;
+    const ll vTiles = (VB + FG_VT - 1) / FG_VT;
+    const ll vs = blockIdx.x % nsplit;
+    const ll ct = blockIdx.x / nsplit;
+    if (ct >= cTiles) return;
+    const ll c0 = ct * FG_CB;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wv = tid >> 6;                   // 8 waves
+
+    __shared__ bf16_t z_smem[FG_CB][E][FG_ROW];
+    __shared__ float corr_s[FG_CB][P][FG_VT];
+    __shared__ bf16_t a_sm[P][KP][FG_CB][2];
+
+    // symmetric accumulation: only the 10 upper tiles (i <= j) live in
+    // registers; the lower triangle is mirrored at the store
+    f32x4 G[10];
+    #pragma unroll
+    for (int t = 0; t < 10; ++t) G[t] = (f32x4)0.f;
+
+    for (ll vt = vs; vt < vTiles; vt += nsplit) {
+        const ll v0 = vt * FG_VT;
+        for (int s = 0; s < NSUBJ; ++s) {
+            // --- stage A k-pairs for this subject's 4 epochs
+            for (int idx = tid; idx < P * TL * FG_CB; idx += 512) {
+                int i2 = idx & 1;
+                int c = (idx >> 1) % FG_CB;
+                int kp = ((idx >> 1) / FG_CB) % KP;
+                int p = (idx >> 1) / (FG_CB * KP);
+                bf16_t val = (bf16_t)0.0f;
+                if (c0 + c < C)
+                    val = A[((ll)(s * P + p) * TL + 2 * kp + i2) * VA
+                            + (s0 + c0 + c)];
+                a_sm[p][kp][c][i2] = val;
+            }
+            __syncthreads();
+            // --- corr: thread = (chalf, p, v); 4 c's per thread
+            {
+                int v = tid & 63;
+                int p = (tid >> 6) & 3;
+                int chalf = tid >> 8;
+                float acc[4] = {0.f, 0.f, 0.f, 0.f};
+                if (v0 + v < VB) {
+                    bf16x2_t bp[KP];
+                    const bf16_t* brow =
+                        B + ((ll)(s * P + p) * TL) * VB + (v0 + v);
+                    #pragma unroll
+                    for (int kp = 0; kp < KP; ++kp) {
+                        bf16x2_t t;
+                        t[0] = *(const __bf16*)&brow[(ll)(2 * kp) * VB];
+                        t[1] = *(const __bf16*)
+                            &brow[(ll)(2 * kp + 1) * VB];
+                        bp[kp] = t;
+                    }
+                    #pragma unroll
+                    for (int kp = 0; kp < KP; ++kp) {
+                        bf16x8 raw =
+                            *(const bf16x8*)&a_sm[p][kp][chalf * 4][0];
+                        #pragma unroll
+                        for (int j = 0; j < 4; ++j) {
+                            short2_t s2 = {raw[2 * j], raw[2 * j + 1]};
+                            acc[j] = __builtin_amdgcn_fdot2_f32_bf16(
+                                __builtin_bit_cast(bf16x2_t, s2),
+                                bp[kp], acc[j], false);
+                        }
+                    }
+                }
+                #pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    corr_s[chalf * 4 + j][p][v] = acc[j];
+            }
+            __syncthreads();
+            // --- normalize (thread = (c, v)) straight into z_smem
+            {
+                int c = wv;            // 8 waves = 8 c's
+                int v = lane;
+                float z[P];
+                float mean = 0.f, sq = 0.f;
+                #pragma unroll
+                for (int p = 0; p < P; ++p) {
+                    z[p] = fisher_z(corr_s[c][p][v]);
+                    mean += z[p]; sq += z[p] * z[p];
+                }
+                mean /= (float)P;
+                float var = sq / (float)P - mean * mean;
+                float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+                #pragma unroll
+                for (int p = 0; p < P; ++p)
+                    z_smem[c][s * P + p][v] =
+                        (bf16_t)((z[p] - mean) * inv);
+            }
+            __syncthreads();
+        }
+        // --- Gram rank-64 update: wave wv owns voxel c0+wv.
+        // z reads of this window finish before the next window's first
+        // barrier, so no extra sync is needed here.
+        {
+            const int frow = lane & 15;
+            #pragma unroll
+            for (int ks = 0; ks < FG_VT / 32; ++ks) {
+                const int fk = 8 * (lane >> 4) + 32 * ks;
+                bf16x8 f[4];
+                #pragma unroll
+                for (int bnd = 0; bnd < 4; ++bnd)
+                    f[bnd] = (bf16x8)(*(const bf16x8_u*)
+                        &z_smem[wv][bnd * 16 + frow][fk]);
+                int t = 0;
+                #pragma unroll
+                for (int i = 0; i < 4; ++i)
+                    #pragma unroll
+                    for (int j = 0; j < 4; ++j)
+                        if (j >= i)
+                            G[t] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    f[i], f[j], G[t], 0, 0, 0), ++t;
+            }
+        }
+        __syncthreads();
+    }
+
+    if (c0 + wv < C) {
+        float* Gc = Gpart + ((vs * C) + c0 + wv) * (ll)(E * E);
+        const int dcol = lane & 15;
+        const int drow = (lane >> 4) * 4;
+        int t = 0;
+        #pragma unroll
+        for (int i = 0; i < 4; ++i)
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                if (j < i) continue;
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    float val = G[t][r];
+                    Gc[(ll)(i * 16 + drow + r) * E + j * 16 + dcol] =
+                        val;
+                    if (i != j)
+                        Gc[(ll)(j * 16 + dcol) * E + i * 16 + drow + r]
+                            = val;
+                }
+                ++t;
+            }
+    }
+}
+
+extern "C" int fcma_fused_gram_supported(ll E, int P, ll L) {
+    return E == 64 && P == 4 && L > 0 && L % 2 == 0 && L <= CN_MAXL;
+}
+
+template <int TL>
+static void launch_fused_t(const void* A, const void* B, float* Gpart,
+                           ll VA, ll VB, ll s0, ll C, int nsplit,
+                           hipStream_t stream) {
+    ll grid = ((C + FG_CB - 1) / FG_CB) * nsplit;
+    hipLaunchKernelGGL((k_fused_corr_gram<TL>), dim3(grid), dim3(512),
+                       0, stream, (const bf16_t*)A, (const bf16_t*)B,
+                       Gpart, VA, VB, s0, C, nsplit);
+}
+
+extern "C" void launch_fcma_fused_corr_gram(
+    const void* A, const void* B, float* Gpart, ll L, ll VA, ll VB,
+    ll s0, ll C, int nsplit, hipStream_t stream) {
+    switch (L) {
+        case 8:  launch_fused_t<8>(A, B, Gpart, VA, VB, s0, C, nsplit,
+                                   stream); break;
+        case 16: launch_fused_t<16>(A, B, Gpart, VA, VB, s0, C, nsplit,
+                                   stream); break;
+        case 24: launch_fused_t<24>(A, B, Gpart, VA, VB, s0, C, nsplit,
+                                   stream); break;
+        case 32: launch_fused_t<32>(A, B, Gpart, VA, VB, s0, C, nsplit,
+                                   stream); break;
+        case 40: launch_fused_t<40>(A, B, Gpart, VA, VB, s0, C, nsplit,
+                                   stream); break;
+        default: break;
+    }
+}
+
+// ===========================================================================
 // k_gram_bf16: G_c = Z_c Z_c^T per voxel with bf16 MFMA.
 // Z: [C, E, V] bf16; G: [C, E, E] fp32.  E % 64 == 0 (host pads).
 // Block = (c, band_i, band_j>=band_i); 4 waves own the 32x32 quadrants.

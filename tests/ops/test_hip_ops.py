@@ -200,3 +200,28 @@ def test_tfa_factor_and_recon(ops):
     R = ops.tfa_recon(X, W, F, 0.5)
     ref_R = (0.5 * (X - F @ W)).reshape(-1)
     assert torch.allclose(R, ref_R, atol=1e-3, rtol=1e-3)
+
+
+def test_fused_corr_gram_native_e64(ops):
+    """The single-kernel corr+gram path (E=64, P=4) against the CPU
+    oracle and the two-kernel product."""
+    from brainiak_amd.ops import load_extension
+    ext = load_extension()
+    assert ext.fcma_fused_gram_native(64, 4, 16)
+    g = torch.Generator().manual_seed(7)
+    E, L, V, P = 64, 16, 300, 4      # V non-multiple of 64 (tail window)
+    A = _zscored_epochs(g, E, L, V, "cpu")
+    Ab = A.to(torch.bfloat16)
+    dev = Ab.cuda().contiguous()
+    G = ops.fcma_fused_gram(dev, dev, 3, 40, P)    # C=40: c-tile tail
+    corr = torch.einsum('elc,elv->cev', Ab.float()[:, :, 3:43], Ab.float())
+    nc = _ref_normalize(corr, P)
+    ref = torch.bmm(nc, nc.transpose(1, 2))
+    assert G.shape == (40, E, E)
+    assert torch.allclose(G.cpu(), ref, atol=1.0, rtol=3e-2)
+    assert torch.allclose(G, G.transpose(1, 2))    # exact symmetry
+    # two-kernel product from the same inputs (same bf16 z, same 64-wide
+    # k-tiling) — tight agreement expected
+    Z = ext.fcma_corr_norm_z(dev, dev, 3, 40, P, 64, None)
+    G2 = ops.fcma_gram_bf16(Z)
+    assert torch.allclose(G.cpu(), G2.cpu(), atol=2e-2, rtol=1e-3)
