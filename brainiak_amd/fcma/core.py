@@ -201,9 +201,12 @@ class CorrelationPipeline:
         if getattr(ext, "fcma_fused_gram_native", None) is not None and \
                 ext.fcma_fused_gram_native(E, self.epochs_per_subj,
                                            self.data.shape[1]) and \
-                not os.environ.get("BRAINIAK_NO_FUSED"):
-            # single-kernel corr+gram: Z never touches HBM, so there is
-            # nothing to double-buffer or overlap
+                os.environ.get("BRAINIAK_FUSED"):
+            # single-kernel corr+gram (opt-in): Z never touches HBM, but
+            # measured on MI355X the B-read amplification at its 8-voxel
+            # c-tile outweighs the saved Z round trip (24.7 vs 16.7
+            # ms/step, profiles/README.md) — the streamed two-kernel
+            # pipeline below with a 64-voxel corr tile wins
             gram = torch.cat([
                 ops.fcma_fused_gram(self.data, self.data2, s, c,
                                     self.epochs_per_subj)

@@ -282,11 +282,19 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot2(
     ll E, ll Lrt, ll VA, ll VB, ll s0, ll C, int Prt, int mode,
     ll zstride) {
     static_assert(TL % 2 == 0, "dot2 corr kernel requires even L");
+    // c-tile width: every B element is register-reused across TCT
+    // selected voxels, so LLC/HBM B traffic scales as 1/TCT; bounded by
+    // LDS (corr tile TCT*P*VT fp32) and VGPRs (acc[TCT])
+    // measured on MI355X: TCT=64 (72 KB LDS -> 8 waves/CU) halves the
+    // corr floor's occupancy and regresses 2x despite 4x less B traffic
+    // - the kernel needs ~16+ waves/CU to hide the B-column load
+    // latency.  TCT=32 (36 KB -> 16 waves/CU) is the measured optimum.
+    constexpr int TCT = (TP > 0 && TP <= 4) ? 32 : CN_CT;
     const int P = TP > 0 ? TP : Prt;
     const ll L = TL;
     (void)Lrt;
     const ll nSubj = E / P;
-    const ll cTiles = (C + CN_CT - 1) / CN_CT;
+    const ll cTiles = (C + TCT - 1) / TCT;
     const ll vTiles = (VB + CN_VT - 1) / CN_VT;
     ll b = blockIdx.x;
     const ll vt = b % vTiles; b /= vTiles;
@@ -294,26 +302,26 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot2(
     const ll ct = b;
     if (ct >= cTiles) return;
 
-    const ll c0 = ct * CN_CT;
+    const ll c0 = ct * TCT;
     const ll v0 = vt * CN_VT;
-    const int CT = (int)min((ll)CN_CT, C - c0);
+    const int CT = (int)min((ll)TCT, C - c0);
     const int VT = (int)min((ll)CN_VT, VB - v0);
     const int tid = threadIdx.x;
 
     extern __shared__ char smem[];
-    // a_tile [P][L/2][CN_CT][2] bf16 (k-pair interleaved), corr fp32
+    // a_tile [P][L/2][TCT][2] bf16 (k-pair interleaved), corr fp32
     bf16_t* a_tile = (bf16_t*)smem;
-    float* corr = (float*)(a_tile + (size_t)P * TL * CN_CT);
+    float* corr = (float*)(a_tile + (size_t)P * TL * TCT);
     constexpr int KP = TL / 2;
 
-    for (int idx = tid; idx < P * (int)L * CN_CT; idx += 256) {
-        int c = idx % CN_CT;
-        int k = (idx / CN_CT) % (int)L;
-        int p = idx / (CN_CT * (int)L);
+    for (int idx = tid; idx < P * (int)L * TCT; idx += 256) {
+        int c = idx % TCT;
+        int k = (idx / TCT) % (int)L;
+        int p = idx / (TCT * (int)L);
         bf16_t val = (bf16_t)0.0f;
         if (c < CT)
             val = A[((s * P + p) * L + k) * VA + (s0 + c0 + c)];
-        a_tile[(((size_t)p * KP + (k >> 1)) * CN_CT + c) * 2 + (k & 1)]
+        a_tile[(((size_t)p * KP + (k >> 1)) * TCT + c) * 2 + (k & 1)]
             = val;
     }
     __syncthreads();
@@ -332,19 +340,19 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot2(
                 t[1] = *(const __bf16*)&brow[(ll)(2 * kp + 1) * VB];
                 bp[kp] = t;
             }
-            float acc[CN_CT];
+            float acc[TCT];
             #pragma unroll
-            for (int c = 0; c < CN_CT; ++c) acc[c] = 0.f;
+            for (int c = 0; c < TCT; ++c) acc[c] = 0.f;
             // explicit ds_read_b128 of 4 c-pairs; the scheduler runs the
             // reads several k-pairs ahead with counted lgkmcnt waits
             // (verified in the ISA) — no manual pipeline needed
             const bf16x8* arow8 = (const bf16x8*)
-                (a_tile + ((size_t)p * KP) * CN_CT * 2);
+                (a_tile + ((size_t)p * KP) * TCT * 2);
             #pragma unroll
             for (int kp = 0; kp < KP; ++kp)
                 #pragma unroll
-                for (int q = 0; q < CN_CT / 4; ++q) {
-                    bf16x8 raw = arow8[kp * (CN_CT / 4) + q];
+                for (int q = 0; q < TCT / 4; ++q) {
+                    bf16x8 raw = arow8[kp * (TCT / 4) + q];
                     #pragma unroll
                     for (int j = 0; j < 4; ++j) {
                         short2_t s2 = {raw[2 * j], raw[2 * j + 1]};
@@ -354,7 +362,7 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot2(
                     }
                 }
             #pragma unroll
-            for (int c = 0; c < CN_CT; ++c)
+            for (int c = 0; c < TCT; ++c)
                 corr[((size_t)c * P + p) * CN_VT + v] = acc[c];
         }
     }
@@ -1055,6 +1063,11 @@ static int corr_variant() {
     return v;
 }
 
+// host mirror of the dot2 kernel's TCT constexpr
+static ll dot2_ct(int P) {
+    return (P == 2 || P == 4) ? 32 : CN_CT;
+}
+
 template <int TP, int TL>
 static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
                                float* fOut, ll E, ll L, ll VA, ll VB,
@@ -1081,14 +1094,28 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
         return;
     }
 #endif
-    ll grid = ceil_div(C, CN_CT) * nSubj * ceil_div(VB, CN_VT);
     if (corr_variant() == 1 && (TL % 2) == 0) {
-        hipLaunchKernelGGL((k_corr_norm_dot2<TP, TL>), dim3(grid),
+        ll gridd = ceil_div(C, dot2_ct(P)) * nSubj * ceil_div(VB, CN_VT);
+        if (smem > 64 * 1024) {
+            // gfx950 has 160 KB LDS but dynamic allocations above the
+            // 64 KB default need the explicit opt-in, once per kernel
+            static bool raised = false;
+            if (!raised) {
+                (void)hipFuncSetAttribute(
+                    reinterpret_cast<const void*>(
+                        &k_corr_norm_dot2<TP, TL>),
+                    hipFuncAttributeMaxDynamicSharedMemorySize,
+                    160 * 1024);
+                raised = true;
+            }
+        }
+        hipLaunchKernelGGL((k_corr_norm_dot2<TP, TL>), dim3(gridd),
                            dim3(256), smem, stream, (const bf16_t*)A,
                            (const bf16_t*)B, (bf16_t*)zOut, fOut, E, L,
                            VA, VB, s0, C, P, mode, zstride);
         return;
     }
+    ll grid = ceil_div(C, CN_CT) * nSubj * ceil_div(VB, CN_VT);
     hipLaunchKernelGGL((k_corr_norm<TP, TL>), dim3(grid), dim3(256),
                        smem, stream, (const bf16_t*)A,
                        (const bf16_t*)B, (bf16_t*)zOut, fOut, E, L,
@@ -1105,8 +1132,9 @@ extern "C" int fcma_corr_norm_smem(ll L, int P) {
     }
 #endif
     if (corr_variant() == 1 && (L % 2) == 0) {
-        size_t smem = (size_t)P * L * CN_CT * sizeof(bf16_t)  // bf16 a
-                    + (size_t)CN_CT * P * CN_VT * sizeof(float);
+        ll ct = dot2_ct(P);
+        size_t smem = (size_t)P * L * ct * sizeof(bf16_t)   // bf16 a
+                    + (size_t)ct * P * CN_VT * sizeof(float);
         return (int)smem;
     }
     size_t smem = (size_t)P * L * CN_CT * sizeof(float)   // fp32 a_tile
