@@ -15,7 +15,7 @@ void launch_fcma_normalize(float*, ll, ll, ll, int, void*);
 void launch_fcma_corr_norm(const void*, const void*, void*, float*, ll, ll,
                            ll, ll, ll, ll, int, int, ll, void*);
 int fcma_corr_norm_smem(ll, int);
-void launch_fcma_gram_bf16(const void*, float*, ll, ll, ll, void*);
+void launch_fcma_gram_bf16(const void*, float*, ll, ll, ll, ll, void*);
 void launch_fcma_gram_f32(const float*, float*, ll, ll, ll, void*);
 void launch_jacobi_eigh(const float*, float*, float*, ll, int, void*);
 void launch_tfa_factor(const float*, const float*, const float*, float*,
@@ -121,10 +121,23 @@ torch::Tensor fcma_gram_bf16(torch::Tensor Z) {
     check_3d(Z, torch::kBFloat16, "Z");
     ll C = Z.size(0), E = Z.size(1), V = Z.size(2);
     TORCH_CHECK(E % 64 == 0, "E must be a multiple of 64 (host pads)");
-    auto G = torch::empty({C, E, E}, Z.options().dtype(torch::kFloat32));
-    launch_fcma_gram_bf16(Z.data_ptr(), G.data_ptr<float>(), C, E, V,
-                          cur_stream());
-    return G;
+    ll eb = E / 64;
+    // V-split so small-C calls still fill 256 CUs (>= ~512 blocks)
+    ll base = C * eb * eb;
+    ll ktAll = (V + 63) / 64;
+    ll nsplit = std::min(ktAll, std::max((ll)1, (511 + base) / base));
+    if (nsplit <= 1) {
+        auto G = torch::empty({C, E, E},
+                              Z.options().dtype(torch::kFloat32));
+        launch_fcma_gram_bf16(Z.data_ptr(), G.data_ptr<float>(), C, E, V,
+                              1, cur_stream());
+        return G;
+    }
+    auto Gp = torch::empty({nsplit, C, E, E},
+                           Z.options().dtype(torch::kFloat32));
+    launch_fcma_gram_bf16(Z.data_ptr(), Gp.data_ptr<float>(), C, E, V,
+                          nsplit, cur_stream());
+    return Gp.sum(0);
 }
 
 torch::Tensor fcma_gram(torch::Tensor corr) {

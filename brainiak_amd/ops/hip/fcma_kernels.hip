@@ -824,13 +824,27 @@ extern "C" void launch_fcma_fused_corr_gram(
 
 __global__ __launch_bounds__(256) void k_gram_bf16(
     const bf16_t* __restrict__ Z, float* __restrict__ G,
-    ll C, ll E, ll V) {
+    ll C, ll E, ll V, ll nsplit) {
+    // nsplit > 1: V is cut into k-tile-aligned ranges, one partial G
+    // per range (host sums) — keeps the chip full when C is small
+    // (MALL-resident Z slabs want 32-voxel chunks; 32 blocks would
+    // leave 7/8 of the CUs idle)
     const ll eb = E / 64;
     ll b = blockIdx.x;
+    const ll ns = b % nsplit; b /= nsplit;
     const ll band_j = b % eb; b /= eb;
     const ll band_i = b % eb; b /= eb;
     const ll c = b;
     if (c >= C || band_j < band_i) return;
+    const ll ktAll = (V + GR_KT - 1) / GR_KT;
+    const ll ktPer = (ktAll + nsplit - 1) / nsplit;
+    const ll kt0 = ns * ktPer;
+    const ll kt1 = min(ktAll, kt0 + ktPer);
+    if (kt0 >= kt1) {   // empty range: zero this partial
+        for (ll i = threadIdx.x; i < E * E; i += 256)
+            G[(ns * C + c) * E * E + i] = 0.0f;
+        return;
+    }
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -878,16 +892,15 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
             *(bf16x8*)&dst[srow][scol + 8 * h] = regs[h];
     };
 
-    const ll kTiles = (V + GR_KT - 1) / GR_KT;
     bf16x8 ri[2], rj[2];
-    issue_loads(Zc, rows_i, 0, ri);
-    if (!diag) issue_loads(Zc, rows_j, 0, rj);
+    issue_loads(Zc, rows_i, kt0 * GR_KT, ri);
+    if (!diag) issue_loads(Zc, rows_j, kt0 * GR_KT, rj);
 
-    for (ll kt = 0; kt < kTiles; ++kt) {
+    for (ll kt = kt0; kt < kt1; ++kt) {
         __syncthreads();              // previous tile's reads complete
         write_tile(zi, ri);
         if (!diag) write_tile(zj, rj);
-        if (kt + 1 < kTiles) {        // issue next tile early
+        if (kt + 1 < kt1) {           // issue next tile early
             issue_loads(Zc, rows_i, (kt + 1) * GR_KT, ri);
             if (!diag) issue_loads(Zc, rows_j, (kt + 1) * GR_KT, rj);
         }
@@ -912,7 +925,7 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
         }
     }
 
-    float* Gc = G + c * E * E;
+    float* Gc = G + (ns * C + c) * E * E;
     const int dcol = lane & 15;
     const int drow = (lane >> 4) * 4;
     const f32x4* accs[4] = {&acc00, &acc01, &acc10, &acc11};
@@ -1203,11 +1216,12 @@ extern "C" void launch_fcma_corr_norm(const void* A, const void* B,
 }
 
 extern "C" void launch_fcma_gram_bf16(const void* Z, float* G, ll C, ll E,
-                                      ll V, hipStream_t stream) {
+                                      ll V, ll nsplit,
+                                      hipStream_t stream) {
     ll eb = E / 64;
-    ll grid = C * eb * eb;
+    ll grid = C * eb * eb * nsplit;
     hipLaunchKernelGGL(k_gram_bf16, dim3(grid), dim3(256), 0, stream,
-                       (const bf16_t*)Z, G, C, E, V);
+                       (const bf16_t*)Z, G, C, E, V, nsplit);
 }
 
 extern "C" void launch_fcma_gram_f32(const float* Z, float* G, ll C, ll E,
