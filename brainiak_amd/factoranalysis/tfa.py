@@ -252,6 +252,72 @@ class TFA:
             final_err[base:] = np.sqrt(self.sample_scaling * dist).ravel()
         return final_err
 
+    def _jacobian_multivariate(self, estimate, unique_R, inds, X, W,
+                               template_centers,
+                               template_centers_mean_cov,
+                               template_widths,
+                               template_widths_mean_var_reci,
+                               data_sigma):
+        """Closed-form Jacobian of ``_residual_multivariate``.
+
+        The reference differentiates by finite differences (~K*(dim+1)+1
+        residual evaluations per NLSS iteration, each an N8+N9 kernel
+        pass); the RBF residual is analytic in centers and widths:
+
+            r[v,t]        = sigma*(X[v,t] - sum_k F[v,k] W[k,t])
+            dF/dc_{k,d}   = F[v,k] * 2 (coord[v,d] - c[k,d]) / w_k
+            dF/dw_k       = F[v,k] * d2[v,k] / w_k**2
+
+        so ONE factor evaluation yields the whole [m, K*(dim+1)] matrix.
+        Parameter order matches ``estimate``: centers.ravel() then
+        widths.
+        """
+        centers = self.get_centers(estimate)
+        widths = self.get_widths(estimate)
+        coords = np.column_stack(
+            [unique_R[d][inds[d]] for d in range(self.n_dim)]).astype(
+                np.float64)
+        K, D = self.K, self.n_dim
+        V, T = X.shape
+        n_par = K * (D + 1)
+        dev = "cuda" if self._use_gpu() else "cpu"
+        co = torch.as_tensor(coords, dtype=torch.float32, device=dev)
+        ce = torch.as_tensor(centers, dtype=torch.float32, device=dev)
+        wd = torch.as_tensor(widths.ravel(), dtype=torch.float32,
+                             device=dev)
+        Wt = torch.as_tensor(W, dtype=torch.float32, device=dev)
+        diff = co[:, None, :] - ce[None, :, :]            # [V, K, D]
+        d2 = (diff * diff).sum(-1)                        # [V, K]
+        F = torch.exp(-d2 / wd[None, :])
+        dF = torch.empty((V, K, D + 1), dtype=torch.float32, device=dev)
+        dF[:, :, :D] = F[:, :, None] * 2.0 * diff / wd[None, :, None]
+        dF[:, :, D] = F * d2 / (wd * wd)[None, :]
+        # J_recon[v,t, k,j] = -sigma * W[k,t] * dF[v,k,j]
+        Jr = -float(data_sigma) * torch.einsum('kt,vkj->vtkj', Wt, dF)
+        recon = V * T
+        other = 0 if template_centers is None else 2 * K
+        J = np.zeros((recon + other, n_par))
+        Jrc = Jr.reshape(recon, K, D + 1).cpu().double().numpy()
+        # column order: centers (k*D + d), then widths (K*D + k)
+        J[:recon, :K * D] = Jrc[:, :, :D].reshape(recon, K * D)
+        J[:recon, K * D:] = Jrc[:, :, D]
+        if other > 0:
+            S = self.sample_scaling
+            for k in range(K):
+                dfk = centers[k] - template_centers[k]
+                cov = from_tri_2_sym(template_centers_mean_cov[k],
+                                     self.n_dim)
+                cov = cov + cov.T - np.diag(np.diag(cov))
+                solved = np.linalg.solve(cov, dfk)
+                e = math.sqrt(max(S * dfk.dot(solved), 1e-30))
+                J[recon + k, k * D:(k + 1) * D] = S * solved / e
+            reci = np.asarray(template_widths_mean_var_reci).ravel()
+            tw = np.asarray(template_widths).ravel()
+            w = widths.ravel()
+            J[recon + K:, K * D:][np.arange(K), np.arange(K)] = \
+                np.sqrt(S * reci) * np.sign(w - tw)
+        return J
+
     def _estimate_centers_widths(self, unique_R, inds, X, W, init_centers,
                                  init_widths, template_centers,
                                  template_widths,
@@ -260,10 +326,12 @@ class TFA:
         init_estimate = np.hstack((init_centers.ravel(),
                                    init_widths.ravel()))
         data_sigma = 1.0 / math.sqrt(2.0) * np.std(X)
-        # on GPU the residual rides the fp32 HIP factor/recon kernels;
-        # the default 2-point step (~1.5e-8) sits below fp32 noise and
-        # yields garbage Jacobians — use a step well above it
-        diff_step = 1e-3 if self._use_gpu() else None
+        # analytic Jacobian (one factor pass) unless the user asked for
+        # a finite-difference scheme explicitly
+        if self.jac in ('2-point', '3-point'):
+            jac = self._jacobian_multivariate
+        else:
+            jac = self.jac
         final_estimate = least_squares(
             self._residual_multivariate, init_estimate,
             args=(unique_R, inds, X, W, template_centers,
@@ -271,7 +339,7 @@ class TFA:
                   template_widths_mean_var_reci, data_sigma),
             method=self.nlss_method, loss=self.nlss_loss,
             bounds=self.bounds, verbose=0, x_scale=self.x_scale,
-            tr_solver=self.tr_solver, diff_step=diff_step)
+            tr_solver=self.tr_solver, jac=jac)
         return final_estimate.x, final_estimate.cost
 
     # -- convergence ---------------------------------------------------------
