@@ -43,16 +43,18 @@ class TFA:
     """
 
     def __init__(self, max_iter=10, threshold=1.0, K=50,
-                 nlss_method='trf', nlss_loss='soft_l1', jac='2-point',
+                 nlss_method='trf', nlss_loss='soft_l1', jac='analytic',
                  x_scale='jac', tr_solver=None, weight_method='rr',
                  upper_ratio=1.8, lower_ratio=0.02, max_num_tr=500,
-                 max_num_voxel=5000, seed=100, verbose=False, device=None):
+                 max_num_voxel=5000, seed=100, verbose=False, device=None,
+                 nlss_max_nfev=10):
         self.miter = max_iter
         self.threshold = threshold
         self.K = K
         self.nlss_method = nlss_method
         self.nlss_loss = nlss_loss
         self.jac = jac
+        self.nlss_max_nfev = nlss_max_nfev
         self.x_scale = x_scale
         self.tr_solver = tr_solver
         self.weight_method = weight_method
@@ -326,12 +328,14 @@ class TFA:
         init_estimate = np.hstack((init_centers.ravel(),
                                    init_widths.ravel()))
         data_sigma = 1.0 / math.sqrt(2.0) * np.std(X)
-        # analytic Jacobian (one factor pass) unless the user asked for
-        # a finite-difference scheme explicitly
-        if self.jac in ('2-point', '3-point'):
-            jac = self._jacobian_multivariate
-        else:
-            jac = self.jac
+        # 'analytic' (default): the closed-form Jacobian — one factor
+        # pass instead of ~K*(dim+1) FD residual evaluations.  FD
+        # schemes remain selectable via jac='2-point'/'3-point'.
+        # nlss_max_nfev bounds the inner trf iterations: the exact
+        # Jacobian keeps finding descent long past the point where the
+        # outer TFA alternation makes the refinement moot.
+        jac = self._jacobian_multivariate if self.jac == 'analytic' \
+            else self.jac
         final_estimate = least_squares(
             self._residual_multivariate, init_estimate,
             args=(unique_R, inds, X, W, template_centers,
@@ -339,7 +343,8 @@ class TFA:
                   template_widths_mean_var_reci, data_sigma),
             method=self.nlss_method, loss=self.nlss_loss,
             bounds=self.bounds, verbose=0, x_scale=self.x_scale,
-            tr_solver=self.tr_solver, jac=jac)
+            tr_solver=self.tr_solver, jac=jac,
+            max_nfev=self.nlss_max_nfev)
         return final_estimate.x, final_estimate.cost
 
     # -- convergence ---------------------------------------------------------
@@ -472,6 +477,7 @@ class TFA:
         return {"max_iter": self.miter, "threshold": self.threshold,
                 "K": self.K, "nlss_method": self.nlss_method,
                 "nlss_loss": self.nlss_loss, "jac": self.jac,
+                "nlss_max_nfev": self.nlss_max_nfev,
                 "x_scale": self.x_scale, "tr_solver": self.tr_solver,
                 "weight_method": self.weight_method,
                 "upper_ratio": self.upper_ratio,
