@@ -38,14 +38,17 @@ def main():
     mine = [torch.randn((args.trs, V), generator=g).to(device)
             for s in range(subjects) if s % world == rank]
 
+    precision = 'bf16' if device.type == "cuda" else 'fp32'
+
     def step(i):
-        isfc_distributed(mine, ctx, summary_statistic='mean')
+        isfc_distributed(mine, ctx, summary_statistic='mean',
+                         precision=precision)
 
     elapsed = timed_steps(step, args.steps, args.warmup, world, device)
     pairs_per_sec = float(V) * V * subjects * args.steps / elapsed
     emit(rank, "isfc_voxel_pairs_per_sec", pairs_per_sec, "pairs/s",
          world, args.steps, args.warmup, elapsed, True, "strong",
-         "fp32", {"model": "isfc_leave_one_out", "num_voxels": V,
+         precision, {"model": "isfc_leave_one_out", "num_voxels": V,
                   "subjects": subjects, "trs": args.trs,
                   "global_batch": subjects, "seq_len": args.trs,
                   "parallelism": f"subject-sharded dp{world}"})

@@ -191,7 +191,7 @@ def isfc(data, targets=None, pairwise=False, summary_statistic=None,
 
 
 def isfc_distributed(local_data, comm, summary_statistic=None,
-                     row_tile=4096, device=None):
+                     row_tile=4096, device=None, precision='fp32'):
     """Subject-sharded leave-one-out ISFC over RCCL/xGMI.
 
     Parameters
@@ -209,6 +209,10 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
     -----------------------
     summary_statistic None  → [n_subjects_total, V, V] stacked ISFCs
     summary_statistic given → [V, V] collapsed ISFC matrix
+
+    precision 'bf16' runs the [T,V]x[T,V] correlation GEMMs on the
+    MFMA bf16 path (~0.4 % relative error on r; fp32 is the default
+    and matches the serial oracle).
     """
     import torch
 
@@ -239,11 +243,16 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
 
     normed_local = [_norm(d) for d in local]
 
+    def _corr(a, b):
+        if precision == 'bf16':
+            return (a.T.to(torch.bfloat16) @ b.to(torch.bfloat16)).float()
+        return a.T @ b                          # [V, V]
+
     if summary_statistic == 'mean':
         acc = torch.zeros((V, V), dtype=torch.float32, device=dev)
         for d, nd in zip(local, normed_local):
             loo = _norm((total - d) / (n_total - 1))
-            m = nd.T @ loo                      # [V, V]
+            m = _corr(nd, loo)
             m = (m + m.T) / 2
             acc += torch.atanh(m.clamp(-1 + 1e-7, 1 - 1e-7))
         acc = comm.all_reduce(acc)
@@ -252,7 +261,7 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
     stacks = []
     for d, nd in zip(local, normed_local):
         loo = _norm((total - d) / (n_total - 1))
-        m = nd.T @ loo
+        m = _corr(nd, loo)
         stacks.append(((m + m.T) / 2).cpu().numpy())
     gathered = comm.all_gather_object(stacks)
     flat = [m for part in gathered for m in part]

@@ -178,3 +178,15 @@ def test_isfc_distributed_matches_serial(tmp_path):
     # self-ISC values where serial's squareform drops/refills them)
     off = ~np.eye(8, dtype=bool)
     assert np.allclose(dist_result[off], serial[off], atol=1e-4)
+
+
+def test_isfc_distributed_bf16_close_to_fp32():
+    import torch
+    from brainiak_amd.parallel import DistContext
+    rng = np.random.RandomState(4)
+    subs = [rng.randn(50, 40).astype(np.float32) for _ in range(5)]
+    ctx = DistContext(device="cpu")
+    a = isfc_distributed(subs, ctx, summary_statistic='mean')
+    b = isfc_distributed(subs, ctx, summary_statistic='mean',
+                         precision='bf16')
+    assert np.allclose(a, b, atol=2e-2)

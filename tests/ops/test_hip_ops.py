@@ -232,3 +232,34 @@ def test_fused_corr_gram_native_e64(ops):
     Z = ext.fcma_corr_norm_z(dA, dB, 3, 40, P, 64, None)
     G2 = ops.fcma_gram_bf16(Z)
     assert torch.allclose(G.cpu(), G2.cpu(), atol=5e-2, rtol=1e-3)
+
+
+@pytest.mark.parametrize("P,L", [(2, 8), (2, 40), (4, 24), (4, 40),
+                                 (8, 16), (8, 32), (16, 16), (16, 40),
+                                 (5, 16), (3, 24)])
+def test_corr_norm_z_all_templates(ops, P, L):
+    """Every (epochs_per_subj, epoch_len) kernel template variant —
+    incl. the generic runtime-P path (P=5, 3) and every padded L —
+    against the fp32 oracle."""
+    from brainiak_amd.ops import load_extension
+    ext = load_extension()
+    g = torch.Generator().manual_seed(100 * P + L)
+    nsubj = 4 if P <= 8 else 3
+    E, V = P * nsubj, 210
+    A = _zscored_epochs(g, E, L - 2, 96, "cpu")
+    B = _zscored_epochs(g, E, L - 2, V, "cpu")
+    assert A.shape[1] == L                 # padded to the named L
+    Ab, Bb = A.to(torch.bfloat16), B.to(torch.bfloat16)
+    Epad = ((E + 63) // 64) * 64
+    Z = ext.fcma_corr_norm_z(Ab.cuda().contiguous(),
+                             Bb.cuda().contiguous(), 7, 33, P, Epad,
+                             None)
+    corr = torch.einsum('elc,elv->cev', Ab.float()[:, :, 7:40],
+                        Bb.float())
+    ref = _ref_normalize(corr, P)
+    got = Z[:, :E, :].float().cpu()
+    assert torch.allclose(got, ref, atol=3e-2, rtol=3e-2), \
+        (P, L, (got - ref).abs().max().item())
+    # padding rows stay zero
+    if Epad != E:
+        assert torch.all(Z[:, E:, :].float().cpu() == 0)
