@@ -258,12 +258,14 @@ def test_corr_norm_z_all_templates(ops, P, L):
                         Bb.float())
     ref = _ref_normalize(corr, P)
     got = Z[:, :E, :].float().cpu()
-    # mask columns whose within-subject variance is at the fp noise
-    # floor: there the z-score is sign(noise) / the var<=0 -> 0 clamp
-    # (sharpest at P=2, where z = sign(z0 - z1) exactly)
-    z = ref.view(ref.shape[0], E // P, P, -1)
-    var = z.var(dim=2, unbiased=False, keepdim=True)
-    ok = (var > 1e-4).expand_as(z).reshape_as(ref)
+    # mask columns whose within-subject FISHER-Z variance is near the
+    # fp noise floor: there the z-score is sign(noise) / the
+    # var<=0 -> 0 clamp (sharpest at P=2, where z = sign(z0 - z1))
+    zf = 0.5 * torch.log((1 + corr).clamp_min(1e-4)
+                         / (1 - corr).clamp_min(1e-4))
+    zf = zf.view(corr.shape[0], E // P, P, -1)
+    var = zf.var(dim=2, unbiased=False, keepdim=True)
+    ok = (var > 1e-6).expand_as(zf).reshape_as(ref)
     assert ok.float().mean() > 0.95
     assert torch.allclose(got[ok], ref[ok], atol=3e-2, rtol=3e-2), \
         (P, L, (got[ok] - ref[ok]).abs().max().item())
