@@ -199,6 +199,9 @@ class Searchlight:
                                self.bcast_var, extra_block_fn_params)
                 results.append((block[0], out))
 
+        return self._gather_and_stitch(results)
+
+    def _gather_and_stitch(self, results):
         global_outputs = self.comm.gather_object(results)
 
         outmat = np.empty(self.mask.shape, dtype=object)
@@ -214,6 +217,35 @@ class Searchlight:
                         + mat.shape[2]]
                     outmat[coords] = mat
         return outmat
+
+    def run_batched_block_function(self, batch_fn,
+                                   extra_block_fn_params=None):
+        """GPU-batched variant of ``run_block_function``: local blocks
+        with identical shapes are STACKED and handed to ``batch_fn`` in
+        one call, so a numeric block function (correlations, conv3d
+        aggregations, ...) runs as a handful of large batched device
+        ops instead of one kernel chain per block.
+
+        batch_fn(subject_stacks, mask_stack, sl_rad, bcast_var, extra)
+            subject_stacks : list (per subject) of [B, bx, by, bz, T]
+            mask_stack     : [B, bx, by, bz] bool
+            returns        : [B, ox, oy, oz] ndarray (one output block
+                             per input block, border trimmed by sl_rad)
+        """
+        groups = {}
+        for idx in range(len(self.blocks)):
+            shape = self.submasks[idx].shape
+            groups.setdefault(shape, []).append(idx)
+        results = []
+        for idxs in groups.values():
+            subj_stacks = [np.stack([sub[i] for i in idxs])
+                           for sub in self.subproblems]
+            mask_stack = np.stack([self.submasks[i] for i in idxs])
+            outs = batch_fn(subj_stacks, mask_stack, self.sl_rad,
+                            self.bcast_var, extra_block_fn_params)
+            for j, i in enumerate(idxs):
+                results.append((self.blocks[i][0], outs[j]))
+        return self._gather_and_stitch(results)
 
     def run_searchlight(self, voxel_fn, pool_size=None):
         """Apply ``voxel_fn`` at every active voxel; returns an object

@@ -125,3 +125,44 @@ def test_mvpa_voxelselector(seeded_rng):
     scores = [s for _, s in results]
     assert scores == sorted(scores, reverse=True)
     assert scores[0] > 0.8
+
+
+def test_batched_block_function_matches_per_block(seeded_rng):
+    """run_batched_block_function == run_block_function on the same
+    numeric block fn."""
+    dim = (10, 11, 9)
+    data = [seeded_rng.rand(*dim, 6).astype(np.float32) for _ in range(2)]
+    mask = seeded_rng.rand(*dim) > 0.2
+    rad = 1
+
+    def per_block(subjects, msk, r, bcast, extra=None):
+        a, b = subjects
+        s = (a.mean(-1) + b.mean(-1))
+        inner = s[r:-r, r:-r, r:-r]
+        im = msk[r:-r, r:-r, r:-r]
+        return np.where(im, inner, np.nan)
+
+    def batched(stacks, masks, r, bcast, extra=None):
+        a, b = stacks
+        s = a.mean(-1) + b.mean(-1)
+        inner = s[:, r:-r, r:-r, r:-r]
+        im = masks[:, r:-r, r:-r, r:-r]
+        return np.where(im, inner, np.nan)
+
+    sl1 = Searchlight(sl_rad=rad, max_blk_edge=4)
+    sl1.distribute(data, mask)
+    sl1.broadcast(None)
+    out1 = sl1.run_block_function(per_block, pool_size=1)
+
+    sl2 = Searchlight(sl_rad=rad, max_blk_edge=4)
+    sl2.distribute(data, mask)
+    sl2.broadcast(None)
+    out2 = sl2.run_batched_block_function(batched)
+
+    for i in range(dim[0]):
+        for j in range(dim[1]):
+            for k in range(dim[2]):
+                v1, v2 = out1[i, j, k], out2[i, j, k]
+                assert (v1 is None) == (v2 is None)
+                if v1 is not None and not (np.isnan(v1) and np.isnan(v2)):
+                    assert np.isclose(v1, v2, equal_nan=True)
