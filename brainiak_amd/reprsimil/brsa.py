@@ -746,6 +746,7 @@ class GBRSA(_BRSACore):
 
         subj_quads = []
         dims = []
+        subj_ctx = []          # (design, Y, X0) for posterior extraction
         for Xi, Di in zip(X, design):
             Dp, Yi, X0, T, V = self._prepare(Di, Xi, None, scan_onsets)
             if self.auto_nuisance:
@@ -758,6 +759,7 @@ class GBRSA(_BRSACore):
                 torch.as_tensor(Xp, dtype=_DT),
                 torch.as_tensor(Yp, dtype=_DT)))
             dims.append((T, V))
+            subj_ctx.append((Dp, Yi, X0))
 
         rng = np.random.RandomState(self.random_state)
         init = (np.eye(C)[:, :rank] * 1.0).ravel() + rng.randn(
@@ -783,4 +785,137 @@ class GBRSA(_BRSACore):
         self.U_ = L @ L.T
         self.C_ = cov2corr(self.U_ + 1e-15 * np.eye(C))
         self._fitted_nll = res.fun
+
+        # per-subject grid-marginalized posterior point estimates
+        # (ref GBRSA keeps nSNR_/rho_/sigma_/beta_ per subject for
+        # transform/score)
+        self.beta_, self.beta0_, self.rho_, self.sigma_ = [], [], [], []
+        self.nSNR_, self.X0_ = [], []
+        self._rho_design_, self._sigma2_design_ = [], []
+        self._rho_X0_, self._sigma2_X0_ = [], []
+        with torch.no_grad():
+            Lp = torch.as_tensor(res.x, dtype=_DT)
+            for quads, (T, V), (Dp, Yi, X0) in zip(subj_quads, dims,
+                                                   subj_ctx):
+                beta, rho, sig2, snr = self._subject_posterior(
+                    Lp, quads, C, V, T, rank, grids)
+                self.beta_.append(beta)
+                self.rho_.append(rho)
+                self.sigma_.append(np.sqrt(sig2))
+                self.nSNR_.append(snr)
+                self.X0_.append(X0)
+                resid0 = Yi - Dp @ beta
+                self.beta0_.append(
+                    np.linalg.lstsq(X0, resid0, rcond=None)[0])
+                rd, sd = _ar1_params(Dp)
+                self._rho_design_.append(rd)
+                self._sigma2_design_.append(sd)
+                r0, s0 = _ar1_params(X0)
+                self._rho_X0_.append(r0)
+                self._sigma2_X0_.append(s0)
         return self
+
+    def _subject_posterior(self, L_params, quads, C, V, T, rank, grids):
+        """Grid-weighted posterior E[β], E[ρ], E[σ²], E[SNR] per voxel
+        under the fitted U (the marginalization ref brsa.py:3390-3672
+        point-estimates the same way)."""
+        (XtX, XtDX, XtFX), (XtY, XtDY, XtFY), (YtY, YtDY, YtFY) = quads
+        s_grid, w_s, rho_grid, w_rho = grids
+        L = L_params.reshape(C, rank)
+        L = L * torch.as_tensor(
+            np.tril(np.ones((C, rank)))[:, :rank], dtype=_DT)
+        U = L @ L.T
+        eye = torch.eye(C, dtype=_DT)
+        rho_t = torch.as_tensor(rho_grid, dtype=_DT)
+        s2 = torch.as_tensor(s_grid ** 2, dtype=_DT)
+        logw = torch.log(torch.as_tensor(
+            np.outer(w_s, w_rho).ravel(), dtype=_DT))
+        A = (XtX[None] - rho_t[:, None, None] * XtDX[None]
+             + (rho_t ** 2)[:, None, None] * XtFX[None])
+        b = (XtY[None] - rho_t[:, None, None] * XtDY[None]
+             + (rho_t ** 2)[:, None, None] * XtFY[None])
+        q = (YtY[None] - rho_t[:, None] * YtDY[None]
+             + (rho_t ** 2)[:, None] * YtFY[None])
+        UA = U[None] @ A
+        M = eye[None, None] + s2[:, None, None, None] * UA[None]
+        Ub = (U[None] @ b)
+        w = torch.linalg.solve(
+            M, Ub[None].expand(len(s_grid), -1, -1, -1))  # [S, R, C, V]
+        corrq = s2[:, None, None] * (b[None] * w).sum(2)  # [S, R, V]
+        quad = (q[None] - corrq).clamp_min(1e-10)
+        sign, logdetM = torch.linalg.slogdet(M)
+        ll = (-0.5 * (T * torch.log(quad / T) + T
+                      - torch.log(1 - rho_t ** 2)[None, :, None]
+                      + logdetM[:, :, None])
+              - 0.5 * T * np.log(2 * np.pi))              # [S, R, V]
+        lw = ll + logw.reshape(len(s_grid), len(rho_grid))[:, :, None]
+        lw = lw.reshape(-1, V)
+        wgt = torch.softmax(lw, dim=0)                    # [S*R, V]
+        S, R = len(s_grid), len(rho_grid)
+        # E[β|y, s, ρ] = s² U (I + s² A U)⁻¹ b = s² w
+        beta_g = (s2[:, None, None, None] * w).reshape(S * R, C, V)
+        beta = (wgt[:, None, :] * beta_g).sum(0).numpy()  # [C, V]
+        rho = (wgt * rho_t.repeat(S)[:, None]).sum(0).numpy()
+        sig2 = (wgt * (quad.reshape(S * R, V) / T)).sum(0).numpy()
+        snr = (wgt * torch.as_tensor(
+            s_grid, dtype=_DT).repeat_interleave(R)[:, None]
+            ).sum(0).numpy()
+        return beta, rho, sig2, snr
+
+    def transform(self, X, y=None, scan_onsets=None):
+        """Decode per-TR condition and nuisance courses for each
+        subject with the AR(1) Kalman/RTS smoother (same machinery as
+        BRSA.transform).  Returns (list of ts, list of ts0)."""
+        self._check_fitted()
+        single = not isinstance(X, list)
+        if single:
+            X = [X]
+        ts_all, ts0_all = [], []
+        for i, Yi in enumerate(X):
+            Y = np.asarray(Yi, dtype=np.float64)
+            C = self.beta_[i].shape[0]
+            W = np.concatenate([self.beta_[i], self.beta0_[i]], axis=0)
+            rho_x = np.concatenate([self._rho_design_[i],
+                                    self._rho_X0_[i]])
+            sig2_x = np.concatenate([self._sigma2_design_[i],
+                                     self._sigma2_X0_[i]])
+            z = _kalman_rts(Y, W, rho_x, sig2_x, self.rho_[i],
+                            self.sigma_[i] ** 2)
+            ts_all.append(z[:, :C])
+            ts0_all.append(z[:, C:])
+        if single:
+            return ts_all[0], ts0_all[0]
+        return ts_all, ts0_all
+
+    def score(self, X, design, scan_onsets=None):
+        """Mean per-voxel marginal log-likelihood of new data under the
+        fitted U (per subject; higher = better)."""
+        self._check_fitted()
+        single = not isinstance(X, list)
+        if single:
+            X = [X]
+            design = [design]
+        grids = self._grids()
+        rank = self.L_.shape[1]
+        out = []
+        with torch.no_grad():
+            Lp = torch.as_tensor(self.L_.ravel(), dtype=_DT)
+            for i, (Yi, Di) in enumerate(zip(X, design)):
+                Y = np.asarray(Yi, dtype=np.float64)
+                X0 = self.X0_[i] if self.X0_[i].shape[0] == Y.shape[0] \
+                    else np.ones((Y.shape[0], 1))
+                Xp = _project_out(np.asarray(Di, dtype=np.float64), X0)
+                Yp = _project_out(Y, X0)
+                T, V = Yp.shape
+                C = Xp.shape[1]
+                quads = _ar1_quadforms(
+                    torch.as_tensor(Xp, dtype=_DT),
+                    torch.as_tensor(Yp, dtype=_DT))
+                nll = self._neg_loglik_marg(Lp, quads, C, V, T, rank,
+                                            grids)
+                out.append(-float(nll) / V)
+        return out[0] if single else out
+
+    def _check_fitted(self):
+        if not hasattr(self, 'U_'):
+            raise ValueError("The model has not been fit yet.")

@@ -247,3 +247,32 @@ def test_kalman_rts_recovers_smooth_latent(seeded_rng):
     err_gls = np.mean((z_gls - z) ** 2)
     assert err_kalman < err_gls
     assert np.corrcoef(z_hat[:, 0], z[:, 0])[0, 1] > 0.8
+
+
+def test_gbrsa_transform_and_score(seeded_rng):
+    """GBRSA decodes per-subject design courses and scores new data
+    (grid-marginalized posterior point estimates + Kalman smoother)."""
+    rng = seeded_rng
+    T, V, C = 100, 30, 4
+
+    def gen():
+        design = (rng.rand(T, C) < 0.25) * rng.randn(T, C)
+        U = np.eye(C) * 0.5 + 0.5
+        beta = np.linalg.cholesky(U) @ rng.randn(C, V)
+        return design @ beta + rng.randn(T, V), design
+
+    Y1, d1 = gen()
+    Y2, d2 = gen()
+    m = GBRSA(auto_nuisance=False, SNR_bins=7, rho_bins=7,
+              random_state=0,
+              minimize_options={'maxiter': 80, 'disp': False})
+    m.fit(X=[Y1, Y2], design=[d1, d2])
+    ts, ts0 = m.transform([Y1, Y2])
+    assert ts[0].shape == (T, C) and len(ts) == 2
+    r = np.mean([np.corrcoef(ts[0][:, c], d1[:, c])[0, 1]
+                 for c in range(C)])
+    assert r > 0.5
+    s_good = m.score([Y1, Y2], [d1, d2])
+    noise = [rng.randn(T, V) * Y1.std(), rng.randn(T, V) * Y2.std()]
+    s_null = m.score(noise, [d1, d2])
+    assert all(g > n for g, n in zip(s_good, s_null))
