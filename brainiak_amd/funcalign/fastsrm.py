@@ -24,7 +24,11 @@ from .srm import DetSRM, NotFittedError, _polar_orthogonal
 
 logger = logging.getLogger(__name__)
 
-__all__ = ["FastSRM"]
+__all__ = ["FastSRM", "assert_array_2axis", "assert_non_empty_list",
+           "assert_valid_index", "check_atlas", "check_imgs",
+           "check_indexes", "check_shared_response", "create_temp_dir",
+           "fast_srm", "get_shape", "lowram_srm", "reduce_data",
+           "reduce_data_single", "safe_encode", "safe_load"]
 
 
 def safe_load(data):
@@ -258,3 +262,219 @@ class FastSRM:
         for k, v in params.items():
             setattr(self, k, v)
         return self
+
+
+# -- public module-level helpers (reference fastsrm.py:32-1073 API) --------
+
+def get_shape(path):
+    """Shape of a saved ``.npy`` array without loading it."""
+    return np.load(path, mmap_mode='r').shape
+
+
+def safe_encode(img):
+    """Stable hash name for an image (array contents or path string)."""
+    import hashlib
+    if isinstance(img, np.ndarray):
+        return hashlib.md5(img.tobytes()).hexdigest()
+    return hashlib.md5(str(img).encode()).hexdigest()
+
+
+def assert_non_empty_list(input_list, list_name):
+    if len(input_list) == 0:
+        raise ValueError("%s is a list of length 0 which is not valid"
+                         % list_name)
+
+
+def assert_array_2axis(array, name_array):
+    if not isinstance(array, np.ndarray):
+        raise ValueError("%s should be of type np.ndarray but is of "
+                         "type %s" % (name_array, type(array)))
+    if array.ndim != 2:
+        raise ValueError("%s must have exactly 2 axes but has %i axes"
+                         % (name_array, array.ndim))
+
+
+def assert_valid_index(indexes, max_value, name_indexes):
+    for i, ind in enumerate(indexes):
+        if ind < 0 or ind >= max_value:
+            raise ValueError(
+                "Index %i of %s has value %i whereas value should be "
+                "between 0 and %i" % (i, name_indexes, ind,
+                                      max_value - 1))
+
+
+def check_indexes(indexes, name):
+    if not (indexes is None or isinstance(indexes, (list, np.ndarray))):
+        raise ValueError("%s should be either a list, an array or None "
+                         "but received type %s" % (name, type(indexes)))
+
+
+def check_atlas(atlas, n_components=None):
+    """Validate a probabilistic/deterministic atlas (array or path);
+    returns its shape, or None for no atlas."""
+    if atlas is None:
+        return None
+    if isinstance(atlas, (str, os.PathLike)):
+        shape = get_shape(atlas)
+    elif isinstance(atlas, np.ndarray):
+        shape = atlas.shape
+    else:
+        raise ValueError(
+            "Atlas is stored using type %s which is neither np.ndarray "
+            "or str" % type(atlas))
+    if len(shape) == 1:          # deterministic label atlas
+        labels = safe_load(atlas)
+        n_sup = int(np.max(labels))
+    elif len(shape) == 2:        # probabilistic
+        n_sup = shape[0]
+    else:
+        raise ValueError("Atlas has %i axes; it should have either 1 "
+                         "or 2 axes" % len(shape))
+    if n_components is not None and n_sup < n_components:
+        raise ValueError(
+            "Number of atlas supervoxels (%i) is less than the number "
+            "of components (%i)" % (n_sup, n_components))
+    return shape
+
+
+def check_imgs(imgs, n_components=None, atlas_shape=None,
+               ignore_nans=False):
+    """Validate input images (array of paths / list of arrays / list of
+    lists); returns (reshaped [subject][session] list, n_subjects,
+    n_sessions)."""
+    subjects, _ = FastSRM._canonicalize(imgs)
+    n_subjects = len(subjects)
+    n_sessions = len(subjects[0])
+    for s in subjects:
+        if len(s) != n_sessions:
+            raise ValueError("All subjects need the same number of "
+                             "sessions")
+    v0 = None
+    for s in subjects:
+        for img in s:
+            arr = safe_load(img)
+            assert_array_2axis(arr, "imgs element")
+            if v0 is None:
+                v0 = arr.shape[0]
+            elif arr.shape[0] != v0:
+                raise ValueError("All images must have the same number "
+                                 "of voxels")
+    return subjects, n_subjects, n_sessions
+
+
+def check_shared_response(shared_response, aggregate="mean",
+                          n_components=None, input_format=None,
+                          n_timeframes=None):
+    """Validate a shared response (array, list of sessions, or list of
+    lists subject x session); returns the list-of-sessions form."""
+    if isinstance(shared_response, np.ndarray):
+        assert_array_2axis(shared_response, "shared_response")
+        out = [shared_response]
+    elif isinstance(shared_response, list):
+        assert_non_empty_list(shared_response, "shared_response")
+        if isinstance(shared_response[0], list):
+            # subject x session: aggregate by mean over subjects
+            n_sess = len(shared_response[0])
+            out = [np.mean([subj[j] for subj in shared_response], axis=0)
+                   for j in range(n_sess)]
+        else:
+            for s in shared_response:
+                assert_array_2axis(s, "shared_response session")
+            out = list(shared_response)
+    else:
+        raise ValueError("shared_response should be an array or a list")
+    if n_components is not None:
+        for s in out:
+            if n_components not in s.shape:
+                raise ValueError("shared response does not match "
+                                 "n_components=%i" % n_components)
+    return out
+
+
+def create_temp_dir(temp_dir):
+    """Create ``temp_dir`` if needed; error if it already exists (use
+    ``.clean()`` between runs)."""
+    if temp_dir is None:
+        return None
+    if not os.path.exists(temp_dir):
+        os.makedirs(temp_dir)
+    else:
+        raise ValueError(
+            "Path %s already exists. When a model is used, filesystem "
+            "should be cleaned by using the .clean() method" % temp_dir)
+
+
+def reduce_data_single(subject_index, session_index, img, atlas=None,
+                       inv_atlas=None, low_ram=False, temp_dir=None):
+    """Atlas-project one [V, T] image -> [T, n_supervoxels] (array, or
+    path when ``low_ram``)."""
+    data = safe_load(img)
+    if inv_atlas is not None:
+        reduced = data.T @ inv_atlas
+    elif atlas is not None:
+        labels = safe_load(atlas).astype(int)
+        n_parcels = int(labels.max())
+        reduced = np.zeros((data.shape[1], n_parcels))
+        for p in range(1, n_parcels + 1):
+            m = labels == p
+            if m.any():
+                reduced[:, p - 1] = data[m].mean(axis=0)
+    else:
+        reduced = data.T.copy()
+    if low_ram and temp_dir is not None:
+        path = os.path.join(
+            temp_dir, "reduced_data_%i_%i.npy"
+            % (subject_index, session_index))
+        np.save(path, reduced)
+        return path
+    return reduced
+
+
+def reduce_data(imgs, atlas, n_jobs=1, low_ram=False, temp_dir=None):
+    """Atlas-project all images -> [n_subjects][n_sessions] reduced
+    data ([T, n_supervoxels] arrays or paths when ``low_ram``)."""
+    subjects, n_subjects, n_sessions = check_imgs(imgs)
+    inv_atlas = None
+    label_atlas = None
+    if atlas is not None:
+        a = safe_load(atlas)
+        if a.ndim == 2:
+            inv_atlas = np.linalg.pinv(a)
+        else:
+            label_atlas = a
+    return [
+        [reduce_data_single(i, j, subjects[i][j], atlas=label_atlas,
+                            inv_atlas=inv_atlas, low_ram=low_ram,
+                            temp_dir=temp_dir)
+         for j in range(n_sessions)]
+        for i in range(n_subjects)]
+
+
+def fast_srm(reduced_data_list, n_iter=10, n_components=None,
+             low_ram=False, seed=0):
+    """Shared response in reduced space (list of [k, T] per session)
+    from [n_subjects][n_sessions] reduced data."""
+    if isinstance(reduced_data_list, np.ndarray) and \
+            reduced_data_list.ndim == 3:
+        reduced_data_list = [[r] for r in reduced_data_list]
+    n_subjects = len(reduced_data_list)
+    n_sessions = len(reduced_data_list[0])
+    # concatenate sessions in time, run DetSRM in the reduced space
+    data = [np.concatenate(
+        [safe_load(reduced_data_list[i][j]) for j in range(n_sessions)],
+        axis=0).T for i in range(n_subjects)]    # [n_sup, T_total]
+    srm = DetSRM(n_iter=n_iter, features=n_components, rand_seed=seed)
+    srm.fit(data)
+    s = srm.s_                                   # [k, T_total]
+    bounds = np.cumsum(
+        [0] + [safe_load(reduced_data_list[0][j]).shape[0]
+               for j in range(n_sessions)])
+    return [s[:, bounds[j]:bounds[j + 1]] for j in range(n_sessions)]
+
+
+def lowram_srm(reduced_data_list, n_iter=10, n_components=None):
+    """Memory-lean variant of ``fast_srm`` (same computation here —
+    288 GB HBM / host RAM makes the reference's disk-streamed variant
+    unnecessary, the API is kept for parity)."""
+    return fast_srm(reduced_data_list, n_iter=n_iter,
+                    n_components=n_components)
