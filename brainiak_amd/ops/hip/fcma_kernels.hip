@@ -444,6 +444,115 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot2(
 }
 
 // ===========================================================================
+// k_corr_norm_dot3: thread-owns-all-epochs form (P <= 4).
+// The dot2 kernel's (chalf,p,v)->normalize->(c,v) split forces the fp32
+// corr tile through LDS (the 4 epochs of one (c,v) column live in 4
+// different waves).  Here ONE thread owns (v, all P epochs): it dots,
+// Fisher-z's, z-scores and stores each c's column entirely in
+// registers — no corr LDS tile (4 KB a-tile only -> max occupancy),
+// no inter-stage barrier, one batched store pass.
+// ===========================================================================
+#define C3_CT 32
+#define C3_VT 256
+
+template <int TP, int TL>
+__global__ __launch_bounds__(256) void k_corr_norm_dot3(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ zOut, ll E, ll Lrt, ll VA, ll VB, ll s0,
+    ll C, int Prt, int mode, ll zstride) {
+    static_assert(TL % 2 == 0 && TP >= 2 && TP <= 4,
+                  "dot3 kernel: even L, P in {2,4}");
+    constexpr int P = TP;
+    constexpr int KP = TL / 2;
+    const ll L = TL;
+    (void)Lrt; (void)Prt;
+    const ll nSubj = E / P;
+    const ll cTiles = (C + C3_CT - 1) / C3_CT;
+    const ll vTiles = (VB + C3_VT - 1) / C3_VT;
+    ll b = blockIdx.x;
+    const ll vt = b % vTiles; b /= vTiles;
+    const ll s = b % nSubj;   b /= nSubj;
+    const ll ct = b;
+    if (ct >= cTiles) return;
+    const ll c0 = ct * C3_CT;
+    const int CT = (int)min((ll)C3_CT, C - c0);
+    const ll v = vt * (ll)C3_VT + threadIdx.x;
+    const int tid = threadIdx.x;
+
+    // a_tile [c][kp][p][2] bf16: one bf16x8 broadcast per (c, kp)
+    // yields the k-pair for all four epochs
+    __shared__ bf16_t a_tile[C3_CT][KP][4][2];
+    for (int idx = tid; idx < C3_CT * KP * P; idx += 256) {
+        int p = idx % P;
+        int kp = (idx / P) % KP;
+        int c = idx / (P * KP);
+        #pragma unroll
+        for (int i2 = 0; i2 < 2; ++i2) {
+            bf16_t val = (bf16_t)0.0f;
+            if (c < CT)
+                val = A[((ll)(s * P + p) * L + 2 * kp + i2) * VA
+                        + (s0 + c0 + c)];
+            a_tile[c][kp][p][i2] = val;
+        }
+    }
+    __syncthreads();
+    if (v >= VB) return;
+
+    // this thread's B columns for all P epochs (source-dtype bf16)
+    bf16x2_t bp[P][KP];
+    #pragma unroll
+    for (int p = 0; p < P; ++p) {
+        const bf16_t* brow = B + ((ll)(s * P + p) * L) * VB + v;
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp) {
+            bf16x2_t t;
+            t[0] = *(const __bf16*)&brow[(ll)(2 * kp) * VB];
+            t[1] = *(const __bf16*)&brow[(ll)(2 * kp + 1) * VB];
+            bp[p][kp] = t;
+        }
+    }
+
+    for (int c = 0; c < CT; ++c) {
+        float acc[P];
+        #pragma unroll
+        for (int p = 0; p < P; ++p) acc[p] = 0.f;
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp) {
+            bf16x8 raw = *(const bf16x8*)&a_tile[c][kp][0][0];
+            #pragma unroll
+            for (int p = 0; p < P; ++p) {
+                short2_t s2 = {raw[2 * p], raw[2 * p + 1]};
+                acc[p] = __builtin_amdgcn_fdot2_f32_bf16(
+                    __builtin_bit_cast(bf16x2_t, s2), bp[p][kp],
+                    acc[p], false);
+            }
+        }
+        float z[P];
+        float mean = 0.f, sq = 0.f;
+        #pragma unroll
+        for (int p = 0; p < P; ++p) {
+            z[p] = fisher_z(acc[p]);
+            mean += z[p]; sq += z[p] * z[p];
+        }
+        mean /= (float)P;
+        float var = sq / (float)P - mean * mean;
+        float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+        if (mode == 3) {
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                if (zstride < 0)
+                    zOut[(size_t)tid] = (bf16_t)((z[p] - mean) * inv);
+        } else {
+            bf16_t* dst = zOut
+                + ((c0 + c) * zstride + s * (ll)P) * VB + v;
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                dst[(size_t)p * VB] = (bf16_t)((z[p] - mean) * inv);
+        }
+    }
+}
+
+// ===========================================================================
 // k_corr_norm_mfma: the MFMA form of k_corr_norm for L <= 32.
 // PMC evidence (profiles/README.md): the VALU form is issue-bound at
 // ~5 instructions per useful FMA; one v_mfma_f32_16x16x32_bf16 computes a
@@ -1071,7 +1180,10 @@ static int corr_variant() {
     static int v = -1;
     if (v < 0) {
         const char* e = getenv("BRAINIAK_CORR_KERNEL");
-        v = (e && strcmp(e, "classic") == 0) ? 0 : 1;
+        v = 2;   // dot3 where applicable, dot2 otherwise (measured
+                 // 14.5 vs 16.5 ms/step no-cv, profiles/README.md)
+        if (e && strcmp(e, "classic") == 0) v = 0;
+        if (e && strcmp(e, "dot2") == 0) v = 1;
     }
     return v;
 }
@@ -1107,7 +1219,18 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
         return;
     }
 #endif
-    if (corr_variant() == 1 && (TL % 2) == 0) {
+    if (corr_variant() == 2 && (TL % 2) == 0 && mode != 1 && mode != 2
+        && (TP == 2 || TP == 4)) {
+        if constexpr (TL % 2 == 0 && TP >= 2 && TP <= 4) {
+            ll grid3 = ceil_div(C, C3_CT) * nSubj * ceil_div(VB, C3_VT);
+            hipLaunchKernelGGL((k_corr_norm_dot3<TP, TL>), dim3(grid3),
+                               dim3(256), 0, stream, (const bf16_t*)A,
+                               (const bf16_t*)B, (bf16_t*)zOut, E, L,
+                               VA, VB, s0, C, P, mode, zstride);
+            return;
+        }
+    }
+    if (corr_variant() >= 1 && (TL % 2) == 0) {
         ll gridd = ceil_div(C, dot2_ct(P)) * nSubj * ceil_div(VB, CN_VT);
         if (smem > 64 * 1024) {
             // gfx950 has 160 KB LDS but dynamic allocations above the
