@@ -452,10 +452,9 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot2(
 // registers — no corr LDS tile (4 KB a-tile only -> max occupancy),
 // no inter-stage barrier, one batched store pass.
 // ===========================================================================
-#define C3_CT 32
 #define C3_VT 256
 
-template <int TP, int TL>
+template <int TP, int TL, int C3_CT>
 __global__ __launch_bounds__(256) void k_corr_norm_dot3(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ zOut, ll E, ll Lrt, ll VA, ll VB, ll s0,
@@ -1222,11 +1221,48 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
     if (corr_variant() == 2 && (TL % 2) == 0 && mode != 1 && mode != 2
         && (TP == 2 || TP == 4)) {
         if constexpr (TL % 2 == 0 && TP >= 2 && TP <= 4) {
-            ll grid3 = ceil_div(C, C3_CT) * nSubj * ceil_div(VB, C3_VT);
-            hipLaunchKernelGGL((k_corr_norm_dot3<TP, TL>), dim3(grid3),
-                               dim3(256), 0, stream, (const bf16_t*)A,
-                               (const bf16_t*)B, (bf16_t*)zOut, E, L,
-                               VA, VB, s0, C, P, mode, zstride);
+            static int ct3 = -1;
+            if (ct3 < 0) {
+                const char* e = getenv("BRAINIAK_DOT3_CT");
+                // measured sweep (profiles/README.md): 32 -> 15.1,
+                // 64 -> 13.8, 128 -> 13.2, 256 -> 12.9, 512 -> 14.1
+                // ms/step no-cv; 128 and 256 tie on the full step,
+                // 128 keeps more blocks in flight for small shards
+                ct3 = e ? atoi(e) : 128;
+                if (ct3 != 32 && ct3 != 64 && ct3 != 256 && ct3 != 512)
+                    ct3 = 128;
+            }
+            ll grid3 = ceil_div(C, ct3) * nSubj * ceil_div(VB, C3_VT);
+            if (ct3 == 64)
+                hipLaunchKernelGGL((k_corr_norm_dot3<TP, TL, 64>),
+                                   dim3(grid3), dim3(256), 0, stream,
+                                   (const bf16_t*)A, (const bf16_t*)B,
+                                   (bf16_t*)zOut, E, L, VA, VB, s0, C,
+                                   P, mode, zstride);
+            else if (ct3 == 128)
+                hipLaunchKernelGGL((k_corr_norm_dot3<TP, TL, 128>),
+                                   dim3(grid3), dim3(256), 0, stream,
+                                   (const bf16_t*)A, (const bf16_t*)B,
+                                   (bf16_t*)zOut, E, L, VA, VB, s0, C,
+                                   P, mode, zstride);
+            else if (ct3 == 512)
+                hipLaunchKernelGGL((k_corr_norm_dot3<TP, TL, 512>),
+                                   dim3(grid3), dim3(256), 0, stream,
+                                   (const bf16_t*)A, (const bf16_t*)B,
+                                   (bf16_t*)zOut, E, L, VA, VB, s0, C,
+                                   P, mode, zstride);
+            else if (ct3 == 256)
+                hipLaunchKernelGGL((k_corr_norm_dot3<TP, TL, 256>),
+                                   dim3(grid3), dim3(256), 0, stream,
+                                   (const bf16_t*)A, (const bf16_t*)B,
+                                   (bf16_t*)zOut, E, L, VA, VB, s0, C,
+                                   P, mode, zstride);
+            else
+                hipLaunchKernelGGL((k_corr_norm_dot3<TP, TL, 32>),
+                                   dim3(grid3), dim3(256), 0, stream,
+                                   (const bf16_t*)A, (const bf16_t*)B,
+                                   (bf16_t*)zOut, E, L, VA, VB, s0, C,
+                                   P, mode, zstride);
             return;
         }
     }
