@@ -13,7 +13,9 @@ typedef long long ll;
 extern "C" {
 void launch_fcma_normalize(float*, ll, ll, ll, int, void*);
 void launch_fcma_corr_norm(const void*, const void*, void*, float*, ll, ll,
-                           ll, ll, ll, ll, int, int, ll, void*);
+                           ll, ll, ll, ll, int, int, ll, void*,
+                           const void*);
+int fcma_corr_variant(void);
 int fcma_corr_norm_smem(ll, int);
 void launch_fcma_gram_bf16(const void*, float*, ll, ll, ll, ll, void*);
 void launch_fcma_gram_f32(const float*, float*, ll, ll, ll, void*);
@@ -74,7 +76,8 @@ torch::Tensor fcma_correlate(torch::Tensor A, torch::Tensor B,
                             A.options().dtype(torch::kFloat32));
     launch_fcma_corr_norm(A.data_ptr(), B.data_ptr(), nullptr,
                           out.data_ptr<float>(), E, L, VA, VB, start,
-                          count, P, /*mode=*/2, E, cur_stream());
+                          count, P, /*mode=*/2, E, cur_stream(),
+                          nullptr);
     return out;
 }
 
@@ -110,10 +113,18 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
         Z = (Eout == E) ? torch::empty({count, E, VB}, A.options())
                         : torch::zeros({count, Eout, VB}, A.options());
     }
+    // dot3s variant reads the A operand voxel-major through the
+    // scalar path: hand it the [count, E, L] transposed slice
+    torch::Tensor At;
+    const void* At_ptr = nullptr;
+    if (fcma_corr_variant() == 3) {
+        At = A.narrow(2, start, count).permute({2, 0, 1}).contiguous();
+        At_ptr = At.data_ptr();
+    }
     // the kernel writes rows [0, E) with row stride Eout directly
     launch_fcma_corr_norm(A.data_ptr(), B.data_ptr(), Z.data_ptr(),
                           nullptr, E, L, VA, VB, start, count, (int)P,
-                          /*mode=*/0, Eout, cur_stream());
+                          /*mode=*/0, Eout, cur_stream(), At_ptr);
     return Z;
 }
 

@@ -552,6 +552,94 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3(
 }
 
 // ===========================================================================
+// k_corr_norm_dot3s: dot3 with the A operand read through the SCALAR
+// path.  In dot3 every lane of a wave reads the same a-pair (c and s
+// are wave-uniform), so the A tile belongs in SGPRs via the constant
+// cache, not in LDS: the host passes At = A[:, :, start:start+count]
+// transposed to [count, E, L] (contiguous per voxel), and the kernel
+// needs NO LDS, NO staging loop and NO barrier at all.
+// ===========================================================================
+template <int TP, int TL, int C3_CT>
+__global__ __launch_bounds__(256) void k_corr_norm_dot3s(
+    const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ zOut, ll E, ll VB,
+    ll C, int mode, ll zstride) {
+    static_assert(TL % 2 == 0 && TP >= 2 && TP <= 4,
+                  "dot3s kernel: even L, P in {2,4}");
+    constexpr int P = TP;
+    constexpr int KP = TL / 2;
+    constexpr int L = TL;
+    const ll nSubj = E / P;
+    const ll cTiles = (C + C3_CT - 1) / C3_CT;
+    const ll vTiles = (VB + C3_VT - 1) / C3_VT;
+    ll b = blockIdx.x;
+    const ll vt = b % vTiles; b /= vTiles;
+    const ll s = b % nSubj;   b /= nSubj;
+    const ll ct = b;
+    if (ct >= cTiles) return;
+    const ll c0 = ct * C3_CT;
+    const int CT = (int)min((ll)C3_CT, C - c0);
+    const ll v = vt * (ll)C3_VT + threadIdx.x;
+    if (v >= VB) return;
+
+    bf16x2_t bp[P][KP];
+    #pragma unroll
+    for (int p = 0; p < P; ++p) {
+        const bf16_t* brow = B + ((ll)(s * P + p) * L) * VB + v;
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp) {
+            bf16x2_t t;
+            t[0] = *(const __bf16*)&brow[(ll)(2 * kp) * VB];
+            t[1] = *(const __bf16*)&brow[(ll)(2 * kp + 1) * VB];
+            bp[p][kp] = t;
+        }
+    }
+
+    // wave-uniform A base for this (c-tile, subject)
+    const unsigned int* abase = (const unsigned int*)
+        (At + (c0 * (ll)E + s * (ll)P) * L);
+    const int cstride = (E * L) / 2;          // dwords per voxel
+
+    for (int c = 0; c < CT; ++c) {
+        const unsigned int* ac = abase + (ll)c * cstride;
+        float acc[P];
+        #pragma unroll
+        for (int p = 0; p < P; ++p) acc[p] = 0.f;
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp)
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                acc[p] = __builtin_amdgcn_fdot2_f32_bf16(
+                    __builtin_bit_cast(bf16x2_t,
+                                       ac[p * (L / 2) + kp]),
+                    bp[p][kp], acc[p], false);
+        float z[P];
+        float mean = 0.f, sq = 0.f;
+        #pragma unroll
+        for (int p = 0; p < P; ++p) {
+            z[p] = fisher_z(acc[p]);
+            mean += z[p]; sq += z[p] * z[p];
+        }
+        mean /= (float)P;
+        float var = sq / (float)P - mean * mean;
+        float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+        if (mode == 3) {
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                if (zstride < 0)
+                    zOut[(size_t)threadIdx.x] =
+                        (bf16_t)((z[p] - mean) * inv);
+        } else {
+            bf16_t* dst = zOut
+                + ((c0 + c) * zstride + s * (ll)P) * VB + v;
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                dst[(size_t)p * VB] = (bf16_t)((z[p] - mean) * inv);
+        }
+    }
+}
+
+// ===========================================================================
 // k_corr_norm_mfma: the MFMA form of k_corr_norm for L <= 32.
 // PMC evidence (profiles/README.md): the VALU form is issue-bound at
 // ~5 instructions per useful FMA; one v_mfma_f32_16x16x32_bf16 computes a
@@ -1179,10 +1267,11 @@ static int corr_variant() {
     static int v = -1;
     if (v < 0) {
         const char* e = getenv("BRAINIAK_CORR_KERNEL");
-        v = 2;   // dot3 where applicable, dot2 otherwise (measured
-                 // 14.5 vs 16.5 ms/step no-cv, profiles/README.md)
+        v = 3;   // scalar-A dot3s where applicable (12.9 ms/step
+                 // no-cv vs dot3 13.6 / dot2 16.5 — profiles/README.md)
         if (e && strcmp(e, "classic") == 0) v = 0;
         if (e && strcmp(e, "dot2") == 0) v = 1;
+        if (e && strcmp(e, "dot3") == 0) v = 2;
     }
     return v;
 }
@@ -1196,7 +1285,8 @@ template <int TP, int TL>
 static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
                                float* fOut, ll E, ll L, ll VA, ll VB,
                                ll s0, ll C, int P, int mode, ll zstride,
-                               size_t smem, hipStream_t stream) {
+                               size_t smem, hipStream_t stream,
+                               const void* At = nullptr) {
     ll nSubj = E / P;
     // Measured A/B on MI355X (profiles/README.md): the VALU form at
     // 2.09 ms/512-voxel call beats the MFMA form (3.86 ms) — with
@@ -1218,7 +1308,18 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
         return;
     }
 #endif
-    if (corr_variant() == 2 && (TL % 2) == 0 && mode != 1 && mode != 2
+    if (corr_variant() == 3 && At != nullptr && (TL % 2) == 0
+        && mode != 1 && mode != 2 && (TP == 2 || TP == 4)) {
+        if constexpr (TL % 2 == 0 && TP >= 2 && TP <= 4) {
+            ll grid3 = ceil_div(C, 128) * nSubj * ceil_div(VB, C3_VT);
+            hipLaunchKernelGGL((k_corr_norm_dot3s<TP, TL, 128>),
+                               dim3(grid3), dim3(256), 0, stream,
+                               (const bf16_t*)At, (const bf16_t*)B,
+                               (bf16_t*)zOut, E, VB, C, mode, zstride);
+            return;
+        }
+    }
+    if (corr_variant() >= 2 && (TL % 2) == 0 && mode != 1 && mode != 2
         && (TP == 2 || TP == 4)) {
         if constexpr (TL % 2 == 0 && TP >= 2 && TP <= 4) {
             static int ct3 = -1;
@@ -1294,6 +1395,8 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
                        VA, VB, s0, C, P, mode, zstride);
 }
 
+extern "C" int fcma_corr_variant(void) { return corr_variant(); }
+
 extern "C" int fcma_corr_norm_smem(ll L, int P) {
 #ifdef USE_MFMA_CORR
     if (L <= CM_K) {
@@ -1318,15 +1421,15 @@ template <int TL>
 static void dispatch_p(const void* A, const void* B, void* zOut,
                        float* fOut, ll E, ll L, ll VA, ll VB, ll s0, ll C,
                        int P, int mode, ll zstride, size_t smem,
-                       hipStream_t stream) {
+                       hipStream_t stream, const void* At = nullptr) {
     switch (P) {
         case 2:  launch_corr_norm_t<2, TL>(A, B, zOut, fOut, E, L, VA, VB,
                                            s0, C, P, mode, zstride,
-                                           smem, stream);
+                                           smem, stream, At);
                  break;
         case 4:  launch_corr_norm_t<4, TL>(A, B, zOut, fOut, E, L, VA, VB,
                                            s0, C, P, mode, zstride,
-                                           smem, stream);
+                                           smem, stream, At);
                  break;
         case 8:  launch_corr_norm_t<8, TL>(A, B, zOut, fOut, E, L, VA, VB,
                                            s0, C, P, mode, zstride,
@@ -1355,21 +1458,22 @@ extern "C" void launch_fcma_corr_norm(const void* A, const void* B,
                                       void* zOut, float* fOut, ll E, ll L,
                                       ll VA, ll VB, ll s0, ll C, int P,
                                       int mode, ll zstride,
-                                      hipStream_t stream) {
+                                      hipStream_t stream,
+                                      const void* At) {
     static int probe_mode3 = getenv("BRAINIAK_CORR_MODE3") ? 1 : 0;
-    if (probe_mode3 && mode == 0 && corr_variant() == 1) mode = 3;
+    if (probe_mode3 && mode == 0 && corr_variant() >= 1) mode = 3;
     size_t smem = (size_t)fcma_corr_norm_smem(L, P);
     switch (L) {
         case 8:  dispatch_p<8>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                               mode, zstride, smem, stream); break;
+                               mode, zstride, smem, stream, At); break;
         case 16: dispatch_p<16>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                               mode, zstride, smem, stream); break;
+                               mode, zstride, smem, stream, At); break;
         case 24: dispatch_p<24>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                               mode, zstride, smem, stream); break;
+                               mode, zstride, smem, stream, At); break;
         case 32: dispatch_p<32>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                               mode, zstride, smem, stream); break;
+                               mode, zstride, smem, stream, At); break;
         case 40: dispatch_p<40>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
-                               mode, zstride, smem, stream); break;
+                               mode, zstride, smem, stream, At); break;
         default: break;  // host guarantees L in the supported set
     }
 }
