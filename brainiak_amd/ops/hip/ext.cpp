@@ -133,10 +133,11 @@ torch::Tensor fcma_gram_bf16(torch::Tensor Z) {
     ll C = Z.size(0), E = Z.size(1), V = Z.size(2);
     TORCH_CHECK(E % 64 == 0, "E must be a multiple of 64 (host pads)");
     ll eb = E / 64;
-    // V-split so small-C calls still fill 256 CUs (>= ~512 blocks)
+    // V-split for latency hiding: PMC shows the kernel 84 % parked on
+    // its serial k-tile loop at 4 blocks/CU — target ~8 blocks/CU
     ll base = C * eb * eb;
     ll ktAll = (V + 63) / 64;
-    ll nsplit = std::min(ktAll, std::max((ll)1, (511 + base) / base));
+    ll nsplit = std::min(ktAll, std::max((ll)1, (2047 + base) / base));
     if (nsplit <= 1) {
         auto G = torch::empty({C, E, E},
                               Z.options().dtype(torch::kFloat32));
