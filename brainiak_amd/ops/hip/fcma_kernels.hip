@@ -6,16 +6,21 @@
 //  - ref src/brainiak/fcma/cython_blas.pyx:118-207 (per-voxel ssyrk)  -> k_gram_bf16 (MFMA) / k_gram_f32
 //
 // Design (per /opt/skills/guides/cdna_hip_programming.md):
-//  * wave64, 256-thread blocks (4 waves).
-//  * k_corr_norm fuses correlation + Fisher-z + within-subject z-score per
-//    (voxel-tile, subject, column-tile): the fp32 correlation tensor stays
-//    in LDS; only bf16 Z (half bytes) or the requested fp32 view reaches HBM.
-//  * k_gram_bf16: G_c = Z_c Z_c^T via v_mfma_f32_16x16x32_bf16, one block
-//    per voxel x 64x64 band pair, double-buffered LDS K-tiles, padded rows.
-//  * epochs_per_subj is a template parameter for the common values so the
-//    per-column z[] arrays stay in registers (runtime-indexed ext arrays
-//    drop to scratch - guide rule 20); a generic two-pass path covers the
-//    rest.
+//  * wave64, 256-thread blocks.
+//  * The correlation+normalization stage ships FOUR measured variants
+//    (profiles/README.md has the ladder): classic (pk_fma),
+//    dot2 (v_dot2c_f32_bf16 + fused normalize/store), dot3
+//    (thread-owns-all-epochs, no corr LDS tile), and the default dot3s
+//    (dot3 with the wave-uniform A operand on the SCALAR path — no LDS,
+//    no barriers).  BRAINIAK_CORR_KERNEL selects a variant at runtime.
+//  * k_gram_bf16: G_c = Z_c Z_c^T via v_mfma_f32_16x16x32_bf16, T14
+//    issue-early K-tiles, V-split partials so small-C calls fill the
+//    256 CUs.  k_fused_corr_gram: the whole chunk in one kernel
+//    (opt-in; measured slower than the streamed pipeline — see notes).
+//  * epochs_per_subj is a template parameter for the common values so
+//    per-column z[] arrays stay in registers (runtime-indexed arrays
+//    drop to scratch - guide rule 20); a generic two-pass path covers
+//    the rest.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
