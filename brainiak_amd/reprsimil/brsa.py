@@ -505,7 +505,11 @@ class BRSA(_BRSACore):
 
         n_nureg = self.n_nureg
         params = None
-        rounds = 2 if self.auto_nuisance else 1
+        # with auto_nuisance the fit/re-estimate alternation runs up to
+        # n_iter rounds, stopping when the joint nll stops improving by
+        # tol (the reference's outer fit loop, ref brsa.py:1309-1529)
+        rounds = max(2, self.n_iter) if self.auto_nuisance else 1
+        prev_nll = None
         for round_i in range(rounds):
             # flat-prior X0 betas → project X and Y off X0's column space
             Xp = _project_out(X_design, X0)
@@ -514,6 +518,10 @@ class BRSA(_BRSACore):
             Y_t = torch.as_tensor(Yp, dtype=_DT)
             params, nll, quads = self._fit_once(X_t, Y_t, C, V, T, rank,
                                                 init=params, gp=gp)
+            if prev_nll is not None and \
+                    abs(prev_nll - nll) <= self.tol * abs(prev_nll):
+                break
+            prev_nll = nll
             if self.auto_nuisance and round_i < rounds - 1:
                 comps, n_nureg = self._residual_nuisance(
                     X_design, Y_data, X0, n_nureg)
