@@ -4,6 +4,7 @@ TRs, CPU, world_size=1 (the plumbing check).  One step = one full
 SRM fit (n_iter EM iterations)."""
 
 import argparse
+import os
 import sys
 from pathlib import Path
 
@@ -25,6 +26,10 @@ def main():
     args = ap.parse_args()
 
     rank, world, device, _ = dist_setup()
+    # tiny-matrix EM: BLAS/torch thread fan-out costs more than it
+    # saves and makes the measurement host-dependent
+    import torch
+    torch.set_num_threads(min(8, os.cpu_count() or 8))
     from brainiak_amd.funcalign.srm import SRM
 
     rng = np.random.RandomState(0)
