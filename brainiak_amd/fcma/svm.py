@@ -152,8 +152,8 @@ def _accuracy_gpu(kernels: torch.Tensor, labels: np.ndarray, num_folds: int,
     from .. import ops
     if ops.has_hip():
         folds = stratified_folds(labels, num_folds)
-        if max(len(tr) for tr, _ in folds) <= 64 and \
-                max(len(te) for _, te in folds) <= 64:
+        if max(len(tr) for tr, _ in folds) <= 128 and \
+                max(len(te) for _, te in folds) <= 128:
             return _accuracy_gpu_hip(kernels, labels, num_folds, C, tol)
     y_np = np.where(labels == classes[1], 1.0, -1.0)
     folds = stratified_folds(labels, num_folds)

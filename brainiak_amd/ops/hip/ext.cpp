@@ -26,7 +26,7 @@ void launch_tfa_recon(const float*, const float*, const float*, float*,
                       ll, ll, int, float, void*);
 void launch_svm_cv(const float*, const float*, const int*, const int*,
                    const int*, const int*, int*, ll, int, int, float,
-                   float, int, void*);
+                   float, int, ll, void*);
 ll fcma_supported_L(ll);
 int fcma_corr_norm_smem(ll, int);
 int fcma_fused_gram_supported(ll, int, ll);
@@ -300,13 +300,17 @@ torch::Tensor svm_cv(torch::Tensor kernels, torch::Tensor y,
                 "idx must be [F,E]");
     auto nt = n_train.to(torch::kInt32).contiguous();
     auto ns = n_test.to(torch::kInt32).contiguous();
+    ll max_n = std::max<ll>(nt.max().item<int>(),
+                            ns.max().item<int>());
+    TORCH_CHECK(max_n <= 128,
+                "svm_cv supports fold sizes up to 128 samples");
     auto correct = torch::zeros({C, F}, kernels.options()
                                             .dtype(torch::kInt32));
     launch_svm_cv(kernels.data_ptr<float>(), y.data_ptr<float>(),
                   train_idx.data_ptr<int>(), test_idx.data_ptr<int>(),
                   nt.data_ptr<int>(), ns.data_ptr<int>(),
                   correct.data_ptr<int>(), C, E, F, (float)Creg,
-                  (float)tol, (int)max_iter, cur_stream());
+                  (float)tol, (int)max_iter, max_n, cur_stream());
     return correct;
 }
 

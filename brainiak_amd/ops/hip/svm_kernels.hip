@@ -5,18 +5,20 @@
 // kernel launch solves every (voxel, fold) dual QP: one workgroup (64
 // threads = 1 wave) per problem, Q staged in LDS, SMO with
 // maximal-violating-pair selection (LIBSVM WSS1), then the fold's test
-// accuracy computed in the same block.  n_train <= 64 (one dual variable
-// per lane).
+// accuracy computed in the same block.  VPL dual variables per lane
+// (variable v = k*64 + lane): VPL=1 covers n_train <= 64, VPL=2 covers
+// n_train <= 128 (Q tile 66 KB LDS, dynamic).
 //
 // Output: correct-prediction counts [C, F] int32; host divides by test
 // counts and averages folds.
 
 #include <hip/hip_runtime.h>
+#include <stdint.h>
+#include <type_traits>
 
 typedef long long ll;
 
-#define SVM_MAXN 64
-
+template <int VPL>
 __global__ __launch_bounds__(64) void k_svm_cv(
     const float* __restrict__ kernels,  // [C, E, E]
     const float* __restrict__ y,        // [E] in {-1, +1}
@@ -26,6 +28,7 @@ __global__ __launch_bounds__(64) void k_svm_cv(
     const int* __restrict__ n_test,     // [F]
     int* __restrict__ correct,          // [C, F]
     ll C, int E, int F, float Creg, float tol, int max_iter) {
+    constexpr int MAXN = VPL * 64;
     const ll blk = blockIdx.x;
     const ll c = blk / F;
     const int f = (int)(blk % F);
@@ -34,9 +37,14 @@ __global__ __launch_bounds__(64) void k_svm_cv(
     const int n = n_train[f];
     const int m = n_test[f];
 
-    __shared__ float Q[SVM_MAXN][SVM_MAXN + 1];
-    __shared__ float ys[SVM_MAXN];
-    __shared__ float alpha_s[SVM_MAXN];
+    // fp32 Q for the one-variable-per-lane case; fp16 Q (0.05 % rel)
+    // for n <= 128 keeps the tile at 33 KB — under the 64 KB dynamic
+    // LDS launch limit and 4 QPs resident per CU
+    using QT = typename std::conditional<VPL == 1, float, _Float16>::type;
+    extern __shared__ char smem_raw[];
+    QT (*Q)[MAXN + 1] = (QT (*)[MAXN + 1])smem_raw;
+    float* ys = (float*)(smem_raw + sizeof(QT) * MAXN * (MAXN + 1));
+    float* alpha_s = ys + MAXN;
     __shared__ int pair[2];
     __shared__ float delta[2];
 
@@ -44,35 +52,52 @@ __global__ __launch_bounds__(64) void k_svm_cv(
     const int* tr = train_idx + (ll)f * E;
 
     // stage labels + Q = y_i y_j K[tr_i][tr_j]
-    float yl = 0.0f;
-    if (lane < n) {
-        yl = y[tr[lane]];
-        ys[lane] = yl;
+    float yl[VPL];
+    #pragma unroll
+    for (int k = 0; k < VPL; ++k) {
+        int v = k * 64 + lane;
+        yl[k] = 0.0f;
+        if (v < n) {
+            yl[k] = y[tr[v]];
+            ys[v] = yl[k];
+        }
     }
     __syncthreads();
-    if (lane < n) {
-        const float* krow = Kc + (ll)tr[lane] * E;
-        for (int j = 0; j < n; ++j)
-            Q[lane][j] = yl * ys[j] * krow[tr[j]];
+    #pragma unroll
+    for (int k = 0; k < VPL; ++k) {
+        int v = k * 64 + lane;
+        if (v < n) {
+            const float* krow = Kc + (ll)tr[v] * E;
+            for (int j = 0; j < n; ++j)
+                Q[v][j] = (QT)(yl[k] * ys[j] * krow[tr[j]]);
+        }
     }
     __syncthreads();
 
-    // SMO: one dual variable per lane
-    float alpha = 0.0f;
-    float grad = -1.0f;           // (Q alpha)_i - 1
-    const bool valid = lane < n;
+    // SMO: VPL dual variables per lane
+    float alpha[VPL], grad[VPL];
+    #pragma unroll
+    for (int k = 0; k < VPL; ++k) { alpha[k] = 0.0f; grad[k] = -1.0f; }
 
     float b = 0.0f;
     for (int iter = 0; iter < max_iter; ++iter) {
-        float score = -yl * grad;
-        bool in_up = valid && ((yl > 0.f && alpha < Creg - 1e-12f) ||
-                               (yl < 0.f && alpha > 1e-12f));
-        bool in_low = valid && ((yl > 0.f && alpha > 1e-12f) ||
-                                (yl < 0.f && alpha < Creg - 1e-12f));
-        // wave argmax over I_up / argmin over I_low
-        float up = in_up ? score : -3.0e38f;
-        float lo = in_low ? score : 3.0e38f;
-        int up_i = lane, lo_i = lane;
+        // per-lane best over its VPL variables, then wave reduce
+        float up = -3.0e38f, lo = 3.0e38f;
+        int up_i = 0, lo_i = 0;
+        #pragma unroll
+        for (int k = 0; k < VPL; ++k) {
+            int v = k * 64 + lane;
+            bool valid = v < n;
+            float score = -yl[k] * grad[k];
+            bool in_up = valid &&
+                ((yl[k] > 0.f && alpha[k] < Creg - 1e-12f) ||
+                 (yl[k] < 0.f && alpha[k] > 1e-12f));
+            bool in_low = valid &&
+                ((yl[k] > 0.f && alpha[k] > 1e-12f) ||
+                 (yl[k] < 0.f && alpha[k] < Creg - 1e-12f));
+            if (in_up && score > up) { up = score; up_i = v; }
+            if (in_low && score < lo) { lo = score; lo_i = v; }
+        }
         #pragma unroll
         for (int d = 32; d > 0; d >>= 1) {
             float u2 = __shfl_xor(up, d);
@@ -86,12 +111,17 @@ __global__ __launch_bounds__(64) void k_svm_cv(
         b = 0.5f * (up + lo);
 
         // publish lane-private alpha so lane 0 can read the pair's values
-        if (valid) alpha_s[lane] = alpha;
+        #pragma unroll
+        for (int k = 0; k < VPL; ++k) {
+            int v = k * 64 + lane;
+            if (v < n) alpha_s[v] = alpha[k];
+        }
         __syncthreads();
         if (lane == 0) {
             int i = up_i, j = lo_i;
             float yi = ys[i], yj = ys[j];
-            float eta = Q[i][i] + Q[j][j] - 2.0f * Q[i][j] * yi * yj;
+            float eta = (float)Q[i][i] + (float)Q[j][j]
+                        - 2.0f * (float)Q[i][j] * yi * yj;
             eta = fmaxf(eta, 1e-12f);
             float t = (up - lo) / eta;
             float ai = alpha_s[i], aj = alpha_s[j];
@@ -106,24 +136,37 @@ __global__ __launch_bounds__(64) void k_svm_cv(
         __syncthreads();
         const int i = pair[0], j = pair[1];
         const float dai = delta[0], daj = delta[1];
-        if (lane == i) alpha += dai;
-        if (lane == j) alpha += daj;
-        if (valid) grad += Q[i][lane] * dai + Q[j][lane] * daj;
+        #pragma unroll
+        for (int k = 0; k < VPL; ++k) {
+            int v = k * 64 + lane;
+            if (v == i) alpha[k] += dai;
+            if (v == j) alpha[k] += daj;
+            if (v < n) grad[k] += (float)Q[i][v] * dai
+                                   + (float)Q[j][v] * daj;
+        }
         __syncthreads();
     }
 
     // publish final alpha, predict test samples
-    if (valid) alpha_s[lane] = alpha * yl;   // coef_i = alpha_i * y_i
+    #pragma unroll
+    for (int k = 0; k < VPL; ++k) {
+        int v = k * 64 + lane;
+        if (v < n) alpha_s[v] = alpha[k] * yl[k];  // coef_i = alpha_i y_i
+    }
     __syncthreads();
     const int* te = test_idx + (ll)f * E;
     int correct_local = 0;
-    if (lane < m) {
-        const float* krow = Kc + (ll)te[lane] * E;
-        float dec = b;
-        for (int i = 0; i < n; ++i)
-            dec = fmaf(alpha_s[i], krow[tr[i]], dec);
-        float yt = y[te[lane]];
-        correct_local = ((dec > 0.f) == (yt > 0.f)) ? 1 : 0;
+    #pragma unroll
+    for (int k = 0; k < VPL; ++k) {
+        int v = k * 64 + lane;
+        if (v < m) {
+            const float* krow = Kc + (ll)te[v] * E;
+            float dec = b;
+            for (int i = 0; i < n; ++i)
+                dec = fmaf(alpha_s[i], krow[tr[i]], dec);
+            float yt = y[te[v]];
+            correct_local += ((dec > 0.f) == (yt > 0.f)) ? 1 : 0;
+        }
     }
     #pragma unroll
     for (int d = 32; d > 0; d >>= 1)
@@ -135,8 +178,19 @@ extern "C" void launch_svm_cv(const float* kernels, const float* y,
                               const int* train_idx, const int* test_idx,
                               const int* n_train, const int* n_test,
                               int* correct, ll C, int E, int F, float Creg,
-                              float tol, int max_iter, hipStream_t stream) {
-    hipLaunchKernelGGL(k_svm_cv, dim3((unsigned)(C * F)), dim3(64), 0,
-                       stream, kernels, y, train_idx, test_idx, n_train,
-                       n_test, correct, C, E, F, Creg, tol, max_iter);
+                              float tol, int max_iter, ll max_n,
+                              hipStream_t stream) {
+    if (max_n <= 64) {
+        size_t smem = (size_t)64 * 65 * 4 + 2 * 64 * 4;
+        hipLaunchKernelGGL(k_svm_cv<1>, dim3((unsigned)(C * F)), dim3(64),
+                           smem, stream, kernels, y, train_idx, test_idx,
+                           n_train, n_test, correct, C, E, F, Creg, tol,
+                           max_iter);
+        return;
+    }
+    size_t smem = (size_t)128 * 129 * 2 + 2 * 128 * 4;   // fp16 Q
+    hipLaunchKernelGGL(k_svm_cv<2>, dim3((unsigned)(C * F)), dim3(64),
+                       smem, stream, kernels, y, train_idx, test_idx,
+                       n_train, n_test, correct, C, E, F, Creg, tol,
+                       max_iter);
 }

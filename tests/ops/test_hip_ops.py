@@ -128,12 +128,15 @@ def test_fused_gram_matches_staged_cpu(ops):
     assert torch.allclose(G.cpu(), ref, atol=1.0, rtol=3e-2)
 
 
-def test_svm_cv_kernel_matches_sklearn(ops):
+@pytest.mark.parametrize("E", [32, 128])
+def test_svm_cv_kernel_matches_sklearn(ops, E):
+    """E=128 exercises the 2-dual-variables-per-lane path
+    (fold sizes > 64)."""
     from sklearn import model_selection, svm as sksvm
 
     from brainiak_amd.fcma.svm import _accuracy_gpu_hip
     g = torch.Generator().manual_seed(11)
-    n, E = 12, 32
+    n = 12
     y = np.array([0, 1] * (E // 2))
     kernels = []
     for i in range(n):
