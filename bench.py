@@ -102,7 +102,9 @@ def main():
     chunk = min(args.chunk, vps)
 
     def one_step(step_idx):
-        start0 = (step_idx * vps) % V
+        # rank-interleaved rotation: ranks score disjoint voxel ranges
+        # each step, exactly how VoxelSelector shards a whole-brain run
+        start0 = ((step_idx * world + rank) * vps) % V
         chunks = []
         done = 0
         while done < vps:
