@@ -1,7 +1,10 @@
 """Kronecker-product triangular solvers (torch).
 
 Equivalents of the reference's TF kron solvers
-(ref src/brainiak/utils/kronecker_solvers.py:6-330): solve
+(ref src/brainiak/utils/kronecker_solvers.py:6-330; the reference's
+``tf_*`` prefix is dropped because these are torch ops — name map:
+``tf_kron_mult`` → ``kron_mult``, ``tf_solve_lower_triangular_kron`` →
+``solve_lower_triangular_kron``, etc.): solve
 ``(L_1 ⊗ L_2 ⊗ ... ⊗ L_n) x = y`` for triangular factors without ever
 materializing the Kronecker product, by applying per-factor triangular
 solves along reshaped axes.  The masked variants handle covariances
@@ -14,6 +17,7 @@ from typing import List
 import torch
 
 __all__ = [
+    "kron_mult",
     "solve_lower_triangular_kron",
     "solve_lower_triangular_masked_kron",
     "solve_upper_triangular_kron",
@@ -40,6 +44,23 @@ def _kron_matmul_solve(L: List[torch.Tensor], X: torch.Tensor,
         Zm = torch.linalg.solve_triangular(T, Zm, upper=upper)
         Zm = Zm.reshape([sizes[i]] + [sizes[j] for j in range(k)
                                       if j != i] + [ncols])
+        Z = torch.movedim(Zm, 0, i)
+    return Z.reshape(n, ncols)
+
+
+def kron_mult(L: List[torch.Tensor], X: torch.Tensor) -> torch.Tensor:
+    """Compute (⊗_i L_i) @ X without materializing the Kronecker
+    product (ref ``tf_kron_mult``)."""
+    sizes = [int(m.shape[0]) for m in L]
+    n = reduce(lambda a, b: a * b, sizes)
+    ncols = X.shape[1]
+    Z = X.reshape(sizes + [ncols])
+    k = len(L)
+    for i, T in enumerate(L):
+        Zm = torch.movedim(Z, i, 0).reshape(sizes[i], -1)
+        Zm = T @ Zm
+        Zm = Zm.reshape([int(T.shape[0])] + [sizes[j] for j in range(k)
+                                             if j != i] + [ncols])
         Z = torch.movedim(Zm, 0, i)
     return Z.reshape(n, ncols)
 
