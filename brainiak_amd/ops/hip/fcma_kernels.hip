@@ -1019,7 +1019,7 @@ extern "C" void launch_fcma_fused_corr_gram(
 // Z: [C, E, V] bf16; G: [C, E, E] fp32.  E % 64 == 0 (host pads).
 // Block = (c, band_i, band_j>=band_i); 4 waves own the 32x32 quadrants.
 // ===========================================================================
-#define GR_KT 64
+#define GR_KT 128
 #define GR_PAD 8
 #define GR_ROW (GR_KT + GR_PAD)
 
@@ -1068,12 +1068,12 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
     // CURRENT tile from LDS, so the HBM latency hides under compute
     // instead of draining at the barrier.
     const int srow = tid >> 2;            // 64 rows, 4 threads each
-    const int scol = (tid & 3) * 16;      // 2 vector loads per thread
+    const int scol = (tid & 3) * 32;      // 4 vector loads per thread
 
     auto issue_loads = [&](const bf16_t* src, ll rows0, ll k0,
-                           bf16x8 regs[2]) {
+                           bf16x8 regs[4]) {
         #pragma unroll
-        for (int h = 0; h < 2; ++h) {
+        for (int h = 0; h < 4; ++h) {
             ll kk = k0 + scol + 8 * h;
             const bf16_t* s = src + (rows0 + srow) * V + kk;
             if (kk + 8 <= V && (((uintptr_t)s) & 3) == 0) {
@@ -1087,13 +1087,13 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
             }
         }
     };
-    auto write_tile = [&](bf16_t dst[64][GR_ROW], bf16x8 regs[2]) {
+    auto write_tile = [&](bf16_t dst[64][GR_ROW], bf16x8 regs[4]) {
         #pragma unroll
-        for (int h = 0; h < 2; ++h)
+        for (int h = 0; h < 4; ++h)
             *(bf16x8*)&dst[srow][scol + 8 * h] = regs[h];
     };
 
-    bf16x8 ri[2], rj[2];
+    bf16x8 ri[4], rj[4];
     issue_loads(Zc, rows_i, kt0 * GR_KT, ri);
     if (!diag) issue_loads(Zc, rows_j, kt0 * GR_KT, rj);
 
