@@ -64,8 +64,7 @@ class VoxelSelector:
         if voxel_unit is None:
             # GPU chunks amortize kernel launches; 64 is the reference's
             # CPU task size (voxelselector.py:89)
-            import torch as _torch
-            voxel_unit = 1024 if _torch.cuda.is_available() else 64
+            voxel_unit = 1024 if torch.cuda.is_available() else 64
         self.num_voxels = raw_data[0].shape[1]
         self.num_voxels2 = (raw_data2[0].shape[1] if raw_data2 is not None
                             else self.num_voxels)

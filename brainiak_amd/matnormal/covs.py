@@ -24,7 +24,6 @@ from ..utils.kronecker_solvers import (
 from .utils import (
     flatten_cholesky_unique,
     unflatten_cholesky_unique,
-    x_tx,
     xx_t,
 )
 

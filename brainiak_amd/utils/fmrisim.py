@@ -19,7 +19,6 @@ fMRI simulation: validation and application", PeerJ 8:e8564.
 """
 
 import logging
-import math
 
 import numpy as np
 from scipy import ndimage, optimize, signal, stats
