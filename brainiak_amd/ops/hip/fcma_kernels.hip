@@ -1411,7 +1411,7 @@ extern "C" int fcma_corr_norm_smem(ll L, int P) {
         return (int)smem;
     }
 #endif
-    if (corr_variant() == 1 && (L % 2) == 0) {
+    if (corr_variant() >= 1 && (L % 2) == 0) {
         ll ct = dot2_ct(P);
         size_t smem = (size_t)P * L * ct * sizeof(bf16_t)   // bf16 a
                     + (size_t)ct * P * CN_VT * sizeof(float);
