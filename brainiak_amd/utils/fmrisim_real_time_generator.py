@@ -187,7 +187,7 @@ def generate_data(outputDir, user_settings):
     for idx in range(data_dict['numTRs']):
         start = time.time()
         brain = noise[:, :, :, idx] + signal[:, :, :, idx]
-        brain_int32 = brain.astype(np.int32)
+        brain_int32 = np.nan_to_num(brain).astype(np.int32)
         output_file = os.path.join(outputDir,
                                    'rt_' + format(idx, '03d') + '.npy')
         np.save(output_file, brain_int32)
