@@ -327,7 +327,10 @@ def _calc_fwhm(volume, mask, voxel_size=[1.0, 1.0, 1.0]):
         d_var[axis] = (np.sum(diffs ** 2)
                        - (np.sum(diffs) ** 2) / n) / (n - 1)
 
-    o_var = -1.0 / (4 * np.log(1 - 0.5 * d_var / v_var))
+    # volumes rougher than white noise push the AFNI estimator's log
+    # argument <= 0; clamp -> FWHM ~ 0 instead of NaN
+    arg = np.clip(1 - 0.5 * d_var / v_var, 1e-6, None)
+    o_var = -1.0 / (4 * np.log(arg))
     fwhm3 = np.sqrt(o_var) * 2 * np.sqrt(2 * np.log(2))
     return np.prod(fwhm3 * np.asarray(voxel_size)) ** (1 / 3)
 
@@ -884,7 +887,8 @@ def compute_signal_change(signal_function, noise_function, noise_dict,
         raise ValueError(
             'noise_function is not the same size as signal_function')
 
-    signal_function = signal_function / np.max(np.abs(signal_function))
+    peak = np.max(np.abs(signal_function))
+    signal_function = signal_function / (peak if peak > 0 else 1.0)
     out = np.zeros(signal_function.shape)
     for v in range(signal_function.shape[1]):
         sig_voxel = signal_function[:, v]
