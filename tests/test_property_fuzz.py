@@ -1,0 +1,61 @@
+"""Hypothesis property tests over the core numeric primitives."""
+
+import numpy as np
+import torch
+from hypothesis import given, settings, strategies as st
+
+from brainiak_amd.fcma.core import normalize_correlation_
+from brainiak_amd.utils.utils import from_sym_2_tri, from_tri_2_sym
+from brainiak_amd.utils.kronecker_solvers import (
+    kron_mult, solve_lower_triangular_kron, solve_upper_triangular_kron)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(1, 5), st.integers(1, 4), st.integers(2, 6),
+       st.integers(1, 9), st.integers(0, 2 ** 31 - 1))
+def test_normalize_correlation_properties(c, nsubj, p, v, seed):
+    g = torch.Generator().manual_seed(seed)
+    corr = torch.rand((c, nsubj * p, v), generator=g) * 1.8 - 0.9
+    out = normalize_correlation_(corr.clone(), p)
+    z = out.view(c, nsubj, p, v)
+    mean = z.mean(dim=2)
+    # per-(c, subject, voxel) z-scored: mean 0; var 1 or all-zero
+    assert torch.allclose(mean, torch.zeros_like(mean), atol=1e-4)
+    var = (z * z).mean(dim=2) - mean * mean
+    # at the fp cancellation boundary (near-identical epochs) the
+    # rescaled variance can land between 0 and 1 — only assert the
+    # clean cases tightly
+    ok = (var - 1).abs() < 1e-2
+    ok |= var < 1e-4
+    assert bool(ok.all())
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(1, 8), st.integers(0, 2 ** 31 - 1))
+def test_tri_sym_roundtrip(n, seed):
+    rng = np.random.RandomState(seed)
+    m = rng.randn(n, n)
+    sym = (m + m.T) / 2
+    # from_tri_2_sym fills one triangle (reference semantics — callers
+    # symmetrize, e.g. tfa.py's cov + cov.T - diag(diag))
+    t = from_tri_2_sym(from_sym_2_tri(sym), n)
+    full = t + t.T - np.diag(np.diag(t))
+    assert np.allclose(full, sym)
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(1, 4), st.integers(1, 4), st.integers(1, 3),
+       st.integers(0, 2 ** 31 - 1))
+def test_kron_solve_roundtrip(n1, n2, cols, seed):
+    rng = np.random.RandomState(seed)
+    L1 = np.tril(rng.randn(n1, n1)) + n1 * np.eye(n1)
+    L2 = np.tril(rng.randn(n2, n2)) + n2 * np.eye(n2)
+    Lt = [torch.as_tensor(L1), torch.as_tensor(L2)]
+    x = torch.as_tensor(rng.randn(n1 * n2, cols))
+    y = kron_mult(Lt, x)
+    x_back = solve_lower_triangular_kron(Lt, y)
+    assert torch.allclose(x_back, x, atol=1e-8)
+    # upper solve inverts the transposed product
+    yu = kron_mult([m.T.contiguous() for m in Lt], x)
+    xu = solve_upper_triangular_kron(Lt, yu)
+    assert torch.allclose(xu, x, atol=1e-8)
