@@ -8,7 +8,7 @@ factoranalysis/tfa_extension.cpp, eventseg/_utils.pyx):
  - ``fcma_normalize_``  — fused Fisher-z + within-subject z-score (N1)
  - ``fcma_correlate``   — MFMA bf16 batched epoch-correlation GEMM (N2)
  - ``fcma_gram``        — MFMA per-voxel [E,E] Gram / syrk (N3)
- - ``fcma_fused_gram``  — the full chunk pipeline in one kernel
+ - ``fcma_fused_gram``  — chunk pipeline; single-kernel for E=64/P=4
  - ``batched_polar``    — batched K×K Jacobi eigensolve → Procrustes
                           polar factor for SRM (srm.py:595-606 class)
  - ``tfa_factor`` / ``tfa_recon`` — TFA RBF factor matrix + residual
@@ -117,8 +117,10 @@ def fcma_gram_bf16(Z: torch.Tensor) -> torch.Tensor:
 
 def fcma_fused_gram(data: torch.Tensor, data2: torch.Tensor, start: int,
                     count: int, epochs_per_subj: int) -> torch.Tensor:
-    """Fused correlate→normalize→Gram for one voxel chunk; the [C, E, V]
-    intermediate never touches HBM."""
+    """Correlate→normalize→Gram for one voxel chunk.  On the single-
+    kernel path (E=64, P=4 — see ``fcma_fused_gram_native``) the
+    [C, E, V] intermediate never touches HBM; other shapes run the
+    two-kernel corr_norm_z + gram_bf16 composite."""
     return _ext().fcma_fused_gram(data, data2, int(start), int(count),
                                   int(epochs_per_subj))
 
