@@ -73,6 +73,10 @@ def main():
     args = ap.parse_args()
 
     rank, world, device, _ = dist_setup()
+    # let MIOpen pick a direct conv algorithm (the default heuristic
+    # lands on im2col for the 7^3 ball kernel — 32 % of device time
+    # in the profile)
+    torch.backends.cudnn.benchmark = True
     from brainiak_amd.parallel import DistContext
     from brainiak_amd.searchlight import Ball, Searchlight
     ctx = DistContext(device=device)
