@@ -250,12 +250,18 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
         return a.T @ b                          # [V, V]
 
     if summary_statistic == 'mean':
+        from . import ops as _ops
+        use_hip = dev.type == 'cuda' and _ops.has_hip()
         acc = torch.zeros((V, V), dtype=torch.float32, device=dev)
         for d, nd in zip(local, normed_local):
             loo = _norm((total - d) / (n_total - 1))
             m = _corr(nd, loo)
-            m = (m + m.T) / 2
-            acc += torch.atanh(m.clamp(-1 + 1e-7, 1 - 1e-7))
+            if use_hip:
+                # fused symmetrize+atanh+accumulate (one HBM pass)
+                _ops.isfc_accum_(acc, m.contiguous())
+            else:
+                m = (m + m.T) / 2
+                acc += torch.atanh(m.clamp(-1 + 1e-7, 1 - 1e-7))
         acc = comm.all_reduce(acc)
         return torch.tanh(acc / n_total).cpu().numpy()
 

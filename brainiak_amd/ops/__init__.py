@@ -34,6 +34,7 @@ __all__ = [
     "fcma_normalize_",
     "has_fused_gram",
     "has_hip",
+    "isfc_accum_",
     "load_extension",
     "masked_log",
     "tfa_factor",
@@ -175,6 +176,12 @@ def tfa_recon(X: torch.Tensor, W: torch.Tensor, F: torch.Tensor,
 # ---------------------------------------------------------------------------
 # small helpers
 # ---------------------------------------------------------------------------
+
+def isfc_accum_(acc: torch.Tensor, M: torch.Tensor) -> torch.Tensor:
+    """In-place acc += atanh(clamp((M + M^T)/2, +-(1-1e-7))) — the ISFC
+    Fisher-mean accumulation fused into one HBM pass."""
+    return _ext().isfc_accum_(acc, M)
+
 
 def masked_log(x: torch.Tensor) -> torch.Tensor:
     """log(x) with x <= 0 → -inf (ref eventseg/_utils.pyx:27-54).

@@ -275,3 +275,15 @@ def test_corr_norm_z_all_templates(ops, P, L):
     # padding rows stay zero
     if Epad != E:
         assert torch.all(Z[:, E:, :].float().cpu() == 0)
+
+
+@pytest.mark.parametrize("V", [64, 150, 333])
+def test_isfc_accum_matches_torch(ops, V):
+    g = torch.Generator().manual_seed(V)
+    M = (torch.randn((V, V), generator=g) * 0.4).cuda().contiguous()
+    acc0 = torch.randn((V, V), generator=g).cuda().contiguous()
+    got = ops.isfc_accum_(acc0.clone(), M)
+    sym = (M + M.T) / 2
+    ref = acc0 + torch.atanh(sym.clamp(-1 + 1e-7, 1 - 1e-7))
+    assert torch.allclose(got, ref, atol=1e-4, rtol=1e-5), \
+        (got - ref).abs().max().item()
