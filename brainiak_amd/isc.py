@@ -244,9 +244,10 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
 
     normed_local = [_norm(d) for d in local]
 
-    def _corr(a, b):
+    def _corr(a, b, keep_bf16=False):
         if precision == 'bf16':
-            return (a.T.to(torch.bfloat16) @ b.to(torch.bfloat16)).float()
+            m = a.T.to(torch.bfloat16) @ b.to(torch.bfloat16)
+            return m if keep_bf16 else m.float()
         return a.T @ b                          # [V, V]
 
     if summary_statistic == 'mean':
@@ -255,9 +256,10 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
         acc = torch.zeros((V, V), dtype=torch.float32, device=dev)
         for d, nd in zip(local, normed_local):
             loo = _norm((total - d) / (n_total - 1))
-            m = _corr(nd, loo)
+            m = _corr(nd, loo, keep_bf16=use_hip)
             if use_hip:
-                # fused symmetrize+atanh+accumulate (one HBM pass)
+                # fused symmetrize+atanh+accumulate (one HBM pass,
+                # reads the GEMM's bf16 output directly)
                 _ops.isfc_accum_(acc, m.contiguous())
             else:
                 m = (m + m.T) / 2

@@ -30,7 +30,7 @@ void launch_svm_cv(const float*, const float*, const int*, const int*,
 ll fcma_supported_L(ll);
 int fcma_corr_norm_smem(ll, int);
 int fcma_fused_gram_supported(ll, int, ll);
-void launch_isfc_accum(float*, const float*, ll, void*);
+void launch_isfc_accum(float*, const void*, ll, int, void*);
 void launch_fcma_fused_corr_gram(const void*, const void*, float*, ll,
                                  ll, ll, ll, ll, int, void*);
 }
@@ -316,16 +316,17 @@ torch::Tensor svm_cv(torch::Tensor kernels, torch::Tensor y,
 }
 
 torch::Tensor isfc_accum_(torch::Tensor acc, torch::Tensor M) {
+    bool m_bf16 = M.scalar_type() == torch::kBFloat16;
     TORCH_CHECK(acc.is_cuda() && M.is_cuda() && acc.is_contiguous()
                 && M.is_contiguous()
                 && acc.scalar_type() == torch::kFloat32
-                && M.scalar_type() == torch::kFloat32
+                && (m_bf16 || M.scalar_type() == torch::kFloat32)
                 && acc.dim() == 2 && M.dim() == 2
                 && acc.size(0) == acc.size(1)
                 && M.sizes() == acc.sizes(),
-                "acc/M must be matching square fp32 GPU matrices");
-    launch_isfc_accum(acc.data_ptr<float>(), M.data_ptr<float>(),
-                      acc.size(0), cur_stream());
+                "acc fp32 / M fp32-or-bf16 square GPU matrices");
+    launch_isfc_accum(acc.data_ptr<float>(), M.data_ptr(),
+                      acc.size(0), m_bf16 ? 1 : 0, cur_stream());
     return acc;
 }
 

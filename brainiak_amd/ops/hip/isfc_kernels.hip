@@ -13,8 +13,10 @@
 // atanh(clamp(x, -1+1e-7, 1-1e-7)).
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 
 typedef long long ll;
+typedef __hip_bfloat16 bf16_t;
 
 #define IT 64     // tile edge
 #define IPAD 1
@@ -25,8 +27,9 @@ __device__ __forceinline__ float atanh_clamped(float x) {
     return 0.5f * (__logf(1.0f + x) - __logf(1.0f - x));
 }
 
+template <typename T>
 __global__ __launch_bounds__(256) void k_isfc_accum(
-    float* __restrict__ acc, const float* __restrict__ M, ll V) {
+    float* __restrict__ acc, const T* __restrict__ M, ll V) {
     const ll tiles = (V + IT - 1) / IT;
     // upper-triangle tile index -> (ti, tj), ti <= tj
     ll b = blockIdx.x;
@@ -47,7 +50,8 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
     #pragma unroll
     for (int r = ty; r < IT; r += 4) {
         ll row = J0 + r, col = I0 + tx;
-        mt[r][tx] = (row < V && col < V) ? M[row * V + col] : 0.0f;
+        mt[r][tx] = (row < V && col < V) ? (float)M[row * V + col]
+                                         : 0.0f;
     }
     __syncthreads();
 
@@ -56,7 +60,7 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
     for (int r = ty; r < IT; r += 4) {
         ll row = I0 + r, col = J0 + tx;
         if (row >= V || col >= V) continue;
-        float a = M[row * V + col];
+        float a = (float)M[row * V + col];
         float s = 0.5f * (a + mt[tx][r]);     // (M + M^T)/2 at (row,col)
         float z = atanh_clamped(s);
         acc[row * V + col] += z;
@@ -67,10 +71,16 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
     }
 }
 
-extern "C" void launch_isfc_accum(float* acc, const float* M, ll V,
-                                  hipStream_t stream) {
+extern "C" void launch_isfc_accum(float* acc, const void* M, ll V,
+                                  int m_is_bf16, hipStream_t stream) {
     ll tiles = (V + IT - 1) / IT;
     ll nblocks = tiles * (tiles + 1) / 2;
-    hipLaunchKernelGGL(k_isfc_accum, dim3((unsigned)nblocks), dim3(256),
-                       0, stream, acc, M, V);
+    if (m_is_bf16)
+        hipLaunchKernelGGL(k_isfc_accum<bf16_t>, dim3((unsigned)nblocks),
+                           dim3(256), 0, stream, acc, (const bf16_t*)M,
+                           V);
+    else
+        hipLaunchKernelGGL(k_isfc_accum<float>, dim3((unsigned)nblocks),
+                           dim3(256), 0, stream, acc, (const float*)M,
+                           V);
 }

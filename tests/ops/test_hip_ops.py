@@ -278,12 +278,15 @@ def test_corr_norm_z_all_templates(ops, P, L):
 
 
 @pytest.mark.parametrize("V", [64, 150, 333])
-def test_isfc_accum_matches_torch(ops, V):
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_isfc_accum_matches_torch(ops, V, dtype):
     g = torch.Generator().manual_seed(V)
-    M = (torch.randn((V, V), generator=g) * 0.4).cuda().contiguous()
+    M = (torch.randn((V, V), generator=g) * 0.4).to(dtype)
+    M = M.cuda().contiguous()
     acc0 = torch.randn((V, V), generator=g).cuda().contiguous()
     got = ops.isfc_accum_(acc0.clone(), M)
-    sym = (M + M.T) / 2
+    Mf = M.float()
+    sym = (Mf + Mf.T) / 2
     ref = acc0 + torch.atanh(sym.clamp(-1 + 1e-7, 1 - 1e-7))
     assert torch.allclose(got, ref, atol=1e-4, rtol=1e-5), \
         (got - ref).abs().max().item()
