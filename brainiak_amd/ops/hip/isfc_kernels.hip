@@ -43,6 +43,7 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
     if (ti >= tiles) return;
 
     __shared__ float mt[IT][IT + IPAD];       // M[J0.., I0..] tile
+    __shared__ float zt[IT][IT + IPAD];       // atanh tile for mirror
     const int tx = threadIdx.x & 63;
     const int ty = threadIdx.x >> 6;          // 4 rows per pass
 
@@ -59,15 +60,24 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
     #pragma unroll
     for (int r = ty; r < IT; r += 4) {
         ll row = I0 + r, col = J0 + tx;
-        if (row >= V || col >= V) continue;
-        float a = (float)M[row * V + col];
-        float s = 0.5f * (a + mt[tx][r]);     // (M + M^T)/2 at (row,col)
-        float z = atanh_clamped(s);
-        acc[row * V + col] += z;
-        // off-diagonal tile pairs: this block also owns the mirror
-        // tile (diag tiles cover their own lower triangle already)
-        if (!diag)
-            acc[col * V + row] += z;
+        float z = 0.0f;
+        if (row < V && col < V) {
+            float a = (float)M[row * V + col];
+            float s = 0.5f * (a + mt[tx][r]);  // (M + M^T)/2 at (row,col)
+            z = atanh_clamped(s);
+            acc[row * V + col] += z;
+        }
+        zt[r][tx] = z;
+    }
+    if (diag) return;   // diag tiles cover their own lower triangle
+    __syncthreads();
+    // mirror tile written COALESCED via the LDS-transposed z tile
+    // (a direct acc[col*V+row] scatter was 3.5x the HBM roofline)
+    #pragma unroll
+    for (int r = ty; r < IT; r += 4) {
+        ll row = J0 + r, col = I0 + tx;
+        if (row < V && col < V)
+            acc[row * V + col] += zt[tx][r];
     }
 }
 
