@@ -46,7 +46,10 @@ def gpu_isc_block_fn(data, msk, rad, bcast_var, extra):
 
 def gpu_isc_batch_fn(stacks, masks, rad, bcast_var, extra):
     """Batched form for Searchlight.run_batched_block_function: the
-    whole same-shape block group rides ONE correlation + ONE conv3d."""
+    whole same-shape block group rides ONE correlation + ONE ball
+    aggregation (ops.stencil3d — direct LDS stencil instead of
+    MIOpen's im2col conv lowering)."""
+    from brainiak_amd import ops
     device = bcast_var
     a = torch.as_tensor(stacks[0], device=device)  # [B, x, y, z, T]
     b = torch.as_tensor(stacks[1], device=device)
@@ -55,8 +58,11 @@ def gpu_isc_batch_fn(stacks, masks, rad, bcast_var, extra):
     denom = (az.norm(dim=-1) * bz.norm(dim=-1)).clamp_min(1e-12)
     corr = (az * bz).sum(-1) / denom               # [B, x, y, z]
     kernel = _ball_kernel(rad, device)
-    ball_mean = torch.nn.functional.conv3d(
-        corr[:, None], kernel)[:, 0]               # [B, ox, oy, oz]
+    if str(device).startswith("cuda") and ops.has_hip():
+        ball_mean = ops.stencil3d(corr.contiguous(), kernel[0, 0])
+    else:
+        ball_mean = torch.nn.functional.conv3d(
+            corr[:, None], kernel)[:, 0]           # [B, ox, oy, oz]
     out = ball_mean.cpu().numpy()
     inner = masks[:, rad:-rad, rad:-rad, rad:-rad] if rad > 0 else masks
     return np.where(inner, out, np.nan)

@@ -35,6 +35,7 @@ __all__ = [
     "has_fused_gram",
     "has_hip",
     "isfc_accum_",
+    "stencil3d",
     "load_extension",
     "masked_log",
     "tfa_factor",
@@ -176,6 +177,12 @@ def tfa_recon(X: torch.Tensor, W: torch.Tensor, F: torch.Tensor,
 # ---------------------------------------------------------------------------
 # small helpers
 # ---------------------------------------------------------------------------
+
+def stencil3d(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Valid [K,K,K] stencil over [B,X,Y,Z] fp32 (K odd, <= 9) — the
+    searchlight ball aggregation without MIOpen's im2col detour."""
+    return _ext().stencil3d(x, w)
+
 
 def isfc_accum_(acc: torch.Tensor, M: torch.Tensor) -> torch.Tensor:
     """In-place acc += atanh(clamp((M + M^T)/2, +-(1-1e-7))) — the ISFC

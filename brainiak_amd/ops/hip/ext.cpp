@@ -31,6 +31,8 @@ ll fcma_supported_L(ll);
 int fcma_corr_norm_smem(ll, int);
 int fcma_fused_gram_supported(ll, int, ll);
 void launch_isfc_accum(float*, const void*, ll, int, void*);
+void launch_stencil3d(const float*, const float*, float*, ll, int, int,
+                      int, int, void*);
 void launch_fcma_fused_corr_gram(const void*, const void*, float*, ll,
                                  ll, ll, ll, ll, int, void*);
 }
@@ -315,6 +317,29 @@ torch::Tensor svm_cv(torch::Tensor kernels, torch::Tensor y,
     return correct;
 }
 
+torch::Tensor stencil3d(torch::Tensor x, torch::Tensor w) {
+    TORCH_CHECK(x.is_cuda() && w.is_cuda() && x.is_contiguous()
+                && w.is_contiguous()
+                && x.scalar_type() == torch::kFloat32
+                && w.scalar_type() == torch::kFloat32
+                && x.dim() == 4 && w.dim() == 3,
+                "x must be [B,X,Y,Z] fp32, w [K,K,K] fp32 on GPU");
+    ll K = w.size(0);
+    TORCH_CHECK(w.size(1) == K && w.size(2) == K && K % 2 == 1 && K <= 9,
+                "w must be cubic with odd K <= 9");
+    int r = (int)(K / 2);
+    ll B = x.size(0);
+    int X = (int)x.size(1), Y = (int)x.size(2), Z = (int)x.size(3);
+    TORCH_CHECK(X > 2 * r && Y > 2 * r && Z > 2 * r,
+                "volume smaller than the kernel");
+    auto out = torch::empty({B, X - 2 * r, Y - 2 * r, Z - 2 * r},
+                            x.options());
+    launch_stencil3d(x.data_ptr<float>(), w.data_ptr<float>(),
+                     out.data_ptr<float>(), B, X, Y, Z, r,
+                     cur_stream());
+    return out;
+}
+
 torch::Tensor isfc_accum_(torch::Tensor acc, torch::Tensor M) {
     bool m_bf16 = M.scalar_type() == torch::kBFloat16;
     TORCH_CHECK(acc.is_cuda() && M.is_cuda() && acc.is_contiguous()
@@ -331,6 +356,8 @@ torch::Tensor isfc_accum_(torch::Tensor acc, torch::Tensor M) {
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("stencil3d", &stencil3d,
+          "direct [K,K,K] valid conv over [B,X,Y,Z] (searchlight ball)");
     m.def("isfc_accum_", &isfc_accum_,
           "acc += atanh(clamp((M + M^T)/2)) fused in one pass");
     m.def("svm_cv", &svm_cv,

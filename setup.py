@@ -23,6 +23,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "fcma_kernels.hip"),
         os.path.join(HIP_DIR, "svm_kernels.hip"),
         os.path.join(HIP_DIR, "isfc_kernels.hip"),
+        os.path.join(HIP_DIR, "stencil_kernels.hip"),
         os.path.join(HIP_DIR, "procrustes.hip"),
         os.path.join(HIP_DIR, "tfa_kernels.hip"),
     ],

@@ -290,3 +290,19 @@ def test_isfc_accum_matches_torch(ops, V, dtype):
     ref = acc0 + torch.atanh(sym.clamp(-1 + 1e-7, 1 - 1e-7))
     assert torch.allclose(got, ref, atol=1e-4, rtol=1e-5), \
         (got - ref).abs().max().item()
+
+
+@pytest.mark.parametrize("shape,r", [((2, 18, 15, 17), 3),
+                                     ((1, 9, 9, 9), 1),
+                                     ((3, 20, 12, 11), 2)])
+def test_stencil3d_matches_conv3d(ops, shape, r):
+    g = torch.Generator().manual_seed(r)
+    x = torch.randn(shape, generator=g).cuda().contiguous()
+    K = 2 * r + 1
+    w = torch.rand((K, K, K), generator=g).cuda().contiguous()
+    w = w * (torch.rand((K, K, K), generator=g).cuda() > 0.3)  # zeros
+    got = ops.stencil3d(x, w)
+    ref = torch.nn.functional.conv3d(x[:, None], w[None, None])[:, 0]
+    assert got.shape == ref.shape
+    assert torch.allclose(got, ref, atol=1e-4, rtol=1e-4), \
+        (got - ref).abs().max().item()
