@@ -17,10 +17,12 @@ Fisher-z/z-score).  Stages:
      diagonal entry has >2 integer digits).
 
 On CUDA (= ROCm) tensors stages 1–3 dispatch to the hand-written
-HIP/CDNA4 kernels in ``brainiak_amd.ops`` (MFMA bf16 tiles for the GEMM
-classes, a fused LDS pass for the normalization).  The torch
-implementations below are the CPU path and the numerics oracle the GPU
-kernels are tested against.
+HIP/CDNA4 kernels in ``brainiak_amd.ops``: stages 1–2 fuse into one
+correlation+normalize kernel (``k_corr_norm_dot3s`` — v_dot2c bf16
+MACs, scalar-path A operand, in-register Fisher-z/z-score/store; see
+docs/kernels.md), stage 3 is the ``v_mfma_f32_16x16x32_bf16`` Gram.
+The torch implementations below are the CPU path and the numerics
+oracle the GPU kernels are tested against.
 """
 
 import os
