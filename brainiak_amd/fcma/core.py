@@ -166,10 +166,11 @@ class CorrelationPipeline:
 
     def chunk_kernel_matrices(self, start: int, count: int,
                               shrink: bool = True) -> torch.Tensor:
-        """Full fused chunk: correlation → normalize → Gram [count, E, E].
+        """Full chunk: correlation → normalize → Gram [count, E, E].
 
-        On gfx950 this is ONE kernel launch (ops.fcma_fused_gram): the
-        [count, E, V] intermediate lives in LDS/registers, never in HBM.
+        On gfx950 ``ops.fcma_fused_gram`` runs the chunk as one kernel
+        for the headline shape (E=64, P=4) and as the two-kernel
+        corr_norm_z + MFMA-Gram composite otherwise.
         """
         if self.device.type == "cuda" and ops.has_hip() and \
                 ops.has_fused_gram():
