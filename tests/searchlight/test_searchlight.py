@@ -166,3 +166,48 @@ def test_batched_block_function_matches_per_block(seeded_rng):
                 assert (v1 is None) == (v2 is None)
                 if v1 is not None and not (np.isnan(v1) and np.isnan(v2)):
                     assert np.isclose(v1, v2, equal_nan=True)
+
+
+def _batched_mean_fn(stacks, masks, rad, bcast, extra=None):
+    s = stacks[0].mean(-1)[:, rad:-rad, rad:-rad, rad:-rad]
+    im = masks[:, rad:-rad, rad:-rad, rad:-rad]
+    return np.where(im, s + bcast, np.nan)
+
+
+def _dist_batched_searchlight(ctx, outfile):
+    rng = np.random.RandomState(6)
+    dim = (9, 10, 8)
+    data = rng.rand(*dim, 4).astype(np.float32)
+    mask = rng.rand(*dim) > 0.25
+    subjects = [data if ctx.rank == 0 else None]
+    sl = Searchlight(sl_rad=1, max_blk_edge=4, comm=ctx)
+    sl.distribute(subjects, mask)
+    sl.broadcast(2.5)
+    out = sl.run_batched_block_function(_batched_mean_fn)
+    if ctx.rank == 0:
+        vol = np.full(dim, np.nan)
+        for idx in np.ndindex(dim):
+            if out[idx] is not None:
+                vol[idx] = out[idx]
+        np.save(outfile, vol)
+
+
+@pytest.mark.slow
+def test_batched_block_distributed_matches_serial(tmp_path):
+    out = str(tmp_path / "volb.npy")
+    spawn_ranks(_dist_batched_searchlight, world_size=2, args=(out,))
+    vol_dist = np.load(out)
+
+    rng = np.random.RandomState(6)
+    dim = (9, 10, 8)
+    data = rng.rand(*dim, 4).astype(np.float32)
+    mask = rng.rand(*dim) > 0.25
+    sl = Searchlight(sl_rad=1, max_blk_edge=4)
+    sl.distribute([data], mask)
+    sl.broadcast(2.5)
+    serial = sl.run_batched_block_function(_batched_mean_fn)
+    vol_serial = np.full(dim, np.nan)
+    for idx in np.ndindex(dim):
+        if serial[idx] is not None:
+            vol_serial[idx] = serial[idx]
+    assert np.allclose(vol_dist, vol_serial, equal_nan=True, rtol=1e-6)
