@@ -100,6 +100,11 @@ def main():
     # this rank's voxel range rotates so steps touch different voxels
     vps = min(args.voxels_per_step, V)
     chunk = min(args.chunk, vps)
+    if device.type != "cuda":
+        # CPU smoke path: the fp32 correlation chunk is [chunk, E, V]
+        # (9 GB at the GPU defaults) — keep host invocations sane
+        vps = min(vps, 512)
+        chunk = min(chunk, 256)
 
     def one_step(step_idx):
         # rank-interleaved rotation: ranks score disjoint voxel ranges
