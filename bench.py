@@ -103,8 +103,8 @@ def main():
     if device.type != "cuda":
         # CPU smoke path: the fp32 correlation chunk is [chunk, E, V]
         # (9 GB at the GPU defaults) — keep host invocations sane
-        vps = min(vps, 512)
-        chunk = min(chunk, 256)
+        vps = min(vps, 256)
+        chunk = min(chunk, 128)
 
     def one_step(step_idx):
         # rank-interleaved rotation: ranks score disjoint voxel ranges
