@@ -37,10 +37,14 @@ def test_bench_single_process():
 @pytest.mark.slow
 def test_bench_two_ranks_gloo():
     env = dict(os.environ)
+    import socket
+    with socket.socket() as s:      # a currently-free rendezvous port
+        s.bind(("127.0.0.1", 0))
+        port = str(s.getsockname()[1])
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29641", str(REPO / "bench.py"), "--gpus",
+         "--master-port", port, str(REPO / "bench.py"), "--gpus",
          "2", "--steps", "1", "--warmup", "0", "--num-voxels", "256",
          "--voxels-per-step", "64", "--chunk", "64"],
         capture_output=True, text=True, timeout=600, cwd=str(REPO),
