@@ -129,3 +129,22 @@ def test_detsrm_gpu(seeded_rng, gpu_device):
     projected = model.transform(data)
     c = np.corrcoef(projected[0].ravel(), projected[1].ravel())[0, 1]
     assert abs(c) > 0.9
+
+
+def test_srm_loads_frozen_v0_1_model():
+    """Backward-compat: the npz schema written by v0.1 keeps loading
+    (the reference's sr_v0_4.npz compat test, ref
+    tests/funcalign/test_srm.py)."""
+    from pathlib import Path
+    path = Path(__file__).parent / "data" / "srm_v0_1.npz"
+    from brainiak_amd.funcalign.srm import load
+    m = load(str(path))
+    assert m.s_.shape == (5, 30)
+    assert len(m.w_) == 3 and m.w_[0].shape == (40, 5)
+    # transform with the loaded bases works
+    rng = np.random.RandomState(7)
+    S = rng.randn(5, 30)
+    q, _ = np.linalg.qr(rng.randn(40, 5))
+    x = q @ S
+    proj = m.transform([x, None, None])
+    assert proj[0].shape == (5, 30)
