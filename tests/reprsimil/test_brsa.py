@@ -106,7 +106,8 @@ def test_brsa_auto_nuisance_runs(seeded_rng):
     Y, design, _ = _gen_brsa_data(seeded_rng, T=80, V=30)
     # add a shared nuisance time course
     Y += np.outer(np.sin(np.arange(80) / 5), seeded_rng.randn(30)) * 3
-    model = BRSA(auto_nuisance=True, n_nureg=2, random_state=0,
+    model = BRSA(auto_nuisance=True, n_nureg=2, n_iter=3,
+                 random_state=0,
                  minimize_options={'maxiter': 100, 'disp': False})
     model.fit(X=Y, design=design)
     assert model.X0_.shape[1] == 3  # 2 PCs + DC
