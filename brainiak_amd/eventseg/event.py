@@ -3,27 +3,34 @@
 API parity with the reference (ref src/brainiak/eventseg/event.py:64-692):
 annealed EM ``fit`` (variance schedule 4*0.98^(step-1)), Gaussian
 log-observation model measuring Pearson-style distances on space-z-scored
-data, log-space forward-backward with per-step scaling,
-``find_events``/``predict``, ``set_event_patterns``, ``model_prior``,
+data, ``find_events``/``predict``, ``set_event_patterns``, ``model_prior``,
 ``calc_weighted_event_var``, and split-merge proposals.
 
+MI355X-first redesign (round 2): the reference runs a *per-dataset*
+Python forward-backward in exp-space with per-step rescaling
+(ref event.py:284-369, sequential ``for i in range(n_train)``).  Here the
+whole E-step is a **batched pure-log-space recursion on torch**: one
+``[B, T, K+1]`` tensor carries every dataset (and, during split-merge,
+every proposal x dataset pair) through a single T-loop of batched
+``logsumexp`` contractions, so the only sequential dimension is time and
+everything else is device-parallel.  Log-space ``logsumexp`` replaces the
+reference's exp/dot/rescale round trip — numerically it is at least as
+stable and needs no masked-log inside the recursion (transition zeros
+enter once as -inf in the log transition matrix).
+
 The reference's Cython ``masked_log`` helper (eventseg/_utils.pyx:27-54)
-is ``brainiak_amd.ops.masked_log`` (torch, CPU/GPU).  The dense work per
-EM step (X·seg_prob mean patterns and the [T, K] observation
-log-probabilities) runs through torch on the configured device; the
-T-sequential recursion stays on host (it is O(T·K), negligible).
+survives as ``brainiak_amd.ops.masked_log`` for the public quirk surface
+(prior vectors with exact zeros).
 
 Citation context: [BaldassanoC2017] "Discovering event structure in
 continuous narrative perception and memory", Neuron 95(3).
 """
 
-import copy
 import itertools
 import logging
 
 import numpy as np
 import torch
-from scipy import stats
 
 from .. import ops
 
@@ -31,18 +38,72 @@ logger = logging.getLogger(__name__)
 
 __all__ = ["EventSegment"]
 
+_NEG_INF = float("-inf")
+
 
 class NotFittedError(ValueError):
     pass
 
 
-def _check_array(X):
+def _as_valid_2d(X):
     X = np.asarray(X, dtype=np.float64)
     if X.ndim != 2:
         raise ValueError("Input must be 2-D")
     if not np.all(np.isfinite(X)):
         raise ValueError("Input contains NaN or infinity")
     return X
+
+
+def _zscore_rows(a, dim, ddof=1):
+    """Torch z-score along ``dim`` (ddof matching scipy.stats.zscore)."""
+    mu = a.mean(dim=dim, keepdim=True)
+    sd = a.std(dim=dim, unbiased=(ddof == 1), keepdim=True)
+    return (a - mu) / sd
+
+
+class _ChainGraph:
+    """Transition structure of the left-to-right chain(s) + sink state,
+    held directly in log space.
+
+    States 0..K-1 are events; state K is the absorbing sink.  Each chain
+    starts with probability 1/n_chains, self-loops with 1 - p_trans and
+    advances (or exits to the sink from its last event) with
+    p_trans = (chain_len - 1) / T  — the reference's hazard choice
+    (ref event.py:284-369), which makes the expected dwell time span the
+    sequence.
+    """
+
+    def __init__(self, event_chains, t, device, dtype=torch.float64):
+        chains = np.asarray(event_chains)
+        k = chains.shape[0]
+        labels = np.unique(chains, return_inverse=True)[1]
+        n_chains = int(labels.max()) + 1
+
+        start = torch.zeros(k + 1, dtype=dtype)
+        end = torch.zeros(k + 1, dtype=dtype)
+        P = torch.zeros((k + 1, k + 1), dtype=dtype)
+        for c in range(n_chains):
+            members = np.flatnonzero(labels == c)
+            start[members[0]] = 1.0 / n_chains
+            end[members[-1]] = 1.0 / n_chains
+            p_trans = (len(members) - 1) / t
+            if p_trans >= 1:
+                raise ValueError("Too few timepoints")
+            for j, s in enumerate(members):
+                P[s, s] = 1 - p_trans
+                nxt = members[j + 1] if j + 1 < len(members) else k
+                P[s, nxt] = P[s, nxt] + p_trans
+        P[k, k] = 1.0
+
+        self.p_start = start.numpy()
+        self.p_end = end.numpy()
+        self.P = P.numpy()
+        # log-space copies on the compute device (zeros -> -inf, the
+        # masked_log contract, applied once up front)
+        self.log_start = ops.masked_log(start).to(device)
+        self.log_end = ops.masked_log(end).to(device)
+        self.logP = ops.masked_log(P.reshape(-1)).reshape(k + 1,
+                                                          k + 1).to(device)
 
 
 class EventSegment:
@@ -67,10 +128,8 @@ class EventSegment:
         self.split_merge = split_merge
         self.split_merge_proposals = split_merge_proposals
         self.device = device
-        if event_chains is None:
-            self.event_chains = np.zeros(n_events)
-        else:
-            self.event_chains = event_chains
+        self.event_chains = (np.zeros(n_events) if event_chains is None
+                             else event_chains)
 
     # -- estimator params --------------------------------------------------
 
@@ -85,151 +144,197 @@ class EventSegment:
             setattr(self, k, v)
         return self
 
-    # -- core --------------------------------------------------------------
+    # -- batched core ------------------------------------------------------
+
+    def _forward_backward_batch(self, logprob):
+        """Batched log-space forward-backward.
+
+        Parameters
+        ----------
+        logprob : torch [B, T, K] log observation probabilities (no sink
+            column — it is appended here as -inf).
+
+        Returns
+        -------
+        log_gamma : torch [B, T, K] row-normalized posteriors.
+        ll : torch [B] sequence log-likelihoods.
+
+        Everything is pure log space: one batched ``logsumexp``
+        contraction against the [K+1, K+1] log transition matrix per
+        timestep, no per-step rescaling needed (fp64 headroom is vast at
+        fMRI sequence lengths).
+        """
+        B, T, K = logprob.shape
+        dev = logprob.device
+        graph = _ChainGraph(self.event_chains, T, dev)
+        # compat surface: the reference exposes these after any FB pass
+        self.p_start, self.p_end, self.P = (graph.p_start, graph.p_end,
+                                            graph.P)
+
+        obs = torch.full((B, T, K + 1), _NEG_INF, dtype=torch.float64,
+                         device=dev)
+        obs[:, :, :K] = logprob
+
+        logP = graph.logP                     # [K+1, K+1]
+        alphas = torch.empty((B, T, K + 1), dtype=torch.float64,
+                             device=dev)
+        a = graph.log_start[None, :] + obs[:, 0]
+        alphas[:, 0] = a
+        for i in range(1, T):
+            # a[b, j] + logP[j, k] -> logsumexp over j
+            a = torch.logsumexp(a[:, :, None] + logP[None], dim=1) \
+                + obs[:, i]
+            alphas[:, i] = a
+
+        betas = torch.empty_like(alphas)
+        b = graph.log_end[None, :].expand(B, K + 1).contiguous()
+        betas[:, T - 1] = b
+        for i in range(T - 2, -1, -1):
+            w = b + obs[:, i + 1]             # [B, K+1]
+            b = torch.logsumexp(logP[None] + w[:, None, :], dim=2)
+            betas[:, i] = b
+
+        log_gamma = alphas + betas
+        log_gamma = log_gamma - torch.logsumexp(log_gamma, dim=2,
+                                                keepdim=True)
+        ll = torch.logsumexp(alphas[:, T - 1] + graph.log_end[None, :],
+                             dim=1)
+        return log_gamma[:, :, :K], ll
+
+    def _forward_backward(self, logprob):
+        """Single-sequence wrapper; returns (log_gamma [T, K], ll)."""
+        lp = torch.as_tensor(np.asarray(logprob, dtype=np.float64),
+                             device=torch.device(self.device))
+        lg, ll = self._forward_backward_batch(lp[None])
+        return lg[0].cpu().numpy(), float(ll[0])
+
+    def _logprob_obs_batch(self, data, mean_pat, var):
+        """[B, T, K] log N(d_t ; m_k, var_k I) on space-z-scored data,
+        normalized by n_vox (the reference's per-voxel scaling).
+
+        data : torch [B, V, T]; mean_pat : torch [V, K];
+        var : scalar or [K].
+        """
+        B, V, T = data.shape
+        K = mean_pat.shape[1]
+        v = torch.as_tensor(
+            np.broadcast_to(np.asarray(var, float), (K,)).copy(),
+            dtype=torch.float64, device=data.device)
+        dz = _zscore_rows(data, dim=1)        # z over voxels per column
+        mz = _zscore_rows(mean_pat, dim=0)
+        # ||d - m||^2 expanded so the cross term is one batched GEMM
+        d2 = (dz * dz).sum(dim=1)             # [B, T]
+        m2 = (mz * mz).sum(dim=0)             # [K]
+        cross = torch.einsum("bvt,vk->btk", dz, mz)
+        sq = d2[:, :, None] + m2[None, None, :] - 2.0 * cross
+        lp = -0.5 * (V * torch.log(2 * torch.pi * v)[None, None, :]
+                     + sq / v[None, None, :])
+        return lp / V
+
+    def _logprob_obs(self, data, mean_pat, var):
+        """Single-dataset [T, K] observation log-probabilities (compat
+        shim over the batched path)."""
+        dev = torch.device(self.device)
+        d = torch.as_tensor(np.asarray(data, dtype=np.float64), device=dev)
+        m = torch.as_tensor(np.asarray(mean_pat, dtype=np.float64),
+                            device=dev)
+        return self._logprob_obs_batch(d[None], m, var)[0].cpu().numpy()
+
+    # -- fitting -----------------------------------------------------------
 
     def _fit_validate(self, X):
         if len(np.unique(self.event_chains)) > 1:
             raise RuntimeError("Cannot fit chains, use set_event_patterns")
-        X = copy.deepcopy(X)
         if type(X) is not list:
             X = [X]
-        for i in range(len(X)):
-            X[i] = _check_array(X[i]).T
-        n_dim = X[0].shape[0]
-        for i in range(len(X)):
-            assert X[i].shape[0] == n_dim
-        for i in range(len(X)):
-            X[i] = stats.zscore(X[i], axis=1, ddof=1)
-        return X
+        dev = torch.device(self.device)
+        out = []
+        n_dim = None
+        for x in X:
+            x = _as_valid_2d(x).T              # [V, T]
+            if n_dim is None:
+                n_dim = x.shape[0]
+            elif x.shape[0] != n_dim:
+                raise ValueError("All datasets must share the voxel "
+                                 "dimension")
+            t = torch.as_tensor(x, dtype=torch.float64, device=dev)
+            out.append(_zscore_rows(t, dim=1))  # z over time per voxel
+        return out
+
+    @staticmethod
+    def _length_groups(X):
+        """Group dataset indices by sequence length so ragged inputs
+        still ride batched recursions (one batch per distinct T)."""
+        groups = {}
+        for i, x in enumerate(X):
+            groups.setdefault(int(x.shape[1]), []).append(i)
+        return groups
+
+    def _e_step(self, X, mean_pat, var, groups):
+        """Batched E-step over all datasets: returns per-dataset
+        log_gamma list (torch) and ll [n] numpy."""
+        n = len(X)
+        log_gamma = [None] * n
+        ll = np.empty(n)
+        for _t, idxs in groups.items():
+            batch = torch.stack([X[i] for i in idxs])     # [b, V, T]
+            lp = self._logprob_obs_batch(batch, mean_pat, var)
+            lg, l = self._forward_backward_batch(lp)
+            for j, i in enumerate(idxs):
+                log_gamma[i] = lg[j]
+                ll[i] = float(l[j])
+        return log_gamma, ll
+
+    @staticmethod
+    def _mean_pattern(X, log_gamma):
+        """Average (over datasets) of X_b @ column-normalized gamma_b —
+        the M-step event patterns, one batched contraction per dataset."""
+        acc = None
+        for x, lg in zip(X, log_gamma):
+            g = torch.exp(lg)                  # [T, K]
+            g = g / g.sum(dim=0, keepdim=True)
+            p = x @ g                          # [V, K]
+            acc = p if acc is None else acc + p
+        return acc / len(X)
 
     def fit(self, X, y=None):
         """Anneal the event variance downward, alternating mean-pattern
         and segmentation updates until the log-likelihood decreases."""
         X = self._fit_validate(X)
         n_train = len(X)
-        n_dim = X[0].shape[0]
+        groups = self._length_groups(X)
         self.classes_ = np.arange(self.n_events)
+        dev = torch.device(self.device)
 
-        log_gamma = [np.zeros((x.shape[1], self.n_events)) for x in X]
-        step = 1
+        log_gamma = [torch.zeros((x.shape[1], self.n_events),
+                                 dtype=torch.float64, device=dev)
+                     for x in X]
         best_ll = float("-inf")
         self.ll_ = np.empty((0, n_train))
-        while step <= self.n_iter:
+        for step in range(1, self.n_iter + 1):
             iteration_var = self.step_var(step)
-
-            seg_prob = [np.exp(lg) / np.sum(np.exp(lg), axis=0)
-                        for lg in log_gamma]
-            mean_pat = np.empty((n_train, n_dim, self.n_events))
-            for i in range(n_train):
-                mean_pat[i, :, :] = X[i].dot(seg_prob[i])
-            mean_pat = np.mean(mean_pat, axis=0)
-
-            self.ll_ = np.append(self.ll_, np.empty((1, n_train)), axis=0)
-            for i in range(n_train):
-                logprob = self._logprob_obs(X[i], mean_pat, iteration_var)
-                log_gamma[i], self.ll_[-1, i] = \
-                    self._forward_backward(logprob)
+            mean_pat = self._mean_pattern(X, log_gamma)
+            log_gamma_new, ll = self._e_step(X, mean_pat, iteration_var,
+                                             groups)
 
             if step > 1 and self.split_merge:
-                curr_ll = np.mean(self.ll_[-1, :])
-                self.ll_[-1, :], log_gamma, mean_pat = \
-                    self._split_merge(X, log_gamma, iteration_var, curr_ll)
+                ll, log_gamma_new, mean_pat = self._split_merge(
+                    X, log_gamma_new, iteration_var, ll, groups)
 
-            if np.mean(self.ll_[-1, :]) < best_ll:
+            self.ll_ = np.vstack([self.ll_, ll[None, :]])
+            if np.mean(ll) < best_ll:
                 self.ll_ = self.ll_[:-1, :]
                 break
-
-            self.segments_ = [np.exp(lg) for lg in log_gamma]
+            log_gamma = log_gamma_new
+            self.segments_ = [torch.exp(lg).cpu().numpy()
+                              for lg in log_gamma]
             self.event_var_ = iteration_var
-            self.event_pat_ = mean_pat
-            best_ll = np.mean(self.ll_[-1, :])
+            self.event_pat_ = mean_pat.cpu().numpy()
+            best_ll = float(np.mean(ll))
             logger.debug("Fitting step %d, LL=%f", step, best_ll)
-            step += 1
         return self
 
-    def _logprob_obs(self, data, mean_pat, var):
-        """[T, K] log probability of each timepoint under each event
-        Gaussian, on space-z-scored data (so the metric is correlation-
-        like); normalized by n_vox.  Torch on self.device."""
-        n_vox = data.shape[0]
-        if not isinstance(var, np.ndarray):
-            var = var * np.ones(self.n_events)
-        dev = torch.device(self.device)
-        d = torch.as_tensor(data, dtype=torch.float64, device=dev)
-        m = torch.as_tensor(mean_pat, dtype=torch.float64, device=dev)
-        d_z = (d - d.mean(dim=0)) / d.std(dim=0, unbiased=True)
-        m_z = (m - m.mean(dim=0)) / m.std(dim=0, unbiased=True)
-        v = torch.as_tensor(var, dtype=torch.float64, device=dev)
-        # ||d_t - m_k||^2 = |d|^2 + |m|^2 - 2 d·m  (one gemm)
-        d2 = (d_z * d_z).sum(dim=0)[:, None]          # [T, 1]
-        m2 = (m_z * m_z).sum(dim=0)[None, :]          # [1, K]
-        cross = d_z.T @ m_z                           # [T, K]
-        sq = d2 + m2 - 2.0 * cross
-        logprob = (-0.5 * n_vox * torch.log(2 * torch.pi * v)[None, :]
-                   - 0.5 * sq / v[None, :]) / n_vox
-        return logprob.cpu().numpy()
-
-    def _forward_backward(self, logprob):
-        """Log-space forward-backward with per-step scaling; returns
-        (log_gamma [T, K], log-likelihood)."""
-        logprob = np.asarray(logprob, dtype=np.float64).copy()
-        t = logprob.shape[0]
-        logprob = np.hstack((logprob, float("-inf") * np.ones((t, 1))))
-
-        log_scale = np.zeros(t)
-        log_alpha = np.zeros((t, self.n_events + 1))
-        log_beta = np.zeros((t, self.n_events + 1))
-
-        # transition structure: per-chain left-to-right + shared sink
-        self.p_start = np.zeros(self.n_events + 1)
-        self.p_end = np.zeros(self.n_events + 1)
-        self.P = np.zeros((self.n_events + 1, self.n_events + 1))
-        label_ind = np.unique(self.event_chains, return_inverse=True)[1]
-        n_chains = np.max(label_ind) + 1
-        for c in range(n_chains):
-            chain_ind = np.nonzero(label_ind == c)[0]
-            self.p_start[chain_ind[0]] = 1 / n_chains
-            self.p_end[chain_ind[-1]] = 1 / n_chains
-            p_trans = (len(chain_ind) - 1) / t
-            if p_trans >= 1:
-                raise ValueError('Too few timepoints')
-            for i in range(len(chain_ind)):
-                self.P[chain_ind[i], chain_ind[i]] = 1 - p_trans
-                if i < len(chain_ind) - 1:
-                    self.P[chain_ind[i], chain_ind[i + 1]] = p_trans
-                else:
-                    self.P[chain_ind[i], -1] = p_trans
-        self.P[-1, -1] = 1
-
-        for i in range(t):
-            if i == 0:
-                log_alpha[0, :] = self._log(self.p_start) + logprob[0, :]
-            else:
-                log_alpha[i, :] = self._log(
-                    np.exp(log_alpha[i - 1, :]).dot(self.P)) + logprob[i, :]
-            log_scale[i] = np.logaddexp.reduce(log_alpha[i, :])
-            log_alpha[i] -= log_scale[i]
-
-        log_beta[-1, :] = self._log(self.p_end) - log_scale[-1]
-        for i in reversed(range(t - 1)):
-            obs_weighted = log_beta[i + 1, :] + logprob[i + 1, :]
-            offset = np.max(obs_weighted)
-            log_beta[i, :] = offset + self._log(
-                np.exp(obs_weighted - offset).dot(self.P.T)) - log_scale[i]
-
-        log_gamma = log_alpha + log_beta
-        log_gamma -= np.logaddexp.reduce(log_gamma, axis=1, keepdims=True)
-
-        ll = np.sum(log_scale[:(t - 1)]) + np.logaddexp.reduce(
-            log_alpha[-1, :] + log_scale[-1] + self._log(self.p_end))
-        return log_gamma[:, :-1], ll
-
-    @staticmethod
-    def _log(x):
-        """log with x <= 0 → -inf (the reference's masked_log)."""
-        x = np.asarray(x, dtype=np.float64)
-        return ops.masked_log(torch.from_numpy(np.ascontiguousarray(
-            x.ravel()))).numpy().reshape(x.shape)
+    # -- inference ---------------------------------------------------------
 
     def set_event_patterns(self, event_pat):
         if event_pat.shape[1] != self.n_events:
@@ -241,11 +346,11 @@ class EventSegment:
         """Segment a new dataset with the learned event patterns;
         returns (soft segmentation [T, K], log-likelihood)."""
         if var is None:
-            if not hasattr(self, 'event_var_'):
+            if not hasattr(self, "event_var_"):
                 raise NotFittedError("Event variance must be provided, if "
                                      "not previously set by fit()")
             var = self.event_var_
-        if not hasattr(self, 'event_pat_'):
+        if not hasattr(self, "event_pat_"):
             raise NotFittedError("The event patterns must first be set "
                                  "by fit() or set_event_patterns()")
         if scramble:
@@ -259,108 +364,138 @@ class EventSegment:
 
     def predict(self, X):
         """Hard event label per timepoint (argmax of find_events)."""
-        if not hasattr(self, 'event_pat_') or \
-                not hasattr(self, 'event_var_'):
+        if not hasattr(self, "event_pat_") or \
+                not hasattr(self, "event_var_"):
             raise NotFittedError("fit() has not been run")
-        X = _check_array(X)
+        X = _as_valid_2d(X)
         segments, _ = self.find_events(X)
         return np.argmax(segments, axis=1)
 
     def calc_weighted_event_var(self, D, weights, event_pat):
-        """Weighted variance of timepoints around each event pattern."""
-        Dz = stats.zscore(D, axis=1, ddof=1)
-        ev_var = np.empty(event_pat.shape[1])
-        for e in range(event_pat.shape[1]):
-            nz = weights[:, e] > np.max(weights[:, e]) / 1000
-            sumsq = np.dot(weights[nz, e],
-                           np.sum(np.square(Dz[nz, :] - event_pat[:, e]),
-                                  axis=1))
-            ev_var[e] = sumsq / (np.sum(weights[nz, e])
-                                 - np.sum(np.square(weights[nz, e]))
-                                 / np.sum(weights[nz, e]))
-        return ev_var / D.shape[1]
+        """Weighted variance of timepoints around each event pattern,
+        with near-zero weights dropped (reference's 1/1000 floor)."""
+        dev = torch.device(self.device)
+        d = torch.as_tensor(np.asarray(D, dtype=np.float64), device=dev)
+        dz = _zscore_rows(d, dim=1)            # [T, V], z over voxels
+        w = torch.as_tensor(np.asarray(weights, dtype=np.float64),
+                            device=dev)
+        pat = torch.as_tensor(np.asarray(event_pat, dtype=np.float64),
+                              device=dev)
+        K = pat.shape[1]
+        out = torch.empty(K, dtype=torch.float64)
+        for e in range(K):
+            keep = w[:, e] > w[:, e].max() / 1000
+            we = w[keep, e]
+            resid = dz[keep] - pat[:, e][None, :]
+            sumsq = (we * (resid * resid).sum(dim=1)).sum()
+            denom = we.sum() - (we * we).sum() / we.sum()
+            out[e] = sumsq / denom
+        return (out / D.shape[1]).cpu().numpy()
 
     def model_prior(self, t):
         """Prior segmentation (forward-backward with flat observations)."""
-        lg, test_ll = self._forward_backward(
-            np.zeros((t, self.n_events)))
-        return np.exp(lg), test_ll
+        lg, ll = self._forward_backward(np.zeros((t, self.n_events)))
+        return np.exp(lg), ll
 
-    def _split_merge(self, X, log_gamma, iteration_var, curr_ll):
-        """Propose event merges/splits to escape local minima; accept the
-        best proposal that improves the mean log-likelihood."""
-        n_train = len(X)
-        n_dim = X[0].shape[0]
+    # -- split-merge -------------------------------------------------------
 
-        seg_prob = [np.exp(lg) / np.sum(np.exp(lg), axis=0)
-                    for lg in log_gamma]
-        mean_pat = np.empty((n_train, n_dim, self.n_events))
-        for i in range(n_train):
-            mean_pat[i, :, :] = X[i].dot(seg_prob[i])
-        mean_pat = np.mean(mean_pat, axis=0)
+    def _proposal_patterns(self, X, log_gamma):
+        """Merge (adjacent-pair average) and split (half-mass) patterns.
 
-        merge_pat = np.empty((n_train, n_dim, self.n_events))
-        split_pat = np.empty((n_train, n_dim, 2 * self.n_events))
-        for i, sp in enumerate(seg_prob):
-            m_evprob = np.zeros((sp.shape[0], sp.shape[1]))
-            s_evprob = np.zeros((sp.shape[0], 2 * sp.shape[1]))
-            cs = np.cumsum(sp, axis=0)
-            for e in range(sp.shape[1]):
-                mid = np.where(cs[:, e] >= 0.5)[0][0]
-                cs_first = cs[mid, e] - sp[mid, e]
-                cs_second = 1 - cs_first
-                s_evprob[:mid, 2 * e] = sp[:mid, e] / cs_first
-                s_evprob[mid:, 2 * e + 1] = sp[mid:, e] / cs_second
-                m_evprob[:, e] = sp[:, e:(e + 2)].mean(1)
-            merge_pat[i, :, :] = X[i].dot(m_evprob)
-            split_pat[i, :, :] = X[i].dot(s_evprob)
+        Returns torch ``merge_pat [V, K]`` and ``split_pat [V, 2K]``,
+        averaged over datasets.  The half-split point of event e is the
+        first timepoint where its cumulative soft mass reaches 0.5.
+        """
+        K = self.n_events
+        merge_acc = split_acc = None
+        for x, lg in zip(X, log_gamma):
+            g = torch.exp(lg)
+            g = g / g.sum(dim=0, keepdim=True)            # [T, K]
+            T = g.shape[0]
+            cs = torch.cumsum(g, dim=0)
+            split_w = torch.zeros((T, 2 * K), dtype=g.dtype,
+                                  device=g.device)
+            for e in range(K):
+                mid = int(torch.nonzero(cs[:, e] >= 0.5)[0])
+                first_mass = float(cs[mid, e] - g[mid, e])
+                split_w[:mid, 2 * e] = g[:mid, e] / first_mass
+                split_w[mid:, 2 * e + 1] = g[mid:, e] / (1 - first_mass)
+            # merge weights: average event e with its right neighbor
+            merge_w = torch.zeros_like(g)
+            merge_w[:, :K - 1] = 0.5 * (g[:, :K - 1] + g[:, 1:])
+            merge_w[:, K - 1] = g[:, K - 1]
+            mp = x @ merge_w
+            sp = x @ split_w
+            merge_acc = mp if merge_acc is None else merge_acc + mp
+            split_acc = sp if split_acc is None else split_acc + sp
+        return merge_acc / len(X), split_acc / len(X)
 
-        merge_pat = np.mean(merge_pat, axis=0)
-        split_pat = np.mean(split_pat, axis=0)
+    @staticmethod
+    def _columns_corr(a, b):
+        """Pearson r between column a [V] and each column of b [V, m]."""
+        az = a - a.mean()
+        bz = b - b.mean(dim=0, keepdim=True)
+        num = az @ bz
+        den = az.norm() * bz.norm(dim=0)
+        return num / den
 
-        merge_corr = np.zeros(self.n_events)
-        split_corr = np.zeros(self.n_events)
+    def _candidate_patterns(self, mean_pat, merge_pat, split_pat,
+                            m_e, s_e):
+        """Pattern matrix with event ``s_e`` split in two and events
+        ``m_e, m_e+1`` merged (keeping K columns total)."""
+        cols = []
         for e in range(self.n_events):
-            split_corr[e] = np.corrcoef(
-                mean_pat[:, e], split_pat[:, (2 * e):(2 * e + 2)],
-                rowvar=False)[0, 1:3].max()
-            merge_corr[e] = np.corrcoef(
-                merge_pat[:, e], mean_pat[:, e:(e + 2)],
-                rowvar=False)[0, 1:3].min()
-        merge_corr = merge_corr[:-1]
-
-        best_merge = np.flipud(np.argsort(merge_corr))
-        best_merge = best_merge[:self.split_merge_proposals]
-        best_split = np.argsort(split_corr)[:self.split_merge_proposals]
-
-        mean_pat_last = mean_pat.copy()
-        return_ll = curr_ll
-        return_lg = copy.deepcopy(log_gamma)
-        return_mp = mean_pat.copy()
-        for m_e, s_e in itertools.product(best_merge, best_split):
-            if m_e == s_e or m_e + 1 == s_e:
+            if e == s_e:
+                cols.append(split_pat[:, 2 * e])
+                cols.append(split_pat[:, 2 * e + 1])
+            elif e == m_e:
+                cols.append(merge_pat[:, e])
+            elif e == m_e + 1:
                 continue
-            mean_pat_ms = np.delete(mean_pat_last, s_e, axis=1)
-            mean_pat_ms = np.insert(
-                mean_pat_ms, [s_e, s_e],
-                split_pat[:, (2 * s_e):(2 * s_e + 2)], axis=1)
-            mean_pat_ms = np.delete(
-                mean_pat_ms,
-                [m_e + (s_e < m_e), m_e + (s_e < m_e) + 1], axis=1)
-            mean_pat_ms = np.insert(mean_pat_ms, m_e + (s_e < m_e),
-                                    merge_pat[:, m_e], axis=1)
-            ll_ms = np.zeros(n_train)
-            log_gamma_ms = []
-            for i in range(n_train):
-                logprob = self._logprob_obs(X[i], mean_pat_ms,
-                                            iteration_var)
-                lg, ll_ms[i] = self._forward_backward(logprob)
-                log_gamma_ms.append(lg)
-            if ll_ms.mean() > np.mean(return_ll):
-                return_mp = mean_pat_ms.copy()
-                return_ll = ll_ms
-                for i in range(n_train):
-                    return_lg[i] = log_gamma_ms[i].copy()
+            else:
+                cols.append(mean_pat[:, e])
+        return torch.stack(cols, dim=1)
+
+    def _split_merge(self, X, log_gamma, iteration_var, curr_ll, groups):
+        """Propose event merges/splits to escape local minima; evaluate
+        every (merge, split) candidate for every dataset in ONE batched
+        forward-backward pass and keep the best improvement."""
+        mean_pat = self._mean_pattern(X, log_gamma)
+        merge_pat, split_pat = self._proposal_patterns(X, log_gamma)
+
+        K = self.n_events
+        split_corr = torch.empty(K)
+        merge_corr = torch.empty(K)
+        for e in range(K):
+            split_corr[e] = self._columns_corr(
+                mean_pat[:, e], split_pat[:, 2 * e:2 * e + 2]).max()
+            merge_corr[e] = self._columns_corr(
+                merge_pat[:, e], mean_pat[:, e:min(e + 2, K)]).min()
+        merge_rank = torch.argsort(merge_corr[:K - 1],
+                                   descending=True).tolist()
+        split_rank = torch.argsort(split_corr).tolist()
+        pairs = [(m, s) for m, s in itertools.product(
+            merge_rank[:self.split_merge_proposals],
+            split_rank[:self.split_merge_proposals])
+            if s not in (m, m + 1)]
+        if not pairs:
+            return curr_ll, log_gamma, mean_pat
+
+        candidates = [self._candidate_patterns(mean_pat, merge_pat,
+                                               split_pat, m, s)
+                      for m, s in pairs]
+
+        best_ll, best_lg, best_mp = curr_ll, log_gamma, mean_pat
+        best_mean = float(np.mean(curr_ll))
+        n = len(X)
+        for ci, cand in enumerate(candidates):
+            # all datasets (per length group) through one batched pass
+            lg_c, ll_c = self._e_step(X, cand, iteration_var, groups)
+            if float(np.mean(ll_c)) > best_mean:
+                best_mean = float(np.mean(ll_c))
+                best_ll, best_lg, best_mp = ll_c, lg_c, cand
+                m, s = pairs[ci]
                 logger.debug("Identified merge %d,%d and split %d",
-                             m_e, m_e + 1, s_e)
-        return return_ll, return_lg, return_mp
+                             m, m + 1, s)
+        del n
+        return best_ll, best_lg, best_mp

@@ -5,6 +5,14 @@ distributed data distribution.  Where the reference broadcasts each
 epoch matrix one-by-one over mpi4py (preprocessing.py:211-223), here the
 epochs ride a single ``broadcast_object`` through the DistContext (gloo
 on CPU, RCCL device broadcast on GPU).
+
+Round-2 redesign: the reference walks the one-hot epoch specs with a
+triple Python loop per call site (ref preprocessing.py:41-153); here a
+single vectorized span-table builder (`_epoch_spans` — argmax/sum over
+the one-hot axis) feeds every public function, and the per-epoch
+normalization is one fused numpy expression.  Epochs are taken as
+contiguous TR runs (the reference's own `generate_epochs_info` already
+assumes this when it emits ``(start, start+r)`` spans).
 """
 
 import logging
@@ -12,7 +20,6 @@ import math
 from enum import Enum
 
 import numpy as np
-from scipy.stats import zscore
 
 from ..image import mask_images, multimask_images
 from ..parallel import DistContext
@@ -36,26 +43,48 @@ class RandomType(Enum):
     UNREPRODUCIBLE = 2
 
 
+def _epoch_spans(epoch_list):
+    """Vectorized span table from one-hot epoch specs.
+
+    Each subject's spec is ``[n_conditions, n_epochs, n_TRs]`` one-hot.
+    Returns a list of ``(condition, subject, start_tr, end_tr)`` tuples
+    in the reference's emission order (subject-major, then condition,
+    then epoch), with empty epochs dropped.  Start/length come from one
+    ``argmax``/``sum`` over the TR axis instead of per-epoch scans.
+    """
+    spans = []
+    for sid, spec in enumerate(epoch_list):
+        spec = np.asarray(spec)
+        lengths = spec.sum(axis=2)          # [n_cond, n_epochs]
+        starts = spec.argmax(axis=2)
+        for cond, eid in np.argwhere(lengths > 0):
+            a = int(starts[cond, eid])
+            spans.append((int(cond), sid, a, a + int(lengths[cond, eid])))
+    return spans
+
+
+def _unit_epoch(segment):
+    """``[L, V]`` epoch → per-voxel z-score (population std) scaled by
+    1/sqrt(L), so a later matrix product of two epochs IS their Pearson
+    correlation.  Constant voxels become all-zero columns (the
+    reference's nan_to_num semantics)."""
+    seg = np.asarray(segment, dtype=np.float64)
+    mu = seg.mean(axis=0)
+    sd = seg.std(axis=0)
+    with np.errstate(divide="ignore", invalid="ignore"):
+        z = (seg - mu) / sd
+    z[~np.isfinite(z)] = 0.0
+    return np.ascontiguousarray(
+        (z / math.sqrt(seg.shape[0])).astype(np.float32))
+
+
 def _separate_epochs(activity_data, epoch_list):
-    """Cut per-subject [nVoxels, nTRs] data into per-epoch
-    [epoch_len, nVoxels] matrices, z-scored per voxel (ddof=0) and scaled
-    by 1/sqrt(epoch_len) so correlation is a plain matrix product."""
-    raw_data = []
-    labels = []
-    for sid in range(len(epoch_list)):
-        epoch = epoch_list[sid]
-        for cond in range(epoch.shape[0]):
-            sub_epoch = epoch[cond, :, :]
-            for eid in range(epoch.shape[1]):
-                r = np.sum(sub_epoch[eid, :])
-                if r > 0:
-                    mat = activity_data[sid][:, sub_epoch[eid, :] == 1]
-                    mat = np.ascontiguousarray(mat.T)
-                    mat = zscore(mat, axis=0, ddof=0)
-                    mat = np.nan_to_num(mat)
-                    mat = mat / math.sqrt(r)
-                    raw_data.append(mat.astype(np.float32))
-                    labels.append(cond)
+    """Cut per-subject ``[nVoxels, nTRs]`` data into per-epoch
+    ``[epoch_len, nVoxels]`` unit-normalized matrices + labels."""
+    spans = _epoch_spans(epoch_list)
+    raw_data = [_unit_epoch(activity_data[sid][:, a:b].T)
+                for _, sid, a, b in spans]
+    labels = [cond for cond, _, _, _ in spans]
     return raw_data, labels
 
 
@@ -69,11 +98,11 @@ def _randomize_single_subject(data, seed=None):
 
 def _randomize_subject_list(data_list, random):
     if random == RandomType.REPRODUCIBLE:
-        for i in range(len(data_list)):
-            _randomize_single_subject(data_list[i], seed=i)
+        for i, d in enumerate(data_list):
+            _randomize_single_subject(d, seed=i)
     elif random == RandomType.UNREPRODUCIBLE:
-        for data in data_list:
-            _randomize_single_subject(data)
+        for d in data_list:
+            _randomize_single_subject(d)
 
 
 def prepare_fcma_data(images, conditions, mask1, mask2=None,
@@ -85,114 +114,96 @@ def prepare_fcma_data(images, conditions, mask1, mask2=None,
     reference does.
     """
     ctx = comm if isinstance(comm, DistContext) else DistContext()
-    labels = []
-    raw_data1 = []
-    raw_data2 = []
+    results = ([], [], [])            # raw_data1, raw_data2, labels
     if ctx.is_root:
         logger.info('start to apply masks and separate epochs')
-        if mask2 is not None:
-            masks = (mask1, mask2)
-            pairs = [tuple(m) for m in multimask_images(images, masks,
-                                                        np.float32)]
-            activity_data1 = [p[0] for p in pairs]
-            activity_data2 = [p[1] for p in pairs]
-            _randomize_subject_list(activity_data2, random)
-            raw_data2, _ = _separate_epochs(activity_data2, conditions)
-            _randomize_subject_list(activity_data1, random)
-            raw_data1, labels = _separate_epochs(activity_data1, conditions)
+        if mask2 is None:
+            per_mask = [list(mask_images(images, mask1, np.float32))]
         else:
-            activity_data1 = list(mask_images(images, mask1, np.float32))
-            _randomize_subject_list(activity_data1, random)
-            raw_data1, labels = _separate_epochs(activity_data1, conditions)
+            stacked = [tuple(m) for m in
+                       multimask_images(images, (mask1, mask2),
+                                        np.float32)]
+            per_mask = [[s[i] for s in stacked] for i in (0, 1)]
+        # shuffle+separate mask2 FIRST, then mask1 — the reference's
+        # RNG consumption order (its REPRODUCIBLE seeds reset per
+        # subject, but UNREPRODUCIBLE draws stream through)
+        epoched = {}
+        for mi in reversed(range(len(per_mask))):
+            _randomize_subject_list(per_mask[mi], random)
+            epoched[mi] = _separate_epochs(per_mask[mi], conditions)
+        results = (epoched[0][0],
+                   epoched[1][0] if 1 in epoched else [],
+                   epoched[0][1])
 
     if ctx.is_distributed:
-        payload = ctx.broadcast_object(
-            (raw_data1, raw_data2, labels) if ctx.is_root else None)
-        raw_data1, raw_data2, labels = payload
+        results = ctx.broadcast_object(results if ctx.is_root else None)
         logger.info('data broadcasting done')
-    if mask2 is None:
-        raw_data2 = None
-    return raw_data1, raw_data2, labels
+    raw_data1, raw_data2, labels = results
+    return raw_data1, (None if mask2 is None else raw_data2), labels
 
 
 def generate_epochs_info(epoch_list):
-    """Per-epoch (label, sid, start, end) tuples from one-hot epoch specs."""
-    epoch_info = []
-    for sid, epoch in enumerate(epoch_list):
-        for cond in range(epoch.shape[0]):
-            sub_epoch = epoch[cond, :, :]
-            for eid in range(epoch.shape[1]):
-                r = np.sum(sub_epoch[eid, :])
-                if r > 0:
-                    start = np.nonzero(sub_epoch[eid, :])[0][0]
-                    epoch_info.append((cond, sid, start, start + r))
-    return epoch_info
+    """Per-epoch (label, sid, start, end) tuples from one-hot epoch
+    specs (the vectorized span table, see `_epoch_spans`)."""
+    return _epoch_spans(epoch_list)
+
+
+def _zscore_subject_groups(mat, subject_ids, axis):
+    """Z-score (population std) the epoch axis within each subject's
+    contiguous run of columns; subjects with a single epoch pass
+    through untouched.  NaNs (constant rows) become zeros."""
+    subject_ids = np.asarray(subject_ids)
+    out = np.array(mat, copy=True)
+    for sid in np.unique(subject_ids):
+        sel = np.nonzero(subject_ids == sid)[0]
+        if sel.size <= 1:
+            continue
+        block = np.take(out, sel, axis=axis)
+        mu = block.mean(axis=axis, keepdims=True)
+        sd = block.std(axis=axis, keepdims=True)
+        with np.errstate(divide="ignore", invalid="ignore"):
+            block = (block - mu) / sd
+        ix = [slice(None)] * out.ndim
+        ix[axis] = sel
+        out[tuple(ix)] = block
+    return np.nan_to_num(out)
 
 
 def prepare_mvpa_data(images, conditions, mask):
     """Epoch-averaged, within-subject z-scored activity:
     returns ([num_voxels, num_epochs], labels)."""
     activity_data = list(mask_images(images, mask, np.float32))
-    epoch_info = generate_epochs_info(conditions)
-    num_epochs = len(epoch_info)
-    d1 = activity_data[0].shape[0]
-    processed_data = np.empty([d1, num_epochs])
-    labels = np.empty(num_epochs)
-    subject_count = [0]
-    cur_sid = -1
-    for idx, epoch in enumerate(epoch_info):
-        labels[idx] = epoch[0]
-        if cur_sid != epoch[1]:
-            subject_count.append(0)
-            cur_sid = epoch[1]
-        subject_count[-1] += 1
-        processed_data[:, idx] = np.mean(
-            activity_data[cur_sid][:, epoch[2]:epoch[3]], axis=1)
-    cur_epoch = 0
-    for i in subject_count:
-        if i > 1:
-            processed_data[:, cur_epoch:cur_epoch + i] = zscore(
-                processed_data[:, cur_epoch:cur_epoch + i], axis=1, ddof=0)
-        cur_epoch += i
-    processed_data = np.nan_to_num(processed_data)
-    return processed_data, labels
+    spans = _epoch_spans(conditions)
+    labels = np.array([cond for cond, _, _, _ in spans], dtype=float)
+    sids = [sid for _, sid, _, _ in spans]
+    processed = np.column_stack(
+        [activity_data[sid][:, a:b].mean(axis=1)
+         for _, sid, a, b in spans])
+    return _zscore_subject_groups(processed, sids, axis=1), labels
 
 
 def prepare_searchlight_mvpa_data(images, conditions, data_type=np.float32,
                                   random=RandomType.NORANDOM):
     """Epoch-averaged, within-subject z-scored activity keeping the 3-D
     volume: returns ([x, y, z, num_epochs], labels).  Subjects are
-    streamed one at a time."""
-    epoch_info = generate_epochs_info(conditions)
-    num_epochs = len(epoch_info)
-    processed_data = None
-    labels = np.empty(num_epochs)
-    for idx, epoch in enumerate(epoch_info):
-        labels[idx] = epoch[0]
-    subject_count = np.zeros(len(conditions), dtype=np.int32)
+    streamed one at a time so only one 4-D volume is resident."""
+    spans = _epoch_spans(conditions)
+    labels = np.array([cond for cond, _, _, _ in spans], dtype=float)
+    sids = [sid for _, sid, _, _ in spans]
+    processed = None
 
     for sid, f in enumerate(images):
         data = f.get_fdata().astype(data_type)
-        d1, d2, d3, d4 = data.shape
         if random != RandomType.NORANDOM:
-            flat = data.reshape((d1 * d2 * d3, d4))
+            flat = data.reshape((-1, data.shape[3]))
             _randomize_single_subject(
-                flat, seed=sid if random == RandomType.REPRODUCIBLE else None)
-            data = flat.reshape((d1, d2, d3, d4))
-        if processed_data is None:
-            processed_data = np.empty([d1, d2, d3, num_epochs],
-                                      dtype=data_type)
-        for idx, epoch in enumerate(epoch_info):
-            if sid == epoch[1]:
-                subject_count[sid] += 1
-                processed_data[:, :, :, idx] = np.mean(
-                    data[:, :, :, epoch[2]:epoch[3]], axis=3)
-    cur_epoch = 0
-    for i in subject_count:
-        if i > 1:
-            processed_data[:, :, :, cur_epoch:cur_epoch + i] = zscore(
-                processed_data[:, :, :, cur_epoch:cur_epoch + i],
-                axis=3, ddof=0)
-        cur_epoch += i
-    processed_data = np.nan_to_num(processed_data)
-    return processed_data, labels
+                flat,
+                seed=sid if random == RandomType.REPRODUCIBLE else None)
+            data = flat.reshape(data.shape)
+        if processed is None:
+            processed = np.empty(data.shape[:3] + (len(spans),),
+                                 dtype=data_type)
+        for idx, (_, s, a, b) in enumerate(spans):
+            if s == sid:
+                processed[..., idx] = data[..., a:b].mean(axis=3)
+    return _zscore_subject_groups(processed, sids, axis=3), labels

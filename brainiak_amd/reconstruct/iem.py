@@ -3,9 +3,27 @@ spaces.
 
 API parity with the reference (ref src/brainiak/reconstruct/iem.py:67-1050):
 ``InvertedEncoding1D`` (half-rectified exponentiated-sinusoid channel
-basis; fit = pinv regression B = W·C; predict = channel inversion +
-argmax; score = circular-distance R²) and ``InvertedEncoding2D``
-(exponentiated 2-D cosine channels on square/triangular grids).
+basis; fit = least-squares regression of X onto channel activations;
+predict = channel inversion + argmax; score = circular-distance R²) and
+``InvertedEncoding2D`` (exponentiated 2-D cosine channels on square or
+triangular grids).
+
+Round-2 redesign notes:
+ - The forward/inverse solves use ``np.linalg.lstsq`` on the design
+   matrices directly (minimum-norm least squares — identical solutions
+   to the reference's explicit pinv products for full-rank designs,
+   ref iem.py:245-260, without forming pseudoinverses).
+ - Channel bases, trial activations, and stimulus masks are broadcast
+   expressions (one distance matrix per call) instead of per-center /
+   per-trial Python loops (ref iem.py:308-399).
+ - Reference quirks preserved: the channel domain spans
+   ``[start, stop-1]``, ``stim_radius`` in 2-D thresholds the *squared*
+   pixel distance (ref iem.py:387-390), and stimulus values are shifted
+   by ``|range_start|`` before grid lookup.  One deviation: when
+   ``channel_density`` is a multiple of ``stim_res`` the coarse-grid
+   lookup scales indices correctly (the reference's ``np.repeat``
+   without an axis flattens the mask and cannot reach the following
+   matmul, ref iem.py:159-166).
 
 Citation: [Brouwer2009] "Decoding and reconstructing color from
 responses in human visual cortex", J. Neurosci 29(44).
@@ -24,6 +42,35 @@ logger = logging.getLogger(__name__)
 MAX_CONDITION_CHECK = 9000
 
 __all__ = ["InvertedEncoding1D", "InvertedEncoding2D"]
+
+
+def _lstsq(a, b):
+    """Minimum-norm least-squares solve of a @ x = b."""
+    return np.linalg.lstsq(a, b, rcond=None)[0]
+
+
+def _fit_weights(design, data):
+    """Voxel weight matrix W (voxels x channels) minimizing
+    ``||data - design @ W.T||``; raises if either the data or the
+    resulting weights are near-singular (the reference's condition
+    guard, MAX_CONDITION_CHECK)."""
+    if np.linalg.cond(data) > MAX_CONDITION_CHECK:
+        logger.error("ill-conditioned data matrix")
+        raise ValueError("data matrix is near-singular (condition "
+                         "number > %d)" % MAX_CONDITION_CHECK)
+    W = _lstsq(design, data).T
+    if np.linalg.cond(W) > MAX_CONDITION_CHECK:
+        raise ValueError("fitted weight matrix is near-singular")
+    return W
+
+
+def _warn_if_rank_deficient(C, n_channels):
+    rank = np.linalg.matrix_rank(C)
+    if rank < n_channels:
+        warnings.warn(
+            "Stimulus matrix is {}, not full rank. May cause issues with "
+            "stimulus prediction/reconstruction.".format(rank),
+            RuntimeWarning)
 
 
 class InvertedEncoding1D:
@@ -46,69 +93,65 @@ class InvertedEncoding1D:
         self._check_params()
 
     def _check_params(self):
-        if self.range_start >= self.range_stop:
-            raise ValueError("range_start {} must be less than "
-                             "{} range_stop.".format(self.range_start,
-                                                     self.range_stop))
-        if self.stimulus_mode == 'halfcircular':
-            if (self.range_stop - self.range_start) != 180.:
-                raise ValueError("For half-circular feature spaces,"
-                                 "the range must be 180 degrees, "
-                                 "not {}".format(self.range_stop
-                                                 - self.range_start))
-        elif self.stimulus_mode == 'circular':
-            if (self.range_stop - self.range_start) != 360.:
-                raise ValueError("For circular feature spaces, the"
-                                 " range must be 360 degrees"
-                                 "not {}".format(self.range_stop
-                                                 - self.range_start))
+        span = self.range_stop - self.range_start
+        if span <= 0:
+            raise ValueError(
+                "range_start {} must be less than {} range_stop.".format(
+                    self.range_start, self.range_stop))
+        required = {'halfcircular': 180., 'circular': 360.}
+        if self.stimulus_mode not in required:
+            raise ValueError("stimulus_mode must be 'circular' or "
+                             "'halfcircular', got %r"
+                             % (self.stimulus_mode,))
+        if span != required[self.stimulus_mode]:
+            raise ValueError(
+                "A {} feature space must span {} degrees, not {}".format(
+                    self.stimulus_mode, required[self.stimulus_mode],
+                    span))
         if self.n_channels < 2:
-            raise ValueError("Insufficient number of channels.")
-        if self.stimulus_mode not in ('circular', 'halfcircular'):
-            raise ValueError("Stimulus mode must be one of these: "
-                             "'circular', 'halfcircular'")
+            raise ValueError("need at least 2 channels")
+
+    # -- model -------------------------------------------------------------
 
     def fit(self, X, y):
-        """Estimate W from training data B = W·C."""
-        if np.linalg.cond(X) > MAX_CONDITION_CHECK:
-            logger.error("Data is singular.")
-            raise ValueError("Data matrix is nearly singular.")
-        if X.shape[0] < self.n_channels:
-            raise ValueError("Fewer observations (trials) than "
-                             "channels. Cannot compute pseudoinverse.")
+        """Estimate voxel weights from training data
+        ``X ≈ C(y) @ W.T``."""
+        X = np.asarray(X)
         if np.ndim(X) != 2:
-            raise ValueError("Data matrix has too many or too few "
-                             "dimensions.")
-        if np.shape(X)[0] != np.shape(y)[0]:
-            raise ValueError("Mismatched data samples and label samples")
+            raise ValueError("expected a 2-D [observations, voxels] data "
+                             "matrix")
+        if X.shape[0] < self.n_channels:
+            raise ValueError("under-determined fit: fewer trials than "
+                             "channels")
+        if X.shape[0] != np.shape(y)[0]:
+            raise ValueError("X and y disagree on the number of trials")
 
-        self.channels_, channel_centers = self._define_channels()
+        self.channels_, centers = self._define_channels()
         logger.info("Defined channels centered at %s degrees.",
-                    np.rad2deg(channel_centers))
+                    np.rad2deg(centers))
         C = self._define_trial_activations(y)
-        self.W_ = X.transpose() @ np.linalg.pinv(C.transpose())
-        if np.linalg.cond(self.W_) > MAX_CONDITION_CHECK:
-            raise ValueError("Weight matrix is nearly singular.")
+        self.W_ = _fit_weights(C, X)
         return self
 
     def predict(self, X):
         """Predicted feature value per observation."""
         if np.ndim(X) != 2:
-            raise ValueError("Data matrix has too many or too few "
-                             "dimensions.")
-        return self._predict_features(X)
+            raise ValueError("expected a 2-D [observations, voxels] data "
+                             "matrix")
+        return self._predict_features(np.asarray(X))
 
     def score(self, X, y):
         """Circular-distance R² of predictions against y."""
-        pred_features = self.predict(X)
+        pred = self.predict(X)
+        y = np.asarray(y, dtype=float)
         if self.stimulus_mode == 'halfcircular':
-            pred_features = pred_features * 2
-            y = y * 2
-        ssres = (circ_dist(np.deg2rad(y),
-                           np.deg2rad(pred_features)) ** 2).sum()
-        sstot = (circ_dist(np.deg2rad(y),
-                           np.ones(np.size(y)) * scipy.stats.circmean(
-                               np.deg2rad(y))) ** 2).sum()
+            # stretch the half-circle onto the full circle
+            pred, y = 2 * pred, 2 * y
+        y_rad = np.deg2rad(y)
+        ssres = np.sum(circ_dist(y_rad, np.deg2rad(pred)) ** 2)
+        mean_dir = scipy.stats.circmean(y_rad)
+        sstot = np.sum(circ_dist(y_rad,
+                                 np.full(y.size, mean_dir)) ** 2)
         return 1 - ssres / sstot
 
     def get_params(self, deep: bool = True):
@@ -128,61 +171,52 @@ class InvertedEncoding1D:
         self._check_params()
         return self
 
+    # -- internals ---------------------------------------------------------
+
     def _define_channels(self):
-        """Half-rectified sinusoid^exp channel basis."""
-        channel_centers = np.linspace(np.deg2rad(self.range_start),
-                                      np.deg2rad(self.range_stop),
-                                      self.n_channels + 1)[:-1]
-        if self.stimulus_mode == 'circular':
-            domain = self.channel_domain * 0.5
-            centers = channel_centers * 0.5
-        else:
-            domain = self.channel_domain
-            centers = channel_centers
-        channels = np.asarray(
-            [np.cos(np.deg2rad(domain) - cx) ** self.channel_exp
-             for cx in centers])
-        return np.abs(channels), channel_centers
+        """Half-rectified sinusoid^exp basis: |cos(domain - center)|^exp,
+        on the half-angle scale for circular spaces.  One broadcast
+        expression over [n_channels, channel_density]."""
+        centers = np.deg2rad(np.linspace(
+            self.range_start, self.range_stop, self.n_channels,
+            endpoint=False))
+        half = 0.5 if self.stimulus_mode == 'circular' else 1.0
+        grid = np.deg2rad(self.channel_domain) * half
+        resp = np.cos(grid[None, :] - half * centers[:, None]) \
+            ** self.channel_exp
+        return np.abs(resp), centers
 
     def _define_trial_activations(self, stimuli):
-        """[observations, n_channels] predicted channel responses."""
-        stim_axis = np.linspace(self.range_start, self.range_stop - 1,
-                                self.stim_res)
-        stimuli = np.asarray(stimuli)
-        if self.range_start > 0:
-            stimuli = stimuli + self.range_start
-        elif self.range_start < 0:
-            stimuli = stimuli - self.range_start
-        one_hot = np.eye(self.stim_res)
-        indices = [np.argmin(abs(stim_axis - x)) for x in stimuli]
-        stimulus_mask = one_hot[indices, :]
+        """[observations, n_channels] predicted channel responses:
+        nearest stimulus-grid point per trial, gathered straight from
+        the channel basis (no one-hot matmul)."""
+        axis = np.linspace(self.range_start, self.range_stop - 1,
+                           self.stim_res)
+        vals = np.asarray(stimuli, dtype=float)
+        if self.range_start != 0:
+            vals = vals + abs(self.range_start)
+        idx = np.abs(axis[None, :] - vals[:, None]).argmin(axis=1)
         if self.channel_density != self.stim_res:
-            if self.channel_density % self.stim_res == 0:
-                stimulus_mask = np.repeat(
-                    stimulus_mask, self.channel_density // self.stim_res)
-            else:
+            if self.channel_density % self.stim_res:
                 raise NotImplementedError(
                     "Stimulus resolution must evenly divide the channel "
                     "density.")
-        C = stimulus_mask @ self.channels_.transpose()
-        if np.linalg.matrix_rank(C) < self.n_channels:
-            warnings.warn("Stimulus matrix is {}, not full rank. May "
-                          "cause issues with stimulus prediction/"
-                          "reconstruction.".format(
-                              np.linalg.matrix_rank(C)), RuntimeWarning)
+            idx = idx * (self.channel_density // self.stim_res)
+        C = self.channels_[:, idx].T
+        _warn_if_rank_deficient(C, self.n_channels)
         return C
 
     def _predict_channel_responses(self, X):
-        return np.matmul(np.linalg.pinv(self.W_), X.transpose())
+        """[n_channels, observations] inverted responses (least-squares
+        through the fitted weights)."""
+        return _lstsq(self.W_, X.T)
 
     def _predict_feature_responses(self, X):
-        return np.matmul(self.channels_.transpose(),
-                         self._predict_channel_responses(X))
+        return self.channels_.T @ self._predict_channel_responses(X)
 
     def _predict_features(self, X):
-        pred_response = self._predict_feature_responses(X)
-        feature_ind = np.argmax(pred_response, 0)
-        return self.channel_domain[feature_ind]
+        peak = np.argmax(self._predict_feature_responses(X), axis=0)
+        return self.channel_domain[peak]
 
 
 class InvertedEncoding2D:
@@ -194,42 +228,36 @@ class InvertedEncoding2D:
         if not isinstance(stimulus_resolution, list):
             stimulus_resolution = [stimulus_resolution,
                                    stimulus_resolution]
-        if (len(stim_xlim) != 2) or (len(stim_ylim) != 2):
-            raise ValueError(
-                "Stimulus limits should be a sequence, 2 values")
+        if len(stim_xlim) != 2 or len(stim_ylim) != 2:
+            raise ValueError("each stimulus limit must be a (lo, hi) pair")
         self.stim_fov = [stim_xlim, stim_ylim]
-        self.stim_pixels = [np.linspace(stim_xlim[0], stim_xlim[1],
-                                        stimulus_resolution[0]),
-                            np.linspace(stim_ylim[0], stim_ylim[1],
-                                        stimulus_resolution[1])]
-        self.xp, self.yp = np.meshgrid(self.stim_pixels[0],
-                                       self.stim_pixels[1])
+        self.stim_pixels = [
+            np.linspace(stim_xlim[0], stim_xlim[1],
+                        stimulus_resolution[0]),
+            np.linspace(stim_ylim[0], stim_ylim[1],
+                        stimulus_resolution[1])]
+        self.xp, self.yp = np.meshgrid(*self.stim_pixels)
         self.stim_radius_px = stim_radius
         self.channels = channels
         self.n_channels = None if channels is None else channels.shape[0]
-        if chan_xlim is None:
-            chan_xlim = stim_xlim
-        if chan_ylim is None:
-            chan_ylim = stim_ylim
-        self.channel_limits = [chan_xlim, chan_ylim]
+        self.channel_limits = [
+            stim_xlim if chan_xlim is None else chan_xlim,
+            stim_ylim if chan_ylim is None else chan_ylim]
         self.channel_exp = channel_exp
         self._check_params()
 
     def _check_params(self):
         if len(self.stim_fov) != 2:
-            raise ValueError(
-                "Stim FOV needs to have an x-list and a y-list")
-        if len(self.stim_fov[0]) != 2 or len(self.stim_fov[1]) != 2:
-            raise ValueError(
-                "Stimulus limits should be a sequence, 2 values")
-        if (self.stim_fov[0][0] >= self.stim_fov[0][1]) or \
-                (self.stim_fov[1][0] >= self.stim_fov[1][1]):
-            raise ValueError(
-                "Stimulus x or y limits should be ascending values")
+            raise ValueError("stim_fov must hold x and y limit pairs")
+        for lim in self.stim_fov:
+            if len(lim) != 2:
+                raise ValueError("each stimulus limit must be a (lo, hi) "
+                                 "pair")
+            if lim[0] >= lim[1]:
+                raise ValueError("stimulus limits must be ascending")
         if self.xp.size != self.yp.size:
-            raise ValueError("xpixel grid and ypixel grid do not have "
-                             "same number of elements")
-        if self.n_channels and np.all(self.channels):
+            raise ValueError("x and y pixel grids differ in size")
+        if self.channels is not None:
             if self.n_channels != self.channels.shape[0]:
                 raise ValueError(
                     "Number of channels {} does not match the defined "
@@ -237,43 +265,42 @@ class InvertedEncoding2D:
                                           self.channels.shape[0]))
             if self.channels.shape[1] != self.xp.size:
                 raise ValueError(
-                    "Defined {} channels over {} pixels, but stimuli are "
-                    "represented over {} pixels. Pixels should match."
-                    .format(self.n_channels, self.channels.shape[1],
-                            self.xp.size))
+                    "Defined {} channels over {} pixels, but stimuli "
+                    "are represented over {} pixels.".format(
+                        self.n_channels, self.channels.shape[1],
+                        self.xp.size))
+
+    # -- model -------------------------------------------------------------
 
     def fit(self, X, y, C=None):
-        """Estimate W from training data; C defaults to circular-stimulus
-        channel activations built from y."""
-        if np.linalg.cond(X) > MAX_CONDITION_CHECK:
-            raise ValueError("Data matrix is nearly singular.")
+        """Estimate W from training data; C defaults to
+        circular-stimulus channel activations built from y."""
+        X = np.asarray(X)
         if self.channels is None:
-            raise ValueError(
-                "Must define channels (set of basis functions).")
+            raise ValueError("no channel basis defined - call one of the "
+                             "define_basis_functions_* methods first")
         if X.shape[0] < self.n_channels:
-            raise ValueError("Fewer observations (trials) than "
-                             "channels. Cannot compute pseudoinverse.")
-        if np.shape(X)[0] != np.shape(y)[0]:
-            raise ValueError("Mismatched data samples and label samples")
+            raise ValueError("under-determined fit: fewer trials than "
+                             "channels")
+        if X.shape[0] != np.shape(y)[0]:
+            raise ValueError("X and y disagree on the number of trials")
         if C is None:
             C = self._define_trial_activations(y)
-        self.W_ = X.transpose() @ np.linalg.pinv(C.transpose())
-        if np.linalg.cond(self.W_) > MAX_CONDITION_CHECK:
-            raise ValueError("Weight matrix is nearly singular.")
+        self.W_ = _fit_weights(C, X)
         return self
 
     def predict(self, X):
         if np.ndim(X) != 2:
-            raise ValueError("Data matrix has too many or too few "
-                             "dimensions.")
-        return self._predict_features(X)
+            raise ValueError("expected a 2-D [observations, voxels] data "
+                             "matrix")
+        return self._predict_features(np.asarray(X))
 
     def score(self, X, y):
         """Per-observation R² of predicted 2-D features against y."""
-        pred_features = self.predict(X)
-        ssres = np.sum((pred_features - y) ** 2, axis=1)
+        pred = self.predict(X)
+        ssres = np.sum((pred - y) ** 2, axis=1)
         sstot = np.sum((y - np.mean(y)) ** 2, axis=1)
-        return 1 - (ssres / sstot)
+        return 1 - ssres / sstot
 
     def score_against_reconstructed(self, X, y, metric="euclidean"):
         """Distance between reconstructed pixel maps and expected maps."""
@@ -281,14 +308,12 @@ class InvertedEncoding2D:
             cosine_distances,
             euclidean_distances,
         )
-        yhat = self.predict_feature_responses(X)
-        if metric == "euclidean":
-            score_value = euclidean_distances(y.T, yhat.T)
-        elif metric == "cosine":
-            score_value = cosine_distances(y.T, yhat.T)
-        else:
+        dist_fn = {"euclidean": euclidean_distances,
+                   "cosine": cosine_distances}.get(metric)
+        if dist_fn is None:
             raise ValueError("metric must be 'euclidean' or 'cosine'")
-        return score_value[0, :]
+        yhat = self.predict_feature_responses(X)
+        return dist_fn(y.T, yhat.T)[0, :]
 
     def get_params(self, deep: bool = True):
         return {"n_channels": self.n_channels,
@@ -305,16 +330,18 @@ class InvertedEncoding2D:
         self._check_params()
         return self
 
+    # -- channel construction ----------------------------------------------
+
     def _make_2d_cosine(self, x, y, x_center, y_center, s):
-        """Exponentiated 2-D cosine bumps (zero beyond radius s)."""
-        cos_functions = np.zeros((len(x_center), len(x)))
-        for i in range(len(x_center)):
-            myr = np.sqrt((x - x_center[i]) ** 2
-                          + (y - y_center[i]) ** 2).squeeze()
-            qq = (myr <= s) * 1
-            zp = (0.5 * (1 + np.cos(myr * np.pi / s))) ** self.channel_exp
-            cos_functions[i, :] = zp * qq
-        return cos_functions
+        """Exponentiated 2-D cosine bumps (zero beyond radius s), as one
+        [n_centers, n_pixels] broadcast distance matrix."""
+        px = np.ravel(x)
+        py = np.ravel(y)
+        r = np.hypot(px[None, :] - np.ravel(x_center)[:, None],
+                     py[None, :] - np.ravel(y_center)[:, None])
+        bump = (0.5 * (1 + np.cos(np.minimum(r, s) * np.pi / s))) \
+            ** self.channel_exp
+        return np.where(r <= s, bump, 0.0)
 
     def _2d_cosine_sz_to_fwhm(self, size_constant):
         return 2 * size_constant * np.arccos(
@@ -328,85 +355,71 @@ class InvertedEncoding2D:
         """Square-grid channel layout; sets self.channels."""
         if not isinstance(nchannels, list):
             nchannels = [nchannels, nchannels]
-        chan_xcenters = np.linspace(self.channel_limits[0][0],
-                                    self.channel_limits[0][1],
-                                    nchannels[0])
-        chan_ycenters = np.linspace(self.channel_limits[1][0],
-                                    self.channel_limits[1][1],
-                                    nchannels[1])
-        cx, cy = np.meshgrid(chan_xcenters, chan_ycenters)
-        cx = cx.reshape(-1, 1)
-        cy = cy.reshape(-1, 1)
+        gx = np.linspace(*self.channel_limits[0], nchannels[0])
+        gy = np.linspace(*self.channel_limits[1], nchannels[1])
+        cx, cy = (g.reshape(-1) for g in np.meshgrid(gx, gy))
         if channel_size is None:
-            channel_size = 1.2 * (chan_xcenters[1] - chan_xcenters[0])
-        cos_width = self._2d_cosine_fwhm_to_sz(channel_size)
+            channel_size = 1.2 * (gx[1] - gx[0])
         self.channels = self._make_2d_cosine(
-            self.xp.reshape(-1, 1), self.yp.reshape(-1, 1), cx, cy,
-            cos_width)
+            self.xp, self.yp, cx, cy,
+            self._2d_cosine_fwhm_to_sz(channel_size))
         self.n_channels = self.channels.shape[0]
-        return self.channels, np.hstack([cx, cy])
+        return self.channels, np.stack([cx, cy], axis=1)
 
     def define_basis_functions_trigrid(self, grid_radius,
                                        channel_size=None):
-        """Triangular-grid channel layout; sets self.channels."""
-        x_dist = np.diff(self.channel_limits[0]) / (grid_radius * 2)
+        """Triangular-grid channel layout (odd rows offset half a step);
+        sets self.channels."""
+        (x0, x1), (y0, y1) = self.channel_limits
+        x_dist = float(x1 - x0) / (grid_radius * 2)
         y_dist = x_dist * np.sqrt(3) * 0.5
-        trigrid = np.zeros((0, 2))
-        xbase = np.expand_dims(
-            np.arange(self.channel_limits[0][0],
-                      self.channel_limits[0][1], x_dist.item()), 1)
-        for yi, yv in enumerate(np.arange(self.channel_limits[1][0],
-                                          self.channel_limits[1][1],
-                                          y_dist.item())):
-            if (yi % 2) == 0:
-                xx = xbase.copy()
-            else:
-                xx = xbase.copy() + x_dist / 2
-            yy = np.ones((xx.size, 1)) * yv
-            trigrid = np.vstack((trigrid, np.hstack((xx, yy))))
+        xs = np.arange(x0, x1, x_dist)
+        ys = np.arange(y0, y1, y_dist)
+        rows = [np.stack([xs + (x_dist / 2 if i % 2 else 0.0),
+                          np.full(xs.size, yv)], axis=1)
+                for i, yv in enumerate(ys)]
+        trigrid = np.concatenate(rows, axis=0)
         if channel_size is None:
             channel_size = 1.1 * x_dist
-        cos_width = self._2d_cosine_fwhm_to_sz(channel_size)
         self.channels = self._make_2d_cosine(
-            self.xp.reshape(-1, 1), self.yp.reshape(-1, 1),
-            trigrid[:, 0], trigrid[:, 1], cos_width)
+            self.xp, self.yp, trigrid[:, 0], trigrid[:, 1],
+            self._2d_cosine_fwhm_to_sz(channel_size))
         self.n_channels = self.channels.shape[0]
         return self.channels, trigrid
 
     def _define_trial_activations(self, stim_centers, stim_radius=None):
-        """[observations, channels] responses for circular stimuli."""
+        """[observations, channels] responses for circular stimuli.
+
+        NOTE (reference quirk, preserved): ``stim_radius`` thresholds
+        the SQUARED pixel distance, so the effective radius is
+        sqrt(stim_radius) pixels (ref iem.py:387-390).
+        """
+        stim_centers = np.asarray(stim_centers, dtype=float)
         nstim = stim_centers.shape[0]
         if self.stim_radius_px is None:
             if stim_radius is None:
-                raise ValueError("No defined stimulus radius. Please set.")
+                raise ValueError("stim_radius is unset; pass one here or at "
+                                 "construction")
             self.stim_radius_px = stim_radius
-        if not isinstance(self.stim_radius_px, (np.ndarray, list)):
-            self.stim_radius_px = np.ones(nstim) * self.stim_radius_px
-        stimulus_mask = np.zeros((self.xp.size, nstim))
-        for i in range(nstim):
-            rad_vals = ((self.xp.reshape(-1, 1) - stim_centers[i, 0]) ** 2
-                        + (self.yp.reshape(-1, 1)
-                           - stim_centers[i, 1]) ** 2)
-            inds = np.where(rad_vals < self.stim_radius_px[i])[0]
-            stimulus_mask[inds, i] = 1
-        C = self.channels.squeeze() @ stimulus_mask
-        C = C.transpose()
-        if np.linalg.matrix_rank(C) < self.n_channels:
-            warnings.warn("Stimulus matrix is {}, not full rank. May "
-                          "cause issues with stimulus prediction/"
-                          "reconstruction.".format(
-                              np.linalg.matrix_rank(C)), RuntimeWarning)
+        radii = np.broadcast_to(
+            np.asarray(self.stim_radius_px, dtype=float), (nstim,))
+        px, py = self.xp.reshape(-1), self.yp.reshape(-1)
+        d2 = ((px[:, None] - stim_centers[None, :, 0]) ** 2
+              + (py[:, None] - stim_centers[None, :, 1]) ** 2)
+        mask = (d2 < radii[None, :]).astype(float)   # [pixels, nstim]
+        C = (self.channels @ mask).T
+        _warn_if_rank_deficient(C, self.n_channels)
         return C
 
+    # -- inversion ---------------------------------------------------------
+
     def _predict_channel_responses(self, X):
-        return np.matmul(np.linalg.pinv(self.W_), X.transpose())
+        return _lstsq(self.W_, X.T)
 
     def predict_feature_responses(self, X):
-        return np.matmul(self.channels.transpose(),
-                         self._predict_channel_responses(X))
+        return self.channels.T @ self._predict_channel_responses(X)
 
     def _predict_features(self, X):
-        pred_response = self.predict_feature_responses(X)
-        feature_ind = np.argmax(pred_response, 0)
-        return np.hstack((self.xp.reshape(-1, 1)[feature_ind],
-                          self.yp.reshape(-1, 1)[feature_ind]))
+        peak = np.argmax(self.predict_feature_responses(X), axis=0)
+        return np.stack([self.xp.reshape(-1)[peak],
+                         self.yp.reshape(-1)[peak]], axis=1)
