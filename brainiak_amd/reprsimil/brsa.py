@@ -15,7 +15,11 @@ to batched C×C forms:
 
     A_v = XᵀK̃⁻¹X,  b_v = XᵀK̃⁻¹y_v,  q_v = y_vᵀK̃⁻¹y_v
 
-with σ_v² profiled out analytically.  ``BRSA`` optimizes the shared
+with σ_v² profiled out analytically.  Multiple runs (``scan_onsets``)
+enter through block-diagonal D/F templates — the AR(1) process restarts
+at every run onset, each run contributes its own log(1−ρ²) determinant
+term, and with ``baseline_single=False`` each run carries its own DC
+baseline regressor (ref brsa.py:969-1001, 310-323).  ``BRSA`` optimizes the shared
 (Cholesky-parameterized, optionally low-rank) covariance U, per-voxel
 SNR and AR coefficients jointly with scipy L-BFGS over the autograd
 gradient; ``GBRSA`` marginalizes per-voxel SNR/ρ over grids
@@ -120,17 +124,50 @@ def Ncomp_SVHT_MG_DLD_approx(X, zscore=True):
     return int(np.sum(sv > thresh))
 
 
-def _ar1_quadforms(X, Y):
-    """The six AR(1) building blocks: for precision I − ρD + ρ²F,
-    every needed quadratic form is a ρ-polynomial in these."""
+def _run_lengths(n_T, scan_onsets=None):
+    """Per-run TR counts from scan onset indices (0-length runs from
+    duplicated onsets are dropped, matching ref brsa.py:969-987)."""
+    if scan_onsets is None:
+        return np.array([n_T], dtype=int)
+    onsets = np.asarray(scan_onsets, dtype=int)
+    assert onsets.ndim == 1 and onsets.min() >= 0 \
+        and onsets.max() <= n_T, \
+        'scan_onsets must be 1-D indices into the time axis'
+    lens = np.diff(np.append(np.sort(onsets), n_T))
+    lens = lens[lens > 0]
+    if 0 not in onsets:
+        # leading segment before the first onset is its own run
+        lens = np.append(onsets.min(), lens) if onsets.min() > 0 else lens
+    return lens.astype(int)
+
+
+def _ar1_quadforms(X, Y, run_TRs=None):
+    """The six AR(1) building blocks: for the block-diagonal (per-run)
+    precision I − ρD + ρ²F, every needed quadratic form is a
+    ρ-polynomial in these.
+
+    Multi-run support: D carries the lag-1 adjacency only WITHIN each
+    run, and F marks each run's interior samples — so one pass over the
+    concatenated data yields the sum of per-run quad forms (the
+    reference's block_diag construction, ref brsa.py:988-1001).
+    """
     T = X.shape[0]
+    if run_TRs is None:
+        run_TRs = np.array([T], dtype=int)
+    bounds = np.cumsum(run_TRs)          # run end indices (exclusive)
+    assert bounds[-1] == T, 'run lengths must sum to the time axis'
+
+    adj = torch.ones(T - 1, dtype=_DT)
+    adj[torch.as_tensor(bounds[:-1] - 1, dtype=torch.long)] = 0.0
     D = torch.zeros((T, T), dtype=_DT)
     idx = torch.arange(T - 1)
-    D[idx, idx + 1] = 1.0
-    D[idx + 1, idx] = 1.0
-    F = torch.diag(torch.cat([torch.zeros(1, dtype=_DT),
-                              torch.ones(T - 2, dtype=_DT),
-                              torch.zeros(1, dtype=_DT)]))
+    D[idx, idx + 1] = adj
+    D[idx + 1, idx] = adj
+    interior = torch.ones(T, dtype=_DT)
+    starts = np.concatenate([[0], bounds[:-1]])
+    interior[torch.as_tensor(starts, dtype=torch.long)] = 0.0
+    interior[torch.as_tensor(bounds - 1, dtype=torch.long)] = 0.0
+    F = torch.diag(interior)
     XtX = X.T @ X
     XtDX = X.T @ D @ X
     XtFX = X.T @ F @ X
@@ -251,13 +288,24 @@ class _BRSACore:
         T, V = Y.shape
         assert X.shape[0] == T, \
             'design matrix and data must have the same number of TRs'
-        del scan_onsets  # single-run AR(1) model (see class docstring)
+        run_TRs = _run_lengths(T, scan_onsets)
 
-        # nuisance: user-provided + DC baseline
-        X0 = np.ones((T, 1))
+        # nuisance: user-provided + DC baseline.  With
+        # baseline_single=False each run carries its own DC regressor
+        # (scanner baseline drifts between runs, ref brsa.py:310-323);
+        # with True a single shared constant.
+        if getattr(self, 'baseline_single', False) or len(run_TRs) == 1:
+            dc = np.ones((T, 1))
+        else:
+            dc = np.zeros((T, len(run_TRs)))
+            start = 0
+            for i, n in enumerate(run_TRs):
+                dc[start:start + n, i] = 1.0
+                start += n
+        X0 = dc
         if nuisance is not None:
-            X0 = np.column_stack([np.asarray(nuisance), X0])
-        return X, Y, X0, T, V
+            X0 = np.column_stack([np.asarray(nuisance), dc])
+        return X, Y, X0, T, V, run_TRs
 
     def _residual_nuisance(self, X, Y, X0, n_nureg):
         """Top PCs of the residual after regressing out [X, X0]."""
@@ -322,7 +370,7 @@ class BRSA(_BRSACore):
 
     @staticmethod
     def _neg_loglik(params, quadX, quadXY, quadYY, C, V, T, rank,
-                    gp=None, tau_range=5.0):
+                    gp=None, tau_range=5.0, n_runs=1):
         """Negative marginal log-likelihood, σ² profiled out.
 
         params = [L_flat (C*rank), log_snr (V), rho_unc (V),
@@ -371,8 +419,10 @@ class BRSA(_BRSACore):
 
         sign, logdetM = torch.linalg.slogdet(M)
         # profiled σ̂² = quad / T
+        # logdet of the per-run AR(1) precision contributes
+        # log(1-rho^2) once PER RUN (ref _calc_LL's n_run factor)
         loglik = -0.5 * (T * torch.log(quad / T) + T
-                         - torch.log(1 - rho ** 2) + logdetM)
+                         - n_runs * torch.log(1 - rho ** 2) + logdetM)
         loglik = loglik - 0.5 * T * np.log(2 * np.pi)
         nll = -loglik.sum()
 
@@ -410,8 +460,10 @@ class BRSA(_BRSACore):
                                         gp['inten_smooth_range'])
         return nll
 
-    def _fit_once(self, X_t, Y_t, C, V, T, rank, init=None, gp=None):
-        quadX, quadXY, quadYY = _ar1_quadforms(X_t, Y_t)
+    def _fit_once(self, X_t, Y_t, C, V, T, rank, init=None, gp=None,
+                  run_TRs=None):
+        n_runs = 1 if run_TRs is None else len(run_TRs)
+        quadX, quadXY, quadYY = _ar1_quadforms(X_t, Y_t, run_TRs)
         nL = C * rank
         if init is None:
             rng = np.random.RandomState(self.random_state)
@@ -441,7 +493,8 @@ class BRSA(_BRSACore):
                 params.grad = None
             loss = self._neg_loglik(params, quadX, quadXY, quadYY, C, V,
                                     T, rank, gp=gp,
-                                    tau_range=self.tau_range)
+                                    tau_range=self.tau_range,
+                                    n_runs=n_runs)
             loss.backward()
             return float(loss.detach()), params.grad.numpy().copy()
 
@@ -496,8 +549,9 @@ class BRSA(_BRSACore):
         if design is None:
             design = y
         assert design is not None, 'design matrix is required'
-        X_design, Y_data, X0, T, V = self._prepare(design, X, nuisance,
-                                                   scan_onsets)
+        X_design, Y_data, X0, T, V, run_TRs = self._prepare(
+            design, X, nuisance, scan_onsets)
+        self._run_TRs_ = run_TRs
         C = X_design.shape[1]
         rank = self.rank if self.rank is not None else C
         rank = min(rank, C)
@@ -517,7 +571,8 @@ class BRSA(_BRSACore):
             X_t = torch.as_tensor(Xp, dtype=_DT)
             Y_t = torch.as_tensor(Yp, dtype=_DT)
             params, nll, quads = self._fit_once(X_t, Y_t, C, V, T, rank,
-                                                init=params, gp=gp)
+                                                init=params, gp=gp,
+                                                run_TRs=run_TRs)
             if prev_nll is not None and \
                     abs(prev_nll - nll) <= self.tol * abs(prev_nll):
                 break
@@ -525,7 +580,10 @@ class BRSA(_BRSACore):
             if self.auto_nuisance and round_i < rounds - 1:
                 comps, n_nureg = self._residual_nuisance(
                     X_design, Y_data, X0, n_nureg)
-                X0 = np.column_stack([comps, np.ones((T, 1))])
+                dc = X0[:, X0.shape[1] - (1 if self.baseline_single
+                                          or len(run_TRs) == 1
+                                          else len(run_TRs)):]
+                X0 = np.column_stack([comps, dc])
 
         # unpack
         nL = C * rank
@@ -635,15 +693,18 @@ class BRSA(_BRSACore):
         Yp = _project_out(Y, X0)
         T, V = Yp.shape
         C = Xp.shape[1]
+        run_TRs = _run_lengths(T, scan_onsets)
         quadX, quadXY, quadYY = _ar1_quadforms(
-            torch.as_tensor(Xp, dtype=_DT), torch.as_tensor(Yp, dtype=_DT))
+            torch.as_tensor(Xp, dtype=_DT), torch.as_tensor(Yp, dtype=_DT),
+            run_TRs)
         params = torch.tensor(np.concatenate([
             self.L_.ravel(), np.log(self.nSNR_),
             np.arctanh(np.clip(self.rho_, -0.999, 0.999))]), dtype=_DT)
         with torch.no_grad():
             nll = self._neg_loglik(params, quadX, quadXY, quadYY, C, V, T,
                                    self.L_.shape[1],
-                                   tau_range=self.tau_range)
+                                   tau_range=self.tau_range,
+                                   n_runs=len(run_TRs))
         return -float(nll) / V
 
     def _check_fitted(self):
@@ -700,7 +761,8 @@ class GBRSA(_BRSACore):
         w_rho = np.ones_like(rho) / len(rho)
         return s, w, rho, w_rho
 
-    def _neg_loglik_marg(self, L_params, quads, C, V, T, rank, grids):
+    def _neg_loglik_marg(self, L_params, quads, C, V, T, rank, grids,
+                         n_runs=1):
         (XtX, XtDX, XtFX), (XtY, XtDY, XtFY), (YtY, YtDY, YtFY) = quads
         s_grid, w_s, rho_grid, w_rho = grids
         L = L_params.reshape(C, rank)
@@ -732,7 +794,7 @@ class GBRSA(_BRSACore):
         quad = (q[None] - corr).clamp_min(1e-10)
         sign, logdetM = torch.linalg.slogdet(M)               # [S, R]
         ll = (-0.5 * (T * torch.log(quad / T) + T
-                      - torch.log(1 - rho_t ** 2)[None, :, None]
+                      - n_runs * torch.log(1 - rho_t ** 2)[None, :, None]
                       + logdetM[:, :, None])
               - 0.5 * T * np.log(2 * np.pi))                  # [S, R, V]
         ll_flat = ll.reshape(-1, V) + logw[:, None]
@@ -752,21 +814,29 @@ class GBRSA(_BRSACore):
         rank = min(self.rank if self.rank is not None else C, C)
         grids = self._grids()
 
+        if scan_onsets is None or not isinstance(scan_onsets, list):
+            scan_onsets = [scan_onsets] * len(X)
+
         subj_quads = []
         dims = []
+        subj_runs = []
         subj_ctx = []          # (design, Y, X0) for posterior extraction
-        for Xi, Di in zip(X, design):
-            Dp, Yi, X0, T, V = self._prepare(Di, Xi, None, scan_onsets)
+        for Xi, Di, onsets in zip(X, design, scan_onsets):
+            Dp, Yi, X0, T, V, run_TRs = self._prepare(Di, Xi, None,
+                                                      onsets)
             if self.auto_nuisance:
                 comps, _ = self._residual_nuisance(Dp, Yi, X0,
                                                    self.n_nureg)
-                X0 = np.column_stack([comps, np.ones((T, 1))])
+                n_dc = (1 if self.baseline_single or len(run_TRs) == 1
+                        else len(run_TRs))
+                X0 = np.column_stack([comps, X0[:, -n_dc:]])
             Xp = _project_out(Dp, X0)
             Yp = _project_out(Yi, X0)
             subj_quads.append(_ar1_quadforms(
                 torch.as_tensor(Xp, dtype=_DT),
-                torch.as_tensor(Yp, dtype=_DT)))
+                torch.as_tensor(Yp, dtype=_DT), run_TRs))
             dims.append((T, V))
+            subj_runs.append(len(run_TRs))
             subj_ctx.append((Dp, Yi, X0))
 
         rng = np.random.RandomState(self.random_state)
@@ -780,8 +850,9 @@ class GBRSA(_BRSACore):
             if params.grad is not None:
                 params.grad = None
             loss = sum(self._neg_loglik_marg(params, quads, C, V, T,
-                                             rank, grids)
-                       for quads, (T, V) in zip(subj_quads, dims))
+                                             rank, grids, n_runs=nr)
+                       for quads, (T, V), nr in zip(subj_quads, dims,
+                                                    subj_runs))
             loss.backward()
             return float(loss.detach()), params.grad.numpy().copy()
 
@@ -916,11 +987,12 @@ class GBRSA(_BRSACore):
                 Yp = _project_out(Y, X0)
                 T, V = Yp.shape
                 C = Xp.shape[1]
+                run_TRs = _run_lengths(T, scan_onsets)
                 quads = _ar1_quadforms(
                     torch.as_tensor(Xp, dtype=_DT),
-                    torch.as_tensor(Yp, dtype=_DT))
+                    torch.as_tensor(Yp, dtype=_DT), run_TRs)
                 nll = self._neg_loglik_marg(Lp, quads, C, V, T, rank,
-                                            grids)
+                                            grids, n_runs=len(run_TRs))
                 out.append(-float(nll) / V)
         return out[0] if single else out
 
