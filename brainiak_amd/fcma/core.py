@@ -40,6 +40,10 @@ __all__ = [
     "stack_epochs",
 ]
 
+# epoch lengths the HIP correlation kernels are templated for;
+# longer epochs fall back to the rocBLAS bmm + normalize path
+_HIP_EPOCH_LENGTHS = (8, 16, 24, 32, 40)
+
 
 def stack_epochs(raw_data: List[np.ndarray], device,
                  dtype=torch.float32) -> torch.Tensor:
@@ -52,8 +56,9 @@ def stack_epochs(raw_data: List[np.ndarray], device,
     L = max(m.shape[0] for m in raw_data)
     if device is not None and torch.device(device).type == "cuda":
         # pad to the HIP kernel's supported epoch lengths; zero rows are
-        # inert for z-scored data
-        for opt in (8, 16, 24, 32, 40):
+        # inert for z-scored data.  L > max(_HIP_EPOCH_LENGTHS) stays
+        # unpadded and routes through the rocBLAS bmm fallback.
+        for opt in _HIP_EPOCH_LENGTHS:
             if L <= opt:
                 L = opt
                 break
@@ -78,7 +83,7 @@ def normalize_correlation_(corr: torch.Tensor, epochs_per_subj: int
     biased variance over each subject's ``epochs_per_subj`` epochs, and
     zero out zero-variance entries.
     """
-    if corr.is_cuda and ops.has_hip():
+    if corr.is_cuda and ops.require_hip():
         ops.fcma_normalize_(corr, epochs_per_subj)
         return corr
     C, E, V = corr.shape
@@ -115,7 +120,7 @@ def _shrink_(gram: torch.Tensor) -> torch.Tensor:
 def gram_matrices(corr_norm: torch.Tensor, shrink: bool = True
                   ) -> torch.Tensor:
     """Per-voxel [E, E] Gram matrices of corr_norm [C, E, V]."""
-    if corr_norm.is_cuda and ops.has_hip():
+    if corr_norm.is_cuda and ops.require_hip():
         gram = ops.fcma_gram(corr_norm)
     else:
         gram = torch.bmm(corr_norm, corr_norm.transpose(1, 2))
@@ -153,10 +158,17 @@ class CorrelationPipeline:
         self.num_epochs = self.data.shape[0]
         self.num_voxels = self.data.shape[2]
         self.num_voxels2 = self.data2.shape[2]
+        # HIP kernels handle bf16 inputs at the templated epoch lengths;
+        # fp32 compute (use_bf16=False) or long epochs route through the
+        # rocBLAS bmm + normalize fallback on device
+        self._hip_path = (
+            self.device.type == "cuda" and self.use_bf16
+            and self.data.shape[1] in _HIP_EPOCH_LENGTHS
+            and ops.require_hip())
 
     def correlate_chunk(self, start: int, count: int) -> torch.Tensor:
         """corr [count, E, V2] fp32 for voxels [start, start+count)."""
-        if self.device.type == "cuda" and ops.has_hip():
+        if self._hip_path:
             return ops.fcma_correlate(self.data, self.data2, start, count)
         a = self.data[:, :, start:start + count].to(torch.float32)
         b = self.data2.to(torch.float32)
@@ -172,8 +184,7 @@ class CorrelationPipeline:
         for the headline shape (E=64, P=4) and as the two-kernel
         corr_norm_z + MFMA-Gram composite otherwise.
         """
-        if self.device.type == "cuda" and ops.has_hip() and \
-                ops.has_fused_gram():
+        if self._hip_path and ops.has_fused_gram():
             gram = ops.fcma_fused_gram(
                 self.data, self.data2, start, count, self.epochs_per_subj)
             if shrink:
@@ -194,8 +205,7 @@ class CorrelationPipeline:
         of the Gram time — measured ≈25 % off the whole FCMA step.
         """
         chunks = list(chunks)
-        if self.device.type != "cuda" or not ops.has_hip() or \
-                len(chunks) <= 1:
+        if not self._hip_path or len(chunks) <= 1:
             return torch.cat([self.chunk_kernel_matrices(s, c, shrink)
                               for s, c in chunks], dim=0)
         E = self.num_epochs

@@ -150,7 +150,7 @@ def _accuracy_gpu(kernels: torch.Tensor, labels: np.ndarray, num_folds: int,
         raise ValueError("GPU batched SVM supports binary labels; got "
                          f"{len(classes)} classes")
     from .. import ops
-    if ops.has_hip():
+    if kernels.is_cuda and ops.require_hip():
         folds = stratified_folds(labels, num_folds)
         if max(len(tr) for tr, _ in folds) <= 128 and \
                 max(len(te) for _, te in folds) <= 128:

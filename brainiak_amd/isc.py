@@ -252,7 +252,7 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
 
     if summary_statistic == 'mean':
         from . import ops as _ops
-        use_hip = dev.type == 'cuda' and _ops.has_hip()
+        use_hip = dev.type == 'cuda' and _ops.require_hip()
         acc = torch.zeros((V, V), dtype=torch.float32, device=dev)
         for d, nd in zip(local, normed_local):
             loo = _norm((total - d) / (n_total - 1))

@@ -28,6 +28,8 @@ import torch
 
 __all__ = [
     "batched_polar",
+    "polar_invsqrt",
+    "require_hip",
     "fcma_correlate",
     "fcma_fused_gram",
     "fcma_gram",
@@ -64,17 +66,28 @@ def load_extension():
 
 
 def has_hip() -> bool:
-    """True iff the HIP extension is loaded.  On a machine with visible
-    GPUs a missing extension is an ERROR, not a fallback."""
-    ext = load_extension()
-    if ext is not None:
+    """Pure capability probe: True iff the HIP extension is importable.
+    Never raises — CPU-only estimator use on a GPU box must not abort
+    just for probing."""
+    return load_extension() is not None
+
+
+def require_hip() -> bool:
+    """Dispatch-site gate: call when a CUDA tensor is about to run
+    through a HIP kernel.  True if the extension is loaded; raises if a
+    GPU is visible but the extension is missing — silently falling back
+    to eager torch there would invalidate every benchmark.  Set
+    BRAINIAK_AMD_ALLOW_NO_HIP=1 to get the eager fallback for
+    debugging."""
+    if load_extension() is not None:
         return True
     if torch.cuda.is_available() and not os.environ.get(
             "BRAINIAK_AMD_ALLOW_NO_HIP"):
         raise RuntimeError(
-            "brainiak_amd.ops._hip_ops is not built but a GPU is visible. "
-            "Run `python setup.py build_ext --inplace` (hipcc, gfx950); "
-            "refusing to fall back to eager torch on the GPU path.")
+            "brainiak_amd.ops._hip_ops is not built but a GPU tensor "
+            "reached a HIP dispatch site. Run `python setup.py "
+            "build_ext --inplace` (hipcc, gfx950); refusing to fall "
+            "back to eager torch on the GPU path.")
     return False
 
 
@@ -103,7 +116,8 @@ def fcma_normalize_(corr: torch.Tensor, epochs_per_subj: int) -> torch.Tensor:
 def fcma_correlate(data: torch.Tensor, data2: torch.Tensor,
                    start: int, count: int) -> torch.Tensor:
     """corr [count, E, V2] fp32 from stacked epochs data/data2 [E, L, V]
-    (bf16 or fp32)."""
+    (bf16 only — fp32 inputs use the CorrelationPipeline torch
+    fallback)."""
     return _ext().fcma_correlate(data, data2, int(start), int(count))
 
 
@@ -147,6 +161,14 @@ def jacobi_eigh(G: torch.Tensor):
     """Batched symmetric eigensolve of G [B, K, K] (K <= 64) → (evals,
     evecs), one wavefront per matrix."""
     return _ext().jacobi_eigh(G)
+
+
+def polar_invsqrt(G: torch.Tensor) -> torch.Tensor:
+    """G^{-1/2} [B, K, K] for a stack of SPD Gram matrices (batched
+    one-workgroup Jacobi eigensolve; K <= 64).  The ragged-batch
+    Procrustes building block: callers keep per-subject [V_i, K] GEMMs
+    and batch only the K x K eigensolves."""
+    return _ext().polar_invsqrt(G)
 
 
 def batched_polar(A: torch.Tensor, perturb: float = 0.001) -> torch.Tensor:

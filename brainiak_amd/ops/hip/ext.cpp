@@ -223,6 +223,22 @@ std::vector<torch::Tensor> jacobi_eigh(torch::Tensor G) {
     return {evals, evecs};
 }
 
+torch::Tensor polar_invsqrt(torch::Tensor G) {
+    // G^{-1/2} for a stack of SPD K x K Gram matrices via the batched
+    // one-workgroup Jacobi eigensolver.  Lets callers with RAGGED
+    // [V_i, K] factors (SRM subjects of different voxel counts) batch
+    // the eigensolve while keeping per-subject GEMMs.
+    TORCH_CHECK(G.is_cuda() && G.dim() == 3 && G.size(1) == G.size(2),
+                "G must be [B,K,K] on GPU");
+    TORCH_CHECK(G.size(1) <= 64, "K must be <= 64");
+    auto Gf = G.to(torch::kFloat32).contiguous();
+    auto ev = jacobi_eigh(Gf);
+    auto lam = ev[0].clamp_min(1e-30);
+    auto Vc = ev[1];
+    return torch::bmm(Vc * lam.rsqrt().unsqueeze(1),
+                      Vc.transpose(1, 2));
+}
+
 torch::Tensor batched_polar(torch::Tensor A, double perturb) {
     TORCH_CHECK(A.is_cuda() && A.dim() == 3, "A must be [B,V,K] on GPU");
     auto Af = A.to(torch::kFloat32).contiguous();
@@ -239,13 +255,8 @@ torch::Tensor batched_polar(torch::Tensor A, double perturb) {
                       diag + perturb);
     }
     auto G = torch::bmm(Af.transpose(1, 2), Af).contiguous();  // [B,K,K]
-    auto ev = jacobi_eigh(G);
-    auto lam = ev[0].clamp_min(1e-30);
-    auto Vc = ev[1];
-    // G^{-1/2} = Vec diag(rsqrt(lam)) Vec^T; W = A G^{-1/2}
-    auto inv_sqrt = torch::bmm(Vc * lam.rsqrt().unsqueeze(1),
-                               Vc.transpose(1, 2));
-    return torch::bmm(Af, inv_sqrt);
+    // W = A G^{-1/2}
+    return torch::bmm(Af, polar_invsqrt(G));
 }
 
 // ---------------------------------------------------------------------------
@@ -382,6 +393,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("jacobi_eigh", &jacobi_eigh, "batched KxK symmetric eigensolve");
     m.def("batched_polar", &batched_polar,
           "batched orthogonal Procrustes polar factor");
+    m.def("polar_invsqrt", &polar_invsqrt,
+          "G^{-1/2} for a stack of SPD KxK Gram matrices");
     m.def("tfa_factor", &tfa_factor, "TFA RBF factor matrix");
     m.def("tfa_recon", &tfa_recon, "TFA residual");
 }
