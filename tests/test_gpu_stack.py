@@ -128,3 +128,105 @@ def test_tfa_gpu_runs_hip_kernels(cuda, seeded_rng):
         corrs[dev] = np.corrcoef(recon.ravel(), X.ravel())[0, 1]
     assert corrs["cuda"] > 0.65
     assert abs(corrs["cuda"] - corrs["cpu"]) < 0.15
+
+
+def test_polar_invsqrt_matches_eigh(cuda, seeded_rng):
+    """Batched Jacobi G^{-1/2} vs torch.linalg.eigh oracle (ragged-batch
+    SRM building block)."""
+    from brainiak_amd import ops
+    B, K = 12, 50
+    A = torch.tensor(seeded_rng.randn(B, 200, K), dtype=torch.float32,
+                     device=cuda)
+    G = torch.bmm(A.transpose(1, 2), A)
+    got = ops.polar_invsqrt(G)
+    evals, evecs = torch.linalg.eigh(G.double())
+    want = ((evecs * evals.clamp_min(1e-30).rsqrt().unsqueeze(1))
+            @ evecs.transpose(1, 2)).float()
+    assert torch.allclose(got, want, atol=1e-4, rtol=1e-3)
+
+
+def test_srm_gpu_batched_procrustes_matches_cpu(cuda, seeded_rng):
+    """SRM fit on cuda (batched polar_invsqrt path) == CPU fit."""
+    from brainiak_amd.funcalign.srm import DetSRM
+    subjects = [seeded_rng.randn(40, 50) for _ in range(4)]
+    fits = {}
+    for dev in ("cpu", "cuda"):
+        m = DetSRM(n_iter=8, features=5, rand_seed=0, device=dev)
+        m.fit([s.copy() for s in subjects])
+        fits[dev] = m.s_
+    assert np.allclose(fits["cpu"], fits["cuda"], atol=1e-5)
+
+
+def _unit_epochs(rng, n, L, V):
+    """Z-scored, 1/sqrt(L)-scaled epochs (the pipeline's input
+    contract — keeps correlations in [-1, 1] so the Fisher-z clamp
+    branch is stable across dtypes)."""
+    out = []
+    for _ in range(n):
+        m = rng.randn(L, V).astype(np.float32)
+        m = (m - m.mean(0)) / m.std(0)
+        out.append((m / np.sqrt(L)).astype(np.float32))
+    return out
+
+
+def test_fcma_fp32_pipeline_on_gpu(cuda, seeded_rng):
+    """use_bf16=False on CUDA must run (rocBLAS fp32 fallback), not
+    raise, and match the CPU fp32 oracle."""
+    from brainiak_amd.fcma.core import CorrelationPipeline
+    raw = _unit_epochs(seeded_rng, 8, 12, 64)
+    gpu = CorrelationPipeline(raw, None, 2, device="cuda",
+                              use_bf16=False)
+    cpu = CorrelationPipeline(raw, None, 2, device="cpu")
+    g_gpu = gpu.chunk_kernel_matrices(0, 16).cpu()
+    g_cpu = cpu.chunk_kernel_matrices(0, 16)
+    assert torch.allclose(g_gpu, g_cpu, atol=1e-3, rtol=1e-3)
+
+
+def test_fcma_long_epochs_on_gpu(cuda, seeded_rng):
+    """Epoch length > 40 TRs must fall back to the rocBLAS path on GPU
+    instead of raising."""
+    from brainiak_amd.fcma.core import CorrelationPipeline
+    raw = _unit_epochs(seeded_rng, 8, 55, 48)
+    gpu = CorrelationPipeline(raw, None, 2, device="cuda")  # bf16 default
+    cpu = CorrelationPipeline(raw, None, 2, device="cpu")
+    g_gpu = gpu.pipelined_kernel_matrices([(0, 24), (24, 24)]).float().cpu()
+    cpu_b = CorrelationPipeline(raw, None, 2, device="cpu", use_bf16=True)
+    g_cpu = cpu_b.pipelined_kernel_matrices([(0, 24), (24, 24)]).float()
+    # same bf16 input quantization on both sides; fp32 accumulate
+    assert torch.allclose(g_gpu, g_cpu, atol=5e-2, rtol=5e-2)
+    # and the fp32 oracle agrees loosely
+    g_ref = cpu.pipelined_kernel_matrices([(0, 24), (24, 24)])
+    assert torch.allclose(g_gpu, g_ref, atol=0.5, rtol=0.2)
+
+
+def test_fcma_bf16_ranking_parity(cuda, seeded_rng):
+    """quantify voxel-ranking churn of the bf16 default vs
+    the fp32 chain at a moderate scale — top-quartile selection must
+    agree almost everywhere."""
+    from brainiak_amd.fcma.core import CorrelationPipeline
+    from brainiak_amd.fcma.svm import cross_validate_voxels
+    E, L, V = 32, 12, 512
+    labels = np.array([e % 2 for e in range(E)])
+    raw = []
+    for e in range(E):
+        m = seeded_rng.randn(L, V).astype(np.float32)
+        # plant label-dependent correlation structure in the first half
+        if e % 2:
+            m[:, :V // 2] += 0.8 * seeded_rng.randn(L, 1)
+        m = (m - m.mean(0)) / m.std(0)
+        raw.append((m / np.sqrt(L)).astype(np.float32))
+    accs = {}
+    for tag, bf16 in (("bf16", True), ("fp32", False)):
+        pipe = CorrelationPipeline(raw, None, 4, device="cuda",
+                                   use_bf16=bf16)
+        kernels = pipe.pipelined_kernel_matrices([(0, V)])
+        accs[tag] = np.asarray(cross_validate_voxels(kernels, labels, 4))
+    # rank correlation between the two accuracy vectors
+    from scipy.stats import spearmanr
+    r = spearmanr(accs["bf16"], accs["fp32"]).statistic
+    assert r > 0.9, r
+    # top-25% selections agree on >= 80% of members
+    k = V // 4
+    top_b = set(np.argsort(accs["bf16"])[-k:])
+    top_f = set(np.argsort(accs["fp32"])[-k:])
+    assert len(top_b & top_f) >= 0.8 * k
