@@ -27,6 +27,7 @@ from scipy import ndimage, optimize, signal, stats
 logger = logging.getLogger(__name__)
 
 __all__ = [
+    'default_brain_template',
     "apply_signal",
     "calc_noise",
     "compute_signal_change",
@@ -739,12 +740,37 @@ def _generate_noise_temporal(stimfunction_tr, tr_duration, dimensions,
     return noise_volume
 
 
+def default_brain_template(dimensions):
+    """Procedural standard-brain intensity template.
+
+    The reference bundles an MNI152 gray-matter template for
+    ``mask_brain`` (ref fmrisim.py:2230); shipping MNI data offline is
+    not possible here, so this builds a deterministic stand-in with the
+    properties mask_brain's threshold detector needs: a superellipsoid
+    head with a bright cortical shell over a dimmer interior, smooth
+    falloff to zero outside, values in [0, 1] with a bimodal histogram.
+    """
+    dims = np.asarray(dimensions[:3], dtype=int)
+    gx, gy, gz = np.meshgrid(
+        *(np.linspace(-1, 1, d) for d in dims), indexing='ij')
+    # head slightly egg-shaped: wider front-back (y), flat-bottomed (z)
+    r = ((gx / 0.82) ** 2 + (gy / 0.94) ** 2
+         + ((gz - 0.08) / 0.78) ** 2) ** 0.5
+    interior = np.clip(1.0 - r, 0, None)
+    core = 0.55 * np.tanh(6 * interior)           # white-matter plateau
+    shell = 0.45 * np.exp(-((r - 0.82) / 0.10) ** 2)   # cortical ring
+    template = np.clip(core + shell, 0, 1)
+    template[r > 1.0] = 0.0
+    return template
+
+
 def mask_brain(volume, template_name=None, mask_threshold=None,
                mask_self=True):
     """Build a (mask, template) pair from a volume (or a stored template).
 
-    No MNI template ships with this package: provide ``template_name``
-    (an .npy file) or keep ``mask_self=True``.
+    With ``mask_self=False`` and no ``template_name`` the procedural
+    :func:`default_brain_template` is used (see its docstring for why
+    no MNI data ships).
     """
     if len(volume.shape) == 1:
         volume = np.ones(volume.astype(int))
@@ -752,9 +778,7 @@ def mask_brain(volume, template_name=None, mask_threshold=None,
     if mask_self is True:
         mask_raw = volume
     elif template_name is None:
-        raise ValueError(
-            'No bundled MNI template in brainiak_amd; supply '
-            'template_name or use mask_self=True')
+        mask_raw = default_brain_template(volume.shape)
     else:
         mask_raw = np.load(template_name)
 

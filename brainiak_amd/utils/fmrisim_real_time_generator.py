@@ -6,9 +6,10 @@ paced in real time, with condition labels and a brain mask — the input
 stream for real-time analysis pipelines.
 
 Deviations: no bundled ROI/template NIfTIs — defaults are generated
-synthetically (two spherical ROIs inside a smooth ellipsoid template);
-DICOM output is unavailable in this stack (pydicom is not installed),
-``save_dicom=True`` raises.
+synthetically (two spherical ROIs inside a smooth ellipsoid template).
+``save_dicom=True`` writes real DICOM Part-10 files (``rt_###.dcm``,
+Explicit-VR LE multi-frame, via the self-contained
+``utils.dicom_minimal`` writer — pydicom is not installed here).
 """
 
 import logging
@@ -96,10 +97,6 @@ def generate_data(outputDir, user_settings):
     """
     data_dict = default_settings.copy()
     data_dict.update(user_settings)
-    if data_dict['save_dicom']:
-        raise NotImplementedError(
-            "DICOM output requires pydicom, which is not available; "
-            "volumes are saved as .npy")
 
     Path(outputDir).mkdir(parents=True, exist_ok=True)
 
@@ -188,9 +185,17 @@ def generate_data(outputDir, user_settings):
         start = time.time()
         brain = noise[:, :, :, idx] + signal[:, :, :, idx]
         brain_int32 = np.nan_to_num(brain).astype(np.int32)
-        output_file = os.path.join(outputDir,
-                                   'rt_' + format(idx, '03d') + '.npy')
-        np.save(output_file, brain_int32)
+        if data_dict['save_dicom']:
+            from .dicom_minimal import write_dicom
+            output_file = os.path.join(
+                outputDir, 'rt_' + format(idx, '03d') + '.dcm')
+            write_dicom(output_file, np.clip(brain_int32, 0, 65535),
+                        instance_number=idx + 1,
+                        tr_seconds=data_dict['trDuration'])
+        else:
+            output_file = os.path.join(
+                outputDir, 'rt_' + format(idx, '03d') + '.npy')
+            np.save(output_file, brain_int32)
         if data_dict['save_realtime']:
             elapsed = time.time() - start
             remaining = data_dict['trDuration'] - elapsed

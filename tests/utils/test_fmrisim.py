@@ -165,3 +165,61 @@ def test_1d_rfs():
     data = sim.generate_1d_rf_responses(rfs, np.array([10, 90, 170]),
                                         180, (0, 179))
     assert data.shape == (20, 3)
+
+
+def test_arma_mle_recovers_coefficients(seeded_rng):
+    """Batched conditional-MLE ARMA(1,1): planted (phi, theta) recovery
+    across a batch — sharper than the moment fallback."""
+    rho, theta = 0.55, 0.3
+    B, n = 8, 1500
+    X = np.zeros((B, n))
+    for b in range(B):
+        e = seeded_rng.randn(n)
+        for t in range(1, n):
+            X[b, t] = rho * X[b, t - 1] + e[t] + theta * e[t - 1]
+    ar, ma = sim._estimate_arma_mle_batch(
+        X - X.mean(axis=1, keepdims=True), 1, 1)
+    assert np.abs(ar.mean() - rho) < 0.08, ar.mean()
+    assert np.abs(ma.mean() - theta) < 0.12, ma.mean()
+    # moment estimator on the same data is no better
+    ar_m = np.array([sim._estimate_ar_ma(X[b], 1, 1)[0][0]
+                     for b in range(B)])
+    assert np.abs(ar.mean() - rho) <= np.abs(ar_m.mean() - rho) + 0.05
+
+
+def test_calc_noise_roundtrip_recovers_noise_dict():
+    """The reference's own oracle (ref tests/utils/test_fmrisim.py):
+    calc_noise on generate_noise output recovers the input noise_dict
+    within tolerance — now with the MLE ARMA path."""
+    np.random.seed(11)
+    dims = np.array([14, 14, 14])
+    stim_tr = np.zeros(60)
+    template = np.zeros(dims)
+    template[4:10, 4:10, 4:10] = 0.9
+    mask = (template > 0).astype(float)
+    target = {'sfnr': 70, 'snr': 35, 'max_activity': 600,
+              'auto_reg_rho': [0.6], 'ma_rho': [0.0], 'matched': 1,
+              'fwhm': 4.0}
+    noise = sim.generate_noise(dims, stim_tr, 2.0, template, mask,
+                               noise_dict=dict(target),
+                               iterations=[10, 5])
+    est = sim.calc_noise(noise, mask, template)
+    assert abs(est['sfnr'] - target['sfnr']) / target['sfnr'] < 0.35
+    assert abs(est['auto_reg_rho'][0] - 0.6) < 0.25, est['auto_reg_rho']
+
+
+def test_default_brain_template_and_mask():
+    tpl = sim.default_brain_template((32, 32, 24))
+    assert tpl.shape == (32, 32, 24)
+    assert tpl.min() >= 0 and tpl.max() <= 1
+    # center is brain, corners are empty
+    assert tpl[16, 16, 12] > 0.3
+    assert tpl[0, 0, 0] == 0 and tpl[-1, -1, -1] == 0
+    # mask_brain with the default template (mask_self=False, no file)
+    vol = np.random.rand(20, 20, 16, 3) * 100
+    mask, template = sim.mask_brain(vol, mask_self=False)
+    assert mask.shape == (20, 20, 16)
+    assert mask.sum() > 50            # a real brain-sized region
+    assert mask[0, 0, 0] == 0
+    # template resampled to the volume grid
+    assert template.shape == (20, 20, 16)
