@@ -46,6 +46,9 @@ def parse_args():
                     help="pipeline chunk (voxels per kernel pass)")
     ap.add_argument("--no-cv", action="store_true",
                     help="skip the SVM CV stage (pipeline only)")
+    ap.add_argument("--fp8", action="store_true",
+                    help="fp8(e4m3) Z tile between corr and Gram "
+                         "(BRAINIAK_FP8=1 equivalent)")
     ap.add_argument("--device", default=None)
     return ap.parse_args()
 
@@ -94,7 +97,8 @@ def main():
     labels = np.array([e % 2 for e in range(E)])
     pipeline = CorrelationPipeline(
         raw, None, args.epochs_per_subj, device=device,
-        use_bf16=device.type == "cuda")
+        use_bf16=device.type == "cuda",
+        z_fp8=args.fp8 if args.fp8 else None)
     del raw
 
     # this rank's voxel range rotates so steps touch different voxels

@@ -47,13 +47,16 @@ _HIP_EPOCH_LENGTHS = (8, 16, 24, 32, 40)
 
 def stack_epochs(raw_data: List[np.ndarray], device,
                  dtype=torch.float32) -> torch.Tensor:
-    """Stack per-epoch [len_e, V] matrices into [E, L, V] (zero-padded).
+    """Stack per-epoch [len_e, V] matrices into [E, L, Vp] (zero-padded).
 
     The inputs are z-scored and 1/sqrt(len)-scaled, so zero padding does
-    not change any correlation dot product.
+    not change any correlation dot product.  On GPU the voxel dimension
+    pads to a multiple of 16 (zero columns are equally inert) so fp8/
+    bf16 Z rows stay 16-byte aligned for vector loads.
     """
     E = len(raw_data)
     L = max(m.shape[0] for m in raw_data)
+    V = raw_data[0].shape[1]
     if device is not None and torch.device(device).type == "cuda":
         # pad to the HIP kernel's supported epoch lengths; zero rows are
         # inert for z-scored data.  L > max(_HIP_EPOCH_LENGTHS) stays
@@ -62,7 +65,7 @@ def stack_epochs(raw_data: List[np.ndarray], device,
             if L <= opt:
                 L = opt
                 break
-    V = raw_data[0].shape[1]
+        V = ((V + 15) // 16) * 16
     out = torch.zeros((E, L, V), dtype=dtype, device=device)
     for e, m in enumerate(raw_data):
         if isinstance(m, torch.Tensor):
@@ -70,7 +73,7 @@ def stack_epochs(raw_data: List[np.ndarray], device,
         else:
             t = torch.as_tensor(np.ascontiguousarray(m),
                                 dtype=torch.float32).to(device).to(dtype)
-        out[e, :m.shape[0], :] = t
+        out[e, :m.shape[0], :m.shape[1]] = t
     return out
 
 
@@ -143,7 +146,8 @@ class CorrelationPipeline:
     """
 
     def __init__(self, raw_data, raw_data2, epochs_per_subj, device=None,
-                 use_bf16: Optional[bool] = None):
+                 use_bf16: Optional[bool] = None,
+                 z_fp8: Optional[bool] = None):
         self.device = torch.device(device) if device is not None else (
             torch.device("cuda") if torch.cuda.is_available()
             else torch.device("cpu"))
@@ -151,13 +155,22 @@ class CorrelationPipeline:
             use_bf16 = self.device.type == "cuda"
         self.use_bf16 = use_bf16
         gemm_dtype = torch.bfloat16 if use_bf16 else torch.float32
+        if z_fp8 is None:
+            z_fp8 = bool(os.environ.get("BRAINIAK_FP8"))
+        # fp8(e4m3) Z tile: halves the normalized-correlation HBM round
+        # trip between the corr and Gram kernels (~3 % RMS quantization
+        # on z-scores; selection-accuracy parity is the tested contract)
+        self.z_fp8 = bool(z_fp8) and self.device.type == "cuda"
         self.epochs_per_subj = epochs_per_subj
         self.data = stack_epochs(raw_data, self.device, gemm_dtype)
         self.data2 = (stack_epochs(raw_data2, self.device, gemm_dtype)
                       if raw_data2 is not None else self.data)
         self.num_epochs = self.data.shape[0]
-        self.num_voxels = self.data.shape[2]
-        self.num_voxels2 = self.data2.shape[2]
+        # logical voxel counts (the stacked tensors may be 16-padded)
+        self.num_voxels = (raw_data[0].shape[1] if raw_data
+                           else self.data.shape[2])
+        self.num_voxels2 = (raw_data2[0].shape[1] if raw_data2
+                            else self.num_voxels)
         # HIP kernels handle bf16 inputs at the templated epoch lengths;
         # fp32 compute (use_bf16=False) or long epochs route through the
         # rocBLAS bmm + normalize fallback on device
@@ -167,14 +180,18 @@ class CorrelationPipeline:
             and ops.require_hip())
 
     def correlate_chunk(self, start: int, count: int) -> torch.Tensor:
-        """corr [count, E, V2] fp32 for voxels [start, start+count)."""
+        """corr [count, E, V2] fp32 for voxels [start, start+count)
+        (V2 = logical voxel count; any 16-pad columns are sliced off)."""
         if self._hip_path:
-            return ops.fcma_correlate(self.data, self.data2, start, count)
-        a = self.data[:, :, start:start + count].to(torch.float32)
-        b = self.data2.to(torch.float32)
-        # [E, C, V] then→ [C, E, V]
-        out = torch.bmm(a.transpose(1, 2), b)
-        return out.transpose(0, 1).contiguous()
+            out = ops.fcma_correlate(self.data, self.data2, start, count)
+        else:
+            a = self.data[:, :, start:start + count].to(torch.float32)
+            b = self.data2.to(torch.float32)
+            # [E, C, V] then→ [C, E, V]
+            out = torch.bmm(a.transpose(1, 2), b).transpose(0, 1)
+        if out.shape[2] != self.num_voxels2:
+            out = out[:, :, :self.num_voxels2]
+        return out.contiguous()
 
     def chunk_kernel_matrices(self, start: int, count: int,
                               shrink: bool = True) -> torch.Tensor:
@@ -230,12 +247,15 @@ class CorrelationPipeline:
         max_count = max(c for _, c in chunks)
         # persistent double-buffered Z workspace: repeated multi-GB
         # allocations churn the caching allocator across streams
+        zdtype = torch.float8_e4m3fn if self.z_fp8 else torch.bfloat16
+        VB = self.data2.shape[2]        # padded voxel count
         if getattr(self, "_zbuf", None) is None or \
                 self._zbuf[0].shape[0] < max_count or \
-                self._zbuf[0].shape[1] != Epad:
+                self._zbuf[0].shape[1] != Epad or \
+                self._zbuf[0].dtype != zdtype:
             self._zbuf = [
-                torch.zeros((max_count, Epad, self.num_voxels2),
-                            dtype=torch.bfloat16, device=self.device)
+                torch.zeros((max_count, Epad, VB),
+                            dtype=zdtype, device=self.device)
                 for _ in range(2)]
         if getattr(self, "_streams", None) is None:
             self._streams = (torch.cuda.Stream(device=self.device),
@@ -253,6 +273,8 @@ class CorrelationPipeline:
                 if skip_gram:
                     g = torch.zeros((z.shape[0], Epad, Epad),
                                     dtype=torch.float32, device=z.device)
+                elif self.z_fp8:
+                    g = ops.fcma_gram_fp8(z)
                 else:
                     g = ops.fcma_gram_bf16(z)
                 if Epad != E:

@@ -126,6 +126,12 @@ def fcma_gram(corr_norm: torch.Tensor) -> torch.Tensor:
     return _ext().fcma_gram(corr_norm)
 
 
+def fcma_gram_fp8(Z: torch.Tensor) -> torch.Tensor:
+    """Per-voxel Gram [C, E, E] fp32 of Z [C, E, V] float8_e4m3fn
+    (E % 64 == 0, V % 16 == 0 — host pads both)."""
+    return _ext().fcma_gram_fp8(Z)
+
+
 def fcma_gram_bf16(Z: torch.Tensor) -> torch.Tensor:
     """Per-voxel Gram [C, E, E] fp32 of Z [C, E, V] bf16 (E % 64 == 0)."""
     return _ext().fcma_gram_bf16(Z)

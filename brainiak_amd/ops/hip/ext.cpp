@@ -18,7 +18,11 @@ void launch_fcma_corr_norm(const void*, const void*, void*, float*, ll, ll,
 int fcma_corr_variant(void);
 int fcma_corr_norm_smem(ll, int);
 void launch_fcma_gram_bf16(const void*, float*, ll, ll, ll, ll, void*);
+void launch_fcma_gram_fp8(const void*, float*, ll, ll, ll, ll, void*);
 void launch_fcma_gram_f32(const float*, float*, ll, ll, ll, void*);
+int fcma_corr_norm_z8_supported(ll, int, ll);
+void launch_fcma_corr_norm_z8(const void*, const void*, void*, ll, ll,
+                              ll, ll, ll, int, void*);
 void launch_jacobi_eigh(const float*, float*, float*, ll, int, void*);
 void launch_tfa_factor(const float*, const float*, const float*, float*,
                        ll, int, void*);
@@ -101,12 +105,15 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
                 "epoch length must be padded to one of {8,16,24,32,40}");
     ll Eout = std::max((ll)padE, E);
     torch::Tensor Z;
+    bool fp8 = false;
     if (out_opt.has_value()) {
         // caller-provided persistent buffer (rows >= E; padding rows
-        // must be pre-zeroed once by the caller)
+        // must be pre-zeroed once by the caller).  A Float8_e4m3fn
+        // buffer selects the fp8 Z path (half the HBM round trip).
         Z = out_opt.value();
+        fp8 = Z.scalar_type() == torch::kFloat8_e4m3fn;
         TORCH_CHECK(Z.is_cuda() && Z.is_contiguous()
-                    && Z.scalar_type() == torch::kBFloat16
+                    && (fp8 || Z.scalar_type() == torch::kBFloat16)
                     && Z.size(0) >= count && Z.size(1) >= Eout
                     && Z.size(2) == VB, "bad out buffer");
         Eout = Z.size(1);
@@ -125,6 +132,17 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
         At_ptr = At.data_ptr();
     }
     // the kernel writes rows [0, E) with row stride Eout directly
+    if (fp8) {
+        TORCH_CHECK(At_ptr != nullptr,
+                    "fp8 Z output needs the dot3s variant "
+                    "(BRAINIAK_CORR_VARIANT=3, the default)");
+        TORCH_CHECK(fcma_corr_norm_z8_supported(E, (int)P, L),
+                    "fp8 Z output needs P in {2,4} and even L");
+        launch_fcma_corr_norm_z8(At_ptr, B.data_ptr(), Z.data_ptr(),
+                                 E, L, VB, count, Eout, (int)P,
+                                 cur_stream());
+        return Z;
+    }
     launch_fcma_corr_norm(A.data_ptr(), B.data_ptr(), Z.data_ptr(),
                           nullptr, E, L, VA, VB, start, count, (int)P,
                           /*mode=*/0, Eout, cur_stream(), At_ptr);
@@ -152,6 +170,31 @@ torch::Tensor fcma_gram_bf16(torch::Tensor Z) {
                            Z.options().dtype(torch::kFloat32));
     launch_fcma_gram_bf16(Z.data_ptr(), Gp.data_ptr<float>(), C, E, V,
                           nsplit, cur_stream());
+    return Gp.sum(0);
+}
+
+torch::Tensor fcma_gram_fp8(torch::Tensor Z) {
+    TORCH_CHECK(Z.is_cuda() && Z.dim() == 3 && Z.is_contiguous()
+                && Z.scalar_type() == torch::kFloat8_e4m3fn,
+                "Z must be contiguous [C,E,V] float8_e4m3fn on GPU");
+    ll C = Z.size(0), E = Z.size(1), V = Z.size(2);
+    TORCH_CHECK(E % 64 == 0, "E must be a multiple of 64 (host pads)");
+    TORCH_CHECK(V % 16 == 0, "V must be a multiple of 16 (host pads)");
+    ll eb = E / 64;
+    ll base = C * eb * eb;
+    ll ktAll = (V + 255) / 256;
+    ll nsplit = std::min(ktAll, std::max((ll)1, (2047 + base) / base));
+    if (nsplit <= 1) {
+        auto G = torch::empty({C, E, E},
+                              Z.options().dtype(torch::kFloat32));
+        launch_fcma_gram_fp8(Z.data_ptr(), G.data_ptr<float>(), C, E, V,
+                             1, cur_stream());
+        return G;
+    }
+    auto Gp = torch::empty({nsplit, C, E, E},
+                           Z.options().dtype(torch::kFloat32));
+    launch_fcma_gram_fp8(Z.data_ptr(), Gp.data_ptr<float>(), C, E, V,
+                         nsplit, cur_stream());
     return Gp.sum(0);
 }
 
@@ -384,6 +427,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("padE"),
           pybind11::arg("out") = pybind11::none());
     m.def("fcma_gram", &fcma_gram, "per-voxel Gram from fp32 [C,E,V]");
+    m.def("fcma_gram_fp8", &fcma_gram_fp8,
+          "per-voxel Gram from fp8(e4m3) Z [C,E,V]");
     m.def("fcma_gram_bf16", &fcma_gram_bf16,
           "per-voxel Gram from bf16 Z [C,E,V]");
     m.def("fcma_fused_gram_native", &fcma_fused_gram_native,

@@ -36,6 +36,32 @@ typedef short bf16x8 __attribute__((ext_vector_type(8)));
 typedef short bf16x8_u __attribute__((ext_vector_type(8), aligned(4)));
 typedef long long ll;
 
+// --- fp8 (OCP e4m3fn) Z-tile path -----------------------------------------
+// The normalized Z values are within-subject z-scores, bounded by
+// sqrt(P-1) (<= 1.74 for the headline P=4), so e4m3's 3-bit mantissa
+// (~3 % RMS quantization) is the only loss; the Gram accumulates in
+// fp32 MFMA.  Halves the Z HBM round trip (profiles/NEXT.md item 1).
+#include <hip/amd_detail/amd_hip_fp8.h>
+typedef unsigned char fp8_t;
+typedef unsigned int uint32x4 __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ fp8_t to_fp8(float v) {
+    __hip_fp8_e4m3 t(v);
+    return t.__x;
+}
+
+// store-cast for the corr kernel's templated output type
+template <typename OT>
+__device__ __forceinline__ OT z_cast(float v);
+template <>
+__device__ __forceinline__ bf16_t z_cast<bf16_t>(float v) {
+    return (bf16_t)v;
+}
+template <>
+__device__ __forceinline__ fp8_t z_cast<fp8_t>(float v) {
+    return to_fp8(v);
+}
+
 // ---------------------------------------------------------------------------
 // MFMA fragment maps, v_mfma_f32_16x16x32_bf16 (gfx950):
 //   A (16x32): lane l, j in 0..7 -> A[l%16][frag_k(l,j)]
@@ -564,10 +590,10 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3(
 // transposed to [count, E, L] (contiguous per voxel), and the kernel
 // needs NO LDS, NO staging loop and NO barrier at all.
 // ===========================================================================
-template <int TP, int TL, int C3_CT>
+template <int TP, int TL, int C3_CT, typename OT = bf16_t>
 __global__ __launch_bounds__(256) void k_corr_norm_dot3s(
     const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
-    bf16_t* __restrict__ zOut, ll E, ll VB,
+    OT* __restrict__ zOut, ll E, ll VB,
     ll C, int mode, ll zstride) {
     static_assert(TL % 2 == 0 && TP >= 2 && TP <= 4,
                   "dot3s kernel: even L, P in {2,4}");
@@ -633,13 +659,13 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3s(
             for (int p = 0; p < P; ++p)
                 if (zstride < 0)
                     zOut[(size_t)threadIdx.x] =
-                        (bf16_t)((z[p] - mean) * inv);
+                        z_cast<OT>((z[p] - mean) * inv);
         } else {
-            bf16_t* dst = zOut
+            OT* dst = zOut
                 + ((c0 + c) * zstride + s * (ll)P) * VB + v;
             #pragma unroll
             for (int p = 0; p < P; ++p)
-                dst[(size_t)p * VB] = (bf16_t)((z[p] - mean) * inv);
+                dst[(size_t)p * VB] = z_cast<OT>((z[p] - mean) * inv);
         }
     }
 }
@@ -1145,6 +1171,129 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
 }
 
 // ===========================================================================
+// k_gram_fp8: G_c = Z_c Z_c^T per voxel with fp8(e4m3) MFMA
+// (v_mfma_f32_16x16x32_fp8_fp8, fp32 accumulate).  Z: [C, E, V] fp8,
+// V % 16 == 0 (host pads voxels); G: [C, E, E] fp32; E % 64 == 0.
+// Same band/V-split decomposition and T14 issue-early staging as
+// k_gram_bf16, but each staged tile carries 256 voxels in the same
+// 16 KB — the Z re-read from HBM halves (profiles/NEXT.md item 1).
+// ===========================================================================
+#define G8_KT 256
+#define G8_PAD 8
+#define G8_ROW (G8_KT + G8_PAD)
+
+__global__ __launch_bounds__(256) void k_gram_fp8(
+    const fp8_t* __restrict__ Z, float* __restrict__ G,
+    ll C, ll E, ll V, ll nsplit) {
+    const ll eb = E / 64;
+    ll b = blockIdx.x;
+    const ll ns = b % nsplit; b /= nsplit;
+    const ll band_j = b % eb; b /= eb;
+    const ll band_i = b % eb; b /= eb;
+    const ll c = b;
+    if (c >= C || band_j < band_i) return;
+    const ll ktAll = (V + G8_KT - 1) / G8_KT;
+    const ll ktPer = (ktAll + nsplit - 1) / nsplit;
+    const ll kt0 = ns * ktPer;
+    const ll kt1 = min(ktAll, kt0 + ktPer);
+    if (kt0 >= kt1) {
+        for (ll i = threadIdx.x; i < E * E; i += 256)
+            G[(ns * C + c) * E * E + i] = 0.0f;
+        return;
+    }
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int w = tid >> 6;
+    const int wr = (w >> 1) * 32;
+    const int wc = (w & 1) * 32;
+
+    __shared__ fp8_t zi[64][G8_ROW];
+    __shared__ fp8_t zj[64][G8_ROW];
+    const bool diag = (band_i == band_j);
+    const fp8_t* Zc = Z + c * E * V;
+    const ll rows_i = band_i * 64;
+    const ll rows_j = band_j * 64;
+
+    f32x4 acc00 = (f32x4)0.f, acc01 = (f32x4)0.f;
+    f32x4 acc10 = (f32x4)0.f, acc11 = (f32x4)0.f;
+
+    // 64 rows x 256 B per tile; 4 threads per row, 4x16 B each.
+    // V % 16 == 0 keeps every global load dword4-aligned.
+    const int srow = tid >> 2;
+    const int scol = (tid & 3) * 64;
+
+    auto issue_loads = [&](const fp8_t* src, ll rows0, ll k0,
+                           uint32x4 regs[4]) {
+        #pragma unroll
+        for (int h = 0; h < 4; ++h) {
+            ll kk = k0 + scol + 16 * h;
+            const fp8_t* sp = src + (rows0 + srow) * V + kk;
+            if (kk + 16 <= V) {
+                regs[h] = *(const uint32x4*)sp;
+            } else {
+                regs[h] = (uint32x4)0u;   // V is 16-padded: whole
+            }                             // vector in or out of range
+        }
+    };
+    auto write_tile = [&](fp8_t dst[64][G8_ROW], uint32x4 regs[4]) {
+        #pragma unroll
+        for (int h = 0; h < 4; ++h)
+            *(uint32x4*)&dst[srow][scol + 16 * h] = regs[h];
+    };
+
+    uint32x4 ri[4], rj[4];
+    issue_loads(Zc, rows_i, kt0 * G8_KT, ri);
+    if (!diag) issue_loads(Zc, rows_j, kt0 * G8_KT, rj);
+
+    for (ll kt = kt0; kt < kt1; ++kt) {
+        __syncthreads();
+        write_tile(zi, ri);
+        if (!diag) write_tile(zj, rj);
+        if (kt + 1 < kt1) {
+            issue_loads(Zc, rows_i, (kt + 1) * G8_KT, ri);
+            if (!diag) issue_loads(Zc, rows_j, (kt + 1) * G8_KT, rj);
+        }
+        __syncthreads();
+        const int frow = lane & 15;
+        #pragma unroll
+        for (int ks = 0; ks < G8_KT / 32; ++ks) {
+            const int fk = 8 * (lane >> 4) + 32 * ks;
+            long fi0 = *(const long*)&zi[wr + frow][fk];
+            long fi1 = *(const long*)&zi[wr + 16 + frow][fk];
+            const fp8_t (*zjs)[G8_ROW] = diag ? zi : zj;
+            long fj0 = *(const long*)&zjs[wc + frow][fk];
+            long fj1 = *(const long*)&zjs[wc + 16 + frow][fk];
+            acc00 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                fi0, fj0, acc00, 0, 0, 0);
+            acc01 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                fi0, fj1, acc01, 0, 0, 0);
+            acc10 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                fi1, fj0, acc10, 0, 0, 0);
+            acc11 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                fi1, fj1, acc11, 0, 0, 0);
+        }
+    }
+
+    float* Gc = G + (ns * C + c) * E * E;
+    const int dcol = lane & 15;
+    const int drow = (lane >> 4) * 4;
+    const f32x4* accs[4] = {&acc00, &acc01, &acc10, &acc11};
+    #pragma unroll
+    for (int q = 0; q < 4; ++q) {
+        int mi = q >> 1, ni = q & 1;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            ll gr = rows_i + wr + 16 * mi + drow + r;
+            ll gc = rows_j + wc + 16 * ni + dcol;
+            float val = (*accs[q])[r];
+            Gc[gr * E + gc] = val;
+            if (!diag || gr != gc) Gc[gc * E + gr] = val;
+        }
+    }
+}
+
+// ===========================================================================
 // k_gram_f32: the same contraction from fp32 via exact-f32 MFMA
 // v_mfma_f32_16x16x4_f32 (A: lane l -> A[l&15][l>>4]).
 // ===========================================================================
@@ -1490,6 +1639,52 @@ extern "C" void launch_fcma_gram_bf16(const void* Z, float* G, ll C, ll E,
     ll grid = C * eb * eb * nsplit;
     hipLaunchKernelGGL(k_gram_bf16, dim3(grid), dim3(256), 0, stream,
                        (const bf16_t*)Z, G, C, E, V, nsplit);
+}
+
+extern "C" void launch_fcma_gram_fp8(const void* Z, float* G, ll C, ll E,
+                                     ll V, ll nsplit,
+                                     hipStream_t stream) {
+    ll eb = E / 64;
+    ll grid = C * eb * eb * nsplit;
+    hipLaunchKernelGGL(k_gram_fp8, dim3(grid), dim3(256), 0, stream,
+                       (const fp8_t*)Z, G, C, E, V, nsplit);
+}
+
+// fp8-output corr+normalize: the dot3s kernel with an e4m3 Z store.
+// P in {2, 4}, even L (the production dot3s envelope); At required.
+extern "C" int fcma_corr_norm_z8_supported(ll E, int P, ll L) {
+    return (P == 2 || P == 4) && (L % 2) == 0 && (E % P) == 0;
+}
+
+extern "C" void launch_fcma_corr_norm_z8(const void* At, const void* B,
+                                         void* zOut, ll E, ll L, ll VB,
+                                         ll C, ll zstride,
+                                         int P, hipStream_t stream) {
+    ll nSubj = E / P;
+    ll grid3 = ceil_div(C, 128) * nSubj * ceil_div(VB, C3_VT);
+    #define Z8_CASE(TP, TL)                                              \
+        hipLaunchKernelGGL((k_corr_norm_dot3s<TP, TL, 128, fp8_t>),      \
+                           dim3(grid3), dim3(256), 0, stream,            \
+                           (const bf16_t*)At, (const bf16_t*)B,          \
+                           (fp8_t*)zOut, E, VB, C, /*mode=*/0, zstride)
+    if (P == 4) {
+        switch (L) {
+            case 8:  Z8_CASE(4, 8);  return;
+            case 16: Z8_CASE(4, 16); return;
+            case 24: Z8_CASE(4, 24); return;
+            case 32: Z8_CASE(4, 32); return;
+            case 40: Z8_CASE(4, 40); return;
+        }
+    } else if (P == 2) {
+        switch (L) {
+            case 8:  Z8_CASE(2, 8);  return;
+            case 16: Z8_CASE(2, 16); return;
+            case 24: Z8_CASE(2, 24); return;
+            case 32: Z8_CASE(2, 32); return;
+            case 40: Z8_CASE(2, 40); return;
+        }
+    }
+    #undef Z8_CASE
 }
 
 extern "C" void launch_fcma_gram_f32(const float* Z, float* G, ll C, ll E,

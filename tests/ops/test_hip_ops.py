@@ -306,3 +306,100 @@ def test_stencil3d_matches_conv3d(ops, shape, r):
     assert got.shape == ref.shape
     assert torch.allclose(got, ref, atol=1e-4, rtol=1e-4), \
         (got - ref).abs().max().item()
+
+
+def test_gram_fp8_identity_asymmetric(ops):
+    """Transpose-detecting fragment-map check for the fp8 MFMA Gram:
+    asymmetric Z, values exactly representable in e4m3."""
+    g = torch.Generator().manual_seed(13)
+    # e4m3-exact values: multiples of 0.25 in [-2, 2]
+    Z = (torch.randint(-8, 9, (2, 64, 128), generator=g).float() / 4.0)
+    Z[0, 0, :] = (torch.arange(128) % 8).float() / 4.0  # asymmetric row
+    Z8 = Z.to(torch.float8_e4m3fn).cuda().contiguous()
+    G = ops.fcma_gram_fp8(Z8)
+    ref = torch.bmm(Z, Z.transpose(1, 2))
+    # inputs exact in e4m3 and fp32 accumulate → exact match expected
+    assert torch.allclose(G.cpu(), ref, atol=1e-3, rtol=1e-5)
+
+
+def test_gram_fp8_matches_fp32_reference(ops):
+    """Random z-score-scale data: fp8 Gram within quantization error of
+    the fp32 oracle (relative to the diagonal scale)."""
+    g = torch.Generator().manual_seed(14)
+    C, E, V = 3, 64, 1024
+    Z = torch.randn((C, E, V), generator=g).float().clamp_(-1.74, 1.74)
+    Z8 = Z.to(torch.float8_e4m3fn)
+    G = ops.fcma_gram_fp8(Z8.cuda().contiguous())
+    Zq = Z8.float()                      # same quantized values
+    ref = torch.bmm(Zq, Zq.transpose(1, 2))
+    assert torch.allclose(G.cpu(), ref, atol=0.5, rtol=1e-3)
+    # and against the unquantized oracle: ~3% of the row-norm scale
+    full = torch.bmm(Z, Z.transpose(1, 2))
+    scale = full.diagonal(dim1=1, dim2=2).mean()
+    assert (G.cpu() - full).abs().max() < 0.12 * scale
+
+
+def test_gram_fp8_vsplit_path(ops):
+    """V large enough to trigger the nsplit partial-sum path."""
+    g = torch.Generator().manual_seed(15)
+    Z = torch.randn((1, 64, 16384), generator=g).float().clamp_(-2, 2)
+    Z8 = Z.to(torch.float8_e4m3fn)
+    G = ops.fcma_gram_fp8(Z8.cuda().contiguous())
+    Zq = Z8.float()
+    ref = torch.bmm(Zq, Zq.transpose(1, 2))
+    assert torch.allclose(G.cpu(), ref, atol=2.0, rtol=1e-3)
+
+
+def test_corr_norm_z_fp8_output(ops):
+    """fp8 Z output of the fused corr+normalize matches the bf16-path
+    values to e4m3 quantization."""
+    g = torch.Generator().manual_seed(16)
+    E, L, V, P = 16, 12, 256, 4
+    A = _zscored_epochs(g, E, L, V, "cpu").to(torch.bfloat16)
+    Acu = A.cuda().contiguous()
+    ext = ops.load_extension()
+    Epad = 64
+    z_bf = torch.zeros((32, Epad, V), dtype=torch.bfloat16,
+                       device="cuda")
+    z_f8 = torch.zeros((32, Epad, V), dtype=torch.float8_e4m3fn,
+                       device="cuda")
+    ext.fcma_corr_norm_z(Acu, Acu, 5, 32, P, Epad, out=z_bf)
+    ext.fcma_corr_norm_z(Acu, Acu, 5, 32, P, Epad, out=z_f8)
+    a = z_bf.float().cpu()
+    b = z_f8.float().cpu()
+    # z-scores bounded by sqrt(P-1); e4m3 relative step is 2^-3
+    assert (a - b).abs().max() < 0.15
+    assert torch.allclose(a, b, atol=0.15, rtol=0.13)
+
+
+def test_fp8_pipeline_end_to_end(ops):
+    """Full chunk pipeline with z_fp8=True: Gram close to the fp32
+    chain, CV selection rankings close to the bf16 default."""
+    import numpy as np
+
+    from brainiak_amd.fcma.core import CorrelationPipeline
+    from brainiak_amd.fcma.svm import cross_validate_voxels
+    rng = np.random.RandomState(21)
+    E, L, V = 32, 12, 500    # V deliberately not a multiple of 16
+    labels = np.array([e % 2 for e in range(E)])
+    raw = []
+    for e in range(E):
+        m = rng.randn(L, V).astype(np.float32)
+        if e % 2:
+            m[:, :V // 2] += 0.8 * rng.randn(L, 1)
+        m = (m - m.mean(0)) / m.std(0)
+        raw.append((m / np.sqrt(L)).astype(np.float32))
+    accs = {}
+    for tag, fp8 in (("bf16", False), ("fp8", True)):
+        pipe = CorrelationPipeline(raw, None, 4, device="cuda",
+                                   z_fp8=fp8)
+        kernels = pipe.pipelined_kernel_matrices([(0, 256), (256, V - 256)])
+        assert kernels.shape[0] == V
+        accs[tag] = np.asarray(cross_validate_voxels(kernels, labels, 4))
+    from scipy.stats import spearmanr
+    r = spearmanr(accs["bf16"], accs["fp8"]).statistic
+    assert r > 0.85, r
+    k = V // 4
+    top_b = set(np.argsort(accs["bf16"])[-k:])
+    top_f = set(np.argsort(accs["fp8"])[-k:])
+    assert len(top_b & top_f) >= 0.75 * k
