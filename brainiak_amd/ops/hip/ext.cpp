@@ -21,6 +21,8 @@ void launch_fcma_gram_bf16(const void*, float*, ll, ll, ll, ll, void*);
 void launch_fcma_gram_fp8(const void*, float*, ll, ll, ll, ll, void*);
 void launch_fcma_gram_f32(const float*, float*, ll, ll, ll, void*);
 int fcma_corr_norm_z8_supported(ll, int, ll);
+void launch_fp8_cvt_probe(const float*, void*, ll, void*);
+void launch_fp8_cvt_probe_sw(const float*, void*, ll, void*);
 void launch_fcma_corr_norm_z8(const void*, const void*, void*, ll, ll,
                               ll, ll, ll, int, void*);
 void launch_jacobi_eigh(const float*, float*, float*, ll, int, void*);
@@ -171,6 +173,20 @@ torch::Tensor fcma_gram_bf16(torch::Tensor Z) {
     launch_fcma_gram_bf16(Z.data_ptr(), Gp.data_ptr<float>(), C, E, V,
                           nsplit, cur_stream());
     return Gp.sum(0);
+}
+
+torch::Tensor debug_fp8_cvt(torch::Tensor x, bool sw) {
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kFloat32);
+    auto xc = x.contiguous();
+    auto o = torch::empty_like(xc, xc.options()
+                                       .dtype(torch::kFloat8_e4m3fn));
+    if (sw)
+        launch_fp8_cvt_probe_sw(xc.data_ptr<float>(), o.data_ptr(),
+                                xc.numel(), cur_stream());
+    else
+        launch_fp8_cvt_probe(xc.data_ptr<float>(), o.data_ptr(),
+                             xc.numel(), cur_stream());
+    return o;
 }
 
 torch::Tensor fcma_gram_fp8(torch::Tensor Z) {
@@ -427,6 +443,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("padE"),
           pybind11::arg("out") = pybind11::none());
     m.def("fcma_gram", &fcma_gram, "per-voxel Gram from fp32 [C,E,V]");
+    m.def("debug_fp8_cvt", &debug_fp8_cvt,
+          "device to_fp8 conversion probe", py::arg("x"),
+          py::arg("sw") = false);
     m.def("fcma_gram_fp8", &fcma_gram_fp8,
           "per-voxel Gram from fp8(e4m3) Z [C,E,V]");
     m.def("fcma_gram_bf16", &fcma_gram_bf16,

@@ -1170,6 +1170,35 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
     }
 }
 
+// debug probe: elementwise to_fp8 over a float array (conversion
+// semantics validation vs torch's e4m3fn — see tests/ops)
+__global__ void k_fp8_cvt_probe(const float* __restrict__ x,
+                                fp8_t* __restrict__ o, ll n) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) o[i] = to_fp8(x[i]);
+}
+
+extern "C" void launch_fp8_cvt_probe(const float* x, void* o, ll n,
+                                     hipStream_t stream) {
+    hipLaunchKernelGGL(k_fp8_cvt_probe, dim3((n + 255) / 256), dim3(256),
+                       0, stream, x, (fp8_t*)o, n);
+}
+
+// the header's software (bit-exact) conversion path, for A/B against
+// the hardware cvt instruction the __hip_fp8_e4m3 ctor uses on gfx950
+__global__ void k_fp8_cvt_probe_sw(const float* __restrict__ x,
+                                   fp8_t* __restrict__ o, ll n) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n)
+        o[i] = __hip_cvt_float_to_fp8(x[i], __HIP_SATFINITE, __HIP_E4M3);
+}
+
+extern "C" void launch_fp8_cvt_probe_sw(const float* x, void* o, ll n,
+                                        hipStream_t stream) {
+    hipLaunchKernelGGL(k_fp8_cvt_probe_sw, dim3((n + 255) / 256),
+                       dim3(256), 0, stream, x, (fp8_t*)o, n);
+}
+
 // ===========================================================================
 // k_gram_fp8: G_c = Z_c Z_c^T per voxel with fp8(e4m3) MFMA
 // (v_mfma_f32_16x16x32_fp8_fp8, fp32 accumulate).  Z: [C, E, V] fp8,

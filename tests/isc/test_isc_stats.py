@@ -1,0 +1,272 @@
+"""Statistical-property battery for the ISC resampling tests.
+
+Ports the reference's oracle *assertions* (ref tests/isc/test_isc.py:
+251-601): signal/noise significance detection across every resampling
+method x pairwise/LOO x summary statistic, seed reproducibility,
+observed==recomputation identities, exact-test enumeration, NaN
+semantics, and the Phipson-Smyth p-value floor.
+"""
+
+import numpy as np
+import pytest
+
+from brainiak_amd.isc import (
+    bootstrap_isc,
+    compute_summary_statistic,
+    isc,
+    permutation_isc,
+    phaseshift_isc,
+    squareform_isfc,
+    timeshift_isc,
+)
+from brainiak_amd.utils.utils import p_from_null, phase_randomize
+
+N_TRS = 100
+N_SUBJ = 12
+
+
+def _signal_noise_data(rng, n_subjects=N_SUBJ, n_TRs=N_TRS):
+    """Voxels 0,1 carry a strong shared signal; voxel 2 is pure noise
+    (the reference's detection fixture, ref test_isc.py:296-303)."""
+    signal = rng.randn(n_TRs)
+    data = rng.randn(n_TRs, 3, n_subjects) * 0.3
+    data[:, 0, :] += signal[:, None]
+    data[:, 1, :] += signal[:, None]
+    return data
+
+
+# -- detection across every method -------------------------------------------
+
+@pytest.mark.parametrize("pairwise", [False, True])
+def test_bootstrap_detects_signal(pairwise):
+    rng = np.random.RandomState(4)
+    data = _signal_noise_data(rng)
+    iscs = isc(data, pairwise=pairwise)
+    assert np.all(iscs[:, :2] > 0.5)
+    observed, ci, p, dist = bootstrap_isc(
+        iscs, pairwise=pairwise, n_bootstraps=200, random_state=42)
+    assert dist.shape == (200, 3)
+    assert p[0] < 0.05 and p[1] < 0.05
+    assert p[2] > 0.01
+    # observed must equal the plain summary recomputation
+    assert np.allclose(observed, isc(data, pairwise=pairwise,
+                                     summary_statistic='median'))
+    # CI brackets the observed for the signal voxels
+    assert ci[0][0] <= observed[0] <= ci[1][0]
+
+
+@pytest.mark.parametrize("pairwise", [False, True])
+def test_permutation_detects_signal(pairwise):
+    rng = np.random.RandomState(1)
+    data = _signal_noise_data(rng)
+    iscs = isc(data, pairwise=pairwise)
+    observed, p, dist = permutation_isc(
+        iscs, pairwise=pairwise, n_permutations=200, random_state=42)
+    assert p[0] < 0.05 and p[1] < 0.05
+    assert p[2] > 0.01
+    assert np.allclose(np.asarray(observed).ravel(),
+                       isc(data, pairwise=pairwise,
+                           summary_statistic='median'))
+
+
+@pytest.mark.parametrize("pairwise", [False, True])
+def test_timeshift_detects_signal(pairwise):
+    rng = np.random.RandomState(2)
+    data = _signal_noise_data(rng)
+    observed, p, dist = timeshift_isc(
+        data, pairwise=pairwise, n_shifts=200, random_state=42)
+    assert p[0] < 0.05 and p[1] < 0.05
+    assert p[2] > 0.01
+    assert np.allclose(observed, isc(data, pairwise=pairwise,
+                                     summary_statistic='median'))
+    # circular-shift null stays near zero
+    assert abs(np.mean(dist[:, 2])) < 0.2
+
+
+@pytest.mark.parametrize("pairwise", [False, True])
+def test_phaseshift_detects_signal(pairwise):
+    rng = np.random.RandomState(3)
+    data = _signal_noise_data(rng)
+    observed, p, dist = phaseshift_isc(
+        data, pairwise=pairwise, n_shifts=200, random_state=42)
+    assert p[0] < 0.05 and p[1] < 0.05
+    assert p[2] > 0.01
+    assert np.allclose(observed, isc(data, pairwise=pairwise,
+                                     summary_statistic='median'))
+
+
+@pytest.mark.parametrize("summary_statistic", ["mean", "median"])
+def test_bootstrap_summary_statistics(summary_statistic):
+    rng = np.random.RandomState(4)
+    data = _signal_noise_data(rng)
+    iscs = isc(data)
+    observed, ci, p, dist = bootstrap_isc(
+        iscs, summary_statistic=summary_statistic, n_bootstraps=100,
+        random_state=0)
+    assert np.allclose(observed, compute_summary_statistic(
+        iscs, summary_statistic=summary_statistic, axis=0))
+    assert p[0] < 0.1 and p[2] > 0.01
+
+
+# -- reproducibility ----------------------------------------------------------
+
+def test_resampling_seed_reproducibility():
+    rng = np.random.RandomState(5)
+    data = _signal_noise_data(rng)
+    iscs = isc(data)
+    dists = []
+    for seed in (42, 42, 7):
+        _, _, _, dist = bootstrap_isc(iscs, n_bootstraps=50,
+                                      random_state=seed)
+        dists.append(dist)
+    assert np.array_equal(dists[0], dists[1])
+    assert not np.array_equal(dists[1], dists[2])
+
+    perms = []
+    for seed in (42, 42, 7):
+        _, _, dist = permutation_isc(iscs, n_permutations=50,
+                                     random_state=seed)
+        perms.append(dist)
+    assert np.array_equal(perms[0], perms[1])
+    assert not np.array_equal(perms[1], perms[2])
+
+
+# -- exact-test enumeration ---------------------------------------------------
+
+def test_permutation_exact_one_sample():
+    """n_permutations >= 2^n triggers exact sign-flip enumeration:
+    the null distribution is deterministic (seed-independent)."""
+    rng = np.random.RandomState(6)
+    data = _signal_noise_data(rng, n_subjects=5)
+    iscs = isc(data)
+    out1 = permutation_isc(iscs, n_permutations=2 ** 5 + 10,
+                           random_state=1)
+    out2 = permutation_isc(iscs, n_permutations=2 ** 5 + 10,
+                           random_state=99)
+    assert out1[2].shape[0] == 2 ** 5
+    assert np.array_equal(out1[2], out2[2])
+    assert np.array_equal(out1[1], out2[1])
+
+
+def test_permutation_exact_two_sample():
+    rng = np.random.RandomState(7)
+    data = _signal_noise_data(rng, n_subjects=4)
+    iscs = isc(data)
+    groups = [0, 0, 1, 1]
+    out1 = permutation_isc(iscs, group_assignment=groups,
+                           n_permutations=30, random_state=3)
+    out2 = permutation_isc(iscs, group_assignment=groups,
+                           n_permutations=30, random_state=8)
+    # 4! = 24 <= 30 → exact
+    assert out1[2].shape[0] == 24
+    assert np.array_equal(out1[2], out2[2])
+
+
+def test_permutation_two_sample_group_difference():
+    """Group A has shared signal, group B is noise → two-sample test
+    flags the difference in the signal voxels."""
+    rng = np.random.RandomState(8)
+    n_per = 8
+    signal = rng.randn(N_TRS)
+    a = rng.randn(N_TRS, 3, n_per) * 0.3
+    a[:, :2, :] += signal[:, None, None]
+    b = rng.randn(N_TRS, 3, n_per)
+    data = np.concatenate([a, b], axis=2)
+    iscs = isc(data)
+    groups = [0] * n_per + [1] * n_per
+    observed, p, dist = permutation_isc(
+        iscs, group_assignment=groups, n_permutations=200,
+        random_state=11, side='two-sided')
+    assert p[0] < 0.05 and p[1] < 0.05
+    assert p[2] > 0.01
+
+
+# -- NaN semantics ------------------------------------------------------------
+
+def test_isc_nan_semantics():
+    """One subject's NaN voxel: tolerate_nans=True keeps the other
+    subjects' values; =False NaNs every pair/LOO touching it (the
+    reference's counting assertions, ref test_isc.py:160-247)."""
+    rng = np.random.RandomState(9)
+    data = _signal_noise_data(rng, n_subjects=6)
+    data[:, 0, 2] = np.nan
+
+    loo_t = isc(data, tolerate_nans=True)
+    loo_f = isc(data, tolerate_nans=False)
+    # subject 2's own LOO value is NaN either way
+    assert np.isnan(loo_t[2, 0])
+    # tolerant: only subject 2 NaN; strict: every subject's LOO mean
+    # includes subject 2 → all NaN in that voxel
+    assert np.sum(np.isnan(loo_t[:, 0])) == 1
+    assert np.sum(np.isnan(loo_f[:, 0])) == 6
+
+    pw_t = isc(data, pairwise=True, tolerate_nans=True)
+    pw_f = isc(data, pairwise=True, tolerate_nans=False)
+    # pairwise: exactly the 5 pairs containing subject 2 are NaN, and
+    # tolerate_nans does not change pairwise values
+    assert np.sum(np.isnan(pw_t[:, 0])) == 5
+    assert np.allclose(pw_t, pw_f, equal_nan=True)
+
+
+# -- p-value floors -----------------------------------------------------------
+
+def test_p_from_null_phipson_smyth_floor():
+    """p-values carry the Phipson-Smyth +1 correction: never 0, never
+    below 1/(n+1) (ref utils.py:862-874)."""
+    rng = np.random.RandomState(10)
+    null = rng.randn(200)
+    p = p_from_null(np.array(10.0), null, side='right', exact=False)
+    assert 0 < p <= 1
+    assert p >= 1 / (200 + 1) - 1e-12
+    p_left = p_from_null(np.array(-10.0), null, side='left', exact=False)
+    assert p_left >= 1 / (200 + 1) - 1e-12
+    p_two = p_from_null(np.array(0.0), null, side='two-sided',
+                        exact=False)
+    assert p_two > 0.5
+
+
+def test_bootstrap_p_floor():
+    rng = np.random.RandomState(12)
+    data = _signal_noise_data(rng)
+    iscs = isc(data)
+    _, _, p, _ = bootstrap_isc(iscs, n_bootstraps=100, random_state=0)
+    assert np.all(p >= 1 / 101 - 1e-12)
+    assert np.all(p <= 1.0)
+
+
+# -- structural identities ----------------------------------------------------
+
+def test_squareform_isfc_roundtrip(seeded_rng):
+    from brainiak_amd.isc import isfc
+    data = seeded_rng.randn(40, 6, 4)
+    isfcs, iscs = isfc(data, vectorize_isfcs=True)
+    assert isfcs.shape == (4, 6 * 5 // 2)
+    assert iscs.shape == (4, 6)
+    mats = squareform_isfc(isfcs, iscs)
+    assert mats.shape == (4, 6, 6)
+    # diagonal carries the ISCs, matrix is symmetric
+    assert np.allclose(np.diagonal(mats, axis1=1, axis2=2), iscs)
+    assert np.allclose(mats, np.transpose(mats, (0, 2, 1)))
+    # back to vectors
+    v2, d2 = squareform_isfc(mats)
+    assert np.allclose(v2, isfcs) and np.allclose(d2, iscs)
+
+
+def test_compute_summary_statistic_identities(seeded_rng):
+    iscs = np.clip(seeded_rng.randn(8, 5) * 0.3, -0.99, 0.99)
+    m = compute_summary_statistic(iscs, 'mean', axis=0)
+    assert np.allclose(m, np.tanh(np.mean(np.arctanh(iscs), axis=0)))
+    md = compute_summary_statistic(iscs, 'median', axis=0)
+    assert np.allclose(md, np.median(iscs, axis=0))
+
+
+def test_phase_randomize_preserves_spectrum(seeded_rng):
+    """Phase randomization preserves each voxel's power spectrum
+    (the timeshift/phaseshift null's defining property)."""
+    data = seeded_rng.randn(64, 3, 2)
+    shuffled = phase_randomize(data, random_state=1)
+    assert shuffled.shape == data.shape
+    orig_power = np.abs(np.fft.rfft(data, axis=0)) ** 2
+    new_power = np.abs(np.fft.rfft(shuffled, axis=0)) ** 2
+    assert np.allclose(orig_power, new_power, rtol=1e-8)
+    assert not np.allclose(shuffled, data)

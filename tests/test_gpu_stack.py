@@ -174,9 +174,12 @@ def test_fcma_fp32_pipeline_on_gpu(cuda, seeded_rng):
     raise, and match the CPU fp32 oracle."""
     from brainiak_amd.fcma.core import CorrelationPipeline
     raw = _unit_epochs(seeded_rng, 8, 12, 64)
-    gpu = CorrelationPipeline(raw, None, 4, device="cuda",
+    raw2 = _unit_epochs(seeded_rng, 8, 12, 48)
+    # cross-correlation (raw2 != raw): the self-voxel r=1 diagonal of a
+    # self-correlation Fisher-clamps and amplifies last-ulp differences
+    gpu = CorrelationPipeline(raw, raw2, 4, device="cuda",
                               use_bf16=False)
-    cpu = CorrelationPipeline(raw, None, 4, device="cpu")
+    cpu = CorrelationPipeline(raw, raw2, 4, device="cpu")
     g_gpu = gpu.chunk_kernel_matrices(0, 16).cpu()
     g_cpu = cpu.chunk_kernel_matrices(0, 16)
     assert torch.allclose(g_gpu, g_cpu, atol=1e-3, rtol=1e-3)
@@ -187,10 +190,11 @@ def test_fcma_long_epochs_on_gpu(cuda, seeded_rng):
     instead of raising."""
     from brainiak_amd.fcma.core import CorrelationPipeline
     raw = _unit_epochs(seeded_rng, 8, 55, 48)
-    gpu = CorrelationPipeline(raw, None, 4, device="cuda")  # bf16 default
-    cpu = CorrelationPipeline(raw, None, 4, device="cpu")
+    raw2 = _unit_epochs(seeded_rng, 8, 55, 40)
+    gpu = CorrelationPipeline(raw, raw2, 4, device="cuda")  # bf16 default
+    cpu = CorrelationPipeline(raw, raw2, 4, device="cpu")
     g_gpu = gpu.pipelined_kernel_matrices([(0, 24), (24, 24)]).float().cpu()
-    cpu_b = CorrelationPipeline(raw, None, 4, device="cpu", use_bf16=True)
+    cpu_b = CorrelationPipeline(raw, raw2, 4, device="cpu", use_bf16=True)
     g_cpu = cpu_b.pipelined_kernel_matrices([(0, 24), (24, 24)]).float()
     # same bf16 input quantization on both sides; fp32 accumulate
     assert torch.allclose(g_gpu, g_cpu, atol=5e-2, rtol=5e-2)
