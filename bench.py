@@ -125,8 +125,13 @@ def main():
         kernels = pipeline.pipelined_kernel_matrices(chunks)
         if not args.no_cv:
             if device.type == "cuda":
-                accs = cross_validate_voxels(kernels, labels,
-                                             args.num_folds)
+                # fp8 Z: the SMO's KKT gap cannot fall below the e4m3
+                # quantization noise floor; tol 1e-2 matches it
+                # (measured: accuracy churn == bf16@1e-2, CV 0.45 ms
+                # vs 55 ms grinding at 1e-3 — profiles/fp8.md)
+                accs = cross_validate_voxels(
+                    kernels, labels, args.num_folds,
+                    tol=1e-2 if args.fp8 else 1e-3)
             else:  # CPU smoke: tiny sklearn sample to keep runtime sane
                 accs = cross_validate_voxels(kernels[:8], labels,
                                              args.num_folds)

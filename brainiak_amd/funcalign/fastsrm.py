@@ -125,9 +125,57 @@ class FastSRM:
 
     # -- fitting -----------------------------------------------------------
 
+    def _validate(self, subjects):
+        """Input-consistency checks (the reference's checker suite,
+        condensed to the load-bearing assertions: ref
+        fastsrm.py:69-576)."""
+        if len(subjects) == 0:
+            raise ValueError("imgs is empty")
+        shapes = [[np.asarray(safe_load(sess)).shape for sess in subj]
+                  for subj in subjects]
+        v0 = shapes[0][0][0]
+        for i, subj in enumerate(shapes):
+            if len(subj) != len(shapes[0]):
+                raise ValueError(
+                    "Subject %d has %d sessions but subject 0 has %d"
+                    % (i, len(subj), len(shapes[0])))
+            for j, sh in enumerate(subj):
+                if len(sh) != 2:
+                    raise ValueError(
+                        "imgs[%d][%d] must be 2-D, got shape %s"
+                        % (i, j, (sh,)))
+                if sh[0] != v0:
+                    raise ValueError(
+                        "Subject %d session %d has %d voxels; expected "
+                        "%d" % (i, j, sh[0], v0))
+                if sh[1] != shapes[0][j][1]:
+                    raise ValueError(
+                        "Session %d timeframe counts differ across "
+                        "subjects (%d vs %d)" % (j, sh[1],
+                                                 shapes[0][j][1]))
+        if self.atlas is not None:
+            atlas = safe_load(self.atlas)
+            n_regions = (atlas.shape[0] if atlas.ndim == 2
+                         else int(atlas.max()))
+            n_vox_atlas = (atlas.shape[1] if atlas.ndim == 2
+                           else atlas.shape[0])
+            if atlas.ndim > 2:
+                raise ValueError("Atlas has %d axes; expected 1 or 2"
+                                 % atlas.ndim)
+            if n_regions < self.n_components:
+                raise ValueError(
+                    "Number of regions in the atlas is lower than the "
+                    "number of components (%d < %d)"
+                    % (n_regions, self.n_components))
+            if n_vox_atlas != v0:
+                raise ValueError(
+                    "Atlas covers %d voxels but data has %d"
+                    % (n_vox_atlas, v0))
+
     def fit(self, imgs):
         """Fit the shared response and per-subject bases."""
         subjects, _ = self._canonicalize(imgs)
+        self._validate(subjects)
         n_subjects = len(subjects)
         n_sessions = len(subjects[0])
         for s in subjects:
@@ -229,6 +277,16 @@ class FastSRM:
         subjects, _ = self._canonicalize(imgs)
         single = isinstance(shared_response, np.ndarray)
         sessions_s = [shared_response] if single else list(shared_response)
+        if sessions_s and isinstance(sessions_s[0], (list, tuple)):
+            # aggregate=None responses: [subjects][sessions] → average
+            # across subjects per session (the reference's
+            # check_shared_response collapse, ref fastsrm.py:365-393)
+            n_sess = len(sessions_s[0])
+            sessions_s = [np.mean([np.asarray(subj[j])
+                                   for subj in sessions_s], axis=0)
+                          for j in range(n_sess)]
+        else:
+            sessions_s = [np.asarray(s) for s in sessions_s]
         # accept [k, T] or [T, k]
         sessions_s = [s if s.shape[0] != self.n_components else s.T
                       for s in sessions_s]

@@ -355,16 +355,21 @@ def test_corr_norm_z_fp8_output(ops):
     values to e4m3 quantization."""
     g = torch.Generator().manual_seed(16)
     E, L, V, P = 16, 12, 256, 4
+    # cross-correlation (A != B): a self-correlation's r=1 diagonal
+    # clamps in fisher_z and its degenerate z-score columns are
+    # rounding-order sensitive ACROSS template instantiations
     A = _zscored_epochs(g, E, L, V, "cpu").to(torch.bfloat16)
+    B = _zscored_epochs(g, E, L, V, "cpu").to(torch.bfloat16)
     Acu = A.cuda().contiguous()
+    Bcu = B.cuda().contiguous()
     ext = ops.load_extension()
     Epad = 64
     z_bf = torch.zeros((32, Epad, V), dtype=torch.bfloat16,
                        device="cuda")
     z_f8 = torch.zeros((32, Epad, V), dtype=torch.float8_e4m3fn,
                        device="cuda")
-    ext.fcma_corr_norm_z(Acu, Acu, 5, 32, P, Epad, out=z_bf)
-    ext.fcma_corr_norm_z(Acu, Acu, 5, 32, P, Epad, out=z_f8)
+    ext.fcma_corr_norm_z(Acu, Bcu, 5, 32, P, Epad, out=z_bf)
+    ext.fcma_corr_norm_z(Acu, Bcu, 5, 32, P, Epad, out=z_f8)
     a = z_bf.float().cpu()
     b = z_f8.float().cpu()
     # z-scores bounded by sqrt(P-1); e4m3 relative step is 2^-3
