@@ -129,7 +129,7 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
     // scalar path: hand it the [count, E, L] transposed slice
     torch::Tensor At;
     const void* At_ptr = nullptr;
-    if (fcma_corr_variant() == 3) {
+    if (fcma_corr_variant() >= 3) {
         At = A.narrow(2, start, count).permute({2, 0, 1}).contiguous();
         At_ptr = At.data_ptr();
     }

@@ -671,6 +671,107 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3s(
 }
 
 // ===========================================================================
+// k_corr_norm_dot3p: dot3s + software-pipelined SCALAR A prefetch.
+// The dot3s c-loop re-loads its wave-uniform A row (P*L bf16 through
+// the constant cache) at the top of every iteration and the compiler
+// parks the wave on that s_load before any MAC issues (PMC: 45 %
+// WAIT_INST, 26 % issue).  Here iteration c+1's A row streams into an
+// alternate register set WHILE c's dot chain and normalize run, so the
+// scalar-load latency hides under compute.
+// ===========================================================================
+template <int TP, int TL, int C3_CT, typename OT = bf16_t>
+__global__ __launch_bounds__(256) void k_corr_norm_dot3p(
+    const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
+    OT* __restrict__ zOut, ll E, ll VB,
+    ll C, int mode, ll zstride) {
+    static_assert(TL % 2 == 0 && TP >= 2 && TP <= 4,
+                  "dot3p kernel: even L, P in {2,4}");
+    constexpr int P = TP;
+    constexpr int KP = TL / 2;
+    constexpr int L = TL;
+    const ll nSubj = E / P;
+    const ll cTiles = (C + C3_CT - 1) / C3_CT;
+    const ll vTiles = (VB + C3_VT - 1) / C3_VT;
+    ll b = blockIdx.x;
+    const ll vt = b % vTiles; b /= vTiles;
+    const ll s = b % nSubj;   b /= nSubj;
+    const ll ct = b;
+    if (ct >= cTiles) return;
+    const ll c0 = ct * C3_CT;
+    const int CT = (int)min((ll)C3_CT, C - c0);
+    const ll v = vt * (ll)C3_VT + threadIdx.x;
+    if (v >= VB) return;
+
+    bf16x2_t bp[P][KP];
+    #pragma unroll
+    for (int p = 0; p < P; ++p) {
+        const bf16_t* brow = B + ((ll)(s * P + p) * L) * VB + v;
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp) {
+            bf16x2_t t;
+            t[0] = *(const __bf16*)&brow[(ll)(2 * kp) * VB];
+            t[1] = *(const __bf16*)&brow[(ll)(2 * kp + 1) * VB];
+            bp[p][kp] = t;
+        }
+    }
+
+    const unsigned int* abase = (const unsigned int*)
+        (At + (c0 * (ll)E + s * (ll)P) * L);
+    const int cstride = (E * L) / 2;          // dwords per voxel
+
+    unsigned int a_cur[P][KP], a_nxt[P][KP];
+    #pragma unroll
+    for (int p = 0; p < P; ++p)
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp)
+            a_cur[p][kp] = abase[p * (L / 2) + kp];
+
+    for (int c = 0; c < CT; ++c) {
+        // issue c+1's scalar loads before any use of c's values
+        if (c + 1 < CT) {
+            const unsigned int* an = abase + (ll)(c + 1) * cstride;
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                #pragma unroll
+                for (int kp = 0; kp < KP; ++kp)
+                    a_nxt[p][kp] = an[p * (L / 2) + kp];
+        }
+        float acc[P];
+        #pragma unroll
+        for (int p = 0; p < P; ++p) acc[p] = 0.f;
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp)
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                acc[p] = __builtin_amdgcn_fdot2_f32_bf16(
+                    __builtin_bit_cast(bf16x2_t, a_cur[p][kp]),
+                    bp[p][kp], acc[p], false);
+        float z[P];
+        float mean = 0.f, sq = 0.f;
+        #pragma unroll
+        for (int p = 0; p < P; ++p) {
+            z[p] = fisher_z(acc[p]);
+            mean += z[p]; sq += z[p] * z[p];
+        }
+        mean /= (float)P;
+        float var = sq / (float)P - mean * mean;
+        float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+        {
+            OT* dst = zOut
+                + ((c0 + c) * zstride + s * (ll)P) * VB + v;
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                dst[(size_t)p * VB] = z_cast<OT>((z[p] - mean) * inv);
+        }
+        #pragma unroll
+        for (int p = 0; p < P; ++p)
+            #pragma unroll
+            for (int kp = 0; kp < KP; ++kp)
+                a_cur[p][kp] = a_nxt[p][kp];
+    }
+}
+
+// ===========================================================================
 // k_corr_norm_mfma: the MFMA form of k_corr_norm for L <= 32.
 // PMC evidence (profiles/README.md): the VALU form is issue-bound at
 // ~5 instructions per useful FMA; one v_mfma_f32_16x16x32_bf16 computes a
@@ -1455,6 +1556,8 @@ static int corr_variant() {
         if (e && strcmp(e, "classic") == 0) v = 0;
         if (e && strcmp(e, "dot2") == 0) v = 1;
         if (e && strcmp(e, "dot3") == 0) v = 2;
+        if (e && strcmp(e, "dot3s") == 0) v = 3;
+        if (e && strcmp(e, "dot3p") == 0) v = 4;
     }
     return v;
 }
@@ -1491,14 +1594,22 @@ static void launch_corr_norm_t(const void* A, const void* B, void* zOut,
         return;
     }
 #endif
-    if (corr_variant() == 3 && At != nullptr && (TL % 2) == 0
+    if (corr_variant() >= 3 && At != nullptr && (TL % 2) == 0
         && mode != 1 && mode != 2 && (TP == 2 || TP == 4)) {
         if constexpr (TL % 2 == 0 && TP >= 2 && TP <= 4) {
             ll grid3 = ceil_div(C, 128) * nSubj * ceil_div(VB, C3_VT);
-            hipLaunchKernelGGL((k_corr_norm_dot3s<TP, TL, 128>),
-                               dim3(grid3), dim3(256), 0, stream,
-                               (const bf16_t*)At, (const bf16_t*)B,
-                               (bf16_t*)zOut, E, VB, C, mode, zstride);
+            if (corr_variant() == 4)
+                hipLaunchKernelGGL((k_corr_norm_dot3p<TP, TL, 128>),
+                                   dim3(grid3), dim3(256), 0, stream,
+                                   (const bf16_t*)At, (const bf16_t*)B,
+                                   (bf16_t*)zOut, E, VB, C, mode,
+                                   zstride);
+            else
+                hipLaunchKernelGGL((k_corr_norm_dot3s<TP, TL, 128>),
+                                   dim3(grid3), dim3(256), 0, stream,
+                                   (const bf16_t*)At, (const bf16_t*)B,
+                                   (bf16_t*)zOut, E, VB, C, mode,
+                                   zstride);
             return;
         }
     }
