@@ -552,6 +552,12 @@ class BRSA(_BRSACore):
         X_design, Y_data, X0, T, V, run_TRs = self._prepare(
             design, X, nuisance, scan_onsets)
         self._run_TRs_ = run_TRs
+        # the BASE regressors (user nuisance + per-run DC) stay fixed;
+        # auto-estimated components are re-derived each round from the
+        # residual against [X, X_base] ONLY (ref brsa.py:1967-1971 —
+        # regressing the previous components out first would erase the
+        # very signal being re-estimated)
+        X_base = X0
         C = X_design.shape[1]
         rank = self.rank if self.rank is not None else C
         rank = min(rank, C)
@@ -579,11 +585,8 @@ class BRSA(_BRSACore):
             prev_nll = nll
             if self.auto_nuisance and round_i < rounds - 1:
                 comps, n_nureg = self._residual_nuisance(
-                    X_design, Y_data, X0, n_nureg)
-                dc = X0[:, X0.shape[1] - (1 if self.baseline_single
-                                          or len(run_TRs) == 1
-                                          else len(run_TRs)):]
-                X0 = np.column_stack([comps, dc])
+                    X_design, Y_data, X_base, n_nureg)
+                X0 = np.column_stack([comps, X_base])
 
         # unpack
         nL = C * rank
