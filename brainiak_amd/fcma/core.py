@@ -42,7 +42,7 @@ __all__ = [
 
 # epoch lengths the HIP correlation kernels are templated for;
 # longer epochs fall back to the rocBLAS bmm + normalize path
-_HIP_EPOCH_LENGTHS = (8, 16, 24, 32, 40)
+_HIP_EPOCH_LENGTHS = (8, 12, 16, 20, 24, 28, 32, 36, 40)
 
 
 def stack_epochs(raw_data: List[np.ndarray], device,

@@ -74,6 +74,18 @@ __device__ __forceinline__ int frag_k(int lane, int j) {
     return 8 * (lane >> 4) + j;
 }
 
+// scale-free Fisher z: log2(num) - log2(den).  The downstream
+// within-subject z-score (z - mean) * rsqrt(var) is invariant to the
+// 0.5*ln2 factor, so the kernels that always z-score skip the multiply
+// (one VALU op per element on an issue-bound kernel).
+__device__ __forceinline__ float fisher_z_unscaled(float r) {
+    float num = 1.0f + r;
+    float den = 1.0f - r;
+    num = (num <= 0.0f) ? 1e-4f : num;
+    den = (den <= 0.0f) ? 1e-4f : den;
+    return __builtin_amdgcn_logf(num) - __builtin_amdgcn_logf(den);
+}
+
 __device__ __forceinline__ float fisher_z(float r) {
     float num = 1.0f + r;
     float den = 1.0f - r;
@@ -561,7 +573,7 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3(
         float mean = 0.f, sq = 0.f;
         #pragma unroll
         for (int p = 0; p < P; ++p) {
-            z[p] = fisher_z(acc[p]);
+            z[p] = fisher_z_unscaled(acc[p]);
             mean += z[p]; sq += z[p] * z[p];
         }
         mean /= (float)P;
@@ -648,7 +660,7 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3s(
         float mean = 0.f, sq = 0.f;
         #pragma unroll
         for (int p = 0; p < P; ++p) {
-            z[p] = fisher_z(acc[p]);
+            z[p] = fisher_z_unscaled(acc[p]);
             mean += z[p]; sq += z[p] * z[p];
         }
         mean /= (float)P;
@@ -750,7 +762,7 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3p(
         float mean = 0.f, sq = 0.f;
         #pragma unroll
         for (int p = 0; p < P; ++p) {
-            z[p] = fisher_z(acc[p]);
+            z[p] = fisher_z_unscaled(acc[p]);
             mean += z[p]; sq += z[p] * z[p];
         }
         mean /= (float)P;
@@ -1742,8 +1754,8 @@ static void dispatch_p(const void* A, const void* B, void* zOut,
 
 // host pads L to one of these (zero rows are inert for z-scored epochs)
 extern "C" ll fcma_supported_L(ll L) {
-    const ll opts[5] = {8, 16, 24, 32, 40};
-    for (int i = 0; i < 5; ++i)
+    const ll opts[9] = {8, 12, 16, 20, 24, 28, 32, 36, 40};
+    for (int i = 0; i < 9; ++i)
         if (L <= opts[i]) return opts[i];
     return -1;
 }
@@ -1759,6 +1771,14 @@ extern "C" void launch_fcma_corr_norm(const void* A, const void* B,
     size_t smem = (size_t)fcma_corr_norm_smem(L, P);
     switch (L) {
         case 8:  dispatch_p<8>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
+                               mode, zstride, smem, stream, At); break;
+        case 12: dispatch_p<12>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
+                               mode, zstride, smem, stream, At); break;
+        case 20: dispatch_p<20>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
+                               mode, zstride, smem, stream, At); break;
+        case 28: dispatch_p<28>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
+                               mode, zstride, smem, stream, At); break;
+        case 36: dispatch_p<36>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
                                mode, zstride, smem, stream, At); break;
         case 16: dispatch_p<16>(A, B, zOut, fOut, E, L, VA, VB, s0, C, P,
                                mode, zstride, smem, stream, At); break;
@@ -1810,17 +1830,25 @@ extern "C" void launch_fcma_corr_norm_z8(const void* At, const void* B,
     if (P == 4) {
         switch (L) {
             case 8:  Z8_CASE(4, 8);  return;
+            case 12: Z8_CASE(4, 12); return;
             case 16: Z8_CASE(4, 16); return;
+            case 20: Z8_CASE(4, 20); return;
             case 24: Z8_CASE(4, 24); return;
+            case 28: Z8_CASE(4, 28); return;
             case 32: Z8_CASE(4, 32); return;
+            case 36: Z8_CASE(4, 36); return;
             case 40: Z8_CASE(4, 40); return;
         }
     } else if (P == 2) {
         switch (L) {
             case 8:  Z8_CASE(2, 8);  return;
+            case 12: Z8_CASE(2, 12); return;
             case 16: Z8_CASE(2, 16); return;
+            case 20: Z8_CASE(2, 20); return;
             case 24: Z8_CASE(2, 24); return;
+            case 28: Z8_CASE(2, 28); return;
             case 32: Z8_CASE(2, 32); return;
+            case 36: Z8_CASE(2, 36); return;
             case 40: Z8_CASE(2, 40); return;
         }
     }
