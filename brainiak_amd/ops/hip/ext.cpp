@@ -161,6 +161,8 @@ torch::Tensor fcma_gram_bf16(torch::Tensor Z) {
     ll base = C * eb * eb;
     ll ktAll = (V + 63) / 64;
     ll nsplit = std::min(ktAll, std::max((ll)1, (2047 + base) / base));
+    if (const char* e = getenv("BRAINIAK_GRAM_NSPLIT"))
+        nsplit = std::min(ktAll, std::max((ll)1, (ll)atoll(e)));
     if (nsplit <= 1) {
         auto G = torch::empty({C, E, E},
                               Z.options().dtype(torch::kFloat32));
