@@ -178,6 +178,13 @@ class CorrelationPipeline:
             self.device.type == "cuda" and self.use_bf16
             and self.data.shape[1] in _HIP_EPOCH_LENGTHS
             and ops.require_hip())
+        # deferred-normalize split (bf16 only): corr writes raw r, the
+        # Gram kernel Fisher-z/z-scores each staged tile in its memory
+        # latency shadow.  BRAINIAK_NO_RAWCORR=1 restores the fused
+        # corr+norm kernel for A/B.
+        self._raw_split = (self._hip_path and not self.z_fp8
+                           and epochs_per_subj in (2, 4)
+                           and not os.environ.get("BRAINIAK_NO_RAWCORR"))
 
     def correlate_chunk(self, start: int, count: int) -> torch.Tensor:
         """corr [count, E, V2] fp32 for voxels [start, start+count)
@@ -258,8 +265,13 @@ class CorrelationPipeline:
                             dtype=zdtype, device=self.device)
                 for _ in range(2)]
         if getattr(self, "_streams", None) is None:
-            self._streams = (torch.cuda.Stream(device=self.device),
-                             torch.cuda.Stream(device=self.device))
+            # corr at high priority: it is the VALU-bound critical path
+            # and the latency-bound gram stretches it ~25 % when the
+            # hardware scheduler treats them equally (rocprof timeline,
+            # profiles/README.md r2)
+            self._streams = (
+                torch.cuda.Stream(device=self.device, priority=-1),
+                torch.cuda.Stream(device=self.device))
         corr_stream, gram_stream = self._streams
         grams = []
         pending = None          # (z, ready-event, buffer index)
@@ -276,7 +288,9 @@ class CorrelationPipeline:
                 elif self.z_fp8:
                     g = ops.fcma_gram_fp8(z)
                 else:
-                    g = ops.fcma_gram_bf16(z)
+                    g = ops.fcma_gram_bf16(
+                        z, norm_P=self.epochs_per_subj
+                        if self._raw_split else 0)
                 if Epad != E:
                     g = g[:, :E, :E].contiguous()
                 done = torch.cuda.Event()
@@ -292,7 +306,8 @@ class CorrelationPipeline:
                     corr_stream.wait_event(buf_free[bidx])
                 z = ext.fcma_corr_norm_z(
                     self.data, self.data2, start, count,
-                    self.epochs_per_subj, Epad, out=self._zbuf[bidx])
+                    self.epochs_per_subj, Epad, out=self._zbuf[bidx],
+                    raw=self._raw_split)
                 ev = torch.cuda.Event()
                 ev.record(corr_stream)
             if pending is not None:

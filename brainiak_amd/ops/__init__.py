@@ -132,9 +132,14 @@ def fcma_gram_fp8(Z: torch.Tensor) -> torch.Tensor:
     return _ext().fcma_gram_fp8(Z)
 
 
-def fcma_gram_bf16(Z: torch.Tensor) -> torch.Tensor:
-    """Per-voxel Gram [C, E, E] fp32 of Z [C, E, V] bf16 (E % 64 == 0)."""
-    return _ext().fcma_gram_bf16(Z)
+def fcma_gram_bf16(Z: torch.Tensor, norm_P: int = 0) -> torch.Tensor:
+    """Per-voxel Gram [C, E, E] fp32 of Z [C, E, V] bf16 (E % 64 == 0).
+
+    norm_P in {2, 4}: Z holds RAW correlations; Fisher-z + the
+    within-subject z-score over norm_P epochs are applied to each
+    staged tile in LDS before the MFMAs (the corr kernel then skips
+    its normalize epilogue — see CorrelationPipeline)."""
+    return _ext().fcma_gram_bf16(Z, int(norm_P))
 
 
 def fcma_fused_gram(data: torch.Tensor, data2: torch.Tensor, start: int,

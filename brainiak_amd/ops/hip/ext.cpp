@@ -22,6 +22,10 @@ void launch_fcma_gram_fp8(const void*, float*, ll, ll, ll, ll, void*);
 void launch_fcma_gram_f32(const float*, float*, ll, ll, ll, void*);
 int fcma_corr_norm_z8_supported(ll, int, ll);
 void launch_fp8_cvt_probe(const float*, void*, ll, void*);
+void launch_fcma_corr_raw(const void*, const void*, void*, ll, ll, ll,
+                          ll, ll, int, void*);
+void launch_fcma_gram_bf16_norm(const void*, float*, ll, ll, ll, ll,
+                                int, void*);
 void launch_fp8_cvt_probe_sw(const float*, void*, ll, void*);
 void launch_fcma_corr_norm_z8(const void*, const void*, void*, ll, ll,
                               ll, ll, ll, int, void*);
@@ -93,7 +97,8 @@ torch::Tensor fcma_correlate(torch::Tensor A, torch::Tensor B,
 torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
                                int64_t start, int64_t count, int64_t P,
                                int64_t padE,
-                               c10::optional<torch::Tensor> out_opt) {
+                               c10::optional<torch::Tensor> out_opt,
+                               bool raw) {
     check_3d(A, torch::kBFloat16, "A");
     check_3d(B, torch::kBFloat16, "B");
     ll E = A.size(0), L = A.size(1), VA = A.size(2), VB = B.size(2);
@@ -134,6 +139,16 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
         At_ptr = At.data_ptr();
     }
     // the kernel writes rows [0, E) with row stride Eout directly
+    if (raw) {
+        TORCH_CHECK(!fp8, "raw correlations are bf16-only (e4m3 cannot "
+                          "resolve r near +-1 before Fisher-z)");
+        TORCH_CHECK(At_ptr != nullptr && (P == 2 || P == 4),
+                    "raw mode needs the dot3s variant and P in {2,4}");
+        launch_fcma_corr_raw(At_ptr, B.data_ptr(), Z.data_ptr(),
+                             E, L, VB, count, Eout, (int)P,
+                             cur_stream());
+        return Z;
+    }
     if (fp8) {
         TORCH_CHECK(At_ptr != nullptr,
                     "fp8 Z output needs the dot3s variant "
@@ -151,7 +166,7 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
     return Z;
 }
 
-torch::Tensor fcma_gram_bf16(torch::Tensor Z) {
+torch::Tensor fcma_gram_bf16(torch::Tensor Z, int64_t norm_P) {
     check_3d(Z, torch::kBFloat16, "Z");
     ll C = Z.size(0), E = Z.size(1), V = Z.size(2);
     TORCH_CHECK(E % 64 == 0, "E must be a multiple of 64 (host pads)");
@@ -160,20 +175,33 @@ torch::Tensor fcma_gram_bf16(torch::Tensor Z) {
     // its serial k-tile loop at 4 blocks/CU — target ~8 blocks/CU
     ll base = C * eb * eb;
     ll ktAll = (V + 63) / 64;
-    ll nsplit = std::min(ktAll, std::max((ll)1, (2047 + base) / base));
+    // measured sweep (profiles/README.md r2): ~8k blocks in flight
+    // beats the old ~2k target by 0.3 ms/step; 16k regresses
+    ll nsplit = std::min(ktAll, std::max((ll)1, (8191 + base) / base));
     if (const char* e = getenv("BRAINIAK_GRAM_NSPLIT"))
         nsplit = std::min(ktAll, std::max((ll)1, (ll)atoll(e)));
+    TORCH_CHECK(norm_P == 0 || norm_P == 2 || norm_P == 4,
+                "norm_P must be 0 (pre-normalized) or 2/4");
+    if (norm_P)
+        TORCH_CHECK(E % 64 == 0 && 64 % norm_P == 0,
+                    "fused normalize needs band-aligned subjects");
+    auto launch = [&](float* g, ll ns) {
+        if (norm_P)
+            launch_fcma_gram_bf16_norm(Z.data_ptr(), g, C, E, V, ns,
+                                       (int)norm_P, cur_stream());
+        else
+            launch_fcma_gram_bf16(Z.data_ptr(), g, C, E, V, ns,
+                                  cur_stream());
+    };
     if (nsplit <= 1) {
         auto G = torch::empty({C, E, E},
                               Z.options().dtype(torch::kFloat32));
-        launch_fcma_gram_bf16(Z.data_ptr(), G.data_ptr<float>(), C, E, V,
-                              1, cur_stream());
+        launch(G.data_ptr<float>(), 1);
         return G;
     }
     auto Gp = torch::empty({nsplit, C, E, E},
                            Z.options().dtype(torch::kFloat32));
-    launch_fcma_gram_bf16(Z.data_ptr(), Gp.data_ptr<float>(), C, E, V,
-                          nsplit, cur_stream());
+    launch(Gp.data_ptr<float>(), nsplit);
     return Gp.sum(0);
 }
 
@@ -259,8 +287,9 @@ torch::Tensor fcma_fused_gram(torch::Tensor A, torch::Tensor B,
         return nsplit == 1 ? G.squeeze(0) : G.sum(0);
     }
     ll Epad = ((E + 63) / 64) * 64;
-    auto Z = fcma_corr_norm_z(A, B, start, count, P, Epad, c10::nullopt);
-    auto G = fcma_gram_bf16(Z);
+    auto Z = fcma_corr_norm_z(A, B, start, count, P, Epad, c10::nullopt,
+                              /*raw=*/false);
+    auto G = fcma_gram_bf16(Z, /*norm_P=*/0);
     if (Epad != E)
         return G.narrow(1, 0, E).narrow(2, 0, E).contiguous();
     return G;
@@ -439,19 +468,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fcma_correlate", &fcma_correlate,
           "raw chunk correlation [count,E,VB] fp32");
     m.def("fcma_corr_norm_z", &fcma_corr_norm_z,
-          "fused corr+norm -> bf16 Z",
+          "fused corr(+norm) -> Z (bf16/fp8; raw=True defers norm)",
           pybind11::arg("A"), pybind11::arg("B"), pybind11::arg("start"),
           pybind11::arg("count"), pybind11::arg("P"),
           pybind11::arg("padE"),
-          pybind11::arg("out") = pybind11::none());
+          pybind11::arg("out") = pybind11::none(),
+          pybind11::arg("raw") = false);
     m.def("fcma_gram", &fcma_gram, "per-voxel Gram from fp32 [C,E,V]");
     m.def("debug_fp8_cvt", &debug_fp8_cvt,
-          "device to_fp8 conversion probe", py::arg("x"),
-          py::arg("sw") = false);
+          "device to_fp8 conversion probe", pybind11::arg("x"),
+          pybind11::arg("sw") = false);
     m.def("fcma_gram_fp8", &fcma_gram_fp8,
           "per-voxel Gram from fp8(e4m3) Z [C,E,V]");
     m.def("fcma_gram_bf16", &fcma_gram_bf16,
-          "per-voxel Gram from bf16 Z [C,E,V]");
+          "per-voxel Gram from bf16 Z [C,E,V]; norm_P>0 applies "
+          "Fisher-z + z-score to raw correlations in-tile",
+          pybind11::arg("Z"), pybind11::arg("norm_P") = 0);
     m.def("fcma_fused_gram_native", &fcma_fused_gram_native,
           "true when the single-kernel corr+gram path covers (E, P, L)");
     m.def("fcma_fused_gram", &fcma_fused_gram,

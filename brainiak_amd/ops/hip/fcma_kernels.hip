@@ -602,7 +602,8 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3(
 // transposed to [count, E, L] (contiguous per voxel), and the kernel
 // needs NO LDS, NO staging loop and NO barrier at all.
 // ===========================================================================
-template <int TP, int TL, int C3_CT, typename OT = bf16_t>
+template <int TP, int TL, int C3_CT, typename OT = bf16_t,
+          bool RAW = false>
 __global__ __launch_bounds__(256) void k_corr_norm_dot3s(
     const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
     OT* __restrict__ zOut, ll E, ll VB,
@@ -656,6 +657,17 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3s(
                     __builtin_bit_cast(bf16x2_t,
                                        ac[p * (L / 2) + kp]),
                     bp[p][kp], acc[p], false);
+        if (RAW) {
+            // raw-correlation store: Fisher-z + z-score are deferred
+            // to the Gram kernel's staging phase (whose waves sit in
+            // memory-latency shadow) — this kernel is VALU-issue bound
+            OT* dst = zOut
+                + ((c0 + c) * zstride + s * (ll)P) * VB + v;
+            #pragma unroll
+            for (int p = 0; p < P; ++p)
+                dst[(size_t)p * VB] = z_cast<OT>(acc[p]);
+            continue;
+        }
         float z[P];
         float mean = 0.f, sq = 0.f;
         #pragma unroll
@@ -1162,6 +1174,13 @@ extern "C" void launch_fcma_fused_corr_gram(
 #define GR_PAD 8
 #define GR_ROW (GR_KT + GR_PAD)
 
+// NP = 0: Z is pre-normalized; NP in {2,4}: Z holds RAW correlations
+// and each staged tile gets Fisher-z + within-subject z-score applied
+// in LDS before the MFMAs read it.  The normalize VALU work runs in
+// this kernel's memory-latency shadow (PMC: 84 % WAIT_ANY) instead of
+// the issue-bound corr kernel.  Subject groups (NP consecutive rows)
+// never straddle the 64-row bands (E % NP == 0, 64 % NP == 0).
+template <int NP = 0>
 __global__ __launch_bounds__(256) void k_gram_bf16(
     const bf16_t* __restrict__ Z, float* __restrict__ G,
     ll C, ll E, ll V, ll nsplit) {
@@ -1236,6 +1255,31 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
     issue_loads(Zc, rows_i, kt0 * GR_KT, ri);
     if (!diag) issue_loads(Zc, rows_j, kt0 * GR_KT, rj);
 
+    auto normalize_tile = [&](bf16_t t[64][GR_ROW]) {
+        // per (subject-group, column): Fisher-z then z-score over the
+        // NP epochs, in place.  256 threads cover the
+        // (64/NP) x GR_KT groups; adjacent threads take adjacent
+        // columns (conflict-free LDS rows)
+        constexpr int NG = (64 / (NP > 0 ? NP : 1)) * GR_KT;
+        for (int g = threadIdx.x; g < NG; g += 256) {
+            const int r0 = (g / GR_KT) * NP;
+            const int col = g % GR_KT;
+            float z[NP > 0 ? NP : 1];
+            float mean = 0.f, sq = 0.f;
+            #pragma unroll
+            for (int p = 0; p < NP; ++p) {
+                z[p] = fisher_z_unscaled((float)t[r0 + p][col]);
+                mean += z[p]; sq += z[p] * z[p];
+            }
+            mean /= (float)NP;
+            float var = sq / (float)NP - mean * mean;
+            float inv = (var <= 0.f) ? 0.f : rsqrtf(var);
+            #pragma unroll
+            for (int p = 0; p < NP; ++p)
+                t[r0 + p][col] = (bf16_t)((z[p] - mean) * inv);
+        }
+    };
+
     for (ll kt = kt0; kt < kt1; ++kt) {
         __syncthreads();              // previous tile's reads complete
         write_tile(zi, ri);
@@ -1245,6 +1289,11 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
             if (!diag) issue_loads(Zc, rows_j, (kt + 1) * GR_KT, rj);
         }
         __syncthreads();              // tile visible
+        if (NP > 0) {
+            normalize_tile(zi);
+            if (!diag) normalize_tile(zj);
+            __syncthreads();          // normalized values visible
+        }
         const int frow = lane & 15;
         #pragma unroll
         for (int ks = 0; ks < GR_KT / 32; ++ks) {
@@ -1797,8 +1846,62 @@ extern "C" void launch_fcma_gram_bf16(const void* Z, float* G, ll C, ll E,
                                       hipStream_t stream) {
     ll eb = E / 64;
     ll grid = C * eb * eb * nsplit;
-    hipLaunchKernelGGL(k_gram_bf16, dim3(grid), dim3(256), 0, stream,
+    hipLaunchKernelGGL(k_gram_bf16<0>, dim3(grid), dim3(256), 0, stream,
                        (const bf16_t*)Z, G, C, E, V, nsplit);
+}
+
+extern "C" void launch_fcma_gram_bf16_norm(const void* Z, float* G,
+                                           ll C, ll E, ll V, ll nsplit,
+                                           int P, hipStream_t stream) {
+    ll eb = E / 64;
+    ll grid = C * eb * eb * nsplit;
+    if (P == 4)
+        hipLaunchKernelGGL(k_gram_bf16<4>, dim3(grid), dim3(256), 0,
+                           stream, (const bf16_t*)Z, G, C, E, V, nsplit);
+    else
+        hipLaunchKernelGGL(k_gram_bf16<2>, dim3(grid), dim3(256), 0,
+                           stream, (const bf16_t*)Z, G, C, E, V, nsplit);
+}
+
+// raw-r variant: dot3s with the normalize deferred (bf16 Z only)
+extern "C" void launch_fcma_corr_raw(const void* At, const void* B,
+                                     void* zOut, ll E, ll L, ll VB,
+                                     ll C, ll zstride, int P,
+                                     hipStream_t stream) {
+    ll nSubj = E / P;
+    ll grid3 = ceil_div(C, 128) * nSubj * ceil_div(VB, C3_VT);
+    #define RAW_CASE(TP, TL)                                             \
+        hipLaunchKernelGGL((k_corr_norm_dot3s<TP, TL, 128, bf16_t,       \
+                                              true>),                   \
+                           dim3(grid3), dim3(256), 0, stream,            \
+                           (const bf16_t*)At, (const bf16_t*)B,          \
+                           (bf16_t*)zOut, E, VB, C, /*mode=*/0, zstride)
+    if (P == 4) {
+        switch (L) {
+            case 8:  RAW_CASE(4, 8);  return;
+            case 12: RAW_CASE(4, 12); return;
+            case 16: RAW_CASE(4, 16); return;
+            case 20: RAW_CASE(4, 20); return;
+            case 24: RAW_CASE(4, 24); return;
+            case 28: RAW_CASE(4, 28); return;
+            case 32: RAW_CASE(4, 32); return;
+            case 36: RAW_CASE(4, 36); return;
+            case 40: RAW_CASE(4, 40); return;
+        }
+    } else if (P == 2) {
+        switch (L) {
+            case 8:  RAW_CASE(2, 8);  return;
+            case 12: RAW_CASE(2, 12); return;
+            case 16: RAW_CASE(2, 16); return;
+            case 20: RAW_CASE(2, 20); return;
+            case 24: RAW_CASE(2, 24); return;
+            case 28: RAW_CASE(2, 28); return;
+            case 32: RAW_CASE(2, 32); return;
+            case 36: RAW_CASE(2, 36); return;
+            case 40: RAW_CASE(2, 40); return;
+        }
+    }
+    #undef RAW_CASE
 }
 
 extern "C" void launch_fcma_gram_fp8(const void* Z, float* G, ll C, ll E,

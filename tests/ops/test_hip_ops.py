@@ -408,3 +408,66 @@ def test_fp8_pipeline_end_to_end(ops):
     top_b = set(np.argsort(accs["bf16"])[-k:])
     top_f = set(np.argsort(accs["fp8"])[-k:])
     assert len(top_b & top_f) >= 0.75 * k
+
+
+def test_gram_deferred_normalize_matches_fused(ops):
+    """Raw-r corr + in-tile normalize gram == fused corr+norm + plain
+    gram (same bf16 Z quantization points differ: raw path quantizes r,
+    fused path quantizes z — compare against the r-quantized oracle)."""
+    import numpy as np
+    g = torch.Generator().manual_seed(23)
+    E, L, V, P = 16, 12, 512, 4
+    A = _zscored_epochs(g, E, L, V, "cpu").to(torch.bfloat16)
+    B = _zscored_epochs(g, E, L, V, "cpu").to(torch.bfloat16)
+    Acu, Bcu = A.cuda().contiguous(), B.cuda().contiguous()
+    ext = ops.load_extension()
+    Epad = 64
+    z_raw = torch.zeros((32, Epad, V), dtype=torch.bfloat16,
+                        device="cuda")
+    ext.fcma_corr_norm_z(Acu, Bcu, 5, 32, P, Epad, out=z_raw, raw=True)
+    G = ops.fcma_gram_bf16(z_raw, norm_P=P)[:, :E, :E]
+
+    # oracle: same r-quantization (bf16), fp32 normalize + gram
+    corr = torch.einsum('elc,elv->cev', A.float()[:, :, 5:37], B.float())
+    rq = corr.to(torch.bfloat16).float()          # raw-r bf16 rounding
+    nc = _ref_normalize(rq, P)
+    ref = torch.bmm(nc, nc.transpose(1, 2))
+    assert torch.allclose(G.cpu(), ref, atol=1.5, rtol=5e-2)
+
+
+def test_pipeline_raw_split_matches_no_raw(ops):
+    """End-to-end: default (raw-split) pipeline grams vs the
+    BRAINIAK_NO_RAWCORR fused path — close at bf16 tolerance, and CV
+    rankings effectively identical."""
+    import os
+
+    import numpy as np
+
+    from brainiak_amd.fcma.core import CorrelationPipeline
+    from brainiak_amd.fcma.svm import cross_validate_voxels
+    rng = np.random.RandomState(31)
+    E, L, V = 32, 12, 512
+    labels = np.array([e % 2 for e in range(E)])
+    raw = []
+    for e in range(E):
+        m = rng.randn(L, V).astype(np.float32)
+        if e % 2:
+            m[:, :V // 2] += 0.8 * rng.randn(L, 1)
+        m = (m - m.mean(0)) / m.std(0)
+        raw.append((m / np.sqrt(L)).astype(np.float32))
+    pipe = CorrelationPipeline(raw, None, 4, device="cuda")
+    assert pipe._raw_split
+    g_split = pipe.pipelined_kernel_matrices([(0, 256), (256, 256)])
+    os.environ["BRAINIAK_NO_RAWCORR"] = "1"
+    try:
+        pipe2 = CorrelationPipeline(raw, None, 4, device="cuda")
+        assert not pipe2._raw_split
+        g_fused = pipe2.pipelined_kernel_matrices([(0, 256), (256, 256)])
+    finally:
+        del os.environ["BRAINIAK_NO_RAWCORR"]
+    a = np.asarray(cross_validate_voxels(g_split, labels, 4))
+    b = np.asarray(cross_validate_voxels(g_fused, labels, 4))
+    from scipy.stats import spearmanr
+    assert spearmanr(a, b).statistic > 0.95
+    assert torch.allclose(g_split.float(), g_fused.float(),
+                          atol=2.0, rtol=5e-2)
