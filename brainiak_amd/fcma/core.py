@@ -265,12 +265,14 @@ class CorrelationPipeline:
                             dtype=zdtype, device=self.device)
                 for _ in range(2)]
         if getattr(self, "_streams", None) is None:
-            # corr at high priority: it is the VALU-bound critical path
-            # and the latency-bound gram stretches it ~25 % when the
-            # hardware scheduler treats them equally (rocprof timeline,
-            # profiles/README.md r2)
+            # equal priorities: a high-priority corr stream SERIALIZES
+            # the two kernels (the scheduler starves the gram stream
+            # while any corr block is pending — measured exactly
+            # corr+gram serial); equal priority lets them co-schedule
+            # (BRAINIAK_CORR_PRIO=-1 re-enables the experiment)
+            prio = int(os.environ.get("BRAINIAK_CORR_PRIO", "0"))
             self._streams = (
-                torch.cuda.Stream(device=self.device, priority=-1),
+                torch.cuda.Stream(device=self.device, priority=prio),
                 torch.cuda.Stream(device=self.device))
         corr_stream, gram_stream = self._streams
         grams = []
