@@ -185,3 +185,142 @@ def test_kron_mult_matches_dense(seeded_rng):
     X = torch.as_tensor(seeded_rng.randn(12, 5))
     ref = torch.kron(A, B) @ X
     assert torch.allclose(kron_mult([A, B], X), ref, atol=1e-10)
+
+
+# -- round-2 depth: per-class density parity vs scipy (the reference's
+#    oracle style, ref tests/matnormal/test_cov.py etc.) --------------------
+
+def _dense_logp(X, R, C):
+    """scipy matrix-normal log density with dense row/col covariances."""
+    from scipy.stats import multivariate_normal
+    n, p = X.shape
+    full = np.kron(C, R)     # vec(X) row-major? use column-major vec
+    v = X.T.ravel()          # vec by columns → kron(C, R)
+    return multivariate_normal.logpdf(v, mean=np.zeros(n * p), cov=full)
+
+
+@pytest.mark.parametrize("covname", [
+    "identity", "isotropic", "diagonal", "diag_gamma", "chol",
+    "invchol", "ar1"])
+def test_every_cov_class_density_parity(covname, seeded_rng):
+    """matnorm_logp with EVERY Cov class == scipy dense density at the
+    class's current parameters."""
+    import torch
+
+    from brainiak_amd.matnormal.covs import (
+        CovAR1,
+        CovDiagonal,
+        CovDiagonalGammaPrior,
+        CovIdentity,
+        CovIsotropic,
+        CovUnconstrainedCholesky,
+        CovUnconstrainedInvCholesky,
+    )
+    from brainiak_amd.matnormal.matnormal_likelihoods import matnorm_logp
+    n, p = 6, 5
+    X = seeded_rng.randn(n, p)
+    row = {
+        "identity": CovIdentity(size=n),
+        "isotropic": CovIsotropic(size=n),
+        "diagonal": CovDiagonal(size=n),
+        "diag_gamma": CovDiagonalGammaPrior(size=n),
+        "chol": CovUnconstrainedCholesky(size=n),
+        "invchol": CovUnconstrainedInvCholesky(size=n),
+        "ar1": CovAR1(size=n),
+    }[covname]
+    col = CovIdentity(size=p)
+    Xt = torch.as_tensor(X)
+    with torch.no_grad():
+        lp = float(matnorm_logp(Xt, row, col))
+        R = np.linalg.inv(
+            row.solve(torch.eye(n, dtype=torch.float64))
+            .detach().numpy())           # dense row covariance
+    ref = _dense_logp(X, R, np.eye(p))
+    assert np.isclose(lp, ref, rtol=1e-6), (lp, ref)
+
+
+def test_marginal_and_conditional_consistency(seeded_rng):
+    """p(X) = p(X | Y) marginalized: check marginal-row logp equals the
+    dense evaluation of the marginal covariance R + A Q A^T."""
+    import torch
+
+    from brainiak_amd.matnormal.covs import (
+        CovIdentity,
+        CovUnconstrainedCholesky,
+    )
+    from brainiak_amd.matnormal.matnormal_likelihoods import (
+        matnorm_logp_marginal_row,
+    )
+    n, p, k = 5, 4, 3
+    X = seeded_rng.randn(n, p)
+    A = seeded_rng.randn(n, k)
+    row = CovIdentity(size=n)
+    col = CovIdentity(size=p)
+    Q = CovUnconstrainedCholesky(size=k)
+    Xt = torch.as_tensor(X)
+    with torch.no_grad():
+        lp = float(matnorm_logp_marginal_row(
+            Xt, row, col, torch.as_tensor(A), Q))
+        Qdense = np.linalg.inv(
+            Q.solve(torch.eye(k, dtype=torch.float64))
+            .detach().numpy())
+    Rm = np.eye(n) + A @ Qdense @ A.T
+    ref = _dense_logp(X, Rm, np.eye(p))
+    assert np.isclose(lp, ref, rtol=1e-6), (lp, ref)
+
+
+def test_cov_optimize_vars_trainable(seeded_rng):
+    """Every Cov class exposes optimizable parameters that autograd can
+    move (the TF get_optimize_vars contract re-expressed)."""
+    import torch
+
+    from brainiak_amd.matnormal.covs import (
+        CovAR1,
+        CovDiagonal,
+        CovIsotropic,
+        CovUnconstrainedCholesky,
+    )
+    from brainiak_amd.matnormal.matnormal_likelihoods import matnorm_logp
+    from brainiak_amd.matnormal.covs import CovIdentity
+    n, p = 6, 4
+    X = torch.as_tensor(seeded_rng.randn(n, p))
+    for cov in (CovIsotropic(size=n), CovDiagonal(size=n),
+                CovUnconstrainedCholesky(size=n), CovAR1(size=n)):
+        params = cov.get_optimize_vars()
+        assert len(params) >= 1
+        lp = matnorm_logp(X, cov, CovIdentity(size=p))
+        lp.backward()
+        grads = [p.grad for p in params if p.grad is not None]
+        assert grads, type(cov).__name__
+        assert all(torch.isfinite(g).all() for g in grads)
+
+
+def test_matnorm_regression_sklearn_api(seeded_rng):
+    from brainiak_amd.matnormal.regression import MatnormalRegression
+    from brainiak_amd.matnormal.covs import CovIdentity
+    n, k, p = 40, 3, 6
+    X = seeded_rng.randn(n, k)
+    B = seeded_rng.randn(k, p) * 2
+    Y = X @ B + 0.1 * seeded_rng.randn(n, p)
+    m = MatnormalRegression(time_cov=CovIdentity(size=n),
+                            space_cov=CovIdentity(size=p))
+    m.fit(X, Y)
+    pred = m.predict(X)
+    assert pred.shape == (n, p)
+    r = np.corrcoef(pred.ravel(), Y.ravel())[0, 1]
+    assert r > 0.98
+    # residual log-density under the fitted model beats a permuted
+    # design (model-comparison direction check)
+    import torch
+    with torch.no_grad():
+        m.beta = torch.as_tensor(m.beta_)
+        lp_good = float(m.logp(torch.as_tensor(X),
+                               torch.as_tensor(Y)))
+        lp_bad = float(m.logp(
+            torch.as_tensor(seeded_rng.permutation(X)),
+            torch.as_tensor(Y)))
+    assert lp_good > lp_bad
+    # MLE decode (calibrate) recovers the design up to noise
+    dec = m.calibrate(Y)
+    r_dec = np.corrcoef(dec.ravel(), np.asarray(X).ravel())[0, 1]
+    assert r_dec > 0.9
