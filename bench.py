@@ -42,8 +42,9 @@ def parse_args():
     ap.add_argument("--epoch-len", type=int, default=12)
     ap.add_argument("--voxels-per-step", type=int, default=4096)
     ap.add_argument("--num-folds", type=int, default=4)
-    ap.add_argument("--chunk", type=int, default=1024,
-                    help="pipeline chunk (voxels per kernel pass)")
+    ap.add_argument("--chunk", type=int, default=512,
+                    help="pipeline chunk (voxels per kernel pass; "
+                         "512 measured best on MI355X, profiles/)")
     ap.add_argument("--no-cv", action="store_true",
                     help="skip the SVM CV stage (pipeline only)")
     ap.add_argument("--fp8", action="store_true",
