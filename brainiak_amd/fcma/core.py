@@ -232,6 +232,8 @@ class CorrelationPipeline:
         if not self._hip_path or len(chunks) <= 1:
             return torch.cat([self.chunk_kernel_matrices(s, c, shrink)
                               for s, c in chunks], dim=0)
+        if self._raw_split and not os.environ.get("BRAINIAK_NO_DUO"):
+            return self._duo_pipeline(chunks, shrink)
         E = self.num_epochs
         Epad = ((E + 63) // 64) * 64
         ext = ops.load_extension()
@@ -322,6 +324,61 @@ class CorrelationPipeline:
         cur.wait_stream(corr_stream)
         for g in grams:   # allocated on gram_stream, consumed on cur
             g.record_stream(cur)
+        gram = torch.cat(grams, dim=0)
+        if shrink:
+            _shrink_(gram)
+        return gram
+
+    def _duo_pipeline(self, chunks, shrink=True):
+        """Single-stream duo launches: each kernel carries the raw-corr
+        blocks of chunk i AND the Gram(+normalize) blocks of chunk i-1
+        in one grid — HIP streams do not co-schedule the two kernels
+        (measured exactly serial), a shared grid forces CU-level
+        co-residency of the VALU-bound and latency-bound waves."""
+        ext = ops.load_extension()
+        E = self.num_epochs
+        Epad = ((E + 63) // 64) * 64
+        VB = self.data2.shape[2]
+        P = self.epochs_per_subj
+        max_count = max(c for _, c in chunks)
+        if getattr(self, "_zbuf", None) is None or \
+                self._zbuf[0].shape[0] < max_count or \
+                self._zbuf[0].shape[1] != Epad or \
+                self._zbuf[0].dtype != torch.bfloat16:
+            self._zbuf = [
+                torch.zeros((max_count, Epad, VB),
+                            dtype=torch.bfloat16, device=self.device)
+                for _ in range(2)]
+        nsplit = int(os.environ.get("BRAINIAK_GRAM_NSPLIT", "0")) or \
+            max(1, (8191 + max_count) // max_count)
+        nsplit = min(nsplit, (VB + 127) // 128)
+
+        grams = []
+        prev = None                     # (z, count)
+        for i, (start, count) in enumerate(chunks):
+            z_i = self._zbuf[i % 2]
+            if prev is None:
+                ext.fcma_corr_norm_z(self.data, self.data2, start,
+                                     count, P, Epad, out=z_i, raw=True)
+            else:
+                zp, cp = prev
+                gp = torch.empty((nsplit, cp, Epad, Epad),
+                                 dtype=torch.float32,
+                                 device=self.device)
+                ext.fcma_corr_gram_duo(self.data, self.data2, start,
+                                       count, P, z_i.narrow(0, 0, count),
+                                       Zprev=zp, Gpart=gp)
+                g = gp.sum(0)
+                if Epad != E:
+                    g = g[:, :E, :E].contiguous()
+                grams.append(g)
+            prev = (z_i.narrow(0, 0, count), count)
+        # trailing gram for the last chunk
+        zp, cp = prev
+        g = ops.fcma_gram_bf16(zp, norm_P=P)
+        if Epad != E:
+            g = g[:, :E, :E].contiguous()
+        grams.append(g)
         gram = torch.cat(grams, dim=0)
         if shrink:
             _shrink_(gram)

@@ -26,6 +26,9 @@ void launch_fcma_corr_raw(const void*, const void*, void*, ll, ll, ll,
                           ll, ll, int, void*);
 void launch_fcma_gram_bf16_norm(const void*, float*, ll, ll, ll, ll,
                                 int, void*);
+void launch_fcma_corr_gram_duo(const void*, const void*, void*, ll, ll,
+                               ll, ll, ll, int, const void*, float*,
+                               ll, ll, ll, ll, void*);
 void launch_fp8_cvt_probe_sw(const float*, void*, ll, void*);
 void launch_fcma_corr_norm_z8(const void*, const void*, void*, ll, ll,
                               ll, ll, ll, int, void*);
@@ -203,6 +206,52 @@ torch::Tensor fcma_gram_bf16(torch::Tensor Z, int64_t norm_P) {
                            Z.options().dtype(torch::kFloat32));
     launch(Gp.data_ptr<float>(), nsplit);
     return Gp.sum(0);
+}
+
+void fcma_corr_gram_duo(torch::Tensor A, torch::Tensor B,
+                        int64_t start, int64_t count, int64_t P,
+                        torch::Tensor Zout,
+                        c10::optional<torch::Tensor> Zprev,
+                        c10::optional<torch::Tensor> Gpart) {
+    // one grid = raw-corr blocks for [start, start+count) + Gram
+    // (+in-register normalize) blocks for the PREVIOUS chunk's Z
+    check_3d(A, torch::kBFloat16, "A");
+    check_3d(B, torch::kBFloat16, "B");
+    ll E = A.size(0), L = A.size(1), VA = A.size(2), VB = B.size(2);
+    TORCH_CHECK(start >= 0 && start + count <= VA, "voxel range");
+    TORCH_CHECK(P == 2 || P == 4, "duo path needs P in {2,4}");
+    TORCH_CHECK(fcma_supported_L(L) == L, "L must be a supported "
+                "template length");
+    TORCH_CHECK(Zout.is_cuda() && Zout.is_contiguous()
+                && Zout.scalar_type() == torch::kBFloat16
+                && Zout.size(0) >= count && Zout.size(2) == VB,
+                "bad Zout");
+    ll Eout = Zout.size(1);
+    auto At = A.narrow(2, start, count).permute({2, 0, 1}).contiguous();
+
+    const void* zprev_ptr = nullptr;
+    float* g_ptr = nullptr;
+    ll Cg = 0, Eg = 0, Vg = 0, nsplit = 1;
+    if (Gpart.has_value()) {
+        TORCH_CHECK(Zprev.has_value(), "Gpart needs Zprev");
+        auto& Zp = Zprev.value();
+        auto& G = Gpart.value();
+        TORCH_CHECK(Zp.is_cuda() && Zp.is_contiguous()
+                    && Zp.scalar_type() == torch::kBFloat16, "bad Zprev");
+        TORCH_CHECK(G.is_cuda() && G.is_contiguous() && G.dim() == 4
+                    && G.scalar_type() == torch::kFloat32, "bad Gpart");
+        Cg = G.size(1); Eg = Zp.size(1); Vg = Zp.size(2);
+        nsplit = G.size(0);
+        TORCH_CHECK(Eg % 64 == 0 && G.size(2) == Eg && G.size(3) == Eg,
+                    "Gpart/Zprev shape mismatch");
+        TORCH_CHECK(Zp.size(0) >= Cg, "Zprev rows < Cg");
+        zprev_ptr = Zp.data_ptr();
+        g_ptr = G.data_ptr<float>();
+    }
+    launch_fcma_corr_gram_duo(At.data_ptr(), B.data_ptr(),
+                              Zout.data_ptr(), E, L, VB, count, Eout,
+                              (int)P, zprev_ptr, g_ptr, Cg, Eg, Vg,
+                              nsplit, cur_stream());
 }
 
 torch::Tensor debug_fp8_cvt(torch::Tensor x, bool sw) {
@@ -480,6 +529,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("sw") = false);
     m.def("fcma_gram_fp8", &fcma_gram_fp8,
           "per-voxel Gram from fp8(e4m3) Z [C,E,V]");
+    m.def("fcma_corr_gram_duo", &fcma_corr_gram_duo,
+          "one-grid raw-corr(chunk i) + gram(chunk i-1) co-residency",
+          pybind11::arg("A"), pybind11::arg("B"), pybind11::arg("start"),
+          pybind11::arg("count"), pybind11::arg("P"),
+          pybind11::arg("Zout"),
+          pybind11::arg("Zprev") = pybind11::none(),
+          pybind11::arg("Gpart") = pybind11::none());
     m.def("fcma_gram_bf16", &fcma_gram_bf16,
           "per-voxel Gram from bf16 Z [C,E,V]; norm_P>0 applies "
           "Fisher-z + z-score to raw correlations in-tile",

@@ -510,8 +510,8 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3(
     (void)Lrt; (void)Prt;
     const ll nSubj = E / P;
     const ll cTiles = (C + C3_CT - 1) / C3_CT;
-    const ll vTiles = (VB + C3_VT - 1) / C3_VT;
     ll b = blockIdx.x;
+    const ll vTiles = (VB + C3_VT - 1) / C3_VT;
     const ll vt = b % vTiles; b /= vTiles;
     const ll s = b % nSubj;   b /= nSubj;
     const ll ct = b;
@@ -604,7 +604,8 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3(
 // ===========================================================================
 template <int TP, int TL, int C3_CT, typename OT = bf16_t,
           bool RAW = false>
-__global__ __launch_bounds__(256) void k_corr_norm_dot3s(
+__device__ __forceinline__ void dot3s_body(
+    ll b,
     const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
     OT* __restrict__ zOut, ll E, ll VB,
     ll C, int mode, ll zstride) {
@@ -616,7 +617,6 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3s(
     const ll nSubj = E / P;
     const ll cTiles = (C + C3_CT - 1) / C3_CT;
     const ll vTiles = (VB + C3_VT - 1) / C3_VT;
-    ll b = blockIdx.x;
     const ll vt = b % vTiles; b /= vTiles;
     const ll s = b % nSubj;   b /= nSubj;
     const ll ct = b;
@@ -694,6 +694,16 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3s(
     }
 }
 
+template <int TP, int TL, int C3_CT, typename OT = bf16_t,
+          bool RAW = false>
+__global__ __launch_bounds__(256) void k_corr_norm_dot3s(
+    const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
+    OT* __restrict__ zOut, ll E, ll VB,
+    ll C, int mode, ll zstride) {
+    dot3s_body<TP, TL, C3_CT, OT, RAW>(blockIdx.x, At, B, zOut, E, VB,
+                                       C, mode, zstride);
+}
+
 // ===========================================================================
 // k_corr_norm_dot3p: dot3s + software-pipelined SCALAR A prefetch.
 // The dot3s c-loop re-loads its wave-uniform A row (P*L bf16 through
@@ -715,8 +725,8 @@ __global__ __launch_bounds__(256) void k_corr_norm_dot3p(
     constexpr int L = TL;
     const ll nSubj = E / P;
     const ll cTiles = (C + C3_CT - 1) / C3_CT;
-    const ll vTiles = (VB + C3_VT - 1) / C3_VT;
     ll b = blockIdx.x;
+    const ll vTiles = (VB + C3_VT - 1) / C3_VT;
     const ll vt = b % vTiles; b /= vTiles;
     const ll s = b % nSubj;   b /= nSubj;
     const ll ct = b;
@@ -1181,7 +1191,8 @@ extern "C" void launch_fcma_fused_corr_gram(
 // the issue-bound corr kernel.  Subject groups (NP consecutive rows)
 // never straddle the 64-row bands (E % NP == 0, 64 % NP == 0).
 template <int NP = 0>
-__global__ __launch_bounds__(256) void k_gram_bf16(
+__device__ __forceinline__ void gram_bf16_body(
+    ll b,
     const bf16_t* __restrict__ Z, float* __restrict__ G,
     ll C, ll E, ll V, ll nsplit) {
     // nsplit > 1: V is cut into k-tile-aligned ranges, one partial G
@@ -1189,7 +1200,6 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
     // (MALL-resident Z slabs want 32-voxel chunks; 32 blocks would
     // leave 7/8 of the CUs idle)
     const ll eb = E / 64;
-    ll b = blockIdx.x;
     const ll ns = b % nsplit; b /= nsplit;
     const ll band_j = b % eb; b /= eb;
     const ll band_i = b % eb; b /= eb;
@@ -1342,6 +1352,45 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
             Gc[gr * E + gc] = val;
             if (!diag || gr != gc) Gc[gc * E + gr] = val;
         }
+    }
+}
+
+template <int NP = 0>
+__global__ __launch_bounds__(256) void k_gram_bf16(
+    const bf16_t* __restrict__ Z, float* __restrict__ G,
+    ll C, ll E, ll V, ll nsplit) {
+    gram_bf16_body<NP>(blockIdx.x, Z, G, C, E, V, nsplit);
+}
+
+// ===========================================================================
+// k_corr_gram_duo: ONE grid carrying BOTH the raw-correlation blocks of
+// chunk i and the Gram(+normalize) blocks of chunk i-1, proportionally
+// interleaved over blockIdx.  HIP streams do not co-schedule these two
+// kernels (measured: exactly serial); a single grid forces CU-level
+// co-residency, pairing the VALU-bound corr waves with the
+// memory-latency-bound Gram waves.
+// ===========================================================================
+template <int TP, int TL, int NP>
+__global__ __launch_bounds__(256) void k_corr_gram_duo(
+    const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ zOut, ll E, ll VB, ll C, ll zstride,
+    const bf16_t* __restrict__ Zprev, float* __restrict__ G,
+    ll Cg, ll Eg, ll Vg, ll nsplit,
+    ll nCorr, ll nGram) {
+    const ll b = blockIdx.x;
+    const ll total = nCorr + nGram;
+    // proportional interleave: block b is a gram block iff the
+    // cumulative gram quota advances at b
+    const ll g_before = (b * nGram) / total;
+    const ll g_at = ((b + 1) * nGram) / total;
+    if (g_at != g_before) {
+        if (G != nullptr)
+            gram_bf16_body<NP>(g_before, Zprev, G, Cg, Eg, Vg, nsplit);
+    } else {
+        ll ci = b - g_before;        // corr block index
+        if (ci < nCorr)
+            dot3s_body<TP, TL, 128, bf16_t, true>(
+                ci, At, B, zOut, E, VB, C, /*mode=*/0, zstride);
     }
 }
 
@@ -1915,6 +1964,60 @@ extern "C" void launch_fcma_corr_raw(const void* At, const void* B,
         }
     }
     #undef RAW_CASE
+}
+
+static ll duo_corr_blocks(ll C, ll E, int P, ll VB) {
+    ll nSubj = E / P;
+    return ceil_div(C, 128) * nSubj * ceil_div(VB, C3_VT);
+}
+
+extern "C" ll fcma_duo_gram_blocks(ll Cg, ll Eg, ll nsplit) {
+    ll eb = Eg / 64;
+    return Cg * eb * eb * nsplit;
+}
+
+extern "C" void launch_fcma_corr_gram_duo(
+    const void* At, const void* B, void* zOut, ll E, ll L, ll VB, ll C,
+    ll zstride, int P,
+    const void* Zprev, float* G, ll Cg, ll Eg, ll Vg, ll nsplit,
+    hipStream_t stream) {
+    ll nCorr = duo_corr_blocks(C, E, P, VB);
+    ll nGram = (G != nullptr) ? fcma_duo_gram_blocks(Cg, Eg, nsplit)
+                              : 0;
+    ll grid = nCorr + nGram;
+    #define DUO_CASE(TP, TL)                                             \
+        hipLaunchKernelGGL((k_corr_gram_duo<TP, TL, TP>), dim3(grid),    \
+                           dim3(256), 0, stream,                         \
+                           (const bf16_t*)At, (const bf16_t*)B,          \
+                           (bf16_t*)zOut, E, VB, C, zstride,             \
+                           (const bf16_t*)Zprev, G, Cg, Eg, Vg, nsplit,  \
+                           nCorr, nGram)
+    if (P == 4) {
+        switch (L) {
+            case 8:  DUO_CASE(4, 8);  return;
+            case 12: DUO_CASE(4, 12); return;
+            case 16: DUO_CASE(4, 16); return;
+            case 20: DUO_CASE(4, 20); return;
+            case 24: DUO_CASE(4, 24); return;
+            case 28: DUO_CASE(4, 28); return;
+            case 32: DUO_CASE(4, 32); return;
+            case 36: DUO_CASE(4, 36); return;
+            case 40: DUO_CASE(4, 40); return;
+        }
+    } else if (P == 2) {
+        switch (L) {
+            case 8:  DUO_CASE(2, 8);  return;
+            case 12: DUO_CASE(2, 12); return;
+            case 16: DUO_CASE(2, 16); return;
+            case 20: DUO_CASE(2, 20); return;
+            case 24: DUO_CASE(2, 24); return;
+            case 28: DUO_CASE(2, 28); return;
+            case 32: DUO_CASE(2, 32); return;
+            case 36: DUO_CASE(2, 36); return;
+            case 40: DUO_CASE(2, 40); return;
+        }
+    }
+    #undef DUO_CASE
 }
 
 extern "C" void launch_fcma_gram_fp8(const void* Z, float* G, ll C, ll E,

@@ -471,3 +471,32 @@ def test_pipeline_raw_split_matches_no_raw(ops):
     assert spearmanr(a, b).statistic > 0.95
     assert torch.allclose(g_split.float(), g_fused.float(),
                           atol=2.0, rtol=5e-2)
+
+
+def test_duo_pipeline_matches_streamed(ops):
+    """Duo-kernel pipeline == two-stream raw-split pipeline (identical
+    kernels, different scheduling)."""
+    import os
+
+    import numpy as np
+
+    from brainiak_amd.fcma.core import CorrelationPipeline
+    rng = np.random.RandomState(41)
+    E, L, V = 32, 12, 520
+    raw = []
+    for _ in range(E):
+        m = rng.randn(L, V).astype(np.float32)
+        m = (m - m.mean(0)) / m.std(0)
+        raw.append((m / np.sqrt(L)).astype(np.float32))
+    chunks = [(0, 200), (200, 200), (400, V - 400)]
+    pipe = CorrelationPipeline(raw, None, 4, device="cuda")
+    g_duo = pipe.pipelined_kernel_matrices(chunks)
+    os.environ["BRAINIAK_NO_DUO"] = "1"
+    try:
+        pipe2 = CorrelationPipeline(raw, None, 4, device="cuda")
+        g_str = pipe2.pipelined_kernel_matrices(chunks)
+    finally:
+        del os.environ["BRAINIAK_NO_DUO"]
+    assert g_duo.shape == g_str.shape == (V, E, E)
+    assert torch.allclose(g_duo.float(), g_str.float(), atol=1e-3,
+                          rtol=1e-4)
