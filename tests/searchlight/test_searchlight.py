@@ -211,3 +211,105 @@ def test_batched_block_distributed_matches_serial(tmp_path):
         if serial[idx] is not None:
             vol_serial[idx] = serial[idx]
     assert np.allclose(vol_dist, vol_serial, equal_nan=True, rtol=1e-6)
+
+
+# -- round-2 depth (ref tests/searchlight/test_searchlight.py:1-307) --------
+
+def _mean_sum_fn(subj, msk, rad, bcast):
+    return float(np.sum([s.mean() for s in subj]))
+
+
+def _token_fn(subj, msk, rad, bcast):
+    return bcast["token"]
+
+
+def _masksum_fn(subj, msk, rad, bcast):
+    return int(msk.sum())
+
+
+def _shape_check_fn(subj, msk, rad, bcast):
+    assert subj[0].shape == tuple(bcast["expect"])
+    return 1.0
+
+
+def _one_fn(subj, msk, rad, bcast):
+    return 1.0
+
+
+def _depth_block_fn(subjects, msk, myrad, bcast, extra):
+    out = np.empty((msk.shape[0] - 2, msk.shape[1] - 2,
+                    msk.shape[2] - 2), dtype=object)
+    out[:] = float(msk.shape[0])
+    return out
+
+
+def test_searchlight_rectangular_blocks(seeded_rng):
+    """Non-cubic volume + block edge not dividing the volume: borders
+    trimmed and stitched correctly."""
+    dims = (9, 7, 11)
+    data = [seeded_rng.rand(*dims, 3) for _ in range(2)]
+    mask = np.ones(dims, dtype=bool)
+    sl = Searchlight(sl_rad=1, max_blk_edge=4)
+    sl.distribute(data, mask)
+    sl.broadcast(None)
+    out = sl.run_searchlight(_mean_sum_fn, pool_size=1)
+    assert out.shape == dims
+    # interior voxels computed; the rad-wide volume border is None
+    # (the reference's block-trim semantics)
+    assert all(out[i, j, k] is not None
+               for i in range(1, dims[0] - 1)
+               for j in range(1, dims[1] - 1)
+               for k in range(1, dims[2] - 1))
+    assert out[0, 0, 0] is None
+
+
+def test_searchlight_bcast_var_reaches_fn(seeded_rng):
+    dims = (5, 5, 5)
+    data = [seeded_rng.rand(*dims, 2)]
+    mask = np.ones(dims, dtype=bool)
+    sl = Searchlight(sl_rad=1)
+    sl.distribute(data, mask)
+    sl.broadcast({"token": 17})
+    out = sl.run_searchlight(_token_fn, pool_size=1)
+    assert out[2, 2, 2] == 17
+
+
+def test_searchlight_diamond_shape(seeded_rng):
+    dims = (7, 7, 7)
+    data = [seeded_rng.rand(*dims, 2)]
+    mask = np.ones(dims, dtype=bool)
+    sl = Searchlight(sl_rad=1, shape=Diamond)
+    sl.distribute(data, mask)
+    sl.broadcast(None)
+    out = sl.run_searchlight(_masksum_fn, pool_size=1)
+    assert out[3, 3, 3] == 7        # diamond: center + 6 neighbours
+
+
+def test_searchlight_sparse_mask(seeded_rng):
+    """Only masked voxels are computed; unmasked stay None; the window
+    handed to voxel_fn is the full (2r+1)^3 sub-volume."""
+    dims = (8, 8, 8)
+    data = [seeded_rng.rand(*dims, 2)]
+    mask = np.zeros(dims, dtype=bool)
+    mask[3, 3, 3] = True
+    mask[4, 5, 4] = True
+    sl = Searchlight(sl_rad=2)
+    sl.distribute(data, mask)
+    sl.broadcast({"expect": (5, 5, 5, 2)})
+    out = sl.run_searchlight(_shape_check_fn, pool_size=1)
+    assert out[3, 3, 3] == 1.0 and out[4, 5, 4] == 1.0
+    assert out[2, 2, 2] is None
+
+
+def test_block_function_api(seeded_rng):
+    """run_block_function hands the user raw halo blocks and stitches
+    the trimmed outputs."""
+    dims = (8, 8, 8)
+    data = [seeded_rng.rand(*dims, 2)]
+    mask = np.ones(dims, dtype=bool)
+    sl = Searchlight(sl_rad=1, max_blk_edge=4)
+    sl.distribute(data, mask)
+    sl.broadcast(None)
+    out = sl.run_block_function(_depth_block_fn, None, pool_size=1)
+    assert out.shape == dims
+    assert out[1, 1, 1] == 6.0      # 4 + 2*rad halo
