@@ -1190,7 +1190,11 @@ extern "C" void launch_fcma_fused_corr_gram(
 // this kernel's memory-latency shadow (PMC: 84 % WAIT_ANY) instead of
 // the issue-bound corr kernel.  Subject groups (NP consecutive rows)
 // never straddle the 64-row bands (E % NP == 0, 64 % NP == 0).
-template <int NP = 0>
+// DZ = dual-band staging needed (E > 64).  With E == 64 every block
+// is its own diagonal band and zj is never read — eliding it halves
+// the static LDS (34.8 -> 17.4 KB) and doubles how many corr blocks
+// can co-reside with gram blocks inside the duo kernel.
+template <int NP = 0, bool DZ = true>
 __device__ __forceinline__ void gram_bf16_body(
     ll b,
     const bf16_t* __restrict__ Z, float* __restrict__ G,
@@ -1222,8 +1226,8 @@ __device__ __forceinline__ void gram_bf16_body(
     const int wc = (w & 1) * 32;
 
     __shared__ bf16_t zi[64][GR_ROW];
-    __shared__ bf16_t zj[64][GR_ROW];
-    const bool diag = (band_i == band_j);
+    __shared__ bf16_t zj[DZ ? 64 : 1][GR_ROW];
+    const bool diag = !DZ || (band_i == band_j);
     const bf16_t* Zc = Z + c * E * V;
     const ll rows_i = band_i * 64;
     const ll rows_j = band_j * 64;
@@ -1359,7 +1363,10 @@ template <int NP = 0>
 __global__ __launch_bounds__(256) void k_gram_bf16(
     const bf16_t* __restrict__ Z, float* __restrict__ G,
     ll C, ll E, ll V, ll nsplit) {
-    gram_bf16_body<NP>(blockIdx.x, Z, G, C, E, V, nsplit);
+    if (E == 64)
+        gram_bf16_body<NP, false>(blockIdx.x, Z, G, C, E, V, nsplit);
+    else
+        gram_bf16_body<NP, true>(blockIdx.x, Z, G, C, E, V, nsplit);
 }
 
 // ===========================================================================
@@ -1370,7 +1377,7 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
 // co-residency, pairing the VALU-bound corr waves with the
 // memory-latency-bound Gram waves.
 // ===========================================================================
-template <int TP, int TL, int NP>
+template <int TP, int TL, int NP, bool DZ = true>
 __global__ __launch_bounds__(256) void k_corr_gram_duo(
     const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ zOut, ll E, ll VB, ll C, ll zstride,
@@ -1385,7 +1392,8 @@ __global__ __launch_bounds__(256) void k_corr_gram_duo(
     const ll g_at = ((b + 1) * nGram) / total;
     if (g_at != g_before) {
         if (G != nullptr)
-            gram_bf16_body<NP>(g_before, Zprev, G, Cg, Eg, Vg, nsplit);
+            gram_bf16_body<NP, DZ>(g_before, Zprev, G, Cg, Eg, Vg,
+                                   nsplit);
     } else {
         ll ci = b - g_before;        // corr block index
         if (ci < nCorr)
@@ -1985,13 +1993,26 @@ extern "C" void launch_fcma_corr_gram_duo(
     ll nGram = (G != nullptr) ? fcma_duo_gram_blocks(Cg, Eg, nsplit)
                               : 0;
     ll grid = nCorr + nGram;
+    const bool dz = (Eg != 64);
     #define DUO_CASE(TP, TL)                                             \
-        hipLaunchKernelGGL((k_corr_gram_duo<TP, TL, TP>), dim3(grid),    \
-                           dim3(256), 0, stream,                         \
-                           (const bf16_t*)At, (const bf16_t*)B,          \
-                           (bf16_t*)zOut, E, VB, C, zstride,             \
-                           (const bf16_t*)Zprev, G, Cg, Eg, Vg, nsplit,  \
-                           nCorr, nGram)
+        do {                                                             \
+            if (dz)                                                      \
+                hipLaunchKernelGGL((k_corr_gram_duo<TP, TL, TP, true>),  \
+                                   dim3(grid), dim3(256), 0, stream,     \
+                                   (const bf16_t*)At,                    \
+                                   (const bf16_t*)B, (bf16_t*)zOut, E,   \
+                                   VB, C, zstride,                       \
+                                   (const bf16_t*)Zprev, G, Cg, Eg, Vg,  \
+                                   nsplit, nCorr, nGram);                \
+            else                                                         \
+                hipLaunchKernelGGL((k_corr_gram_duo<TP, TL, TP, false>), \
+                                   dim3(grid), dim3(256), 0, stream,     \
+                                   (const bf16_t*)At,                    \
+                                   (const bf16_t*)B, (bf16_t*)zOut, E,   \
+                                   VB, C, zstride,                       \
+                                   (const bf16_t*)Zprev, G, Cg, Eg, Vg,  \
+                                   nsplit, nCorr, nGram);                \
+        } while (0)
     if (P == 4) {
         switch (L) {
             case 8:  DUO_CASE(4, 8);  return;
