@@ -1359,14 +1359,14 @@ __device__ __forceinline__ void gram_bf16_body(
     }
 }
 
-template <int NP = 0>
+// DZ resolved at LAUNCH (host knows E): a runtime branch would leave
+// BOTH bodies' static __shared__ arrays allocated in one kernel
+// (52 KB instead of 17/35 — measured occupancy 3 vs 5)
+template <int NP = 0, bool DZ = true>
 __global__ __launch_bounds__(256) void k_gram_bf16(
     const bf16_t* __restrict__ Z, float* __restrict__ G,
     ll C, ll E, ll V, ll nsplit) {
-    if (E == 64)
-        gram_bf16_body<NP, false>(blockIdx.x, Z, G, C, E, V, nsplit);
-    else
-        gram_bf16_body<NP, true>(blockIdx.x, Z, G, C, E, V, nsplit);
+    gram_bf16_body<NP, DZ>(blockIdx.x, Z, G, C, E, V, nsplit);
 }
 
 // ===========================================================================
@@ -1916,8 +1916,14 @@ extern "C" void launch_fcma_gram_bf16(const void* Z, float* G, ll C, ll E,
                                       hipStream_t stream) {
     ll eb = E / 64;
     ll grid = C * eb * eb * nsplit;
-    hipLaunchKernelGGL(k_gram_bf16<0>, dim3(grid), dim3(256), 0, stream,
-                       (const bf16_t*)Z, G, C, E, V, nsplit);
+    if (E == 64)
+        hipLaunchKernelGGL((k_gram_bf16<0, false>), dim3(grid),
+                           dim3(256), 0, stream, (const bf16_t*)Z, G, C,
+                           E, V, nsplit);
+    else
+        hipLaunchKernelGGL((k_gram_bf16<0, true>), dim3(grid),
+                           dim3(256), 0, stream, (const bf16_t*)Z, G, C,
+                           E, V, nsplit);
 }
 
 extern "C" void launch_fcma_gram_bf16_norm(const void* Z, float* G,
@@ -1925,12 +1931,18 @@ extern "C" void launch_fcma_gram_bf16_norm(const void* Z, float* G,
                                            int P, hipStream_t stream) {
     ll eb = E / 64;
     ll grid = C * eb * eb * nsplit;
-    if (P == 4)
-        hipLaunchKernelGGL(k_gram_bf16<4>, dim3(grid), dim3(256), 0,
-                           stream, (const bf16_t*)Z, G, C, E, V, nsplit);
-    else
-        hipLaunchKernelGGL(k_gram_bf16<2>, dim3(grid), dim3(256), 0,
-                           stream, (const bf16_t*)Z, G, C, E, V, nsplit);
+    #define GRAM_NORM_CASE(NP, DZ)                                       \
+        hipLaunchKernelGGL((k_gram_bf16<NP, DZ>), dim3(grid),            \
+                           dim3(256), 0, stream, (const bf16_t*)Z, G,    \
+                           C, E, V, nsplit)
+    if (E == 64) {
+        if (P == 4) GRAM_NORM_CASE(4, false);
+        else        GRAM_NORM_CASE(2, false);
+    } else {
+        if (P == 4) GRAM_NORM_CASE(4, true);
+        else        GRAM_NORM_CASE(2, true);
+    }
+    #undef GRAM_NORM_CASE
 }
 
 // raw-r variant: dot3s with the normalize deferred (bf16 Z only)
