@@ -148,3 +148,103 @@ def test_srm_loads_frozen_v0_1_model():
     x = q @ S
     proj = m.transform([x, None, None])
     assert proj[0].shape == (5, 30)
+
+
+# -- round-2 depth (ref tests/funcalign/test_srm.py:1-319) -------------------
+
+def test_srm_batched_polar_many_matches_loop(seeded_rng):
+    """The ragged-batch Procrustes helper equals per-matrix polar
+    factors."""
+    import torch
+
+    from brainiak_amd.funcalign.srm import (
+        _polar_orthogonal,
+        _polar_orthogonal_many,
+    )
+    mats = [torch.tensor(seeded_rng.randn(v, 5)) for v in (30, 17, 44)]
+    batched = _polar_orthogonal_many([m.clone() for m in mats],
+                                     perturb=0.001)
+    for m, w in zip(mats, batched):
+        ref = _polar_orthogonal(m.clone(), perturb=0.001)
+        assert torch.allclose(w, ref, atol=1e-8)
+        # orthonormal columns
+        assert torch.allclose(w.T @ w, torch.eye(5, dtype=w.dtype),
+                              atol=1e-8)
+
+
+def test_srm_w_orthogonality_and_shapes(seeded_rng):
+    from brainiak_amd.funcalign.srm import SRM
+    data = [seeded_rng.randn(30 + 5 * i, 40) for i in range(3)]
+    m = SRM(n_iter=6, features=4, rand_seed=0)
+    m.fit(data)
+    assert m.s_.shape == (4, 40)
+    for i, w in enumerate(m.w_):
+        assert w.shape == (30 + 5 * i, 4)
+        assert np.allclose(w.T @ w, np.eye(4), atol=1e-8)
+    assert m.sigma_s_.shape == (4, 4)
+    # sigma_s symmetric positive semi-definite
+    assert np.allclose(m.sigma_s_, m.sigma_s_.T)
+    assert np.all(np.linalg.eigvalsh(m.sigma_s_) > -1e-10)
+    assert m.rho2_.shape == (3,)
+    assert np.all(m.rho2_ > 0)
+
+
+def test_srm_transform_errors(seeded_rng):
+    from brainiak_amd.funcalign.srm import SRM, NotFittedError
+    data = [seeded_rng.randn(20, 30) for _ in range(3)]
+    m = SRM(n_iter=4, features=3, rand_seed=0)
+    with pytest.raises(NotFittedError):
+        m.transform(data)
+    m.fit(data)
+    with pytest.raises(ValueError):
+        m.transform(data[:2])     # subject count mismatch
+    with pytest.raises(ValueError):
+        m.transform_subject(seeded_rng.randn(20, 25))  # TR mismatch
+
+
+def test_detsrm_matches_reference_complexity_contract(seeded_rng):
+    """DetSRM with orthonormal planted W and noiseless data recovers
+    X_i = W_i S exactly-ish."""
+    from brainiak_amd.funcalign.srm import DetSRM
+    k, t = 3, 40
+    S = seeded_rng.randn(k, t)
+    data, ws = [], []
+    for _ in range(3):
+        w = np.linalg.qr(seeded_rng.randn(25, k))[0]
+        ws.append(w)
+        data.append(w @ S)
+    m = DetSRM(n_iter=30, features=k, rand_seed=0)
+    m.fit(data)
+    for i in range(3):
+        recon = m.w_[i] @ m.s_
+        assert np.allclose(recon, data[i], atol=1e-3)
+
+
+def test_srm_features_exceed_samples_error(seeded_rng):
+    from brainiak_amd.funcalign.srm import SRM
+    data = [seeded_rng.randn(30, 5) for _ in range(2)]   # 5 TRs < 10
+    with pytest.raises(ValueError):
+        SRM(n_iter=3, features=10).fit(data)
+
+
+def test_rsrm_outliers_land_in_s(seeded_rng):
+    """RSRM: spike outliers are absorbed by the sparse S_i term, not
+    the shared response (the model's defining property)."""
+    from brainiak_amd.funcalign.rsrm import RSRM
+    k, t, v = 3, 50, 40
+    R = seeded_rng.randn(k, t)
+    data = []
+    for _ in range(3):
+        w = np.linalg.qr(seeded_rng.randn(v, k))[0]
+        x = w @ R + 0.01 * seeded_rng.randn(v, t)
+        data.append(x)
+    # subject 0 gets heavy sparse corruption
+    data[0][5, ::7] += 20.0
+    m = RSRM(n_iter=12, features=k, gamma=1.0, rand_seed=0)
+    m.fit(data)
+    # the outlier entries appear in s_[0]
+    assert np.abs(m.s_[0][5, ::7]).mean() > 1.0
+    # other entries stay mostly sparse-zero
+    mask = np.ones_like(m.s_[0], dtype=bool)
+    mask[5, ::7] = False
+    assert np.abs(m.s_[0][mask]).mean() < 0.5
