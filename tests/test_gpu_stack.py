@@ -234,3 +234,43 @@ def test_fcma_bf16_ranking_parity(cuda, seeded_rng):
     top_b = set(np.argsort(accs["bf16"])[-k:])
     top_f = set(np.argsort(accs["fp32"])[-k:])
     assert len(top_b & top_f) >= 0.8 * k
+
+
+def test_tfa_torch_lm_recovers_centers(cuda, seeded_rng):
+    """GPU TFA (torch-LM NLSS) recovers planted RBF centers as well as
+    the scipy trf path."""
+    import os
+
+    from brainiak_amd.factoranalysis.tfa import TFA
+    V, T, K = 3000, 60, 4
+    R = (seeded_rng.rand(V, 3) * 30).astype(np.float64)
+    centers = np.array([[7, 7, 7], [22, 8, 20], [8, 22, 15],
+                        [22, 22, 25]], dtype=np.float64)
+    widths = np.full(K, 16.0)
+    d2 = ((R[:, None, :] - centers[None, :, :]) ** 2).sum(-1)
+    F = np.exp(-d2 / widths[None, :])
+    W = seeded_rng.randn(K, T) * 2
+    X = F @ W + 0.05 * seeded_rng.randn(V, T)
+
+    def fit(env):
+        for k, v in env.items():
+            os.environ[k] = v
+        try:
+            t = TFA(K=K, max_iter=6, verbose=False, device="cuda")
+            t.fit(X, R)
+            return t.get_centers(t.local_posterior_)
+        finally:
+            for k in env:
+                del os.environ[k]
+
+    def match_err(est):
+        from scipy.spatial import distance
+        from scipy.optimize import linear_sum_assignment
+        cost = distance.cdist(centers, est)
+        r, c = linear_sum_assignment(cost)
+        return cost[r, c].mean()
+
+    err_lm = match_err(fit({}))                      # torch-LM default
+    err_scipy = match_err(fit({"BRAINIAK_TFA_SCIPY": "1"}))
+    assert err_lm < 5.0, err_lm          # voxel grid is 30 wide
+    assert err_lm < err_scipy + 2.0      # no worse than scipy + slack
