@@ -181,3 +181,66 @@ def test_brsa_transform_requires_fit(seeded_rng):
         BRSA().transform(Y)
     with pytest.raises(ValueError):
         BRSA().score(Y, design)
+
+
+# -- GBRSA grid/marginalization properties ----------------------------------
+
+def test_gbrsa_grids_properties():
+    m = GBRSA(SNR_bins=11, rho_bins=8)
+    s, w, rho, w_rho = m._grids()
+    assert s.shape == (11,) and np.all(s > 0)
+    assert np.isclose(w.sum(), 1.0)
+    assert rho.shape == (8,)
+    assert rho.min() >= -0.95 and rho.max() <= 0.95
+    assert np.isclose(w_rho.sum(), 1.0)
+    # exponential prior decreasing in SNR
+    assert np.all(np.diff(w) < 0)
+    # lognormal variant
+    m2 = GBRSA(SNR_prior='lognorm', SNR_bins=11, logS_range=0.5)
+    s2, w2, *_ = m2._grids()
+    assert np.isclose(w2.sum(), 1.0)
+    assert np.argmax(w2) not in (0, 10)   # interior mode
+    # uniform variant
+    m3 = GBRSA(SNR_prior='unif', SNR_bins=7)
+    _, w3, *_ = m3._grids()
+    assert np.allclose(w3, w3[0])
+
+
+def test_gbrsa_posterior_snr_tracks_truth(seeded_rng):
+    """Voxels with planted high SNR get higher marginal-posterior SNR
+    estimates than low-SNR voxels."""
+    rng = seeded_rng
+    T, V, C = 150, 40, 3
+    U = np.eye(C)
+    design = rng.randn(T, C)
+    snr = np.ones(V)
+    snr[:V // 2] = 3.0
+    snr[V // 2:] = 0.3
+    sigma = np.ones(V)
+    beta = np.linalg.cholesky(U) @ rng.randn(C, V) * (snr * sigma)
+    Y = design @ beta + rng.randn(T, V)
+    m = GBRSA(auto_nuisance=False, random_state=0, SNR_bins=9,
+              rho_bins=4, minimize_options={'maxiter': 50,
+                                            'disp': False})
+    m.fit(X=Y, design=design)
+    est = m.nSNR_[0]
+    assert est[:V // 2].mean() > 1.5 * est[V // 2:].mean()
+
+
+def test_gbrsa_transform_decodes_design(seeded_rng):
+    rng = seeded_rng
+    T, V, C = 140, 50, 3
+    design = rng.randn(T, C)
+    for c in range(C):
+        design[:, c] = np.convolve(design[:, c], np.ones(5) / 5,
+                                   mode='same')
+    beta = rng.randn(C, V) * 2
+    Y = design @ beta + rng.randn(T, V) * 0.6
+    m = GBRSA(auto_nuisance=False, random_state=0, SNR_bins=7,
+              rho_bins=4, minimize_options={'maxiter': 60,
+                                            'disp': False})
+    m.fit(X=Y, design=design)
+    ts, ts0 = m.transform(Y)
+    assert ts.shape == (T, C)
+    rs = [np.corrcoef(ts[:, c], design[:, c])[0, 1] for c in range(C)]
+    assert np.mean(rs) > 0.3, rs
