@@ -53,3 +53,71 @@ def test_collectives_world2(tmp_path):
     assert z["bcast"] == 0
     assert np.allclose(z["fused0"], 1.0)        # 0 + 1
     assert np.allclose(z["fused1"], 1.0)
+
+
+# -- round-2 depth -----------------------------------------------------------
+
+def _collectives_entry(ctx, outfile):
+    import numpy as np
+    # all_reduce on numpy
+    x = np.full(4, float(ctx.rank + 1))
+    summed = ctx.all_reduce(x, op="sum")
+    # fused many: mixed shapes/dtypes ride one collective
+    a = np.arange(3, dtype=np.float32) * (ctx.rank + 1)
+    b = np.array([ctx.rank], dtype=np.int64)
+    fa, fb = ctx.all_reduce_many([a, b])
+    # broadcast object from rank 1
+    obj = {"tag": ctx.rank} if ctx.rank == 1 else None
+    got = ctx.broadcast_object(obj, src=1)
+    # gather_object to root
+    gathered = ctx.gather_object(ctx.rank * 10, dst=0)
+    # scatter from root
+    part = ctx.scatter_object(
+        [f"part{i}" for i in range(ctx.world_size)]
+        if ctx.is_root else None)
+    if ctx.is_root:
+        np.save(outfile, np.array([
+            summed[0], fa[1], fb[0], got["tag"],
+            gathered[1], float(part == "part0")]))
+
+
+def test_collectives_two_ranks(tmp_path):
+    from brainiak_amd.parallel import spawn_ranks
+    out = str(tmp_path / "coll.npy")
+    spawn_ranks(_collectives_entry, world_size=2, args=(out,))
+    import numpy as np
+    vals = np.load(out)
+    assert vals[0] == 3.0          # 1 + 2
+    assert vals[1] == 3.0          # 1*1 + 1*2 fused
+    assert vals[2] == 1.0          # 0 + 1 int64 kept
+    assert vals[3] == 1            # broadcast from rank 1
+    assert vals[4] == 10           # gathered rank 1's value
+    assert vals[5] == 1.0          # scatter delivered part0 to root
+
+
+def test_serial_context_noop_semantics():
+    import numpy as np
+    from brainiak_amd.parallel import DistContext
+    ctx = DistContext(device="cpu")
+    assert not ctx.is_distributed and ctx.is_root
+    x = np.arange(5.0)
+    assert ctx.all_reduce(x) is x
+    assert ctx.all_gather(x) == [x]
+    assert ctx.broadcast_object({"a": 1}) == {"a": 1}
+    assert ctx.gather_object(7) == [7]
+    assert ctx.scatter_object([3]) == 3
+    ctx.barrier()                  # no-op, must not hang
+    assert ctx.shard(10) == slice(0, 10)
+    assert ctx.owner_of(5, 10) == 0
+
+
+def test_shard_slices_cover_and_balance():
+    from brainiak_amd.parallel import shard_slices
+    for n, k in ((10, 3), (7, 7), (5, 8), (100, 8)):
+        sl = shard_slices(n, k)
+        covered = []
+        for s in sl:
+            covered.extend(range(*s.indices(n)))
+        assert covered == list(range(n))
+        sizes = [s.stop - s.start for s in sl]
+        assert max(sizes) - min(sizes) <= 1
