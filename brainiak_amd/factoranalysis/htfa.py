@@ -71,81 +71,81 @@ class HTFA(TFA):
 
     # -- convergence on the global template ---------------------------------
 
+    def _template_delta(self):
+        """(max |Δ|, mse, relative energy) between prior and posterior
+        templates over the mean block."""
+        a = self.global_prior_[:self.prior_size]
+        b = self.global_posterior_[:self.prior_size]
+        d = a - b
+        return (float(np.abs(d).max()), float(np.mean(d ** 2)),
+                float(np.sum(d ** 2) / max(np.sum(b ** 2), 1e-30)))
+
     def _converged(self):
-        prior = self.global_prior_[0:self.prior_size]
-        posterior = self.global_posterior_[0:self.prior_size]
-        diff = prior - posterior
-        max_diff = np.max(np.fabs(diff))
+        max_diff, mse, ratio = self._template_delta()
         if self.verbose:
-            _, mse = self._mse_converged()
-            diff_ratio = np.sum(diff ** 2) / np.sum(posterior ** 2)
             logger.info('htfa prior posterior max diff %f mse %f '
-                        'diff_ratio %f', max_diff, mse, diff_ratio)
+                        'diff_ratio %f', max_diff, mse, ratio)
         return (max_diff <= self.threshold), max_diff
 
     def _mse_converged(self):
-        prior = self.global_prior_[0:self.prior_size]
-        posterior = self.global_posterior_[0:self.prior_size]
-        mse = np.mean((prior - posterior) ** 2)
+        _, mse, _ = self._template_delta()
         return (mse <= self.threshold), mse
 
     # -- MAP update ----------------------------------------------------------
 
-    @staticmethod
-    def _map_update(prior_mean, prior_cov, global_cov_scaled,
-                    new_observation):
-        """Gaussian MAP combination of the prior mean with the mean of the
-        per-subject observations."""
-        common = np.linalg.inv(prior_cov + global_cov_scaled)
-        observation_mean = np.mean(new_observation, axis=1)
-        posterior_mean = prior_cov.dot(common.dot(observation_mean)) + \
-            global_cov_scaled.dot(common.dot(prior_mean))
-        posterior_cov = prior_cov.dot(common.dot(global_cov_scaled))
-        return posterior_mean, posterior_cov
-
     def _map_update_posterior(self):
+        """One BATCHED Gaussian MAP update of all K factors.
+
+        The gathered per-subject posteriors reshape to
+        ``centers [n_subj, K, D]`` / ``widths [n_subj, K]``; the
+        per-factor center update
+            μ* = Σp (Σp + Σg)⁻¹ x̄  +  Σg (Σp + Σg)⁻¹ μp
+            Σ* = Σp (Σp + Σg)⁻¹ Σg
+        runs as one [K, D, D] batched inverse + einsum instead of the
+        reference's per-k / per-subject index walk
+        (ref htfa.py:246-341).
+        """
         self.global_posterior_ = self.global_prior_.copy()
-        prior_centers = self.get_centers(self.global_prior_)
-        prior_widths = self.get_widths(self.global_prior_)
-        prior_centers_mean_cov = self.get_centers_mean_cov(
-            self.global_prior_)
-        prior_widths_mean_var = self.get_widths_mean_var(
-            self.global_prior_)
-        center_size = self.K * self.n_dim
-        posterior_size = center_size + self.K
-        for k in np.arange(self.K):
-            next_centers = np.zeros((self.n_dim, self.n_subj))
-            next_widths = np.zeros(self.n_subj)
-            for s in np.arange(self.n_subj):
-                center_start = s * posterior_size
-                width_start = center_start + center_size
-                start_idx = center_start + k * self.n_dim
-                end_idx = center_start + (k + 1) * self.n_dim
-                next_centers[:, s] = \
-                    self.gather_posterior[start_idx:end_idx].copy()
-                next_widths[s] = self.gather_posterior[width_start + k]
+        K, D = self.K, self.n_dim
+        stacked = self.gather_posterior.reshape(self.n_subj,
+                                                self.prior_size)
+        obs_centers = stacked[:, :K * D].reshape(self.n_subj, K, D)
+        obs_widths = stacked[:, K * D:K * (D + 1)]
 
-            cov = from_tri_2_sym(prior_centers_mean_cov[k], self.n_dim)
-            cov = cov + cov.T - np.diag(np.diag(cov))
-            posterior_mean, posterior_cov = self._map_update(
-                prior_centers[k].T.copy(), cov,
-                self.global_centers_cov_scaled, next_centers)
-            self.global_posterior_[k * self.n_dim:(k + 1) * self.n_dim] = \
-                posterior_mean.T
-            start_idx = int(self.map_offset[2]) + k * self.cov_vec_size
-            end_idx = int(self.map_offset[2]) + (k + 1) * self.cov_vec_size
-            self.global_posterior_[start_idx:end_idx] = \
-                from_sym_2_tri(posterior_cov)
+        mu_p = self.get_centers(self.global_prior_)        # [K, D]
+        w_p = self.get_widths(self.global_prior_).ravel()  # [K]
+        tri = self.get_centers_mean_cov(self.global_prior_)
+        cov_p = np.stack([from_tri_2_sym(tri[k], D) for k in range(K)])
+        cov_p = cov_p + np.transpose(cov_p, (0, 2, 1)) \
+            - np.einsum('kij,ij->kij', cov_p, np.eye(D))
+        wvar_p = np.asarray(self.get_widths_mean_var(
+            self.global_prior_), dtype=np.float64).ravel()
 
-            pw_var = float(prior_widths_mean_var[k])
-            common = 1.0 / (pw_var + self.global_widths_var_scaled)
-            observation_mean = np.mean(next_widths)
-            tmp = common * self.global_widths_var_scaled
-            self.global_posterior_[int(self.map_offset[1]) + k] = \
-                pw_var * common * observation_mean + \
-                tmp * float(prior_widths[k])
-            self.global_posterior_[int(self.map_offset[3]) + k] = \
-                pw_var * tmp
+        sig_g = self.global_centers_cov_scaled                # [D, D]
+        common = np.linalg.inv(cov_p + sig_g[None])           # [K, D, D]
+        x_bar = obs_centers.mean(axis=0)                      # [K, D]
+        mu_star = np.einsum('kij,kj->ki', cov_p,
+                            np.einsum('kij,kj->ki', common, x_bar)) \
+            + np.einsum('ij,kj->ki', sig_g,
+                        np.einsum('kij,kj->ki', common, mu_p))
+        cov_star = cov_p @ common @ sig_g[None]               # [K, D, D]
+
+        wv_g = self.global_widths_var_scaled
+        denom = wvar_p + wv_g
+        w_star = (wvar_p * obs_widths.mean(axis=0)
+                  + wv_g * w_p) / denom
+        wvar_star = wvar_p * wv_g / denom
+
+        self.global_posterior_[:K * D] = mu_star.ravel()
+        self.global_posterior_[int(self.map_offset[1]):
+                               int(self.map_offset[1]) + K] = w_star
+        cov_tri = np.concatenate([from_sym_2_tri(cov_star[k])
+                                  for k in range(K)])
+        self.global_posterior_[int(self.map_offset[2]):
+                               int(self.map_offset[2])
+                               + K * self.cov_vec_size] = cov_tri
+        self.global_posterior_[int(self.map_offset[3]):
+                               int(self.map_offset[3]) + K] = wvar_star
         return self
 
     def _assign_posterior(self):
@@ -172,24 +172,22 @@ class HTFA(TFA):
     # -- subject metadata ----------------------------------------------------
 
     def _get_subject_info(self, n_local_subj, data):
-        max_sample_tr = np.zeros(n_local_subj).astype(int)
-        max_sample_voxel = np.zeros(n_local_subj).astype(int)
-        for idx in np.arange(n_local_subj):
-            nvoxel, ntr = data[idx].shape
-            max_sample_voxel[idx] = min(self.max_voxel,
-                                        int(self.voxel_ratio * nvoxel))
-            max_sample_tr[idx] = min(self.max_tr,
-                                     int(self.tr_ratio * ntr))
-        return max_sample_tr, max_sample_voxel
+        """Per-subject voxel/TR subsampling caps (ratio-scaled, clipped
+        at max_voxel/max_tr — the reference's sizing rule)."""
+        caps_tr = [min(self.max_tr, int(self.tr_ratio * d.shape[1]))
+                   for d in data]
+        caps_vox = [min(self.max_voxel,
+                        int(self.voxel_ratio * d.shape[0]))
+                    for d in data]
+        return (np.asarray(caps_tr, dtype=int),
+                np.asarray(caps_vox, dtype=int))
 
     def _get_weight_size(self, data, n_local_subj):
-        weight_size = np.zeros(1).astype(int)
-        local_weight_offset = np.zeros(n_local_subj).astype(int)
-        for idx, subj_data in enumerate(data):
-            if idx > 0:
-                local_weight_offset[idx] = weight_size[0]
-            weight_size[0] += self.K * subj_data.shape[1]
-        return weight_size, local_weight_offset
+        """Flattened-weight total size and per-subject offsets."""
+        sizes = np.asarray([self.K * d.shape[1] for d in data],
+                           dtype=int)
+        offsets = np.concatenate([[0], np.cumsum(sizes)[:-1]])
+        return np.array([sizes.sum()]), offsets
 
     def _init_prior_posterior(self, ctx, R, n_local_subj):
         if ctx.is_root:
@@ -210,24 +208,18 @@ class HTFA(TFA):
         return self
 
     def _update_weight(self, data, R, n_local_subj, local_weight_offset):
-        for s, subj_data in enumerate(data):
-            base = s * self.prior_size
-            centers = self.local_posterior_[
-                base:base + self.K * self.n_dim].reshape(
-                    (self.K, self.n_dim))
-            widths = self.local_posterior_[
-                base + self.K * self.n_dim:base + self.prior_size].reshape(
-                    (self.K, 1))
-            unique_R, inds = self.get_unique_R(R[s])
+        """Full-resolution ridge weights per subject from the final
+        posterior factors, packed into the flat local_weights_ buffer."""
+        del local_weight_offset, n_local_subj
+        posts = self.local_posterior_.reshape(-1, self.prior_size)
+        pieces = []
+        for subj_data, coords, post in zip(data, R, posts):
+            centers = self.get_centers(post)
+            widths = self.get_widths(post)
+            unique_R, inds = self.get_unique_R(coords)
             F = self.get_factors(unique_R, inds, centers, widths)
-            start_idx = local_weight_offset[s]
-            if s == n_local_subj - 1:
-                self.local_weights_[start_idx:] = \
-                    self.get_weights(subj_data, F).ravel()
-            else:
-                end_idx = local_weight_offset[s + 1]
-                self.local_weights_[start_idx:end_idx] = \
-                    self.get_weights(subj_data, F).ravel()
+            pieces.append(self.get_weights(subj_data, F).ravel())
+        self.local_weights_ = np.concatenate(pieces)
         return self
 
     # -- main loop ------------------------------------------------------------
@@ -293,27 +285,22 @@ class HTFA(TFA):
         return self
 
     def _check_input(self, X, R):
-        if not isinstance(X, list):
-            raise TypeError("Input data should be a list")
-        if not isinstance(R, list):
-            raise TypeError("Coordinates should be a list")
-        if len(X) < 1:
-            raise ValueError("Need at leat one subject to train the model."
-                             " Got {0:d}".format(len(X)))
-        for idx, x in enumerate(X):
-            if not isinstance(x, np.ndarray):
-                raise TypeError("Each subject data should be an array")
-            if x.ndim != 2:
-                raise TypeError("Each subject data should be 2D array")
-            if not isinstance(R[idx], np.ndarray):
+        if not isinstance(X, list) or not isinstance(R, list):
+            raise TypeError("X and R must be lists (one entry per "
+                            "local subject)")
+        if not X:
+            raise ValueError("need at least one local subject")
+        for idx, (x, r) in enumerate(zip(X, R)):
+            if not (isinstance(x, np.ndarray) and x.ndim == 2):
                 raise TypeError(
-                    "Each scanner coordinate matrix should be an array")
-            if R[idx].ndim != 2:
+                    "X[%d] must be a 2-D [voxels, TRs] array" % idx)
+            if not (isinstance(r, np.ndarray) and r.ndim == 2):
                 raise TypeError(
-                    "Each scanner coordinate matrix should be 2D array")
-            if x.shape[0] != R[idx].shape[0]:
+                    "R[%d] must be a 2-D [voxels, dims] array" % idx)
+            if x.shape[0] != r.shape[0]:
                 raise TypeError(
-                    "n_voxel should be the same in X[idx] and R[idx]")
+                    "X[%d] and R[%d] disagree on the voxel count "
+                    "(%d vs %d)" % (idx, idx, x.shape[0], r.shape[0]))
         return self
 
     def fit(self, X, R):
