@@ -947,11 +947,15 @@ class GBRSA(_BRSACore):
     def transform(self, X, y=None, scan_onsets=None):
         """Decode per-TR condition and nuisance courses for each
         subject with the AR(1) Kalman/RTS smoother (same machinery as
-        BRSA.transform).  Returns (list of ts, list of ts0)."""
+        BRSA.transform); ``scan_onsets`` (shared, or a per-subject
+        list) restarts the smoother at each run boundary.
+        Returns (list of ts, list of ts0)."""
         self._check_fitted()
         single = not isinstance(X, list)
         if single:
             X = [X]
+        if scan_onsets is None or not isinstance(scan_onsets, list):
+            scan_onsets = [scan_onsets] * len(X)
         ts_all, ts0_all = [], []
         for i, Yi in enumerate(X):
             Y = np.asarray(Yi, dtype=np.float64)
@@ -961,8 +965,12 @@ class GBRSA(_BRSACore):
                                     self._rho_X0_[i]])
             sig2_x = np.concatenate([self._sigma2_design_[i],
                                      self._sigma2_X0_[i]])
-            z = _kalman_rts(Y, W, rho_x, sig2_x, self.rho_[i],
-                            self.sigma_[i] ** 2)
+            lens = _run_lengths(Y.shape[0], scan_onsets[i])
+            bounds = np.concatenate([[0], np.cumsum(lens)])
+            z = np.concatenate([
+                _kalman_rts(Y[bounds[j]:bounds[j + 1]], W, rho_x,
+                            sig2_x, self.rho_[i], self.sigma_[i] ** 2)
+                for j in range(len(lens))], axis=0)
             ts_all.append(z[:, :C])
             ts0_all.append(z[:, C:])
         if single:
