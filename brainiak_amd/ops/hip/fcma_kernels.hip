@@ -1945,19 +1945,110 @@ extern "C" void launch_fcma_gram_bf16_norm(const void* Z, float* G,
     #undef GRAM_NORM_CASE
 }
 
+// raw-r dot3s with TWO voxels per thread: the raw kernel is
+// memory-latency bound (PMC r2: 67 % WAIT_ANY) — doubling the
+// per-thread streams doubles outstanding B-loads/Z-stores per wave.
+template <int TP, int TL, int C3_CT>
+__global__ __launch_bounds__(256) void k_corr_raw_v2(
+    const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ zOut, ll E, ll VB,
+    ll C, ll zstride) {
+    static_assert(TL % 2 == 0 && TP >= 2 && TP <= 4, "even L, P 2/4");
+    constexpr int P = TP;
+    constexpr int KP = TL / 2;
+    constexpr int L = TL;
+    constexpr int VT2 = 2 * C3_VT;
+    const ll nSubj = E / P;
+    const ll cTiles = (C + C3_CT - 1) / C3_CT;
+    const ll vTiles = (VB + VT2 - 1) / VT2;
+    ll b = blockIdx.x;
+    const ll vt = b % vTiles; b /= vTiles;
+    const ll s = b % nSubj;   b /= nSubj;
+    const ll ct = b;
+    if (ct >= cTiles) return;
+    const ll c0 = ct * C3_CT;
+    const int CT = (int)min((ll)C3_CT, C - c0);
+    const ll v0 = vt * (ll)VT2 + threadIdx.x;
+    const ll v1 = v0 + C3_VT;
+    const bool has1 = v1 < VB;
+    if (v0 >= VB) return;
+
+    bf16x2_t bp0[P][KP], bp1[P][KP];
+    #pragma unroll
+    for (int p = 0; p < P; ++p) {
+        const bf16_t* brow = B + ((ll)(s * P + p) * L) * VB;
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp) {
+            bf16x2_t t0, t1;
+            t0[0] = *(const __bf16*)&brow[(ll)(2 * kp) * VB + v0];
+            t0[1] = *(const __bf16*)&brow[(ll)(2 * kp + 1) * VB + v0];
+            bp0[p][kp] = t0;
+            ll vv = has1 ? v1 : v0;
+            t1[0] = *(const __bf16*)&brow[(ll)(2 * kp) * VB + vv];
+            t1[1] = *(const __bf16*)&brow[(ll)(2 * kp + 1) * VB + vv];
+            bp1[p][kp] = t1;
+        }
+    }
+
+    const unsigned int* abase = (const unsigned int*)
+        (At + (c0 * (ll)E + s * (ll)P) * L);
+    const int cstride = (E * L) / 2;
+
+    for (int c = 0; c < CT; ++c) {
+        const unsigned int* ac = abase + (ll)c * cstride;
+        float acc0[P], acc1[P];
+        #pragma unroll
+        for (int p = 0; p < P; ++p) { acc0[p] = 0.f; acc1[p] = 0.f; }
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp)
+            #pragma unroll
+            for (int p = 0; p < P; ++p) {
+                bf16x2_t a = __builtin_bit_cast(
+                    bf16x2_t, ac[p * (L / 2) + kp]);
+                acc0[p] = __builtin_amdgcn_fdot2_f32_bf16(
+                    a, bp0[p][kp], acc0[p], false);
+                acc1[p] = __builtin_amdgcn_fdot2_f32_bf16(
+                    a, bp1[p][kp], acc1[p], false);
+            }
+        bf16_t* dst = zOut + ((c0 + c) * zstride + s * (ll)P) * VB;
+        #pragma unroll
+        for (int p = 0; p < P; ++p) {
+            dst[(size_t)p * VB + v0] = (bf16_t)acc0[p];
+            if (has1)
+                dst[(size_t)p * VB + v1] = (bf16_t)acc1[p];
+        }
+    }
+}
+
 // raw-r variant: dot3s with the normalize deferred (bf16 Z only)
 extern "C" void launch_fcma_corr_raw(const void* At, const void* B,
                                      void* zOut, ll E, ll L, ll VB,
                                      ll C, ll zstride, int P,
                                      hipStream_t stream) {
     ll nSubj = E / P;
-    ll grid3 = ceil_div(C, 128) * nSubj * ceil_div(VB, C3_VT);
+    static int vpt = -1;
+    if (vpt < 0) {
+        const char* e = getenv("BRAINIAK_CORR_VPT");
+        vpt = (e && atoi(e) == 2) ? 2 : 1;
+    }
+    ll grid3 = ceil_div(C, 128) * nSubj
+             * ceil_div(VB, (ll)(vpt * C3_VT));
     #define RAW_CASE(TP, TL)                                             \
-        hipLaunchKernelGGL((k_corr_norm_dot3s<TP, TL, 128, bf16_t,       \
-                                              true>),                   \
-                           dim3(grid3), dim3(256), 0, stream,            \
-                           (const bf16_t*)At, (const bf16_t*)B,          \
-                           (bf16_t*)zOut, E, VB, C, /*mode=*/0, zstride)
+        do {                                                             \
+            if (vpt == 2)                                                \
+                hipLaunchKernelGGL((k_corr_raw_v2<TP, TL, 128>),         \
+                                   dim3(grid3), dim3(256), 0, stream,    \
+                                   (const bf16_t*)At,                    \
+                                   (const bf16_t*)B, (bf16_t*)zOut, E,   \
+                                   VB, C, zstride);                      \
+            else                                                         \
+                hipLaunchKernelGGL((k_corr_norm_dot3s<TP, TL, 128,       \
+                                                      bf16_t, true>),    \
+                                   dim3(grid3), dim3(256), 0, stream,    \
+                                   (const bf16_t*)At,                    \
+                                   (const bf16_t*)B, (bf16_t*)zOut, E,   \
+                                   VB, C, /*mode=*/0, zstride);          \
+        } while (0)
     if (P == 4) {
         switch (L) {
             case 8:  RAW_CASE(4, 8);  return;
