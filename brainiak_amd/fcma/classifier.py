@@ -61,14 +61,13 @@ class Classifier:
         """corr [num_samples, P, V2] for voxels [start, start+P) of X1."""
         num_samples = len(X1)
         assert num_samples > 0, \
-            'at least one sample is needed for correlation computation'
+            'need at least one (data1, data2) sample to correlate'
         num_voxels1 = X1[0].shape[1]
         num_voxels2 = X2[0].shape[1]
         assert num_voxels1 * num_voxels2 == self.num_features_, \
-            'the number of features provided by the input data ' \
-            'does not match the number of features defined in the model'
+            'input voxel-pair count differs from the fitted feature count'
         assert X1[0].shape[0] == X2[0].shape[0], \
-            'the numbers of TRs of X1 and X2 are not identical'
+            'X1 and X2 disagree on TR count'
         if num_processed_voxels is None:
             num_processed_voxels = num_voxels1
         a = self._stack(X1)[:, :, start_voxel:start_voxel
@@ -137,7 +136,7 @@ class Classifier:
 
     def fit(self, X, y, num_training_samples=None):
         assert len(X) == len(y), \
-            'the number of samples must be equal to the number of labels'
+            'sample count and label count differ'
         for x in X:
             assert len(x) == 2, \
                 'there must be two parts for each correlation computation'
