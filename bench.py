@@ -35,14 +35,18 @@ def parse_args():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     # defaults sized so the timed region is long enough for the
-    # driver's gpu-busy sampling to land inside it (~0.9 s on MI355X)
-    ap.add_argument("--steps", type=int, default=75)
-    ap.add_argument("--warmup", type=int, default=8)
+    # driver's gpu-busy sampling to land inside it (~2.6 s on MI355X
+    # at the whole-brain step default)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=4)
     ap.add_argument("--num-voxels", type=int, default=34470)
     ap.add_argument("--subjects", type=int, default=16)
     ap.add_argument("--epochs-per-subj", type=int, default=4)
     ap.add_argument("--epoch-len", type=int, default=12)
-    ap.add_argument("--voxels-per-step", type=int, default=4096)
+    ap.add_argument("--voxels-per-step", type=int, default=None,
+                    help="default: the whole brain (--num-voxels) — one "
+                         "step = one complete whole-brain selection "
+                         "pass per GPU")
     ap.add_argument("--num-folds", type=int, default=4)
     ap.add_argument("--chunk", type=int, default=512,
                     help="pipeline chunk (voxels per kernel pass; "
@@ -105,7 +109,7 @@ def main():
     del raw
 
     # this rank's voxel range rotates so steps touch different voxels
-    vps = min(args.voxels_per_step, V)
+    vps = min(args.voxels_per_step or V, V)
     chunk = min(args.chunk, vps)
     if device.type != "cuda":
         # CPU smoke path: the fp32 correlation chunk is [chunk, E, V]
