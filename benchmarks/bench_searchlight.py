@@ -45,26 +45,31 @@ def gpu_isc_block_fn(data, msk, rad, bcast_var, extra):
 
 
 def gpu_isc_batch_fn(stacks, masks, rad, bcast_var, extra):
-    """Batched form for Searchlight.run_batched_block_function: the
-    whole same-shape block group rides ONE correlation + ONE ball
-    aggregation (ops.stencil3d — direct LDS stencil instead of
-    MIOpen's im2col conv lowering)."""
+    """Batched device form for run_batched_block_function_device: the
+    whole same-shape block group (already resident on device) rides
+    ONE correlation + ONE ball aggregation (ops.stencil3d — direct LDS
+    stencil instead of MIOpen's im2col conv lowering)."""
     from brainiak_amd import ops
     device = bcast_var
-    a = torch.as_tensor(stacks[0], device=device)  # [B, x, y, z, T]
-    b = torch.as_tensor(stacks[1], device=device)
+    a = (stacks[0] if torch.is_tensor(stacks[0])
+         else torch.as_tensor(stacks[0], device=device)).float()
+    b = (stacks[1] if torch.is_tensor(stacks[1])
+         else torch.as_tensor(stacks[1], device=device)).float()
     az = a - a.mean(-1, keepdim=True)
     bz = b - b.mean(-1, keepdim=True)
     denom = (az.norm(dim=-1) * bz.norm(dim=-1)).clamp_min(1e-12)
     corr = (az * bz).sum(-1) / denom               # [B, x, y, z]
-    kernel = _ball_kernel(rad, device)
-    if str(device).startswith("cuda") and ops.has_hip():
+    kernel = _ball_kernel(rad, a.device)
+    if a.is_cuda and ops.require_hip():
         ball_mean = ops.stencil3d(corr.contiguous(), kernel[0, 0])
     else:
         ball_mean = torch.nn.functional.conv3d(
             corr[:, None], kernel)[:, 0]           # [B, ox, oy, oz]
-    out = ball_mean.cpu().numpy()
     inner = masks[:, rad:-rad, rad:-rad, rad:-rad] if rad > 0 else masks
+    if torch.is_tensor(inner):
+        return torch.where(inner, ball_mean,
+                           torch.full_like(ball_mean, float("nan")))
+    out = ball_mean.cpu().numpy()
     return np.where(inner, out, np.nan)
 
 
@@ -99,7 +104,11 @@ def main():
     sl.broadcast(device)
 
     def step(i):
-        sl.run_batched_block_function(gpu_isc_batch_fn)
+        if device.type == "cuda":
+            sl.run_batched_block_function_device(gpu_isc_batch_fn,
+                                                 device)
+        else:
+            sl.run_batched_block_function(gpu_isc_batch_fn)
 
     elapsed = timed_steps(step, args.steps, args.warmup, world, device)
     centers = int(mask.sum())

@@ -247,6 +247,53 @@ class Searchlight:
                 results.append((self.blocks[i][0], outs[j]))
         return self._gather_and_stitch(results)
 
+    def run_batched_block_function_device(self, batch_fn, device,
+                                          extra_block_fn_params=None):
+        """Device-resident form of ``run_batched_block_function``: this
+        rank's block stacks are uploaded to ``device`` ONCE (on first
+        call after ``distribute``) and stay resident — subsequent runs
+        slice the cached tensors instead of re-staging per-block numpy
+        copies through pageable H2D.  288 GB of HBM makes whole-shard
+        residency the natural layout (measured: the per-step host
+        staging was 98 % of the batched searchlight wall time).
+
+        batch_fn(subject_stacks, mask_stack, sl_rad, bcast_var, extra)
+            receives torch tensors on ``device`` ([B, bx, by, bz, T]
+            per subject, [B, bx, by, bz] bool) and must return a
+            [B, ox, oy, oz] torch or numpy array (border trimmed by
+            sl_rad).
+        """
+        import torch
+
+        key = str(device)
+        cache = getattr(self, "_device_group_cache", None)
+        if cache is None or cache[0] != key:
+            groups = {}
+            for idx in range(len(self.blocks)):
+                shape = self.submasks[idx].shape
+                groups.setdefault(shape, []).append(idx)
+            staged = []
+            for idxs in groups.values():
+                subj_stacks = [
+                    torch.as_tensor(np.stack([sub[i] for i in idxs]))
+                    .to(device) for sub in self.subproblems]
+                mask_stack = torch.as_tensor(
+                    np.stack([self.submasks[i] for i in idxs])).to(
+                        device)
+                staged.append((idxs, subj_stacks, mask_stack))
+            cache = (key, staged)
+            self._device_group_cache = cache
+
+        results = []
+        for idxs, subj_stacks, mask_stack in cache[1]:
+            outs = batch_fn(subj_stacks, mask_stack, self.sl_rad,
+                            self.bcast_var, extra_block_fn_params)
+            if isinstance(outs, torch.Tensor):
+                outs = outs.cpu().numpy()
+            for j, i in enumerate(idxs):
+                results.append((self.blocks[i][0], outs[j]))
+        return self._gather_and_stitch(results)
+
     def run_searchlight(self, voxel_fn, pool_size=None):
         """Apply ``voxel_fn`` at every active voxel; returns an object
         volume (None where inactive / in the trimmed border)."""

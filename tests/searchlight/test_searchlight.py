@@ -313,3 +313,35 @@ def test_block_function_api(seeded_rng):
     out = sl.run_block_function(_depth_block_fn, None, pool_size=1)
     assert out.shape == dims
     assert out[1, 1, 1] == 6.0      # 4 + 2*rad halo
+
+
+def _identity_batch_fn(stacks, masks, rad, bcast, extra):
+    import numpy as np
+    a = stacks[0]
+    if hasattr(a, "cpu"):
+        a = a.cpu().numpy()
+    # mean over time of the inner block
+    out = a.mean(-1)[:, rad:-rad, rad:-rad, rad:-rad]
+    return out
+
+
+def test_device_batched_matches_host_batched(seeded_rng):
+    """run_batched_block_function_device (cpu 'device') == the host
+    path, including the resident-cache reuse on a second call."""
+    dims = (10, 12, 10)
+    data = [seeded_rng.rand(*dims, 5).astype(np.float32)]
+    mask = np.ones(dims, dtype=bool)
+    sl = Searchlight(sl_rad=1, max_blk_edge=5)
+    sl.distribute(data, mask)
+    sl.broadcast(None)
+    host = sl.run_batched_block_function(_identity_batch_fn)
+    dev1 = sl.run_batched_block_function_device(_identity_batch_fn,
+                                                "cpu")
+    dev2 = sl.run_batched_block_function_device(_identity_batch_fn,
+                                                "cpu")   # cached path
+    for a, b in ((host, dev1), (dev1, dev2)):
+        mism = [(i, j, k) for i in range(dims[0])
+                for j in range(dims[1]) for k in range(dims[2])
+                if not np.isclose(float(a[i, j, k] or 0),
+                                  float(b[i, j, k] or 0))]
+        assert not mism, mism[:5]
