@@ -80,7 +80,7 @@ def main():
     ap.add_argument("--dims", type=int, nargs=3, default=[91, 109, 91])
     ap.add_argument("--trs", type=int, default=40)
     ap.add_argument("--rad", type=int, default=3)
-    ap.add_argument("--max-blk-edge", type=int, default=12)
+    ap.add_argument("--max-blk-edge", type=int, default=24)
     args = ap.parse_args()
 
     rank, world, device, _ = dist_setup()
