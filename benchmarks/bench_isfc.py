@@ -41,8 +41,11 @@ def main():
     precision = 'bf16' if device.type == "cuda" else 'fp32'
 
     def step(i):
+        # device-resident result (the [V,V] D2H would dominate; GPU
+        # consumers keep it in HBM — isc.py return_tensor)
         isfc_distributed(mine, ctx, summary_statistic='mean',
-                         precision=precision)
+                         precision=precision,
+                         return_tensor=device.type == "cuda")
 
     elapsed = timed_steps(step, args.steps, args.warmup, world, device)
     pairs_per_sec = float(V) * V * subjects * args.steps / elapsed

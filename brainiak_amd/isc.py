@@ -191,7 +191,8 @@ def isfc(data, targets=None, pairwise=False, summary_statistic=None,
 
 
 def isfc_distributed(local_data, comm, summary_statistic=None,
-                     row_tile=4096, device=None, precision='fp32'):
+                     row_tile=4096, device=None, precision='fp32',
+                     return_tensor=False):
     """Subject-sharded leave-one-out ISFC over RCCL/xGMI.
 
     Parameters
@@ -213,7 +214,9 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
 
     precision 'bf16' runs the [T,V]x[T,V] correlation GEMMs on the
     MFMA bf16 path (~0.4 % relative error on r; fp32 is the default
-    and matches the serial oracle).
+    and matches the serial oracle).  ``return_tensor=True`` keeps the
+    collapsed result on the compute device (the D2H of a [V, V]
+    matrix dominates the GPU step otherwise).
     """
     import torch
 
@@ -254,18 +257,46 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
         from . import ops as _ops
         use_hip = dev.type == 'cuda' and _ops.require_hip()
         acc = torch.zeros((V, V), dtype=torch.float32, device=dev)
-        for d, nd in zip(local, normed_local):
-            loo = _norm((total - d) / (n_total - 1))
-            m = _corr(nd, loo, keep_bf16=use_hip)
-            if use_hip:
-                # fused symmetrize+atanh+accumulate (one HBM pass,
-                # reads the GEMM's bf16 output directly)
-                _ops.isfc_accum_(acc, m.contiguous())
-            else:
+        if use_hip:
+            # subjects stream through the fused sym+atanh+accumulate
+            # kernel in stacks of up to 4: the [V, V] accumulator's
+            # read-modify-write happens once per stack instead of once
+            # per subject (acc traffic dominates at V=50k).  GEMMs
+            # write straight into the stack buffer slices — no copies.
+            nb = min(4, max(1, len(local)))
+            stackbuf = torch.empty((nb, V, V), dtype=torch.bfloat16,
+                                   device=dev)
+            pairs = list(zip(local, normed_local))
+            j = 0
+            for i, (d, nd) in enumerate(pairs):
+                loo = _norm((total - d) / (n_total - 1))
+                torch.matmul(nd.T.to(torch.bfloat16),
+                             loo.to(torch.bfloat16),
+                             out=stackbuf[j])
+                j += 1
+                if j == nb or i == len(pairs) - 1:
+                    _ops.isfc_accum_(acc, stackbuf[:j])
+                    j = 0
+            del stackbuf
+        else:
+            for d, nd in zip(local, normed_local):
+                loo = _norm((total - d) / (n_total - 1))
+                m = _corr(nd, loo)
                 m = (m + m.T) / 2
                 acc += torch.atanh(m.clamp(-1 + 1e-7, 1 - 1e-7))
         acc = comm.all_reduce(acc)
-        return torch.tanh(acc / n_total).cpu().numpy()
+        out = torch.tanh(acc / n_total)
+        if return_tensor:
+            # device-resident result for GPU consumers: the pageable
+            # D2H of a [50k, 50k] fp32 matrix costs ~1 s — more than
+            # the whole on-device computation (profiles/README.md r2)
+            return out
+        if out.is_cuda:
+            host = torch.empty_like(out, device="cpu",
+                                    pin_memory=True)
+            host.copy_(out)
+            return host.numpy()
+        return out.cpu().numpy()
 
     stacks = []
     for d, nd in zip(local, normed_local):

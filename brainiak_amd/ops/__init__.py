@@ -219,7 +219,9 @@ def stencil3d(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
 
 def isfc_accum_(acc: torch.Tensor, M: torch.Tensor) -> torch.Tensor:
     """In-place acc += atanh(clamp((M + M^T)/2, +-(1-1e-7))) — the ISFC
-    Fisher-mean accumulation fused into one HBM pass."""
+    Fisher-mean accumulation fused into one HBM pass.  M may be a
+    [B, V, V] subject stack: the batch streams through while the acc
+    tile loads/stores once."""
     return _ext().isfc_accum_(acc, M)
 
 

@@ -43,7 +43,7 @@ void launch_svm_cv(const float*, const float*, const int*, const int*,
 ll fcma_supported_L(ll);
 int fcma_corr_norm_smem(ll, int);
 int fcma_fused_gram_supported(ll, int, ll);
-void launch_isfc_accum(float*, const void*, ll, int, void*);
+void launch_isfc_accum(float*, const void*, ll, ll, int, void*);
 void launch_stencil3d(const float*, const float*, float*, ll, int, int,
                       int, int, void*);
 void launch_fcma_fused_corr_gram(const void*, const void*, float*, ll,
@@ -491,17 +491,21 @@ torch::Tensor stencil3d(torch::Tensor x, torch::Tensor w) {
 }
 
 torch::Tensor isfc_accum_(torch::Tensor acc, torch::Tensor M) {
+    // M: [V, V] or a [B, V, V] subject stack (the batch amortizes the
+    // acc read-modify-write across B matrices in one pass)
     bool m_bf16 = M.scalar_type() == torch::kBFloat16;
+    ll B = (M.dim() == 3) ? M.size(0) : 1;
+    ll V = acc.size(0);
     TORCH_CHECK(acc.is_cuda() && M.is_cuda() && acc.is_contiguous()
                 && M.is_contiguous()
                 && acc.scalar_type() == torch::kFloat32
                 && (m_bf16 || M.scalar_type() == torch::kFloat32)
-                && acc.dim() == 2 && M.dim() == 2
-                && acc.size(0) == acc.size(1)
-                && M.sizes() == acc.sizes(),
-                "acc fp32 / M fp32-or-bf16 square GPU matrices");
+                && acc.dim() == 2 && acc.size(1) == V
+                && (M.dim() == 2 || M.dim() == 3)
+                && M.size(-1) == V && M.size(-2) == V,
+                "acc fp32 [V,V] / M fp32-or-bf16 [V,V] or [B,V,V]");
     launch_isfc_accum(acc.data_ptr<float>(), M.data_ptr(),
-                      acc.size(0), m_bf16 ? 1 : 0, cur_stream());
+                      V, B, m_bf16 ? 1 : 0, cur_stream());
     return acc;
 }
 

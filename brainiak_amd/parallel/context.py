@@ -131,7 +131,9 @@ class DistContext:
     # -- tensor helpers ----------------------------------------------------
 
     def _as_tensor(self, array) -> Tuple[torch.Tensor, bool]:
-        """Return (tensor on comm device, was_numpy)."""
+        """Return (tensor on comm device, was_numpy).  nccl wants CUDA
+        tensors; gloo wants CPU — device tensors are staged through
+        host for gloo collectives (and restored by _return_like)."""
         if isinstance(array, np.ndarray):
             t = torch.from_numpy(np.ascontiguousarray(array))
             was_numpy = True
@@ -140,6 +142,8 @@ class DistContext:
             was_numpy = False
         if self.backend == "nccl" and not t.is_cuda:
             t = t.to(self.device)
+        elif self.backend == "gloo" and t.is_cuda:
+            t = t.cpu()
         return t.contiguous(), was_numpy
 
     def _return_like(self, t: torch.Tensor, like, was_numpy: bool):
