@@ -264,14 +264,14 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
             # per subject (acc traffic dominates at V=50k).  GEMMs
             # write straight into the stack buffer slices — no copies.
             nb = min(4, max(1, len(local)))
-            stackbuf = torch.empty((nb, V, V), dtype=torch.bfloat16,
-                                   device=dev)
+            gdt = (torch.bfloat16 if precision == 'bf16'
+                   else torch.float32)
+            stackbuf = torch.empty((nb, V, V), dtype=gdt, device=dev)
             pairs = list(zip(local, normed_local))
             j = 0
             for i, (d, nd) in enumerate(pairs):
                 loo = _norm((total - d) / (n_total - 1))
-                torch.matmul(nd.T.to(torch.bfloat16),
-                             loo.to(torch.bfloat16),
+                torch.matmul(nd.T.to(gdt), loo.to(gdt),
                              out=stackbuf[j])
                 j += 1
                 if j == nb or i == len(pairs) - 1:
