@@ -12,6 +12,7 @@ Explicit-VR LE multi-frame, via the self-contained
 ``utils.dicom_minimal`` writer — pydicom is not installed here).
 """
 
+import datetime
 import logging
 import os
 import time
@@ -22,6 +23,11 @@ import numpy as np
 from . import fmrisim as sim
 
 logger = logging.getLogger(__name__)
+
+# reference-parity module attribute: the generator's start timestamp
+# (the reference stamps DICOM acquisition times relative to it,
+# ref fmrisim_real_time_generator.py:49)
+script_datetime = datetime.datetime.now()
 
 __all__ = ["generate_data"]
 
