@@ -1377,7 +1377,7 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
 // co-residency, pairing the VALU-bound corr waves with the
 // memory-latency-bound Gram waves.
 // ===========================================================================
-template <int TP, int TL, int NP, bool DZ = true>
+template <int TP, int TL, int NP, bool DZ = true, int CT = 128>
 __global__ __launch_bounds__(256) void k_corr_gram_duo(
     const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ zOut, ll E, ll VB, ll C, ll zstride,
@@ -1397,7 +1397,7 @@ __global__ __launch_bounds__(256) void k_corr_gram_duo(
     } else {
         ll ci = b - g_before;        // corr block index
         if (ci < nCorr)
-            dot3s_body<TP, TL, 128, bf16_t, true>(
+            dot3s_body<TP, TL, CT, bf16_t, true>(
                 ci, At, B, zOut, E, VB, C, /*mode=*/0, zstride);
     }
 }
@@ -2077,9 +2077,18 @@ extern "C" void launch_fcma_corr_raw(const void* At, const void* B,
     #undef RAW_CASE
 }
 
+static int duo_ct(void) {
+    static int ct = -1;
+    if (ct < 0) {
+        const char* e = getenv("BRAINIAK_DUO_CT");
+        ct = (e && atoi(e) == 64) ? 64 : 128;
+    }
+    return ct;
+}
+
 static ll duo_corr_blocks(ll C, ll E, int P, ll VB) {
     ll nSubj = E / P;
-    return ceil_div(C, 128) * nSubj * ceil_div(VB, C3_VT);
+    return ceil_div(C, duo_ct()) * nSubj * ceil_div(VB, C3_VT);
 }
 
 extern "C" ll fcma_duo_gram_blocks(ll Cg, ll Eg, ll nsplit) {
@@ -2097,24 +2106,19 @@ extern "C" void launch_fcma_corr_gram_duo(
                               : 0;
     ll grid = nCorr + nGram;
     const bool dz = (Eg != 64);
+    const bool ct64 = duo_ct() == 64;
+    #define DUO_ONE(TP, TL, DZV, CTV)                                    \
+        hipLaunchKernelGGL((k_corr_gram_duo<TP, TL, TP, DZV, CTV>),      \
+                           dim3(grid), dim3(256), 0, stream,             \
+                           (const bf16_t*)At, (const bf16_t*)B,          \
+                           (bf16_t*)zOut, E, VB, C, zstride,             \
+                           (const bf16_t*)Zprev, G, Cg, Eg, Vg,          \
+                           nsplit, nCorr, nGram)
     #define DUO_CASE(TP, TL)                                             \
         do {                                                             \
-            if (dz)                                                      \
-                hipLaunchKernelGGL((k_corr_gram_duo<TP, TL, TP, true>),  \
-                                   dim3(grid), dim3(256), 0, stream,     \
-                                   (const bf16_t*)At,                    \
-                                   (const bf16_t*)B, (bf16_t*)zOut, E,   \
-                                   VB, C, zstride,                       \
-                                   (const bf16_t*)Zprev, G, Cg, Eg, Vg,  \
-                                   nsplit, nCorr, nGram);                \
-            else                                                         \
-                hipLaunchKernelGGL((k_corr_gram_duo<TP, TL, TP, false>), \
-                                   dim3(grid), dim3(256), 0, stream,     \
-                                   (const bf16_t*)At,                    \
-                                   (const bf16_t*)B, (bf16_t*)zOut, E,   \
-                                   VB, C, zstride,                       \
-                                   (const bf16_t*)Zprev, G, Cg, Eg, Vg,  \
-                                   nsplit, nCorr, nGram);                \
+            if (dz)         DUO_ONE(TP, TL, true, 128);                  \
+            else if (ct64)  DUO_ONE(TP, TL, false, 64);                  \
+            else            DUO_ONE(TP, TL, false, 128);                 \
         } while (0)
     if (P == 4) {
         switch (L) {
