@@ -16,13 +16,17 @@ Fisher-z/z-score).  Stages:
      reference's magnitude shrink (10**(2-digits) when the leading
      diagonal entry has >2 integer digits).
 
-On CUDA (= ROCm) tensors stages 1–3 dispatch to the hand-written
-HIP/CDNA4 kernels in ``brainiak_amd.ops``: stages 1–2 fuse into one
-correlation+normalize kernel (``k_corr_norm_dot3s`` — v_dot2c bf16
-MACs, scalar-path A operand, in-register Fisher-z/z-score/store; see
-docs/kernels.md), stage 3 is the ``v_mfma_f32_16x16x32_bf16`` Gram.
-The torch implementations below are the CPU path and the numerics
-oracle the GPU kernels are tested against.
+On CUDA (= ROCm) tensors the stages dispatch to the hand-written
+HIP/CDNA4 kernels in ``brainiak_amd.ops``.  The round-2 default is the
+DEFERRED-NORMALIZE split: ``k_corr_norm_dot3s<RAW>`` stores raw
+correlations (v_dot2c bf16 MACs, scalar-path A operand), the Gram
+kernel applies Fisher-z + the z-score to each staged tile IN REGISTERS
+inside its memory-latency shadow, and ``k_corr_gram_duo`` carries the
+corr blocks of chunk i and the Gram blocks of chunk i-1 in ONE grid so
+the VALU-bound and latency-bound waves co-reside (docs/kernels.md).
+``z_fp8=True`` stores Z as OCP e4m3 (profiles/fp8.md).  The torch
+implementations below are the CPU path and the numerics oracle the GPU
+kernels are tested against.
 """
 
 import os
