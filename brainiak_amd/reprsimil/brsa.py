@@ -157,16 +157,20 @@ def _ar1_quadforms(X, Y, run_TRs=None):
     bounds = np.cumsum(run_TRs)          # run end indices (exclusive)
     assert bounds[-1] == T, 'run lengths must sum to the time axis'
 
-    adj = torch.ones(T - 1, dtype=_DT)
-    adj[torch.as_tensor(bounds[:-1] - 1, dtype=torch.long)] = 0.0
-    D = torch.zeros((T, T), dtype=_DT)
-    idx = torch.arange(T - 1)
+    dev = X.device
+    adj = torch.ones(T - 1, dtype=_DT, device=dev)
+    adj[torch.as_tensor(bounds[:-1] - 1, dtype=torch.long,
+                        device=dev)] = 0.0
+    D = torch.zeros((T, T), dtype=_DT, device=dev)
+    idx = torch.arange(T - 1, device=dev)
     D[idx, idx + 1] = adj
     D[idx + 1, idx] = adj
-    interior = torch.ones(T, dtype=_DT)
+    interior = torch.ones(T, dtype=_DT, device=dev)
     starts = np.concatenate([[0], bounds[:-1]])
-    interior[torch.as_tensor(starts, dtype=torch.long)] = 0.0
-    interior[torch.as_tensor(bounds - 1, dtype=torch.long)] = 0.0
+    interior[torch.as_tensor(starts, dtype=torch.long,
+                             device=dev)] = 0.0
+    interior[torch.as_tensor(bounds - 1, dtype=torch.long,
+                             device=dev)] = 0.0
     F = torch.diag(interior)
     XtX = X.T @ X
     XtDX = X.T @ D @ X
@@ -282,6 +286,10 @@ def _kalman_rts(Y, W, rho_x, sig2_x, rho_e, sig2_e):
 class _BRSACore:
     """Shared plumbing for BRSA / GBRSA."""
 
+    def _dev(self):
+        d = getattr(self, "device", None)
+        return torch.device(d) if d is not None else torch.device("cpu")
+
     def _prepare(self, X, Y, nuisance, scan_onsets):
         Y = np.asarray(Y, dtype=np.float64)
         X = np.asarray(X, dtype=np.float64)
@@ -341,7 +349,7 @@ class BRSA(_BRSACore):
                  anneal_speed=10, GP_space=False, GP_inten=False,
                  space_smooth_range=None, inten_smooth_range=None,
                  tau_range=5.0, tau2_prior=prior_GP_var_inv_gamma,
-                 eta=0.0001):
+                 eta=0.0001, device=None):
         self.n_iter = n_iter
         self.rank = rank
         self.auto_nuisance = auto_nuisance
@@ -365,6 +373,7 @@ class BRSA(_BRSACore):
         self.tau_range = tau_range
         self.tau2_prior = tau2_prior
         self.eta = eta
+        self.device = device
 
     # -- likelihood --------------------------------------------------------
 
@@ -387,9 +396,10 @@ class BRSA(_BRSACore):
         (XtY, XtDY, XtFY) = quadXY
         (YtY, YtDY, YtFY) = quadYY
 
+        dev = XtX.device
         nL = C * rank
         L_flat = params[:nL].reshape(C, rank)
-        tril_mask = torch.ones((C, rank), dtype=torch.bool)
+        tril_mask = torch.ones((C, rank), dtype=torch.bool, device=dev)
         for i in range(C):
             for j in range(rank):
                 if j > i:
@@ -409,7 +419,7 @@ class BRSA(_BRSACore):
         q = YtY - rho * YtDY + rho ** 2 * YtFY                 # [V]
 
         U = L @ L.T
-        eye = torch.eye(C, dtype=_DT)
+        eye = torch.eye(C, dtype=_DT, device=dev)
         M = eye[None] + snr2[:, None, None] * (U[None] @ A)    # [V, C, C]
         # solve M w = U b  → quadratic correction b' w
         Ub = (U[None] @ b[:, :, None])                         # [V, C, 1]
@@ -442,7 +452,8 @@ class BRSA(_BRSACore):
                                   + gp['inten_diff2'] / l2_inten) / 2.0)
         else:
             K_major = torch.exp(-gp['dist2'] / l2_space / 2.0)
-        K = K_major + gp['eta'] * torch.eye(V, dtype=_DT)
+        K = K_major + gp['eta'] * torch.eye(V, dtype=_DT,
+                                            device=dev)
         Lk = torch.linalg.cholesky(K)
         y = log_snr[:, None]
         invK_y = torch.cholesky_solve(y, Lk)[:, 0]
@@ -475,16 +486,18 @@ class BRSA(_BRSACore):
             if gp is not None:
                 # start with a small length scale (≈ voxel size, the
                 # reference's choice, brsa.py:1406-1412)
-                d2 = gp['dist2'].numpy()
+                d2 = gp['dist2'].cpu().numpy()
                 off = d2[np.tril_indices_from(d2, k=-1)]
                 c0 = [np.log(max(np.min(off), 1e-2))]
                 if gp['inten_diff2'] is not None:
-                    i2 = gp['inten_diff2'].numpy()
+                    i2 = gp['inten_diff2'].cpu().numpy()
                     ioff = i2[np.tril_indices_from(i2, k=-1)]
                     c0.append(np.log(max(np.percentile(ioff, 2), 0.5)))
                 init = np.concatenate([init, c0])
 
-        params = torch.tensor(init, dtype=_DT, requires_grad=True)
+        dev = X_t.device
+        params = torch.tensor(init, dtype=_DT, device=dev,
+                              requires_grad=True)
 
         def val_and_grad(theta):
             with torch.no_grad():
@@ -496,7 +509,8 @@ class BRSA(_BRSACore):
                                     tau_range=self.tau_range,
                                     n_runs=n_runs)
             loss.backward()
-            return float(loss.detach()), params.grad.numpy().copy()
+            return (float(loss.detach().cpu()),
+                    params.grad.cpu().numpy().copy())
 
         res = minimize(val_and_grad, init, jac=True, method=self.optimizer,
                        options=self.minimize_options)
@@ -528,8 +542,10 @@ class BRSA(_BRSACore):
             if isr is None:
                 isr = np.max(inten_diff2) ** 0.5 / 2.0
         return {
-            'dist2': torch.as_tensor(dist2, dtype=_DT),
-            'inten_diff2': (torch.as_tensor(inten_diff2, dtype=_DT)
+            'dist2': torch.as_tensor(dist2, dtype=_DT,
+                                     device=self._dev()),
+            'inten_diff2': (torch.as_tensor(inten_diff2, dtype=_DT,
+                                            device=self._dev())
                             if inten_diff2 is not None else None),
             'space_smooth_range': float(ssr),
             'inten_smooth_range': (float(isr) if isr is not None
@@ -574,8 +590,8 @@ class BRSA(_BRSACore):
             # flat-prior X0 betas → project X and Y off X0's column space
             Xp = _project_out(X_design, X0)
             Yp = _project_out(Y_data, X0)
-            X_t = torch.as_tensor(Xp, dtype=_DT)
-            Y_t = torch.as_tensor(Yp, dtype=_DT)
+            X_t = torch.as_tensor(Xp, dtype=_DT, device=self._dev())
+            Y_t = torch.as_tensor(Yp, dtype=_DT, device=self._dev())
             params, nll, quads = self._fit_once(X_t, Y_t, C, V, T, rank,
                                                 init=params, gp=gp,
                                                 run_TRs=run_TRs)
@@ -614,11 +630,11 @@ class BRSA(_BRSACore):
                 self.lGPinten_ = float(
                     np.exp(params[nL + 2 * V + 1])) ** 0.5
                 K_major = np.exp(
-                    -(gp['dist2'].numpy() / l2_space
-                      + gp['inten_diff2'].numpy() / self.lGPinten_ ** 2)
+                    -(gp['dist2'].cpu().numpy() / l2_space
+                      + gp['inten_diff2'].cpu().numpy() / self.lGPinten_ ** 2)
                     / 2.0)
             else:
-                K_major = np.exp(-gp['dist2'].numpy() / l2_space / 2.0)
+                K_major = np.exp(-gp['dist2'].cpu().numpy() / l2_space / 2.0)
             K = K_major + self.eta * np.eye(V)
             y_snr = log_snr
             y_invK_y = float(y_snr @ np.linalg.solve(K, y_snr))
@@ -628,22 +644,24 @@ class BRSA(_BRSACore):
         # posterior-mean betas and noise sigma (given point estimates)
         with torch.no_grad():
             quadX, quadXY, quadYY = quads
-            snr2 = torch.as_tensor(self.nSNR_ ** 2, dtype=_DT)
-            rho_t = torch.as_tensor(rho, dtype=_DT)
+            dev = quadX[0].device
+            snr2 = torch.as_tensor(self.nSNR_ ** 2, dtype=_DT,
+                                   device=dev)
+            rho_t = torch.as_tensor(rho, dtype=_DT, device=dev)
             A = (quadX[0][None] - rho_t[:, None, None] * quadX[1][None]
                  + (rho_t ** 2)[:, None, None] * quadX[2][None])
             b = (quadXY[0].T - rho_t[:, None] * quadXY[1].T
                  + (rho_t ** 2)[:, None] * quadXY[2].T)
             q = quadYY[0] - rho_t * quadYY[1] + rho_t ** 2 * quadYY[2]
-            U_t = torch.as_tensor(self.U_, dtype=_DT)
-            eye = torch.eye(C, dtype=_DT)
+            U_t = torch.as_tensor(self.U_, dtype=_DT, device=dev)
+            eye = torch.eye(C, dtype=_DT, device=dev)
             M = eye[None] + snr2[:, None, None] * (U_t[None] @ A)
             Ub = (U_t[None] @ b[:, :, None])
             w = torch.linalg.solve(M, Ub)[:, :, 0]
             quad = (q - snr2 * (b * w).sum(1)).clamp_min(1e-10)
-            sigma2 = (quad / T).numpy()
+            sigma2 = (quad / T).cpu().numpy()
             # E[β|y] = snr² U (I + snr² A U)⁻¹ b  (per voxel)
-            beta = (snr2[:, None] * w).numpy().T        # [C, V]
+            beta = (snr2[:, None] * w).cpu().numpy().T  # [C, V]
         self.sigma_ = np.sqrt(sigma2)
         self.beta_ = beta
         # nuisance loadings + AR(1) stats of the training time courses,
@@ -698,11 +716,13 @@ class BRSA(_BRSACore):
         C = Xp.shape[1]
         run_TRs = _run_lengths(T, scan_onsets)
         quadX, quadXY, quadYY = _ar1_quadforms(
-            torch.as_tensor(Xp, dtype=_DT), torch.as_tensor(Yp, dtype=_DT),
+            torch.as_tensor(Xp, dtype=_DT, device=self._dev()),
+            torch.as_tensor(Yp, dtype=_DT, device=self._dev()),
             run_TRs)
         params = torch.tensor(np.concatenate([
             self.L_.ravel(), np.log(self.nSNR_),
-            np.arctanh(np.clip(self.rho_, -0.999, 0.999))]), dtype=_DT)
+            np.arctanh(np.clip(self.rho_, -0.999, 0.999))]), dtype=_DT,
+            device=self._dev())
         with torch.no_grad():
             nll = self._neg_loglik(params, quadX, quadXY, quadYY, C, V, T,
                                    self.L_.shape[1],
@@ -726,7 +746,7 @@ class GBRSA(_BRSACore):
                  baseline_single=False, logS_range=1.0, SNR_prior='exp',
                  SNR_bins=21, rho_bins=20, tol=1e-4,
                  optimizer='L-BFGS-B', minimize_options=None,
-                 random_state=None, anneal_speed=10):
+                 random_state=None, anneal_speed=10, device=None):
         self.n_iter = n_iter
         self.rank = rank
         self.auto_nuisance = auto_nuisance
@@ -744,6 +764,7 @@ class GBRSA(_BRSACore):
             {'maxiter': 150, 'disp': False}
         self.random_state = random_state
         self.anneal_speed = anneal_speed
+        self.device = device
 
     def _grids(self):
         """SNR and rho grids with prior weights (ref brsa.py:4089-4165)."""
@@ -767,18 +788,20 @@ class GBRSA(_BRSACore):
     def _neg_loglik_marg(self, L_params, quads, C, V, T, rank, grids,
                          n_runs=1):
         (XtX, XtDX, XtFX), (XtY, XtDY, XtFY), (YtY, YtDY, YtFY) = quads
+        dev = XtX.device
         s_grid, w_s, rho_grid, w_rho = grids
         L = L_params.reshape(C, rank)
         mask = torch.as_tensor(
-            np.tril(np.ones((C, rank)))[:, :rank], dtype=_DT)
+            np.tril(np.ones((C, rank)))[:, :rank], dtype=_DT,
+            device=dev)
         L = L * mask
         U = L @ L.T
-        eye = torch.eye(C, dtype=_DT)
+        eye = torch.eye(C, dtype=_DT, device=dev)
 
-        rho_t = torch.as_tensor(rho_grid, dtype=_DT)
-        s2 = torch.as_tensor(s_grid ** 2, dtype=_DT)
+        rho_t = torch.as_tensor(rho_grid, dtype=_DT, device=dev)
+        s2 = torch.as_tensor(s_grid ** 2, dtype=_DT, device=dev)
         logw = torch.log(torch.as_tensor(
-            np.outer(w_s, w_rho).ravel(), dtype=_DT))
+            np.outer(w_s, w_rho).ravel(), dtype=_DT, device=dev))
 
         A = (XtX[None] - rho_t[:, None, None] * XtDX[None]
              + (rho_t ** 2)[:, None, None] * XtFX[None])     # [R, C, C]
@@ -836,8 +859,9 @@ class GBRSA(_BRSACore):
             Xp = _project_out(Dp, X0)
             Yp = _project_out(Yi, X0)
             subj_quads.append(_ar1_quadforms(
-                torch.as_tensor(Xp, dtype=_DT),
-                torch.as_tensor(Yp, dtype=_DT), run_TRs))
+                torch.as_tensor(Xp, dtype=_DT, device=self._dev()),
+                torch.as_tensor(Yp, dtype=_DT, device=self._dev()),
+                run_TRs))
             dims.append((T, V))
             subj_runs.append(len(run_TRs))
             subj_ctx.append((Dp, Yi, X0))
@@ -845,7 +869,8 @@ class GBRSA(_BRSACore):
         rng = np.random.RandomState(self.random_state)
         init = (np.eye(C)[:, :rank] * 1.0).ravel() + rng.randn(
             C * rank) * 0.01
-        params = torch.tensor(init, dtype=_DT, requires_grad=True)
+        params = torch.tensor(init, dtype=_DT, device=self._dev(),
+                              requires_grad=True)
 
         def val_and_grad(theta):
             with torch.no_grad():
@@ -857,7 +882,8 @@ class GBRSA(_BRSACore):
                        for quads, (T, V), nr in zip(subj_quads, dims,
                                                     subj_runs))
             loss.backward()
-            return float(loss.detach()), params.grad.numpy().copy()
+            return (float(loss.detach().cpu()),
+                    params.grad.cpu().numpy().copy())
 
         res = minimize(val_and_grad, init, jac=True,
                        method=self.optimizer,
@@ -876,7 +902,7 @@ class GBRSA(_BRSACore):
         self._rho_design_, self._sigma2_design_ = [], []
         self._rho_X0_, self._sigma2_X0_ = [], []
         with torch.no_grad():
-            Lp = torch.as_tensor(res.x, dtype=_DT)
+            Lp = torch.as_tensor(res.x, dtype=_DT, device=self._dev())
             for quads, (T, V), (Dp, Yi, X0) in zip(subj_quads, dims,
                                                    subj_ctx):
                 beta, rho, sig2, snr = self._subject_posterior(
@@ -902,16 +928,18 @@ class GBRSA(_BRSACore):
         under the fitted U (the marginalization ref brsa.py:3390-3672
         point-estimates the same way)."""
         (XtX, XtDX, XtFX), (XtY, XtDY, XtFY), (YtY, YtDY, YtFY) = quads
+        dev = XtX.device
         s_grid, w_s, rho_grid, w_rho = grids
         L = L_params.reshape(C, rank)
         L = L * torch.as_tensor(
-            np.tril(np.ones((C, rank)))[:, :rank], dtype=_DT)
+            np.tril(np.ones((C, rank)))[:, :rank], dtype=_DT,
+            device=dev)
         U = L @ L.T
-        eye = torch.eye(C, dtype=_DT)
-        rho_t = torch.as_tensor(rho_grid, dtype=_DT)
-        s2 = torch.as_tensor(s_grid ** 2, dtype=_DT)
+        eye = torch.eye(C, dtype=_DT, device=dev)
+        rho_t = torch.as_tensor(rho_grid, dtype=_DT, device=dev)
+        s2 = torch.as_tensor(s_grid ** 2, dtype=_DT, device=dev)
         logw = torch.log(torch.as_tensor(
-            np.outer(w_s, w_rho).ravel(), dtype=_DT))
+            np.outer(w_s, w_rho).ravel(), dtype=_DT, device=dev))
         A = (XtX[None] - rho_t[:, None, None] * XtDX[None]
              + (rho_t ** 2)[:, None, None] * XtFX[None])
         b = (XtY[None] - rho_t[:, None, None] * XtDY[None]
@@ -936,12 +964,12 @@ class GBRSA(_BRSACore):
         S, R = len(s_grid), len(rho_grid)
         # E[β|y, s, ρ] = s² U (I + s² A U)⁻¹ b = s² w
         beta_g = (s2[:, None, None, None] * w).reshape(S * R, C, V)
-        beta = (wgt[:, None, :] * beta_g).sum(0).numpy()  # [C, V]
-        rho = (wgt * rho_t.repeat(S)[:, None]).sum(0).numpy()
-        sig2 = (wgt * (quad.reshape(S * R, V) / T)).sum(0).numpy()
+        beta = (wgt[:, None, :] * beta_g).sum(0).cpu().numpy()
+        rho = (wgt * rho_t.repeat(S)[:, None]).sum(0).cpu().numpy()
+        sig2 = (wgt * (quad.reshape(S * R, V) / T)).sum(0).cpu().numpy()
         snr = (wgt * torch.as_tensor(
-            s_grid, dtype=_DT).repeat_interleave(R)[:, None]
-            ).sum(0).numpy()
+            s_grid, dtype=_DT, device=dev).repeat_interleave(R)[:, None]
+            ).sum(0).cpu().numpy()
         return beta, rho, sig2, snr
 
     def transform(self, X, y=None, scan_onsets=None):
@@ -989,7 +1017,8 @@ class GBRSA(_BRSACore):
         rank = self.L_.shape[1]
         out = []
         with torch.no_grad():
-            Lp = torch.as_tensor(self.L_.ravel(), dtype=_DT)
+            Lp = torch.as_tensor(self.L_.ravel(), dtype=_DT,
+                                 device=self._dev())
             for i, (Yi, Di) in enumerate(zip(X, design)):
                 Y = np.asarray(Yi, dtype=np.float64)
                 X0 = self.X0_[i] if self.X0_[i].shape[0] == Y.shape[0] \
@@ -1000,11 +1029,12 @@ class GBRSA(_BRSACore):
                 C = Xp.shape[1]
                 run_TRs = _run_lengths(T, scan_onsets)
                 quads = _ar1_quadforms(
-                    torch.as_tensor(Xp, dtype=_DT),
-                    torch.as_tensor(Yp, dtype=_DT), run_TRs)
+                    torch.as_tensor(Xp, dtype=_DT, device=self._dev()),
+                    torch.as_tensor(Yp, dtype=_DT, device=self._dev()),
+                    run_TRs)
                 nll = self._neg_loglik_marg(Lp, quads, C, V, T, rank,
                                             grids, n_runs=len(run_TRs))
-                out.append(-float(nll) / V)
+                out.append(-float(nll.cpu()) / V)
         return out[0] if single else out
 
     def _check_fitted(self):

@@ -318,3 +318,45 @@ def test_srm_distributed_cuda_matches_serial(cuda, tmp_path, seeded_rng):
     resid = np.linalg.norm(proj - serial.s_.T) / np.linalg.norm(
         serial.s_.T)
     assert resid < 0.05, resid
+
+
+def test_brsa_gpu_matches_cpu(cuda, seeded_rng):
+    """BRSA fit with device='cuda' (fp64 quad forms + autograd on
+    gfx950) recovers the same covariance as the CPU fit."""
+    from brainiak_amd.reprsimil.brsa import BRSA
+    from brainiak_amd.utils.utils import cov2corr
+    rng = seeded_rng
+    T, V, C = 150, 60, 5
+    U = np.eye(C) * 0.5
+    U[0, 1] = U[1, 0] = 0.4
+    design = rng.randn(T, C)
+    beta = np.linalg.cholesky(U + 1e-9 * np.eye(C)) @ rng.randn(C, V)
+    Y = design @ beta + rng.randn(T, V) * 0.7 + 5.0
+    fits = {}
+    for dev in ("cpu", "cuda"):
+        m = BRSA(auto_nuisance=False, random_state=0, device=dev,
+                 minimize_options={'maxiter': 200, 'disp': False})
+        m.fit(X=Y.copy(), design=design.copy())
+        fits[dev] = m
+    off = ~np.eye(C, dtype=bool)
+    # both recover the planted structure and agree with each other
+    for dev in fits:
+        r = np.corrcoef(fits[dev].C_[off], cov2corr(U)[off])[0, 1]
+        assert r > 0.6, (dev, r)
+    assert np.allclose(fits["cpu"].C_, fits["cuda"].C_, atol=0.05)
+    assert np.allclose(fits["cpu"].beta_, fits["cuda"].beta_,
+                       atol=0.05, rtol=0.05)
+
+
+def test_gbrsa_gpu_runs(cuda, seeded_rng):
+    from brainiak_amd.reprsimil.brsa import GBRSA
+    rng = seeded_rng
+    T, V, C = 100, 30, 3
+    design = rng.randn(T, C)
+    Y = design @ (rng.randn(C, V) * 2) + rng.randn(T, V)
+    m = GBRSA(auto_nuisance=False, random_state=0, SNR_bins=5,
+              rho_bins=4, device="cuda",
+              minimize_options={'maxiter': 40, 'disp': False})
+    m.fit(X=Y, design=design)
+    assert m.U_.shape == (C, C)
+    assert np.isfinite(m.score(Y, design))
