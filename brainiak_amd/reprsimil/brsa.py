@@ -480,7 +480,7 @@ class BRSA(_BRSACore):
             rng = np.random.RandomState(self.random_state)
             init = np.concatenate([
                 (np.eye(C)[:, :rank]
-                 * np.sqrt(np.trace(quadX[0].numpy()) / C / T)).ravel()
+                 * np.sqrt(float(torch.trace(quadX[0]).cpu()) / C / T)).ravel()
                 + rng.randn(nL) * 0.01,
                 np.zeros(V), np.zeros(V)])
             if gp is not None:
