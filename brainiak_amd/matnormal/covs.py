@@ -49,15 +49,44 @@ def _param(value):
 
 
 class CovBase(abc.ABC):
-    """Base class for residual covariances (logdet / solve / params)."""
+    """Base class for residual covariances (logdet / solve / params).
+
+    ``to(device)`` moves every tensor attribute (templates, optimizable
+    parameters) so the whole likelihood graph runs on that device —
+    call it BEFORE handing parameters to an optimizer.
+    """
 
     def __init__(self, size):
         self.size = size
+        self._device = torch.device("cpu")
+
+    def to(self, device):
+        device = torch.device(device)
+
+        def _move(val):
+            moved = val.detach().to(device)
+            moved.requires_grad_(val.requires_grad)
+            return moved
+
+        for name, val in list(self.__dict__.items()):
+            if torch.is_tensor(val):
+                setattr(self, name, _move(val))
+            elif isinstance(val, list) and val and \
+                    all(torch.is_tensor(v) for v in val):
+                setattr(self, name, [_move(v) for v in val])
+        self._device = device
+        return self
+
+    def _zero(self):
+        return torch.zeros((), dtype=_DT, device=self._device)
+
+    def _eye(self):
+        return torch.eye(self.size, dtype=_DT, device=self._device)
 
     @property
     def logp(self):
         """Regularization log-prob (0 unless a subclass adds a prior)."""
-        return torch.zeros((), dtype=_DT)
+        return self._zero()
 
     @abc.abstractmethod
     def get_optimize_vars(self):
@@ -73,7 +102,7 @@ class CovBase(abc.ABC):
 
     @property
     def _prec(self):
-        return self.solve(torch.eye(self.size, dtype=_DT))
+        return self.solve(self._eye())
 
     @property
     def _cov(self):
@@ -85,7 +114,7 @@ class CovIdentity(CovBase):
 
     @property
     def logdet(self):
-        return torch.zeros((), dtype=_DT)
+        return self._zero()
 
     def get_optimize_vars(self):
         return []
@@ -95,11 +124,11 @@ class CovIdentity(CovBase):
 
     @property
     def _prec(self):
-        return torch.eye(self.size, dtype=_DT)
+        return self._eye()
 
     @property
     def _cov(self):
-        return torch.eye(self.size, dtype=_DT)
+        return self._eye()
 
 
 class CovAR1(CovBase):
@@ -298,6 +327,7 @@ class CovKroneckerFactored(CovBase):
     def __init__(self, sizes, Sigmas=None, mask=None):
         if not isinstance(sizes, list):
             raise TypeError("sizes is not a list")
+        self._device = torch.device("cpu")
         self.sizes = sizes
         self.nfactors = len(sizes)
         self.size = int(np.prod(np.array(sizes), dtype=np.int64))
@@ -325,14 +355,14 @@ class CovKroneckerFactored(CovBase):
     def logdet(self):
         if self.mask is None:
             n_list = torch.as_tensor([m.shape[0] for m in self.L],
-                                     dtype=_DT)
+                                     dtype=_DT, device=self._device)
             n_prod = torch.prod(n_list)
             logdets = torch.stack([
                 torch.sum(torch.log(torch.diagonal(m))) for m in self.L])
             return 2.0 * torch.sum(logdets * n_prod / n_list)
         n_list = [m.shape[0] for m in self.L]
         mask_reshaped = self.mask.reshape(n_list)
-        logdet = torch.zeros((), dtype=_DT)
+        logdet = self._zero()
         for i in range(self.nfactors):
             dims = [d for d in range(self.nfactors) if d != i]
             counts = mask_reshaped.sum(dim=dims).to(_DT)

@@ -27,17 +27,19 @@ class MNRSA:
     """Y ~ MN(0, Σ_t + XLLᵀXᵀ + X₀X₀ᵀ, Σ_s), U = LLᵀ; see docstring."""
 
     def __init__(self, time_cov, space_cov, n_nureg=5,
-                 optimizer="L-BFGS-B", optCtrl=None):
+                 optimizer="L-BFGS-B", optCtrl=None, device=None):
         self.n_T = time_cov.size
         self.n_V = space_cov.size
         self.n_nureg = n_nureg
         self.optMethod = optimizer
         self.optCtrl = optCtrl if optCtrl is not None else {}
-        self.X_0 = torch.randn((self.n_T, n_nureg),
-                               dtype=_DT).requires_grad_(True)
+        self.device = torch.device(device) if device is not None \
+            else torch.device("cpu")
+        self.X_0 = torch.randn((self.n_T, n_nureg), dtype=_DT,
+                               device=self.device).requires_grad_(True)
         self.train_variables = [self.X_0]
-        self.time_cov = time_cov
-        self.space_cov = space_cov
+        self.time_cov = time_cov.to(self.device)
+        self.space_cov = space_cov.to(self.device)
         self.train_variables.extend(self.time_cov.get_optimize_vars())
         self.train_variables.extend(self.space_cov.get_optimize_vars())
 
@@ -49,23 +51,27 @@ class MNRSA:
         """X = brain data [T, V], y = design [T, C] (sklearn-style
         argument order, flipped internally like the reference)."""
         X_design, Y_brain = y, X
-        X_design = torch.as_tensor(np.asarray(X_design), dtype=_DT)
-        Y_brain = torch.as_tensor(np.asarray(Y_brain), dtype=_DT)
+        X_design = torch.as_tensor(np.asarray(X_design), dtype=_DT,
+                                   device=self.device)
+        Y_brain = torch.as_tensor(np.asarray(Y_brain), dtype=_DT,
+                                  device=self.device)
         self.n_c = X_design.shape[1]
 
         if naive_init:
             from sklearn.linear_model import LinearRegression
             m = LinearRegression(fit_intercept=False)
-            m.fit(X=X_design.numpy(), y=Y_brain.numpy())
+            m.fit(X=X_design.cpu().numpy(), y=Y_brain.cpu().numpy())
             self.naive_U_ = np.cov(m.coef_.T)
             # regularize in case the naive estimate is singular
             naive = self.naive_U_ + 1e-9 * np.eye(self.n_c)
             self.L_flat = flatten_cholesky_unique(
-                np.linalg.cholesky(naive)).clone().requires_grad_(True)
+                np.linalg.cholesky(naive)).clone().to(
+                    self.device).requires_grad_(True)
         else:
             chol_flat_size = (self.n_c * (self.n_c + 1)) // 2
-            self.L_flat = torch.randn(chol_flat_size,
-                                      dtype=_DT).requires_grad_(True)
+            self.L_flat = torch.randn(
+                chol_flat_size, dtype=_DT,
+                device=self.device).requires_grad_(True)
         self.train_variables = self.train_variables + [self.L_flat]
 
         def lossfn(theta):
@@ -80,14 +86,15 @@ class MNRSA:
         for var, val in zip(self.train_variables, unpacked):
             with torch.no_grad():
                 var.copy_(val)
-        L = self.L.detach().numpy()
+        L = self.L.detach().cpu().numpy()
         self.U_ = L.dot(L.T)
         self.C_ = cov2corr(self.U_)
         return self
 
     def logp(self, X, Y):
         """MNRSA log-likelihood (marginal over the mapping)."""
-        rsa_cov = CovIdentity(size=self.n_c + self.n_nureg)
+        rsa_cov = CovIdentity(size=self.n_c + self.n_nureg).to(
+            self.device)
         x_stack = torch.cat([X @ self.L, self.X_0], dim=1)
         return (self.time_cov.logp + self.space_cov.logp + rsa_cov.logp
                 + matnorm_logp_marginal_row(

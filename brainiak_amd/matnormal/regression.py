@@ -19,11 +19,13 @@ class MatnormalRegression:
     """Y ~ MN(Xβ, time_cov, space_cov); see module docstring."""
 
     def __init__(self, time_cov, space_cov, optimizer="L-BFGS-B",
-                 optCtrl=None):
+                 optCtrl=None, device=None):
         self.optMethod = optimizer
         self.optCtrl = optCtrl if optCtrl is not None else {}
-        self.time_cov = time_cov
-        self.space_cov = space_cov
+        self.device = torch.device(device) if device is not None \
+            else torch.device("cpu")
+        self.time_cov = time_cov.to(self.device)
+        self.space_cov = space_cov.to(self.device)
         self.n_t = time_cov.size
         self.n_v = space_cov.size
 
@@ -33,8 +35,10 @@ class MatnormalRegression:
 
     def fit(self, X, y, naive_init=True):
         """Fit β and covariance params to design X [T, C], data y [T, V]."""
-        X = torch.as_tensor(np.asarray(X), dtype=_DT)
-        y = torch.as_tensor(np.asarray(y), dtype=_DT)
+        X = torch.as_tensor(np.asarray(X), dtype=_DT,
+                            device=self.device)
+        y = torch.as_tensor(np.asarray(y), dtype=_DT,
+                            device=self.device)
         self.n_c = X.shape[1]
 
         if naive_init:
@@ -44,7 +48,8 @@ class MatnormalRegression:
                 beta_init = torch.linalg.solve(X.T @ sigma_inv_x,
                                                X.T @ sigma_inv_y)
         else:
-            beta_init = torch.randn((self.n_c, self.n_v), dtype=_DT)
+            beta_init = torch.randn((self.n_c, self.n_v), dtype=_DT,
+                                    device=self.device)
         self.beta = beta_init.clone().detach().requires_grad_(True)
 
         self.train_variables = [self.beta]
@@ -65,7 +70,7 @@ class MatnormalRegression:
         for var, val in zip(self.train_variables, unpacked):
             with torch.no_grad():
                 var.copy_(val)
-        self.beta_ = self.beta.detach().numpy()
+        self.beta_ = self.beta.detach().cpu().numpy()
         return self
 
     def predict(self, X):
@@ -77,10 +82,12 @@ class MatnormalRegression:
             raise RuntimeError(
                 "More conditions than voxels! System is singular, "
                 "cannot decode.")
-        Y = torch.as_tensor(np.asarray(Y), dtype=_DT)
-        beta = torch.as_tensor(self.beta_, dtype=_DT)
+        Y = torch.as_tensor(np.asarray(Y), dtype=_DT,
+                            device=self.device)
+        beta = torch.as_tensor(self.beta_, dtype=_DT,
+                               device=self.device)
         with torch.no_grad():
             Sigma_s_btrp = self.space_cov.solve(beta.T)
-            Y_Sigma_Btrp = (Y @ Sigma_s_btrp).numpy()
-            B_Sigma_Btrp = (beta @ Sigma_s_btrp).numpy()
+            Y_Sigma_Btrp = (Y @ Sigma_s_btrp).cpu().numpy()
+            B_Sigma_Btrp = (beta @ Sigma_s_btrp).cpu().numpy()
         return np.linalg.solve(B_Sigma_Btrp.T, Y_Sigma_Btrp.T).T

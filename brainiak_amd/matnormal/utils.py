@@ -68,7 +68,8 @@ def unflatten_cholesky_unique(L_flat):
 
 
 def pack_trainable_vars(trainable_vars):
-    return torch.cat([tv.reshape(-1).detach() for tv in trainable_vars])
+    return torch.cat([tv.reshape(-1).detach().cpu()
+                      for tv in trainable_vars])
 
 
 def unpack_trainable_vars(x, trainable_vars):
@@ -96,6 +97,7 @@ def make_val_and_grad(lossfn, train_vars):
             g = var.grad
             grads.append(torch.zeros_like(var) if g is None else g)
         packed = torch.cat([g.reshape(-1) for g in grads])
-        return float(loss.detach()), packed.detach().numpy()
+        return (float(loss.detach().cpu()),
+                packed.detach().cpu().numpy())
 
     return val_and_grad

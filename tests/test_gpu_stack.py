@@ -360,3 +360,36 @@ def test_gbrsa_gpu_runs(cuda, seeded_rng):
     m.fit(X=Y, design=design)
     assert m.U_.shape == (C, C)
     assert np.isfinite(m.score(Y, design))
+
+
+def test_matnormal_regression_gpu_matches_cpu(cuda, seeded_rng):
+    from brainiak_amd.matnormal.covs import CovAR1, CovDiagonal
+    from brainiak_amd.matnormal.regression import MatnormalRegression
+    n, k, p = 60, 3, 10
+    X = seeded_rng.randn(n, k)
+    B = seeded_rng.randn(k, p) * 2
+    Y = X @ B + 0.1 * seeded_rng.randn(n, p)
+    fits = {}
+    for dev in ("cpu", "cuda"):
+        m = MatnormalRegression(time_cov=CovAR1(size=n),
+                                space_cov=CovDiagonal(size=p),
+                                device=dev)
+        m.fit(X, Y)
+        fits[dev] = m.beta_
+    assert np.allclose(fits["cpu"], fits["cuda"], atol=1e-3, rtol=1e-3)
+
+
+def test_mnrsa_gpu_runs(cuda, seeded_rng):
+    from brainiak_amd.matnormal.covs import CovIdentity
+    from brainiak_amd.matnormal.mnrsa import MNRSA
+    n_t, n_v, n_c = 50, 20, 4
+    design = seeded_rng.randn(n_t, n_c)
+    U = np.eye(n_c)
+    beta = np.linalg.cholesky(U) @ seeded_rng.randn(n_c, n_v)
+    Y = design @ beta + 0.3 * seeded_rng.randn(n_t, n_v)
+    m = MNRSA(time_cov=CovIdentity(size=n_t),
+              space_cov=CovIdentity(size=n_v), device="cuda",
+              optCtrl={"options": {"maxiter": 60}})
+    m.fit(X=Y, y=design)
+    assert m.U_.shape == (n_c, n_c)
+    assert np.isfinite(m.U_).all()
