@@ -162,7 +162,8 @@ class CovAR1(CovBase):
     @property
     def logdet(self):
         rho = 2 * torch.sigmoid(self.rho_unc) - 1
-        run_sizes = torch.as_tensor(self.run_sizes, dtype=_DT)
+        run_sizes = torch.as_tensor(self.run_sizes, dtype=_DT,
+                                    device=self._device)
         return torch.sum(2 * run_sizes * self.log_sigma
                          - torch.log(1 - rho ** 2))
 

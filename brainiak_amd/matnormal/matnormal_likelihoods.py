@@ -34,7 +34,8 @@ def solve_det_marginal(x, sigma, A, Q):
     Atrp_Sinv = A.T @ sigma._prec
     prod_term = torch.cholesky_solve(Atrp_Sinv, lemma_factor)
     solve = sigma.solve(
-        scaled_I(1.0, sigma.size, dtype=x.dtype) - A @ prod_term) @ x
+        scaled_I(1.0, sigma.size, dtype=x.dtype,
+                 device=x.device) - A @ prod_term) @ x
     return solve, logdet
 
 
@@ -46,7 +47,8 @@ def solve_det_conditional(x, sigma, A, Q):
     Atrp_Sinv = A.T @ sigma._prec
     prod_term = torch.cholesky_solve(Atrp_Sinv, lemma_factor)
     solve = sigma.solve(
-        scaled_I(1.0, sigma.size, dtype=x.dtype) + A @ prod_term) @ x
+        scaled_I(1.0, sigma.size, dtype=x.dtype,
+                 device=x.device) + A @ prod_term) @ x
     return solve, logdet
 
 
