@@ -2020,6 +2020,86 @@ __global__ __launch_bounds__(256) void k_corr_raw_v2(
     }
 }
 
+// raw-r dot3s with ADJACENT voxel pairs per thread: every B load and
+// Z store becomes a 4-byte (bf16x2) access — half the memory
+// instructions for the same bytes (probe: instruction-rate vs
+// byte-rate limited stores).
+template <int TP, int TL, int C3_CT>
+__global__ __launch_bounds__(256) void k_corr_raw_pair(
+    const bf16_t* __restrict__ At, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ zOut, ll E, ll VB,
+    ll C, ll zstride) {
+    static_assert(TL % 2 == 0 && TP >= 2 && TP <= 4, "even L, P 2/4");
+    constexpr int P = TP;
+    constexpr int KP = TL / 2;
+    constexpr int L = TL;
+    constexpr int VT2 = 2 * C3_VT;
+    const ll nSubj = E / P;
+    const ll cTiles = (C + C3_CT - 1) / C3_CT;
+    const ll vTiles = (VB + VT2 - 1) / VT2;
+    ll b = blockIdx.x;
+    const ll vt = b % vTiles; b /= vTiles;
+    const ll s = b % nSubj;   b /= nSubj;
+    const ll ct = b;
+    if (ct >= cTiles) return;
+    const ll c0 = ct * C3_CT;
+    const int CT = (int)min((ll)C3_CT, C - c0);
+    const ll v0 = vt * (ll)VT2 + 2 * threadIdx.x;   // adjacent pair
+    if (v0 + 1 >= VB) {
+        if (v0 >= VB) return;
+        // odd tail voxel: fall through scalar (VB is 16-padded so
+        // this branch is dead in production, kept for generality)
+    }
+
+    // B loads: one bf16x2 per (p, kp, row) covering (v0, v0+1)
+    bf16x2_t bp[P][KP][2];           // [..][2] = the two L rows
+    #pragma unroll
+    for (int p = 0; p < P; ++p) {
+        const bf16_t* brow = B + ((ll)(s * P + p) * L) * VB + v0;
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp) {
+            bp[p][kp][0] = *(const bf16x2_t*)&brow[(ll)(2 * kp) * VB];
+            bp[p][kp][1] = *(const bf16x2_t*)&brow[(ll)(2 * kp + 1)
+                                                   * VB];
+        }
+    }
+
+    const unsigned int* abase = (const unsigned int*)
+        (At + (c0 * (ll)E + s * (ll)P) * L);
+    const int cstride = (E * L) / 2;
+
+    for (int c = 0; c < CT; ++c) {
+        const unsigned int* ac = abase + (ll)c * cstride;
+        float acc0[P], acc1[P];
+        #pragma unroll
+        for (int p = 0; p < P; ++p) { acc0[p] = 0.f; acc1[p] = 0.f; }
+        #pragma unroll
+        for (int kp = 0; kp < KP; ++kp)
+            #pragma unroll
+            for (int p = 0; p < P; ++p) {
+                bf16x2_t a = __builtin_bit_cast(
+                    bf16x2_t, ac[p * (L / 2) + kp]);
+                // a = (A[2kp], A[2kp+1]); B pair rows give
+                // (B[2kp][v0], B[2kp][v0+1]) etc. — regroup:
+                bf16x2_t b0, b1;
+                b0[0] = bp[p][kp][0][0]; b0[1] = bp[p][kp][1][0];
+                b1[0] = bp[p][kp][0][1]; b1[1] = bp[p][kp][1][1];
+                acc0[p] = __builtin_amdgcn_fdot2_f32_bf16(
+                    a, b0, acc0[p], false);
+                acc1[p] = __builtin_amdgcn_fdot2_f32_bf16(
+                    a, b1, acc1[p], false);
+            }
+        bf16_t* dst = zOut + ((c0 + c) * zstride + s * (ll)P) * VB + v0;
+        #pragma unroll
+        for (int p = 0; p < P; ++p) {
+            bf16x2_t z2;
+            z2[0] = (__bf16)acc0[p];
+            z2[1] = (__bf16)acc1[p];
+            *(bf16x2_t*)&dst[(size_t)p * VB] = z2;
+        }
+    }
+}
+
 // raw-r variant: dot3s with the normalize deferred (bf16 Z only)
 extern "C" void launch_fcma_corr_raw(const void* At, const void* B,
                                      void* zOut, ll E, ll L, ll VB,
@@ -2029,13 +2109,21 @@ extern "C" void launch_fcma_corr_raw(const void* At, const void* B,
     static int vpt = -1;
     if (vpt < 0) {
         const char* e = getenv("BRAINIAK_CORR_VPT");
-        vpt = (e && atoi(e) == 2) ? 2 : 1;
+        vpt = 1;
+        if (e && atoi(e) == 2) vpt = 2;
+        if (e && atoi(e) == 3) vpt = 3;   // adjacent-pair variant
     }
     ll grid3 = ceil_div(C, 128) * nSubj
-             * ceil_div(VB, (ll)(vpt * C3_VT));
+             * ceil_div(VB, (ll)((vpt > 1 ? 2 : 1) * C3_VT));
     #define RAW_CASE(TP, TL)                                             \
         do {                                                             \
-            if (vpt == 2)                                                \
+            if (vpt == 3)                                                \
+                hipLaunchKernelGGL((k_corr_raw_pair<TP, TL, 128>),       \
+                                   dim3(grid3), dim3(256), 0, stream,    \
+                                   (const bf16_t*)At,                    \
+                                   (const bf16_t*)B, (bf16_t*)zOut, E,   \
+                                   VB, C, zstride);                      \
+            else if (vpt == 2)                                           \
                 hipLaunchKernelGGL((k_corr_raw_v2<TP, TL, 128>),         \
                                    dim3(grid3), dim3(256), 0, stream,    \
                                    (const bf16_t*)At,                    \
