@@ -16,6 +16,7 @@ K*(n_dim+1) doubles — latency-, not bandwidth-bound).
 import logging
 
 import numpy as np
+import torch
 from scipy.optimize import linear_sum_assignment
 from scipy.spatial import distance
 
@@ -213,9 +214,23 @@ class HTFA(TFA):
         del local_weight_offset, n_local_subj
         posts = self.local_posterior_.reshape(-1, self.prior_size)
         pieces = []
+        on_dev = self._use_gpu() and R and R[0].shape[1] == 3
         for subj_data, coords, post in zip(data, R, posts):
             centers = self.get_centers(post)
             widths = self.get_widths(post)
+            if on_dev:
+                # factor matrix, ridge beta and weight solve all stay
+                # on device: ONE [V, T] upload and one [K, T] download
+                # per subject (the numpy path re-derived np.var and
+                # round-tripped F through the host)
+                Ft = self._get_factors_dev(
+                    np.ascontiguousarray(coords, dtype=np.float64),
+                    centers, widths)
+                Xt = torch.as_tensor(subj_data, dtype=torch.float32,
+                                     device="cuda")
+                W = self._get_weights_dev(Xt, Ft)
+                pieces.append(W.double().cpu().numpy().ravel())
+                continue
             unique_R, inds = self.get_unique_R(coords)
             F = self.get_factors(unique_R, inds, centers, widths)
             pieces.append(self.get_weights(subj_data, F).ravel())

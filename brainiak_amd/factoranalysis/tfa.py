@@ -203,31 +203,33 @@ class TFA:
             [unique_R[d][inds[d]] for d in range(self.n_dim)]).astype(
                 np.float64)
         if self._use_gpu() and self.n_dim == 3:
-            F = ops.tfa_factor(
-                torch.as_tensor(centers, dtype=torch.float32,
-                                device="cuda"),
-                torch.as_tensor(widths.ravel(), dtype=torch.float32,
-                                device="cuda"),
-                torch.as_tensor(coords, dtype=torch.float32,
-                                device="cuda"))
-            return F.double().cpu().numpy()
+            return self._get_factors_dev(coords, centers,
+                                         widths).double().cpu().numpy()
         d2 = distance.cdist(coords, centers, 'sqeuclidean')
         return np.exp(-d2 / widths.ravel()[None, :])
+
+    def _get_factors_dev(self, coords, centers, widths):
+        """Device-resident factor matrix (fp32 cuda) — lets callers
+        chain into the weight solve without a host round trip."""
+        return ops.tfa_factor(
+            torch.as_tensor(centers, dtype=torch.float32,
+                            device="cuda"),
+            torch.as_tensor(widths.ravel(), dtype=torch.float32,
+                            device="cuda"),
+            torch.as_tensor(coords, dtype=torch.float32,
+                            device="cuda"))
 
     def get_weights(self, data, F):
         """Ridge ('rr') or OLS weights W [K, n_tr] (device GEMMs when
         a GPU is available — the [V, K] normal-equation products are
         the only O(V) work here)."""
-        beta = np.var(data)
         if self._use_gpu():
             Ft = torch.as_tensor(F, dtype=torch.float32, device="cuda")
             Xt = torch.as_tensor(data, dtype=torch.float32,
                                  device="cuda")
-            G = Ft.T @ Ft
-            if self.weight_method == 'rr':
-                G = G + float(beta) * torch.eye(self.K, device="cuda")
-            W = torch.linalg.solve(G, Ft.T @ Xt)
-            return W.double().cpu().numpy()
+            return self._get_weights_dev(
+                Xt, Ft).double().cpu().numpy()
+        beta = np.var(data)
         trans_F = F.T.copy()
         if self.weight_method == 'rr':
             W = np.linalg.solve(trans_F.dot(F)
@@ -236,6 +238,17 @@ class TFA:
         else:
             W = np.linalg.solve(trans_F.dot(F), trans_F.dot(data))
         return W
+
+    def _get_weights_dev(self, Xt, Ft):
+        """Ridge/OLS weight solve entirely on device.  The ridge beta
+        (= var of the data, ddof 0, matching np.var) is computed from
+        the already-uploaded tensor — the CPU np.var over the full
+        [V, T] array was half of HTFA's _update_weight wall."""
+        G = Ft.T @ Ft
+        if self.weight_method == 'rr':
+            beta = Xt.double().var(correction=0)
+            G = G + beta.float() * torch.eye(self.K, device=Ft.device)
+        return torch.linalg.solve(G, Ft.T @ Xt)
 
     def _recon_err(self, X, F, W, data_sigma):
         """Scaled flattened residual data_sigma*(X - F·W) (N9)."""
