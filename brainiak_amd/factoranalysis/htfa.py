@@ -226,8 +226,7 @@ class HTFA(TFA):
                 Ft = self._get_factors_dev(
                     np.ascontiguousarray(coords, dtype=np.float64),
                     centers, widths)
-                Xt = torch.as_tensor(subj_data, dtype=torch.float32,
-                                     device="cuda")
+                Xt = self._to_dev_f32(subj_data)
                 W = self._get_weights_dev(Xt, Ft)
                 pieces.append(W.double().cpu().numpy().ravel())
                 continue

@@ -208,25 +208,30 @@ class TFA:
         d2 = distance.cdist(coords, centers, 'sqeuclidean')
         return np.exp(-d2 / widths.ravel()[None, :])
 
+    @staticmethod
+    def _to_dev_f32(a):
+        """numpy -> fp32 cuda with the dtype cast on DEVICE: as_tensor
+        with dtype=float32 converts fp64 arrays on the host first
+        (single-threaded, and doubles the work before an 80 MB/subject
+        upload — measured as the top cost of an HTFA step)."""
+        t = torch.from_numpy(np.ascontiguousarray(a))
+        return t.to("cuda", non_blocking=False).float()
+
     def _get_factors_dev(self, coords, centers, widths):
         """Device-resident factor matrix (fp32 cuda) — lets callers
         chain into the weight solve without a host round trip."""
         return ops.tfa_factor(
-            torch.as_tensor(centers, dtype=torch.float32,
-                            device="cuda"),
-            torch.as_tensor(widths.ravel(), dtype=torch.float32,
-                            device="cuda"),
-            torch.as_tensor(coords, dtype=torch.float32,
-                            device="cuda"))
+            self._to_dev_f32(np.asarray(centers)),
+            self._to_dev_f32(np.asarray(widths).ravel()),
+            self._to_dev_f32(coords))
 
     def get_weights(self, data, F):
         """Ridge ('rr') or OLS weights W [K, n_tr] (device GEMMs when
         a GPU is available — the [V, K] normal-equation products are
         the only O(V) work here)."""
         if self._use_gpu():
-            Ft = torch.as_tensor(F, dtype=torch.float32, device="cuda")
-            Xt = torch.as_tensor(data, dtype=torch.float32,
-                                 device="cuda")
+            Ft = self._to_dev_f32(F)
+            Xt = self._to_dev_f32(data)
             return self._get_weights_dev(
                 Xt, Ft).double().cpu().numpy()
         beta = np.var(data)
@@ -379,8 +384,8 @@ class TFA:
         coords = np.column_stack(
             [unique_R[d][inds[d]] for d in range(D)]).astype(np.float32)
         co = torch.as_tensor(coords, device=dev)
-        Xt = torch.as_tensor(X, dtype=torch.float32, device=dev)
-        Wt = torch.as_tensor(W, dtype=torch.float32, device=dev)
+        Xt = self._to_dev_f32(X)
+        Wt = self._to_dev_f32(W)
         lb = torch.as_tensor(self.bounds[0], dtype=torch.float32,
                              device=dev)
         ub = torch.as_tensor(self.bounds[1], dtype=torch.float32,
@@ -466,15 +471,23 @@ class TFA:
             Jw = J * wgt[:, None]
             A = Jw.T @ Jw                                 # [n, n] GEMM
             g = Jw.T @ rw_
+            # the [n_par, n_par] damped solve runs on the HOST: n_par
+            # is ~80, and a hipSOLVER call at that size costs ~1.2 ms
+            # of launch/sync overhead per damping try (1500+ tries per
+            # HTFA fit) vs ~50 us in fp64 numpy after a 26 KB download
+            A_h = A.cpu().numpy().astype(np.float64)
+            g_h = g.cpu().numpy().astype(np.float64)
+            dA_h = np.clip(np.diag(A_h), 1e-12, None)
             improved = False
             for _try in range(6):
-                damp = A + lam * torch.diag(torch.diagonal(A)
-                                            .clamp_min(1e-12))
+                damp = A_h + lam * np.diag(dA_h)
                 try:
-                    delta = torch.linalg.solve(damp, -g)
-                except RuntimeError:
+                    delta_h = np.linalg.solve(damp, -g_h)
+                except np.linalg.LinAlgError:
                     lam *= 10
                     continue
+                delta = torch.as_tensor(delta_h, dtype=torch.float32,
+                                        device=dev)
                 cand = (theta + delta).clamp(lb, ub)
                 rc_, _ = residual_and_jac(cand, want_jac=False)
                 c2 = soft_l1_cost(rc_)
