@@ -228,7 +228,7 @@ class HTFA(TFA):
                     centers, widths)
                 Xt = self._to_dev_f32(subj_data)
                 W = self._get_weights_dev(Xt, Ft)
-                pieces.append(W.double().cpu().numpy().ravel())
+                pieces.append(W.ravel())
                 continue
             unique_R, inds = self.get_unique_R(coords)
             F = self.get_factors(unique_R, inds, centers, widths)

@@ -232,8 +232,7 @@ class TFA:
         if self._use_gpu():
             Ft = self._to_dev_f32(F)
             Xt = self._to_dev_f32(data)
-            return self._get_weights_dev(
-                Xt, Ft).double().cpu().numpy()
+            return self._get_weights_dev(Xt, Ft)
         beta = np.var(data)
         trans_F = F.T.copy()
         if self.weight_method == 'rr':
@@ -245,15 +244,22 @@ class TFA:
         return W
 
     def _get_weights_dev(self, Xt, Ft):
-        """Ridge/OLS weight solve entirely on device.  The ridge beta
-        (= var of the data, ddof 0, matching np.var) is computed from
-        the already-uploaded tensor — the CPU np.var over the full
-        [V, T] array was half of HTFA's _update_weight wall."""
+        """Ridge/OLS weight solve: the O(V) normal-equation GEMMs run
+        on device; the [K, K] solve runs on the HOST in fp64 (at
+        K~20-50 a hipSOLVER call is ~1.2 ms of launch/sync overhead vs
+        ~50 us in numpy, and every caller downloads W anyway).  The
+        ridge beta (= var of the data, ddof 0, matching np.var) comes
+        from the already-uploaded tensor — the CPU np.var over the
+        full [V, T] array was half of HTFA's _update_weight wall.
+        Returns W as a float64 numpy array."""
         G = Ft.T @ Ft
+        B = Ft.T @ Xt
         if self.weight_method == 'rr':
             beta = Xt.double().var(correction=0)
             G = G + beta.float() * torch.eye(self.K, device=Ft.device)
-        return torch.linalg.solve(G, Ft.T @ Xt)
+        G_h = G.cpu().numpy().astype(np.float64)
+        B_h = B.cpu().numpy().astype(np.float64)
+        return np.linalg.solve(G_h, B_h)
 
     def _recon_err(self, X, F, W, data_sigma):
         """Scaled flattened residual data_sigma*(X - F·W) (N9)."""
