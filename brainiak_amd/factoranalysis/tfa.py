@@ -132,7 +132,7 @@ class TFA:
         """K-means centers + max-sigma widths.
 
         The clustering only SEEDS the NLSS (centers are refined every
-        iteration), so it runs single-init on a <=20k-voxel subsample —
+        iteration), so it runs single-init on a <=8k-voxel subsample —
         full-brain 10-restart Lloyd was 32 % of an entire HTFA fit
         (profiles/htfa_cprofile.txt) for identical end fits."""
         from sklearn.cluster import KMeans
