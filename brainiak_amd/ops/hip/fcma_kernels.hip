@@ -660,21 +660,12 @@ __device__ __forceinline__ void dot3s_body(
         if (RAW) {
             // raw-correlation store: Fisher-z + z-score are deferred
             // to the Gram kernel's staging phase (whose waves sit in
-            // memory-latency shadow) — this kernel is VALU-issue bound.
-            // Nontemporal: Z streams through exactly once (one write
-            // here, one read in the Gram), so keep it out of L2.
+            // memory-latency shadow) — this kernel is VALU-issue bound
             OT* dst = zOut
                 + ((c0 + c) * zstride + s * (ll)P) * VB + v;
             #pragma unroll
-            for (int p = 0; p < P; ++p) {
-                OT zv = z_cast<OT>(acc[p]);
-                if constexpr (sizeof(OT) == 2)
-                    __builtin_nontemporal_store(
-                        __builtin_bit_cast(unsigned short, zv),
-                        (unsigned short*)&dst[(size_t)p * VB]);
-                else
-                    dst[(size_t)p * VB] = zv;
-            }
+            for (int p = 0; p < P; ++p)
+                dst[(size_t)p * VB] = z_cast<OT>(acc[p]);
             continue;
         }
         float z[P];
@@ -1268,9 +1259,7 @@ __device__ __forceinline__ void gram_bf16_body(
             ll kk = k0 + scol + ((NP > 0) ? 0 : 8 * h);
             const bf16_t* s = src + (rows0 + srow + r) * V + kk;
             if (kk + 8 <= V && (((uintptr_t)s) & 3) == 0) {
-                // nontemporal: each Z element is read exactly once
-                regs[h] = (bf16x8)__builtin_nontemporal_load(
-                    (const bf16x8_u*)s);
+                regs[h] = (bf16x8)(*(const bf16x8_u*)s);
             } else {
                 bf16_t tmp[8];
                 #pragma unroll
@@ -1502,8 +1491,7 @@ __global__ __launch_bounds__(256) void k_gram_fp8(
             ll kk = k0 + scol + 16 * h;
             const fp8_t* sp = src + (rows0 + srow) * V + kk;
             if (kk + 16 <= V) {
-                regs[h] = __builtin_nontemporal_load(
-                    (const uint32x4*)sp);
+                regs[h] = *(const uint32x4*)sp;
             } else {
                 regs[h] = (uint32x4)0u;   // V is 16-padded: whole
             }                             // vector in or out of range
@@ -2107,9 +2095,7 @@ __global__ __launch_bounds__(256) void k_corr_raw_pair(
             bf16x2_t z2;
             z2[0] = (__bf16)acc0[p];
             z2[1] = (__bf16)acc1[p];
-            __builtin_nontemporal_store(
-                __builtin_bit_cast(unsigned int, z2),
-                (unsigned int*)&dst[(size_t)p * VB]);
+            *(bf16x2_t*)&dst[(size_t)p * VB] = z2;
         }
     }
 }
