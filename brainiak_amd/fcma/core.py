@@ -333,12 +333,20 @@ class CorrelationPipeline:
             _shrink_(gram)
         return gram
 
-    def _duo_pipeline(self, chunks, shrink=True):
+    def _duo_pipeline(self, chunks, shrink=True, consumer=None):
         """Single-stream duo launches: each kernel carries the raw-corr
         blocks of chunk i AND the Gram(+normalize) blocks of chunk i-1
         in one grid — HIP streams do not co-schedule the two kernels
         (measured exactly serial), a shared grid forces CU-level
-        co-residency of the VALU-bound and latency-bound waves."""
+        co-residency of the VALU-bound and latency-bound waves.
+
+        consumer, if given, is called as consumer(g, start, count)
+        right after each chunk's [count, E, E] Gram is enqueued (in
+        chunk order, stream-ordered on the current stream) and the
+        grams are NOT accumulated — the pipeline returns None.  This
+        is how the voxel selector overlaps the per-chunk SVM CV with
+        the remaining duo sweep.  shrink is left to the consumer in
+        that mode."""
         ext = ops.load_extension()
         E = self.num_epochs
         Epad = ((E + 63) // 64) * 64
@@ -375,13 +383,19 @@ class CorrelationPipeline:
                 g = gp.sum(0)
                 if Epad != E:
                     g = g[:, :E, :E].contiguous()
-                grams.append(g)
+                if consumer is not None:
+                    consumer(g, *chunks[i - 1])
+                else:
+                    grams.append(g)
             prev = (z_i.narrow(0, 0, count), count)
         # trailing gram for the last chunk
         zp, cp = prev
         g = ops.fcma_gram_bf16(zp, norm_P=P)
         if Epad != E:
             g = g[:, :E, :E].contiguous()
+        if consumer is not None:
+            consumer(g, *chunks[-1])
+            return None
         grams.append(g)
         gram = torch.cat(grams, dim=0)
         if shrink:

@@ -106,37 +106,61 @@ def smo_batch_train(K: torch.Tensor, y: torch.Tensor, C: float = 1.0,
     return alpha, b
 
 
+class FoldPlan:
+    """Device-resident stratified-fold index tables, built once and
+    reused for every chunk's batched-SMO launch (the per-call rebuild
+    plus the max-fold-size .item() sync would otherwise serialize the
+    overlapped CV stream against the duo sweep)."""
+
+    def __init__(self, labels: np.ndarray, num_folds: int, device):
+        labels = np.asarray(labels)
+        E = len(labels)
+        classes = np.unique(labels)
+        if len(classes) != 2:
+            raise ValueError("GPU batched SVM supports binary labels; "
+                             f"got {len(classes)} classes")
+        y_np = np.where(labels == classes[1], 1.0,
+                        -1.0).astype(np.float32)
+        folds = stratified_folds(labels, num_folds)
+        F = len(folds)
+        train_idx = np.zeros((F, E), dtype=np.int32)
+        test_idx = np.zeros((F, E), dtype=np.int32)
+        n_train = np.zeros(F, dtype=np.int32)
+        n_test = np.zeros(F, dtype=np.int32)
+        for f, (tr, te) in enumerate(folds):
+            train_idx[f, :len(tr)] = tr
+            test_idx[f, :len(te)] = te
+            n_train[f] = len(tr)
+            n_test[f] = len(te)
+        self.max_n = int(max(n_train.max(), n_test.max()))
+        self.y = torch.as_tensor(y_np, device=device)
+        self.train_idx = torch.as_tensor(train_idx, device=device)
+        self.test_idx = torch.as_tensor(test_idx, device=device)
+        self.n_train = torch.as_tensor(n_train, device=device)
+        self.n_test = torch.as_tensor(n_test, device=device)
+        self.n_test_f = self.n_test.to(torch.float32)
+
+
+def svm_cv_device(kernels: torch.Tensor, plan: FoldPlan, C: float,
+                  tol: float) -> torch.Tensor:
+    """Batched-SMO CV on the CURRENT stream, returning the per-voxel
+    mean accuracy as a DEVICE tensor — no host sync, so it can be
+    enqueued on a side stream while the duo sweep continues."""
+    from .. import ops
+    correct = ops.svm_cv(
+        kernels.to(torch.float32).contiguous(),
+        plan.y, plan.train_idx, plan.test_idx,
+        plan.n_train, plan.n_test, C=C, tol=tol,
+        max_n=plan.max_n)
+    return (correct.to(torch.float32) / plan.n_test_f).mean(dim=1)
+
+
 def _accuracy_gpu_hip(kernels: torch.Tensor, labels: np.ndarray,
                       num_folds: int, C: float, tol: float) -> np.ndarray:
     """Whole-CV-in-one-launch path: one wavefront per (voxel, fold) QP
     (ops.svm_cv HIP kernel)."""
-    from .. import ops
-    device = kernels.device
-    n_vox, E, _ = kernels.shape
-    classes = np.unique(labels)
-    y_np = np.where(labels == classes[1], 1.0, -1.0).astype(np.float32)
-    folds = stratified_folds(labels, num_folds)
-    F = len(folds)
-    train_idx = np.zeros((F, E), dtype=np.int32)
-    test_idx = np.zeros((F, E), dtype=np.int32)
-    n_train = np.zeros(F, dtype=np.int32)
-    n_test = np.zeros(F, dtype=np.int32)
-    for f, (tr, te) in enumerate(folds):
-        train_idx[f, :len(tr)] = tr
-        test_idx[f, :len(te)] = te
-        n_train[f] = len(tr)
-        n_test[f] = len(te)
-    correct = ops.svm_cv(
-        kernels.to(torch.float32).contiguous(),
-        torch.as_tensor(y_np, device=device),
-        torch.as_tensor(train_idx, device=device),
-        torch.as_tensor(test_idx, device=device),
-        torch.as_tensor(n_train, device=device),
-        torch.as_tensor(n_test, device=device),
-        C=C, tol=tol)
-    acc_per_fold = correct.to(torch.float32) / \
-        torch.as_tensor(n_test, device=device, dtype=torch.float32)
-    return acc_per_fold.mean(dim=1).cpu().numpy()
+    plan = FoldPlan(labels, num_folds, kernels.device)
+    return svm_cv_device(kernels, plan, C, tol).cpu().numpy()
 
 
 def _accuracy_gpu(kernels: torch.Tensor, labels: np.ndarray, num_folds: int,

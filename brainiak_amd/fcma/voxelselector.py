@@ -16,6 +16,7 @@ latency hiding pointless; removing the master frees a GPU.
 """
 
 import logging
+import os
 from typing import Optional
 
 import numpy as np
@@ -116,13 +117,18 @@ class VoxelSelector:
                 count = min(self.voxel_unit, my.stop - start)
                 chunks.append((start, count))
                 start += count
-            kernels = pipeline.pipelined_kernel_matrices(chunks)
+            scores = None
             if use_gpu_svm:
-                scores = cross_validate_voxels(kernels, self.labels,
-                                               self.num_folds, C=C,
-                                               tol=tol)
-            else:
-                scores = self._sklearn_cv(clf, kernels.cpu().numpy())
+                scores = self._overlapped_cv(pipeline, chunks, C, tol)
+            if scores is None:
+                kernels = pipeline.pipelined_kernel_matrices(chunks)
+                if use_gpu_svm:
+                    scores = cross_validate_voxels(kernels, self.labels,
+                                                   self.num_folds, C=C,
+                                                   tol=tol)
+                else:
+                    scores = self._sklearn_cv(clf,
+                                              kernels.cpu().numpy())
             results.extend((my.start + i, float(scores[i]))
                            for i in range(my.stop - my.start))
         else:
@@ -143,6 +149,48 @@ class VoxelSelector:
             results = [r for part in gathered for r in part]
         results.sort(key=lambda t: (-t[1], t[0]))
         return results
+
+    def _overlapped_cv(self, pipeline, chunks, C, tol):
+        """Duo sweep with the per-chunk SVM CV enqueued on a SIDE
+        stream: the SMO grid for chunk i (512 x folds wavefronts, tiny
+        next to the duo grid) fills CU slots as the duo kernel for
+        chunk i+1 drains, instead of paying the whole CV serially
+        after the sweep (~6 % of a whole-brain pass).  Accuracy stays
+        on device until one download at the end.  Returns None when
+        the duo path or the fold sizes do not apply, and the caller
+        falls back to the batched post-pass CV."""
+        if (not pipeline._hip_path or not pipeline._raw_split
+                or len(chunks) <= 1
+                or os.environ.get("BRAINIAK_NO_DUO")
+                or os.environ.get("BRAINIAK_NO_CV_OVERLAP")):
+            return None
+        from .core import _shrink_
+        from .svm import FoldPlan, svm_cv_device
+        try:
+            plan = FoldPlan(self.labels, self.num_folds,
+                            pipeline.device)
+        except ValueError:
+            return None
+        if plan.max_n > 128:
+            return None
+        dev = pipeline.device
+        cv_stream = torch.cuda.Stream(device=dev)
+        accs = []
+
+        def consume(g, start, count):
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream(dev))
+            with torch.cuda.stream(cv_stream):
+                cv_stream.wait_event(ev)
+                _shrink_(g)
+                accs.append(svm_cv_device(g, plan, C, tol))
+                # g was allocated on the main stream; keep the
+                # allocator from recycling it under the CV reads
+                g.record_stream(cv_stream)
+
+        pipeline._duo_pipeline(chunks, consumer=consume)
+        torch.cuda.current_stream(dev).wait_stream(cv_stream)
+        return torch.cat(accs).cpu().numpy()
 
     def _score_chunk(self, pipeline, clf, start, count, precomputed, C,
                      tol, use_gpu_svm):

@@ -158,14 +158,17 @@ def fcma_fused_gram(data: torch.Tensor, data2: torch.Tensor, start: int,
 
 def svm_cv(kernels: torch.Tensor, y: torch.Tensor, train_idx, test_idx,
            n_train, n_test, C: float = 1.0, tol: float = 1e-3,
-           max_iter: int = 10000) -> torch.Tensor:
+           max_iter: int = 10000, max_n: int = -1) -> torch.Tensor:
     """Batched precomputed-kernel SVC k-fold CV fully on device.
 
     One wavefront per (voxel, fold) dual QP (SMO, WSS1); returns
     correct-prediction counts [C, F] int32.  n_train/n_test <= 64.
+    max_n > 0 skips the fold-size .item() sync (pass
+    max(n_train.max(), n_test.max()) computed on the host) so the
+    launch can be enqueued on a side stream without draining it.
     """
     return _ext().svm_cv(kernels, y, train_idx, test_idx, n_train, n_test,
-                         float(C), float(tol), int(max_iter))
+                         float(C), float(tol), int(max_iter), int(max_n))
 
 
 def jacobi_eigh(G: torch.Tensor):

@@ -500,3 +500,37 @@ def test_duo_pipeline_matches_streamed(ops):
     assert g_duo.shape == g_str.shape == (V, E, E)
     assert torch.allclose(g_duo.float(), g_str.float(), atol=1e-3,
                           rtol=1e-4)
+
+
+def test_overlapped_cv_matches_post_pass(ops):
+    """VoxelSelector's side-stream per-chunk CV == the single batched
+    post-pass CV (same kernels, same SMO; only the scheduling and the
+    per-chunk shrink ordering differ)."""
+    import os
+
+    import numpy as np
+
+    from brainiak_amd.fcma.voxelselector import VoxelSelector
+    rng = np.random.RandomState(17)
+    E, L, V, P = 32, 12, 640, 4
+    raw = []
+    for _ in range(E):
+        m = rng.randn(L, V).astype(np.float32)
+        m = (m - m.mean(0)) / m.std(0)
+        raw.append((m / np.sqrt(L)).astype(np.float32))
+    labels = np.tile([0, 1], E // 2)
+    sel = VoxelSelector(labels, P, 4, raw, voxel_unit=256,
+                        device="cuda")
+    from sklearn import svm
+    clf = svm.SVC(kernel='precomputed', shrinking=False, C=1.0)
+    res_overlap = sel.run(clf)
+    os.environ["BRAINIAK_NO_CV_OVERLAP"] = "1"
+    try:
+        res_post = VoxelSelector(labels, P, 4, raw, voxel_unit=256,
+                                 device="cuda").run(clf)
+    finally:
+        del os.environ["BRAINIAK_NO_CV_OVERLAP"]
+    assert len(res_overlap) == len(res_post) == V
+    a = dict(res_overlap)
+    b = dict(res_post)
+    assert all(abs(a[v] - b[v]) < 1e-6 for v in a)
