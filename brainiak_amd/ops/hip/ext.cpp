@@ -439,7 +439,8 @@ torch::Tensor tfa_recon(torch::Tensor X, torch::Tensor W, torch::Tensor F,
 torch::Tensor svm_cv(torch::Tensor kernels, torch::Tensor y,
                      torch::Tensor train_idx, torch::Tensor test_idx,
                      torch::Tensor n_train, torch::Tensor n_test,
-                     double Creg, double tol, int64_t max_iter) {
+                     double Creg, double tol, int64_t max_iter,
+                     int64_t max_n_hint) {
     check_3d(kernels, torch::kFloat32, "kernels");
     ll C = kernels.size(0);
     int E = (int)kernels.size(1);
@@ -453,8 +454,11 @@ torch::Tensor svm_cv(torch::Tensor kernels, torch::Tensor y,
                 "idx must be [F,E]");
     auto nt = n_train.to(torch::kInt32).contiguous();
     auto ns = n_test.to(torch::kInt32).contiguous();
-    ll max_n = std::max<ll>(nt.max().item<int>(),
-                            ns.max().item<int>());
+    // max_n_hint lets the caller skip the .item() device sync (needed
+    // when this launch is enqueued on a side stream mid-pipeline)
+    ll max_n = max_n_hint > 0
+        ? max_n_hint
+        : std::max<ll>(nt.max().item<int>(), ns.max().item<int>());
     TORCH_CHECK(max_n <= 128,
                 "svm_cv supports fold sizes up to 128 samples");
     auto correct = torch::zeros({C, F}, kernels.options()
