@@ -174,7 +174,8 @@ class VoxelSelector:
         if plan.max_n > 128:
             return None
         dev = pipeline.device
-        cv_stream = torch.cuda.Stream(device=dev)
+        prio = int(os.environ.get("BRAINIAK_CV_PRIO", "0"))
+        cv_stream = torch.cuda.Stream(device=dev, priority=prio)
         accs = []
 
         def consume(g, start, count):
