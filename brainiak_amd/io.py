@@ -1,6 +1,6 @@
 """Volume / label IO (API parity: ref src/brainiak/io.py:39-168).
 
-Backed by the self-contained NIfTI-1 implementation in
+Backed by the self-contained NIfTI reader/writer in
 ``brainiak_amd.nifti`` instead of nibabel.
 """
 
@@ -20,42 +20,42 @@ __all__ = [
     "save_as_nifti_file",
 ]
 
-
-def load_images_from_dir(in_dir: Union[str, Path], suffix: str = "nii.gz"):
-    """Lazily load all images in ``in_dir`` with ``suffix``, sorted by name."""
-    in_dir = Path(in_dir)
-    files = sorted(in_dir.glob("*" + suffix))
-    for f in files:
-        yield nifti.load(str(f))
+PathLike = Union[str, Path]
 
 
-def load_images(image_paths: Iterable[Union[str, Path]]):
-    """Lazily load images from an iterable of paths."""
-    for path in image_paths:
-        yield nifti.load(str(path))
+def _read_volume(source: PathLike):
+    return nifti.load(str(source))
 
 
-def load_boolean_mask(path: Union[str, Path],
+def load_images_from_dir(in_dir: PathLike, suffix: str = "nii.gz"):
+    """Lazily load every ``*suffix`` volume in ``in_dir`` (name order)."""
+    matches = sorted(p for p in Path(in_dir).iterdir()
+                     if p.name.endswith(suffix))
+    return load_images(matches)
+
+
+def load_images(image_paths: Iterable[PathLike]):
+    """Lazily load volumes from an iterable of paths."""
+    return map(_read_volume, image_paths)
+
+
+def load_boolean_mask(path: PathLike,
                       predicate: Callable[[np.ndarray], np.ndarray] = None
                       ) -> np.ndarray:
-    """Load a boolean mask volume; default predicate is truthiness."""
-    img = nifti.load(str(path))
-    data = img.get_fdata()
-    if predicate is not None:
-        mask = predicate(data)
-    else:
-        mask = data.astype(bool)
-    return mask
+    """Load a volume as a boolean mask; ``predicate`` (default:
+    truthiness) maps the voxel data to booleans."""
+    voxels = _read_volume(path).get_fdata()
+    keep = predicate(voxels) if predicate is not None else voxels != 0
+    return np.asarray(keep, dtype=bool)
 
 
-def load_labels(path: Union[str, Path]) -> List[SingleConditionSpec]:
-    """Load condition labels from an .npy file of one-hot epoch arrays."""
-    condition_specs = np.load(str(path))
-    return [c.view(SingleConditionSpec) for c in condition_specs]
+def load_labels(path: PathLike) -> List[SingleConditionSpec]:
+    """Condition labels: an .npy stack of one-hot epoch arrays."""
+    stack = np.load(str(path))
+    return [spec.view(SingleConditionSpec) for spec in stack]
 
 
 def save_as_nifti_file(data: np.ndarray, affine: np.ndarray,
-                       path: Union[str, Path]) -> None:
-    """Save an array + affine as a NIfTI-1 file."""
-    image = nifti.NiftiImage(data, affine)
-    nifti.save(image, str(path))
+                       path: PathLike) -> None:
+    """Write ``data`` with ``affine`` as a NIfTI-1 volume."""
+    nifti.save(nifti.NiftiImage(data, affine), str(path))
