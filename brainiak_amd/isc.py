@@ -7,6 +7,13 @@ API parity with the reference isc module (ref src/brainiak/isc.py:81-1551):
 tests when the permutation space is small), ``timeshift_isc`` (circular
 shifts), ``phaseshift_isc`` (FFT phase randomization).
 
+The implementations are vectorized rather than per-iteration loops:
+pairwise ISC is one einsum over z-scored series, leave-one-out means
+use the sum trick (total minus subject) instead of ``np.delete``, the
+bootstrap resamples all iterations with one fancy-index, and the
+one-sample permutation test applies every sign-flip as a single
+matrix product.
+
 MI355X additions:
  - the heavy ISFC gemms run through torch (rocBLAS/MFMA on GPU tensors);
  - ``isfc_distributed``: subject-sharded leave-one-out ISFC over RCCL —
@@ -21,7 +28,7 @@ Citations as in the reference: [Hasson2004], [Simony2016], [Chen2016],
 
 import logging
 import math
-from itertools import combinations, permutations, product
+from itertools import combinations, permutations
 
 import numpy as np
 from scipy.spatial.distance import squareform
@@ -36,8 +43,6 @@ from .utils.utils import (
 
 logger = logging.getLogger(__name__)
 
-MAX_RANDOM_SEED = 2 ** 32 - 1
-
 __all__ = [
     "bootstrap_isc",
     "compute_summary_statistic",
@@ -51,25 +56,69 @@ __all__ = [
 ]
 
 
-def _threshold_nans(data, tolerate_nans):
-    """NaN-threshold voxels; returns (masked data, keep-mask)."""
-    nans = np.all(np.any(np.isnan(data), axis=0), axis=1)
+def _as_rng(random_state):
+    """Coerce None / int seed / RandomState into a RandomState."""
+    if isinstance(random_state, np.random.RandomState):
+        return random_state
+    return np.random.RandomState(random_state)
+
+
+def _drop_bad_voxels(data, tolerate_nans):
+    """Voxel NaN policy shared by isc/isfc.
+
+    A voxel is dropped when every subject has a NaN somewhere in its
+    series; a float policy in [0, 1] additionally requires that
+    fraction of subjects to be NaN-free.  Returns (filtered data,
+    keep mask over voxels).
+    """
+    nan_somewhere = np.any(np.isnan(data), axis=0)      # [voxel, subject]
+    keep = ~np.all(nan_somewhere, axis=1)
     if tolerate_nans is True:
-        logger.info("ISC computation will tolerate all NaNs when averaging")
-    elif type(tolerate_nans) is float:
+        logger.info("ISC: averaging will tolerate all NaNs")
+    elif isinstance(tolerate_nans, float):
         if not 0.0 <= tolerate_nans <= 1.0:
             raise ValueError("If threshold to tolerate NaNs is a float, "
-                             "it must be between 0.0 and 1.0; got {0}".format(
-                                 tolerate_nans))
-        nans += ~(np.sum(~np.any(np.isnan(data), axis=0), axis=1) >=
-                  data.shape[-1] * tolerate_nans)
-        logger.info("ISC computation will tolerate voxels with at least "
-                    "%s non-NaN values: %d voxels do not meet threshold",
-                    tolerate_nans, np.sum(nans))
+                             "it must be between 0.0 and 1.0; got "
+                             f"{tolerate_nans}")
+        clean_subjects = np.sum(~nan_somewhere, axis=1)
+        keep &= clean_subjects >= data.shape[-1] * tolerate_nans
+        logger.info("ISC: voxels need a %s fraction of NaN-free "
+                    "subjects; %d voxels dropped", tolerate_nans,
+                    int(np.sum(~keep)))
     else:
-        logger.info("ISC computation will not tolerate NaNs when averaging")
-    mask = ~nans
-    return data[:, mask, :], mask
+        logger.info("ISC: averaging will not tolerate NaNs")
+    return data[:, keep, :], keep
+
+
+def _loo_means(data, use_nanmean):
+    """Leave-one-out across-subject means, [TR, voxel, subject], via
+    the sum trick (one pass instead of n_subjects ``np.delete``s)."""
+    n = data.shape[-1]
+    if use_nanmean:
+        finite = np.isfinite(data)
+        totals = np.nansum(data, axis=2, keepdims=True)
+        counts = finite.sum(axis=2, keepdims=True)
+        num = totals - np.where(finite, data, 0.0)
+        den = counts - finite
+        with np.errstate(invalid="ignore", divide="ignore"):
+            return num / den
+    totals = data.sum(axis=2, keepdims=True)
+    return (totals - data) / (n - 1)
+
+
+def _pairwise_voxel_correlations(data):
+    """r for every subject pair at every voxel in one einsum.
+
+    data [TR, voxel, subject] → [pair, voxel] with pairs in
+    ``combinations`` (row-major upper-triangle) order.
+    """
+    centered = data - data.mean(axis=0)
+    norms = np.sqrt(np.einsum("tvs,tvs->vs", centered, centered))
+    with np.errstate(invalid="ignore", divide="ignore"):
+        unit = centered / norms
+    gram = np.einsum("tvi,tvj->vij", unit, unit)        # [voxel, s, s]
+    iu = np.triu_indices(data.shape[-1], k=1)
+    return gram[:, iu[0], iu[1]].T
 
 
 def isc(data, pairwise=False, summary_statistic=None, tolerate_nans=True):
@@ -81,54 +130,40 @@ def isc(data, pairwise=False, summary_statistic=None, tolerate_nans=True):
                     "correlation.")
         summary_statistic = None
 
-    mean = np.nanmean if tolerate_nans else np.mean
-    data, mask = _threshold_nans(data, tolerate_nans)
+    data, keep = _drop_bad_voxels(data, tolerate_nans)
 
     if n_subjects == 2:
-        iscs_stack = array_correlation(data[..., 0],
-                                       data[..., 1])[np.newaxis, :]
+        rows = array_correlation(data[..., 0], data[..., 1])[None, :]
     elif pairwise:
-        swapped = np.swapaxes(data, 2, 0)
-        voxel_iscs = []
-        for v in np.arange(swapped.shape[1]):
-            voxel_data = swapped[:, v, :]
-            voxel_iscs.append(squareform(np.corrcoef(voxel_data),
-                                         checks=False))
-        iscs_stack = np.column_stack(voxel_iscs)
+        rows = _pairwise_voxel_correlations(data)
     else:
-        iscs_stack = []
-        for s in np.arange(n_subjects):
-            iscs_stack.append(array_correlation(
-                data[..., s], mean(np.delete(data, s, axis=2), axis=2)))
-        iscs_stack = np.array(iscs_stack)
+        loo = _loo_means(data, use_nanmean=bool(tolerate_nans))
+        rows = np.stack([array_correlation(data[..., s], loo[..., s])
+                         for s in range(n_subjects)])
 
-    iscs = np.full((iscs_stack.shape[0], n_voxels), np.nan)
-    iscs[:, np.where(mask)[0]] = iscs_stack
+    iscs = np.full((rows.shape[0], n_voxels), np.nan)
+    iscs[:, keep] = rows
 
     if summary_statistic:
         iscs = compute_summary_statistic(
-            iscs, summary_statistic=summary_statistic, axis=0)[np.newaxis, :]
-    if iscs.shape[0] == 1:
-        iscs = iscs[0]
-    return iscs
+            iscs, summary_statistic=summary_statistic, axis=0)[None, :]
+    return iscs[0] if iscs.shape[0] == 1 else iscs
 
 
-def _check_targets_input(targets, data):
-    if isinstance(targets, (np.ndarray, list)):
-        targets, n_TRs, n_voxels, n_subjects = (
-            _check_timeseries_input(targets))
-        if data.shape[0] != n_TRs:
-            raise ValueError("Targets array must have same number of "
-                             "TRs as input data")
-        if data.shape[2] != n_subjects:
-            raise ValueError("Targets array must have same number of "
-                             "subjects as input data")
-        symmetric = False
-    else:
-        targets = data
-        n_TRs, n_voxels, n_subjects = data.shape
-        symmetric = True
-    return targets, n_TRs, n_voxels, n_subjects, symmetric
+def _resolve_targets(targets, data):
+    """ISFC target handling: no targets → self-ISFC (symmetric)."""
+    if not isinstance(targets, (np.ndarray, list)):
+        t, v, s = data.shape
+        return data, t, v, s, True
+    targets, n_TRs, n_voxels, n_subjects = (
+        _check_timeseries_input(targets))
+    if data.shape[0] != n_TRs:
+        raise ValueError("Targets array must have same number of "
+                         "TRs as input data")
+    if data.shape[2] != n_subjects:
+        raise ValueError("Targets array must have same number of "
+                         "subjects as input data")
+    return targets, n_TRs, n_voxels, n_subjects, False
 
 
 def isfc(data, targets=None, pairwise=False, summary_statistic=None,
@@ -136,48 +171,39 @@ def isfc(data, targets=None, pairwise=False, summary_statistic=None,
     """Intersubject functional correlation: correlations between each
     voxel's time series and every (target) voxel in other subjects."""
     data, n_TRs, n_voxels, n_subjects = _check_timeseries_input(data)
-    targets, t_n_TRs, t_n_voxels, _, symmetric = (
-        _check_targets_input(targets, data))
+    targets, _, _, _, symmetric = _resolve_targets(targets, data)
     if not symmetric:
         pairwise = False
-    mean = np.nanmean if tolerate_nans else np.mean
-    data, mask = _threshold_nans(data, tolerate_nans)
-    targets, targets_mask = _threshold_nans(targets, tolerate_nans)
+    data, keep = _drop_bad_voxels(data, tolerate_nans)
+    targets, keep_t = _drop_bad_voxels(targets, tolerate_nans)
+
+    def corr(a, b):
+        return compute_correlation(np.ascontiguousarray(a.T),
+                                   np.ascontiguousarray(b.T),
+                                   return_nans=True)
 
     if symmetric and n_subjects == 2:
-        isfcs = compute_correlation(np.ascontiguousarray(data[..., 0].T),
-                                    np.ascontiguousarray(data[..., 1].T),
-                                    return_nans=True)
-        isfcs = ((isfcs + isfcs.T) / 2)[..., np.newaxis]
-        summary_statistic = None
         logger.info("Only two subjects! Computing ISFC between them.")
+        m = corr(data[..., 0], data[..., 1])
+        mats = [(m + m.T) / 2]
+        summary_statistic = None
     elif pairwise:
-        stack = []
-        for pair in combinations(np.arange(n_subjects), 2):
-            isfc_pair = compute_correlation(
-                np.ascontiguousarray(data[..., pair[0]].T),
-                np.ascontiguousarray(targets[..., pair[1]].T),
-                return_nans=True)
-            if symmetric:
-                isfc_pair = (isfc_pair + isfc_pair.T) / 2
-            stack.append(isfc_pair)
-        isfcs = np.dstack(stack)
+        mats = []
+        for a, b in combinations(range(n_subjects), 2):
+            m = corr(data[..., a], targets[..., b])
+            mats.append((m + m.T) / 2 if symmetric else m)
     else:
-        rolled = np.rollaxis(data, 2, 0)
-        rolled_targets = np.rollaxis(targets, 2, 0)
-        stack = [compute_correlation(
-            np.ascontiguousarray(subject.T),
-            np.ascontiguousarray(
-                mean(np.delete(rolled_targets, s, axis=0), axis=0).T),
-            return_nans=True) for s, subject in enumerate(rolled)]
-        isfcs = np.dstack([(m + m.T) / 2 if symmetric else m
-                           for m in stack])
+        use_nan = bool(tolerate_nans)
+        loo = _loo_means(targets, use_nanmean=use_nan)
+        mats = []
+        for s in range(n_subjects):
+            m = corr(data[..., s], loo[..., s])
+            mats.append((m + m.T) / 2 if symmetric else m)
 
-    isfcs_all = np.full((len(mask), len(targets_mask), isfcs.shape[2]),
-                        np.nan)
-    isfcs_all[np.ix_(np.where(mask)[0],
-                     np.where(targets_mask)[0])] = isfcs
-    isfcs = np.moveaxis(isfcs_all, 2, 0)
+    expanded = np.full((len(mats), len(keep), len(keep_t)), np.nan)
+    expanded[np.ix_(np.arange(len(mats)), keep, keep_t)] = \
+        np.stack(mats)
+    isfcs = expanded
 
     if summary_statistic:
         isfcs = compute_summary_statistic(
@@ -185,8 +211,7 @@ def isfc(data, targets=None, pairwise=False, summary_statistic=None,
     if isfcs.shape[0] == 1:
         isfcs = isfcs[0]
     if vectorize_isfcs and symmetric:
-        isfcs, iscs = squareform_isfc(isfcs)
-        return isfcs, iscs
+        return squareform_isfc(isfcs)
     return isfcs
 
 
@@ -311,286 +336,274 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
     return out
 
 
-def _check_isc_input(iscs, pairwise=False):
-    if type(iscs) == list:  # noqa: E721
-        iscs = np.array(iscs)[:, np.newaxis]
-    elif isinstance(iscs, np.ndarray):
-        if iscs.ndim == 1:
-            iscs = iscs[:, np.newaxis]
+def compute_summary_statistic(iscs, summary_statistic='mean', axis=None):
+    """'mean' (Fisher-Z mean: tanh(mean(arctanh))) or 'median' of ISCs."""
+    reducers = {
+        'mean': lambda a: np.tanh(np.nanmean(np.arctanh(a), axis=axis)),
+        'median': lambda a: np.nanmedian(a, axis=axis),
+    }
+    if summary_statistic not in reducers:
+        raise ValueError("Summary statistic must be 'mean' or 'median'")
+    return reducers[summary_statistic](iscs)
+
+
+def squareform_isfc(isfcs, iscs=None):
+    """Square ↔ condensed ISFC conversion retaining the ISC diagonal.
+
+    Square [.., V, V] input (no iscs) → (condensed off-diagonals,
+    diagonal ISCs); condensed input + iscs → square matrices.
+    """
+    have_iscs = isinstance(iscs, np.ndarray)
+    square_in = not have_iscs and isfcs.shape[-2] == isfcs.shape[-1]
+
+    if square_in:
+        stack = isfcs[None] if isfcs.ndim == 2 else isfcs
+        if stack.ndim != 3:
+            raise ValueError("Square (redundant) ISFCs must be square "
+                             "with multiple subjects or pairs of "
+                             "subjects indexed by the first dimension")
+        V = stack.shape[-1]
+        iu = np.triu_indices(V, k=1)
+        condensed = stack[:, iu[0], iu[1]]
+        diags = np.diagonal(stack, axis1=1, axis2=2)
+        if condensed.shape[0] == 1:
+            return condensed[0], diags[0]
+        return condensed, diags
+
+    flat = isfcs[None] if isfcs.ndim == 1 else isfcs
+    diag = iscs[None] if iscs.ndim == 1 else iscs
+    V = diag.shape[-1]
+    iu = np.triu_indices(V, k=1)
+    squares = np.zeros((flat.shape[0], V, V), dtype=flat.dtype)
+    squares[:, iu[0], iu[1]] = flat
+    squares += np.transpose(squares, (0, 2, 1))
+    squares[:, np.arange(V), np.arange(V)] = diag
+    return squares[0] if squares.shape[0] == 1 else squares
+
+
+# --------------------------------------------------------------------------
+# nonparametric statistics
+# --------------------------------------------------------------------------
+
+def _coerce_iscs(iscs, pairwise):
+    """Normalize ISC input to [row, voxel] and infer subject count."""
+    iscs = np.asarray(iscs)
+    if iscs.ndim == 1:
+        iscs = iscs[:, None]
     if pairwise:
         try:
-            test_square = squareform(iscs[:, 0], force='tomatrix')
-            n_subjects = test_square.shape[0]
+            n_subjects = squareform(iscs[:, 0],
+                                    force='tomatrix').shape[0]
         except ValueError:
             raise ValueError("For pairwise input, ISCs must be the "
                              "vectorized triangle of a square matrix.")
     else:
         n_subjects = iscs.shape[0]
-    n_voxels = iscs.shape[1]
-    logger.info("Assuming %d subjects with and %d voxel(s) or ROI(s) in "
-                "bootstrap ISC test.", n_subjects, n_voxels)
-    return iscs, n_subjects, n_voxels
-
-
-def compute_summary_statistic(iscs, summary_statistic='mean', axis=None):
-    """'mean' (Fisher-Z mean: tanh(mean(arctanh))) or 'median' of ISCs."""
-    if summary_statistic not in ('mean', 'median'):
-        raise ValueError("Summary statistic must be 'mean' or 'median'")
-    if summary_statistic == 'mean':
-        return np.tanh(np.nanmean(np.arctanh(iscs), axis=axis))
-    return np.nanmedian(iscs, axis=axis)
-
-
-def squareform_isfc(isfcs, iscs=None):
-    """Square ↔ condensed ISFC conversion retaining the ISC diagonal."""
-    if not type(iscs) == np.ndarray and \
-            isfcs.shape[-2] == isfcs.shape[-1]:  # noqa: E721
-        if isfcs.ndim == 2:
-            isfcs = isfcs[np.newaxis, ...]
-        if isfcs.ndim == 3:
-            iscs = np.diagonal(isfcs, axis1=1, axis2=2)
-            isfcs = np.vstack([squareform(m, checks=False)[np.newaxis, :]
-                               for m in isfcs])
-        else:
-            raise ValueError("Square (redundant) ISFCs must be square "
-                             "with multiple subjects or pairs of subjects "
-                             "indexed by the first dimension")
-        if isfcs.shape[0] == iscs.shape[0] == 1:
-            isfcs, iscs = isfcs[0], iscs[0]
-        return isfcs, iscs
-    else:
-        if isfcs.ndim == iscs.ndim == 1:
-            isfcs, iscs = isfcs[np.newaxis, :], iscs[np.newaxis, :]
-        stack = []
-        for isfc_row, isc_row in zip(isfcs, iscs):
-            sq = squareform(isfc_row, checks=False)
-            np.fill_diagonal(sq, isc_row)
-            stack.append(sq[np.newaxis, ...])
-        isfcs = np.vstack(stack)
-        if isfcs.shape[0] == 1:
-            isfcs = isfcs[0]
-        return isfcs
+    logger.info("Nonparametric ISC test: %d subjects, %d voxel(s)/"
+                "ROI(s)", n_subjects, iscs.shape[1])
+    return iscs, n_subjects, iscs.shape[1]
 
 
 def bootstrap_isc(iscs, pairwise=False, summary_statistic='median',
                   n_bootstraps=1000, ci_percentile=95, side='right',
                   random_state=None):
-    """One-sample subject-level bootstrap with the Hall-Wilson shift."""
-    iscs, n_subjects, n_voxels = _check_isc_input(iscs, pairwise=pairwise)
+    """One-sample subject-level bootstrap with the Hall-Wilson shift.
+
+    All ``n_bootstraps`` subject resamples are drawn up front and the
+    summary statistic is applied to the whole [boot, subject, voxel]
+    stack at once.  Resampled pairs of a subject with itself (pairwise
+    mode) carry no information and are excluded as NaN.
+    """
+    iscs, n_subjects, n_voxels = _coerce_iscs(iscs, pairwise)
     if summary_statistic not in ('mean', 'median'):
         raise ValueError("Summary statistic must be 'mean' or 'median'")
     observed = compute_summary_statistic(
         iscs, summary_statistic=summary_statistic, axis=0)
 
-    distribution = []
-    for _ in np.arange(n_bootstraps):
-        if isinstance(random_state, np.random.RandomState):
-            prng = random_state
-        else:
-            prng = np.random.RandomState(random_state)
-        subject_sample = sorted(prng.choice(np.arange(n_subjects),
-                                            size=n_subjects))
-        if pairwise:
-            isc_sample = []
-            for voxel_iscs in iscs.T:
-                voxel_iscs = squareform(voxel_iscs, force='tomatrix')
-                np.fill_diagonal(voxel_iscs, 1)
-                voxel_sample = voxel_iscs[subject_sample, :][:,
-                                                             subject_sample]
-                voxel_sample = squareform(voxel_sample, checks=False)
-                voxel_sample[voxel_sample == 1.] = np.nan
-                isc_sample.append(voxel_sample)
-            isc_sample = np.column_stack(isc_sample)
-        else:
-            isc_sample = iscs[subject_sample, :]
-        distribution.append(compute_summary_statistic(
-            isc_sample, summary_statistic=summary_statistic, axis=0))
-        random_state = np.random.RandomState(
-            prng.randint(0, MAX_RANDOM_SEED, dtype=np.int64))
+    rng = _as_rng(random_state)
+    draws = np.sort(rng.randint(0, n_subjects,
+                                size=(n_bootstraps, n_subjects)), axis=1)
 
-    distribution = np.array(distribution)
-    ci = (np.percentile(distribution, (100 - ci_percentile) / 2, axis=0),
-          np.percentile(distribution,
-                        ci_percentile + (100 - ci_percentile) / 2, axis=0))
-    shifted = distribution - observed
-    p = p_from_null(observed, shifted, side=side, exact=False, axis=0)
+    if pairwise:
+        # square per-voxel matrices once, then each bootstrap is a
+        # row/column gather + upper-triangle read
+        V = n_voxels
+        iu = np.triu_indices(n_subjects, k=1)
+        sq = np.zeros((n_subjects, n_subjects, V))
+        sq[iu[0], iu[1]] = iscs
+        sq += np.transpose(sq, (1, 0, 2))
+        gathered = sq[draws[:, :, None],
+                      draws[:, None, :]]        # [boot, s, s, V]
+        samples = gathered[:, iu[0], iu[1]]     # [boot, pair, V]
+        self_pair = draws[:, iu[0]] == draws[:, iu[1]]
+        samples[self_pair] = np.nan
+    else:
+        samples = iscs[draws]                   # [boot, subject, voxel]
+
+    distribution = compute_summary_statistic(
+        samples, summary_statistic=summary_statistic, axis=1)
+
+    half_alpha = (100 - ci_percentile) / 2
+    ci = (np.percentile(distribution, half_alpha, axis=0),
+          np.percentile(distribution, 100 - half_alpha, axis=0))
+    p = p_from_null(observed, distribution - observed, side=side,
+                    exact=False, axis=0)
     return observed, ci, p, distribution
 
 
-def _check_group_assignment(group_assignment, n_subjects):
-    if type(group_assignment) == list:  # noqa: E721
-        pass
-    elif type(group_assignment) == np.ndarray:  # noqa: E721
+class _GroupLayout:
+    """Two-group bookkeeping for permutation_isc: label order, the
+    pairwise label matrix, and the observed group selector."""
+
+    def __init__(self, assignment, n_subjects, pairwise):
+        self.assignment = assignment
+        self.n_subjects = n_subjects
+        uniques = np.unique(assignment) if assignment else np.array([])
+        self.n_groups = max(1, len(uniques))
+        if self.n_groups > 2:
+            raise ValueError("This test is not valid for more than "
+                             f"2 groups! (got {self.n_groups})")
+        self.labels = uniques if self.n_groups == 2 else None
+        self.matrix = None
+        self.selector = None
+        if self.n_groups == 2:
+            a = np.asarray(assignment)
+            if pairwise:
+                # label matrix: within-group pairs carry the group
+                # label, cross-group pairs are NaN
+                lab = np.where(a[:, None] == a[None, :],
+                               np.broadcast_to(a, (n_subjects,
+                                                   n_subjects)),
+                               np.nan).astype(float)
+                np.fill_diagonal(lab, np.nan)
+                self.matrix = lab
+                self.selector = squareform(lab, checks=False)
+            else:
+                self.selector = a
+        elif pairwise:
+            self.matrix = np.ones((n_subjects, n_subjects))
+
+
+def _normalize_assignment(group_assignment, n_subjects):
+    if isinstance(group_assignment, np.ndarray):
         group_assignment = group_assignment.tolist()
-    else:
+    elif not isinstance(group_assignment, list):
         logger.info("No group assignment provided, "
                     "performing one-sample test.")
+        group_assignment = None
     if group_assignment and len(group_assignment) != n_subjects:
-        raise ValueError("Group assignments ({0}) "
-                         "do not match number of subjects ({1})!".format(
-                             len(group_assignment), n_subjects))
+        raise ValueError(
+            f"Group assignments ({len(group_assignment)}) do not "
+            f"match number of subjects ({n_subjects})!")
     return group_assignment
 
 
-def _get_group_parameters(group_assignment, n_subjects, pairwise=False):
-    gp = {'group_assignment': group_assignment, 'n_subjects': n_subjects,
-          'group_labels': None, 'groups': None, 'sorter': None,
-          'unsorter': None, 'group_matrix': None, 'group_selector': None}
-    if group_assignment and len(np.unique(group_assignment)) == 2:
-        gp['n_groups'] = 2
-        group_labels = np.unique(group_assignment)
-        groups = {group_labels[0]: group_assignment.count(group_labels[0]),
-                  group_labels[1]: group_assignment.count(group_labels[1])}
-        if pairwise:
-            sorter = np.array(group_assignment).argsort()
-            unsorter = sorter.argsort()
-            ul = np.full((groups[group_labels[0]],) * 2, group_labels[0])
-            ur = np.full((groups[group_labels[0]],
-                          groups[group_labels[1]]), np.nan)
-            ll = np.full((groups[group_labels[1]],
-                          groups[group_labels[0]]), np.nan)
-            lr = np.full((groups[group_labels[1]],) * 2, group_labels[1])
-            group_matrix = np.vstack((np.hstack((ul, ur)),
-                                      np.hstack((ll, lr))))
-            np.fill_diagonal(group_matrix, np.nan)
-            gp['group_matrix'] = group_matrix
-            gp['group_selector'] = squareform(
-                group_matrix[unsorter, :][:, unsorter], checks=False)
-            gp['sorter'] = sorter
-            gp['unsorter'] = unsorter
-        else:
-            gp['group_selector'] = group_assignment
-        gp['groups'] = groups
-        gp['group_labels'] = group_labels
-    elif not group_assignment or len(np.unique(group_assignment)) == 1:
-        gp['n_groups'] = 1
-        if pairwise:
-            gp['group_matrix'] = np.ones((n_subjects, n_subjects))
-    elif len(np.unique(group_assignment)) > 2:
-        raise ValueError("This test is not valid for more than "
-                         "2 groups! (got {0})".format(
-                             len(np.unique(group_assignment))))
-    else:
-        raise ValueError("Invalid group assignments!")
-    return gp
+def _group_difference(iscs, selector, labels, summary_statistic):
+    sel = np.asarray(selector)
+    first = compute_summary_statistic(
+        iscs[sel == labels[0]], summary_statistic=summary_statistic,
+        axis=0)
+    second = compute_summary_statistic(
+        iscs[sel == labels[1]], summary_statistic=summary_statistic,
+        axis=0)
+    return first - second
 
 
-def _permute_one_sample_iscs(iscs, group_parameters, i, pairwise=False,
-                             summary_statistic='median', group_matrix=None,
-                             exact_permutations=None, prng=None):
-    if exact_permutations:
-        sign_flipper = np.array(exact_permutations[i])
-    else:
-        sign_flipper = prng.choice(
-            [-1, 1], size=group_parameters['n_subjects'], replace=True)
-    if pairwise:
-        matrix_flipped = (group_parameters['group_matrix'] * sign_flipper
-                          * sign_flipper[:, np.newaxis])
-        sign_flipper = squareform(matrix_flipped, checks=False)
-    isc_flipped = iscs * sign_flipper[:, np.newaxis]
-    return compute_summary_statistic(
-        isc_flipped, summary_statistic=summary_statistic, axis=0)
+def _sign_flip_rows(layout, flips, pairwise):
+    """Per-permutation row multipliers from subject sign flips.
+
+    flips [n_perm, n_subjects] → [n_perm, n_rows]: in pairwise mode a
+    pair's sign is the product of its two subjects' signs.
+    """
+    if not pairwise:
+        return flips
+    iu = np.triu_indices(layout.n_subjects, k=1)
+    return flips[:, iu[0]] * flips[:, iu[1]]
 
 
-def _permute_two_sample_iscs(iscs, group_parameters, i, pairwise=False,
-                             summary_statistic='median',
-                             exact_permutations=None, prng=None):
-    if exact_permutations:
-        group_shuffler = np.array(exact_permutations[i])
-    elif pairwise:
-        group_shuffler = prng.permutation(np.arange(
-            len(np.array(group_parameters['group_assignment'])[
-                group_parameters['sorter']])))
-    else:
-        group_shuffler = prng.permutation(np.arange(
-            len(group_parameters['group_assignment'])))
-
-    if pairwise:
-        group_shuffled = group_parameters['group_matrix'][
-            group_shuffler, :][:, group_shuffler]
-        group_selector = squareform(
-            group_shuffled[group_parameters['unsorter'], :]
-            [:, group_parameters['unsorter']], checks=False)
-    else:
-        group_selector = np.array(
-            group_parameters['group_assignment'])[group_shuffler]
-
-    labels = group_parameters['group_labels']
-    return (compute_summary_statistic(
-        iscs[group_selector == labels[0], :],
-        summary_statistic=summary_statistic, axis=0)
-        - compute_summary_statistic(
-            iscs[group_selector == labels[1], :],
-            summary_statistic=summary_statistic, axis=0))
-
-
-def permutation_isc(iscs, group_assignment=None, pairwise=False,  # noqa: C901
+def permutation_isc(iscs, group_assignment=None, pairwise=False,
                     summary_statistic='median', n_permutations=1000,
                     side='right', random_state=None):
-    """One-sample (sign-flip) / two-sample (label-shuffle) permutation test;
-    exact enumeration when the permutation space fits in n_permutations."""
-    iscs, n_subjects, n_voxels = _check_isc_input(iscs, pairwise=pairwise)
+    """One-sample (sign-flip) / two-sample (label-shuffle) permutation
+    test; exact enumeration when the permutation space fits in
+    n_permutations."""
+    iscs, n_subjects, n_voxels = _coerce_iscs(iscs, pairwise)
     if summary_statistic not in ('mean', 'median'):
         raise ValueError("Summary statistic must be 'mean' or 'median'")
-    group_assignment = _check_group_assignment(group_assignment, n_subjects)
-    gp = _get_group_parameters(group_assignment, n_subjects,
-                               pairwise=pairwise)
+    assignment = _normalize_assignment(group_assignment, n_subjects)
+    layout = _GroupLayout(assignment, n_subjects, pairwise)
+    rng = _as_rng(random_state)
 
-    if gp['n_groups'] == 1:
-        if n_permutations < 2 ** n_subjects:
-            exact_permutations = None
-        else:
-            exact_permutations = list(product([-1, 1], repeat=n_subjects))
-            n_permutations = 2 ** n_subjects
-    else:
-        if n_permutations < math.factorial(n_subjects):
-            exact_permutations = None
-        else:
-            exact_permutations = list(permutations(np.arange(
-                len(group_assignment))))
-            n_permutations = math.factorial(n_subjects)
+    one_sample = layout.n_groups == 1
+    space = 2 ** n_subjects if one_sample \
+        else math.factorial(n_subjects)
+    exact = n_permutations >= space
 
-    if gp['n_groups'] == 1:
+    if one_sample:
         observed = compute_summary_statistic(
-            iscs, summary_statistic=summary_statistic,
-            axis=0)[np.newaxis, :]
+            iscs, summary_statistic=summary_statistic, axis=0)[None, :]
+        if exact:
+            # every sign pattern via the bits of 0..2^n-1
+            codes = np.arange(space)[:, None] >> np.arange(n_subjects)
+            flips = 1.0 - 2.0 * (codes & 1)
+            n_permutations = space
+        else:
+            flips = rng.choice([-1.0, 1.0],
+                               size=(n_permutations, n_subjects))
+        row_signs = _sign_flip_rows(layout, flips, pairwise)
+        # [perm, row, voxel] in one broadcast multiply
+        flipped = iscs[None] * row_signs[:, :, None]
+        distribution = compute_summary_statistic(
+            flipped, summary_statistic=summary_statistic, axis=1)
     else:
-        labels = gp['group_labels']
-        observed = np.array(
-            compute_summary_statistic(
-                iscs[np.asarray(gp['group_selector']) == labels[0], :],
-                summary_statistic=summary_statistic, axis=0)
-            - compute_summary_statistic(
-                iscs[np.asarray(gp['group_selector']) == labels[1], :],
-                summary_statistic=summary_statistic, axis=0))
-
-    distribution = []
-    for i in np.arange(n_permutations):
-        if exact_permutations:
-            prng = None
-        elif isinstance(random_state, np.random.RandomState):
-            prng = random_state
+        observed = np.array(_group_difference(
+            iscs, layout.selector, layout.labels, summary_statistic))
+        if exact:
+            orders = list(permutations(range(n_subjects)))
+            n_permutations = space
         else:
-            prng = np.random.RandomState(random_state)
-        if gp['n_groups'] == 1:
-            isc_sample = _permute_one_sample_iscs(
-                iscs, gp, i, pairwise=pairwise,
-                summary_statistic=summary_statistic,
-                exact_permutations=exact_permutations, prng=prng)
-        else:
-            isc_sample = _permute_two_sample_iscs(
-                iscs, gp, i, pairwise=pairwise,
-                summary_statistic=summary_statistic,
-                exact_permutations=exact_permutations, prng=prng)
-        distribution.append(isc_sample)
-        if not exact_permutations:
-            random_state = np.random.RandomState(
-                prng.randint(0, MAX_RANDOM_SEED, dtype=np.int64))
+            orders = [rng.permutation(n_subjects)
+                      for _ in range(n_permutations)]
+        rows = []
+        base = np.asarray(layout.assignment)
+        for order in orders:
+            if pairwise:
+                shuffled_matrix = layout.matrix[np.ix_(order, order)]
+                selector = squareform(shuffled_matrix, checks=False)
+            else:
+                selector = base[np.asarray(order)]
+            rows.append(_group_difference(
+                iscs, selector, layout.labels, summary_statistic))
+        distribution = np.array(rows)
 
-    distribution = np.array(distribution)
-    p = p_from_null(observed, distribution, side=side,
-                    exact=bool(exact_permutations), axis=0)
+    p = p_from_null(observed, distribution, side=side, exact=exact,
+                    axis=0)
     return observed, p, distribution
+
+
+def _null_from_resamples(data, observed, n_iter, side, draw_fn):
+    """Shared shell of the timeshift/phaseshift tests: build the null
+    distribution row by row from draw_fn(iteration) and convert to p."""
+    rows = [np.asarray(draw_fn(i)).ravel() for i in range(n_iter)]
+    distribution = np.vstack(rows)
+    p = p_from_null(observed, distribution, side=side, exact=False,
+                    axis=0)
+    return observed, p, distribution
+
+
+def _loo_null_isc(surrogate, data, summary_statistic, tolerate_nans):
+    """Leave-one-out null: each surrogate subject vs the UNshifted
+    mean of the others."""
+    totals = data.sum(axis=2)
+    n = data.shape[-1]
+    cols = []
+    for s in range(n):
+        others_mean = (totals - data[..., s]) / (n - 1)
+        cols.append(isc(np.dstack((surrogate[..., s], others_mean)),
+                        pairwise=False, summary_statistic=None,
+                        tolerate_nans=tolerate_nans))
+    return compute_summary_statistic(
+        np.dstack(cols), summary_statistic=summary_statistic, axis=2)
 
 
 def timeshift_isc(data, pairwise=False, summary_statistic='median',
@@ -601,46 +614,20 @@ def timeshift_isc(data, pairwise=False, summary_statistic='median',
     observed = isc(data, pairwise=pairwise,
                    summary_statistic=summary_statistic,
                    tolerate_nans=tolerate_nans)
+    rng = _as_rng(random_state)
 
-    rolled = np.rollaxis(data, 2, 0) if pairwise else data
-
-    distribution = []
-    for _ in np.arange(n_shifts):
-        if isinstance(random_state, np.random.RandomState):
-            prng = random_state
-        else:
-            prng = np.random.RandomState(random_state)
-        shifts = prng.choice(np.arange(n_TRs), size=n_subjects,
-                             replace=True)
+    def one_draw(_):
+        shifts = rng.randint(0, n_TRs, size=n_subjects)
+        rolled = np.dstack([np.roll(data[..., s], shifts[s], axis=0)
+                            for s in range(n_subjects)])
         if pairwise:
-            shifted_data = []
-            for subject, shift in zip(rolled, shifts):
-                shifted_data.append(np.concatenate(
-                    (subject[-shift:, :], subject[:-shift, :])))
-            shifted_data = np.dstack(shifted_data)
-            shifted_isc = isc(shifted_data, pairwise=True,
-                              summary_statistic=summary_statistic,
-                              tolerate_nans=tolerate_nans)
-        else:
-            shifted_isc = []
-            for s, shift in enumerate(shifts):
-                shifted_subject = np.concatenate(
-                    (data[-shift:, :, s], data[:-shift, :, s]))
-                nonshifted_mean = np.mean(np.delete(data, s, 2), axis=2)
-                loo_isc = isc(np.dstack((shifted_subject, nonshifted_mean)),
-                              pairwise=False, summary_statistic=None,
-                              tolerate_nans=tolerate_nans)
-                shifted_isc.append(loo_isc)
-            shifted_isc = compute_summary_statistic(
-                np.dstack(shifted_isc),
-                summary_statistic=summary_statistic, axis=2)
-        distribution.append(shifted_isc)
-        random_state = np.random.RandomState(
-            prng.randint(0, MAX_RANDOM_SEED, dtype=np.int64))
+            return isc(rolled, pairwise=True,
+                       summary_statistic=summary_statistic,
+                       tolerate_nans=tolerate_nans)
+        return _loo_null_isc(rolled, data, summary_statistic,
+                             tolerate_nans)
 
-    distribution = np.array(distribution).reshape(n_shifts, n_voxels)
-    p = p_from_null(observed, distribution, side=side, exact=False, axis=0)
-    return observed, p, distribution
+    return _null_from_resamples(data, observed, n_shifts, side, one_draw)
 
 
 def phaseshift_isc(data, pairwise=False, summary_statistic='median',
@@ -651,36 +638,15 @@ def phaseshift_isc(data, pairwise=False, summary_statistic='median',
     observed = isc(data, pairwise=pairwise,
                    summary_statistic=summary_statistic,
                    tolerate_nans=tolerate_nans)
+    rng = _as_rng(random_state)
 
-    distribution = []
-    for _ in np.arange(n_shifts):
-        if isinstance(random_state, np.random.RandomState):
-            prng = random_state
-        else:
-            prng = np.random.RandomState(random_state)
-        shifted_data = phase_randomize(data, random_state=prng)
+    def one_draw(_):
+        surrogate = phase_randomize(data, random_state=rng)
         if pairwise:
-            shifted_isc = isc(shifted_data, pairwise=True,
-                              summary_statistic=summary_statistic,
-                              tolerate_nans=tolerate_nans)
-        else:
-            rolled = np.rollaxis(shifted_data, 2, 0)
-            shifted_isc = []
-            for s, shifted_subject in enumerate(rolled):
-                nonshifted_mean = np.mean(np.delete(data, s, axis=2),
-                                          axis=2)
-                loo_isc = isc(np.dstack((shifted_subject,
-                                         nonshifted_mean)),
-                              pairwise=False, summary_statistic=None,
-                              tolerate_nans=tolerate_nans)
-                shifted_isc.append(loo_isc)
-            shifted_isc = compute_summary_statistic(
-                np.dstack(shifted_isc),
-                summary_statistic=summary_statistic, axis=2)
-        distribution.append(shifted_isc)
-        random_state = np.random.RandomState(
-            prng.randint(0, MAX_RANDOM_SEED, dtype=np.int64))
+            return isc(surrogate, pairwise=True,
+                       summary_statistic=summary_statistic,
+                       tolerate_nans=tolerate_nans)
+        return _loo_null_isc(surrogate, data, summary_statistic,
+                             tolerate_nans)
 
-    distribution = np.array(distribution).reshape(n_shifts, n_voxels)
-    p = p_from_null(observed, distribution, side=side, exact=False, axis=0)
-    return observed, p, distribution
+    return _null_from_resamples(data, observed, n_shifts, side, one_draw)
