@@ -35,6 +35,20 @@ def _is_precomputed_svm(clf):
         return False
 
 
+def _split_pairs(X):
+    """Unzip [(data1, data2), ...] with the WIDER member first (the
+    reference orients so voxels1 >= voxels2); returns (X1, X2, v1, v2)."""
+    for sample in X:
+        if len(sample) != 2:
+            raise AssertionError('there must be two parts for each '
+                                 'correlation computation')
+    left, right = zip(*X)
+    v_left, v_right = left[0].shape[1], right[0].shape[1]
+    if v_left < v_right:
+        return right, left, v_right, v_left
+    return left, right, v_left, v_right
+
+
 class Classifier:
     """Train/predict on FCMA correlation features with any sklearn-style
     classifier; see module docstring for the contract."""
@@ -45,9 +59,11 @@ class Classifier:
         self.num_processed_voxels = num_processed_voxels
         self.epochs_per_subj = epochs_per_subj
         self.num_digits_ = 0
-        self.device = torch.device(device) if device is not None else (
-            torch.device("cuda") if torch.cuda.is_available()
-            else torch.device("cpu"))
+        if device is not None:
+            self.device = torch.device(device)
+        else:
+            self.device = torch.device(
+                "cuda" if torch.cuda.is_available() else "cpu")
 
     # -- internals ---------------------------------------------------------
 
@@ -59,110 +75,100 @@ class Classifier:
     def _prepare_correlation_data(self, X1, X2, start_voxel=0,
                                   num_processed_voxels=None):
         """corr [num_samples, P, V2] for voxels [start, start+P) of X1."""
-        num_samples = len(X1)
-        assert num_samples > 0, \
-            'need at least one (data1, data2) sample to correlate'
-        num_voxels1 = X1[0].shape[1]
-        num_voxels2 = X2[0].shape[1]
-        assert num_voxels1 * num_voxels2 == self.num_features_, \
-            'input voxel-pair count differs from the fitted feature count'
-        assert X1[0].shape[0] == X2[0].shape[0], \
-            'X1 and X2 disagree on TR count'
-        if num_processed_voxels is None:
-            num_processed_voxels = num_voxels1
-        a = self._stack(X1)[:, :, start_voxel:start_voxel
-                            + num_processed_voxels]
-        b = self._stack(X2)
-        return torch.bmm(a.transpose(1, 2), b)
+        if not len(X1):
+            raise AssertionError('need at least one (data1, data2) '
+                                 'sample to correlate')
+        v1, v2 = X1[0].shape[1], X2[0].shape[1]
+        if v1 * v2 != self.num_features_:
+            raise AssertionError('input voxel-pair count differs from '
+                                 'the fitted feature count')
+        if X1[0].shape[0] != X2[0].shape[0]:
+            raise AssertionError('X1 and X2 disagree on TR count')
+        stop = v1 if num_processed_voxels is None \
+            else start_voxel + num_processed_voxels
+        a = self._stack(X1)[:, :, start_voxel:stop]
+        return torch.bmm(a.transpose(1, 2), self._stack(X2))
 
     def _normalize(self, corr, norm_unit):
         """Fisher-z + z-score over each ``norm_unit`` samples (the
         reference reshapes to [1, S, P*V] and calls the N1 kernel)."""
-        if norm_unit > 1:
-            S, d2, d3 = corr.shape
-            flat = corr.reshape(1, S, d2 * d3)
-            normalize_correlation_(flat, norm_unit)
-            return flat.reshape(S, d2, d3)
-        return corr
+        if norm_unit <= 1:
+            return corr
+        S = corr.shape[0]
+        flat = corr.reshape(1, S, -1)
+        normalize_correlation_(flat, norm_unit)
+        return flat.reshape(corr.shape)
 
-    @staticmethod
-    def _leading_digits(value):
-        return len(str(int(value)))
-
-    def _compute_kernel_matrix_in_portion(self, X1, X2):
-        kernel = torch.zeros((self.num_samples_, self.num_samples_),
-                             dtype=torch.float32, device=self.device)
-        sr = 0
-        row_length = self.num_processed_voxels
-        normalized = None
-        while sr < self.num_voxels_:
-            if row_length >= self.num_voxels_ - sr:
-                row_length = self.num_voxels_ - sr
-            corr = self._prepare_correlation_data(X1, X2, sr, row_length)
-            normalized = self._normalize(corr, self.epochs_per_subj)
-            flat = normalized.reshape(self.num_samples_, -1)
-            kernel += flat @ flat.T
-            sr += row_length
-        self.num_digits_ = self._leading_digits(float(kernel[0, 0]))
+    def _apply_shrink(self, kernel):
+        """Record the lead-entry digit count and rescale matrices whose
+        magnitudes would swamp the SVM (reference num_digits_ logic)."""
+        self.num_digits_ = len(str(int(float(kernel[0, 0]))))
         if self.num_digits_ > 2:
             kernel *= 10.0 ** (2 - self.num_digits_)
-        return kernel, normalized
+        return kernel
+
+    def _portioned_kernel(self, X1, X2):
+        """Gram matrix accumulated ``num_processed_voxels`` rows at a
+        time; returns (kernel, last portion's normalized features)."""
+        n = self.num_samples_
+        kernel = torch.zeros((n, n), dtype=torch.float32,
+                             device=self.device)
+        normalized = None
+        portions = list(range(0, self.num_voxels_,
+                              self.num_processed_voxels))
+        for begin in portions:
+            width = min(self.num_processed_voxels,
+                        self.num_voxels_ - begin)
+            corr = self._prepare_correlation_data(X1, X2, begin, width)
+            normalized = self._normalize(corr, self.epochs_per_subj)
+            flat = normalized.reshape(n, -1)
+            kernel += flat @ flat.T
+        return self._apply_shrink(kernel), normalized
 
     def _generate_training_data(self, X1, X2, num_training_samples):
         if not _is_precomputed_svm(self.clf):
             corr = self._prepare_correlation_data(X1, X2)
-            normalized = self._normalize(corr, self.epochs_per_subj)
-            data = normalized.reshape(self.num_samples_, self.num_features_)
+            features = self._normalize(corr, self.epochs_per_subj)
             self.training_data_ = None
-            return data.cpu().numpy()
-        if self.num_processed_voxels < self.num_voxels_:
-            if num_training_samples is None:
-                raise RuntimeError(
-                    'the kernel matrix will be computed portion by '
-                    'portion, the test samples must be predefined by '
-                    'specifying num_training_samples')
-            if num_training_samples >= self.num_samples_:
-                raise ValueError('the number of training samples must be '
-                                 'smaller than the number of total samples')
-        data, normalized = self._compute_kernel_matrix_in_portion(X1, X2)
-        if self.num_processed_voxels >= self.num_voxels_:
-            self.training_data_ = normalized.reshape(
-                self.num_samples_, self.num_features_).cpu().numpy()
-        else:
-            self.training_data_ = None
-        return data.cpu().numpy()
+            return features.reshape(self.num_samples_,
+                                    self.num_features_).cpu().numpy()
+        partial = self.num_processed_voxels < self.num_voxels_
+        if partial and num_training_samples is None:
+            raise RuntimeError(
+                'the kernel matrix will be computed portion by '
+                'portion, the test samples must be predefined by '
+                'specifying num_training_samples')
+        if partial and num_training_samples >= self.num_samples_:
+            raise ValueError('the number of training samples must be '
+                             'smaller than the number of total samples')
+        kernel, normalized = self._portioned_kernel(X1, X2)
+        self.training_data_ = None if partial else normalized.reshape(
+            self.num_samples_, self.num_features_).cpu().numpy()
+        return kernel.cpu().numpy()
 
     # -- estimator API -----------------------------------------------------
 
     def fit(self, X, y, num_training_samples=None):
-        assert len(X) == len(y), \
-            'sample count and label count differ'
-        for x in X:
-            assert len(x) == 2, \
-                'there must be two parts for each correlation computation'
-        X1, X2 = zip(*X)
-        if not _is_precomputed_svm(self.clf) and \
-                num_training_samples is not None:
+        if len(X) != len(y):
+            raise AssertionError('sample count and label count differ')
+        X1, X2, v1, v2 = _split_pairs(X)
+        if num_training_samples is not None and \
+                not _is_precomputed_svm(self.clf):
             num_training_samples = None
             logger.warning(
                 'num_training_samples should not be set for classifiers '
                 'other than SVM with precomputed kernels')
-        num_voxels1 = X1[0].shape[1]
-        num_voxels2 = X2[0].shape[1]
-        if num_voxels1 < num_voxels2:
-            X1, X2 = X2, X1
-            num_voxels1, num_voxels2 = num_voxels2, num_voxels1
-        self.num_voxels_ = num_voxels1
-        self.num_features_ = num_voxels1 * num_voxels2
+        self.num_voxels_ = v1
+        self.num_features_ = v1 * v2
         self.num_samples_ = len(X1)
 
         data = self._generate_training_data(X1, X2, num_training_samples)
 
         if num_training_samples is not None:
+            split = num_training_samples
             self.test_raw_data_ = None
-            self.test_data_ = data[num_training_samples:,
-                                   0:num_training_samples]
-            data = data[0:num_training_samples, 0:num_training_samples]
+            self.test_data_ = data[split:, :split]
+            data = data[:split, :split]
         self.clf = self.clf.fit(data, y[0:num_training_samples])
         if num_training_samples is None:
             self.test_raw_data_ = None
@@ -170,39 +176,30 @@ class Classifier:
         return self
 
     def _build_test_data(self, X):
-        for x in X:
-            assert len(x) == 2, \
-                'there must be two parts for each correlation computation'
-        X1, X2 = zip(*X)
-        num_voxels1 = X1[0].shape[1]
-        num_voxels2 = X2[0].shape[1]
-        if num_voxels1 < num_voxels2:
-            X1, X2 = X2, X1
-            num_voxels1, num_voxels2 = num_voxels2, num_voxels1
-        assert self.num_features_ == num_voxels1 * num_voxels2, \
-            'the number of features does not match the model'
-        num_test_samples = len(X1)
+        X1, X2, v1, v2 = _split_pairs(X)
+        if self.num_features_ != v1 * v2:
+            raise AssertionError(
+                'the number of features does not match the model')
         self.test_raw_data_ = X
         corr = self._prepare_correlation_data(X1, X2)
-        normalized = self._normalize(corr, num_test_samples)
+        normalized = self._normalize(corr, len(X1))
         self.test_data_ = self._prepare_test_data(normalized)
 
     def _prepare_test_data(self, corr_data):
-        num_test_samples = corr_data.shape[0]
-        assert num_test_samples > 0, 'at least one test sample is needed'
-        if _is_precomputed_svm(self.clf):
-            assert self.training_data_ is not None, \
-                'when using precomputed kernel of SVM, ' \
-                'all training data must be provided'
-            train = torch.as_tensor(self.training_data_,
-                                    device=self.device)
-            flat = corr_data.reshape(num_test_samples, self.num_features_)
-            data = flat @ train.T
-            if self.num_digits_ > 2:
-                data = data * 10.0 ** (2 - self.num_digits_)
-            return data.cpu().numpy()
-        return corr_data.reshape(num_test_samples,
-                                 self.num_features_).cpu().numpy()
+        n_test = corr_data.shape[0]
+        if n_test < 1:
+            raise AssertionError('at least one test sample is needed')
+        flat = corr_data.reshape(n_test, self.num_features_)
+        if not _is_precomputed_svm(self.clf):
+            return flat.cpu().numpy()
+        if self.training_data_ is None:
+            raise AssertionError('when using precomputed kernel of SVM, '
+                                 'all training data must be provided')
+        train = torch.as_tensor(self.training_data_, device=self.device)
+        similarity = flat @ train.T
+        if self.num_digits_ > 2:
+            similarity = similarity * 10.0 ** (2 - self.num_digits_)
+        return similarity.cpu().numpy()
 
     def predict(self, X=None):
         if X is not None:
@@ -210,18 +207,11 @@ class Classifier:
         return self.clf.predict(self.test_data_)
 
     def _is_equal_to_test_raw_data(self, X):
-        if self.test_raw_data_ is None or \
-                len(X) != len(self.test_raw_data_):
+        cached = self.test_raw_data_
+        if cached is None or len(X) != len(cached):
             return False
-        X1, X2 = zip(*X)
-        c1, c2 = zip(*self.test_raw_data_)
-        for new, old in zip(X1, c1):
-            if not np.array_equal(new, old):
-                return False
-        for new, old in zip(X2, c2):
-            if not np.array_equal(new, old):
-                return False
-        return True
+        return all(np.array_equal(a, b) and np.array_equal(c, d)
+                   for (a, c), (b, d) in zip(X, cached))
 
     def decision_function(self, X=None):
         if X is not None and not self._is_equal_to_test_raw_data(X):
