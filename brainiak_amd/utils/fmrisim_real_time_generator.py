@@ -71,25 +71,23 @@ def _default_roi(dimensions, centre_frac, radius=2):
 def _generate_ROIs(ROI, stimfunc, noise, scale_percentage, data_dict):
     """Evoked-response signal volume for one ROI, scaled to a percent
     signal change against the noise."""
-    idx_list = np.where(ROI == 1)
-    idxs = np.column_stack(idx_list).astype(int)
-    voxels = int(ROI.sum())
+    vx, vy, vz = np.nonzero(ROI == 1)
+    voxels = len(vx)
 
-    if data_dict['multivariate_pattern'] is True:
-        pattern = np.random.rand(voxels).reshape((voxels, 1))
+    if data_dict['multivariate_pattern']:
+        per_voxel = np.random.rand(1, voxels)      # random multivoxel map
     else:
-        pattern = np.ones((voxels, 1))
+        per_voxel = np.ones((1, voxels))           # uniform response
+    weights = stimfunc * per_voxel
 
-    weights = np.tile(stimfunc, voxels) * pattern.T
-    temporal_res = 1 / data_dict['trDuration']
-    signal_func = sim.convolve_hrf(stimfunction=weights,
-                                   tr_duration=data_dict['trDuration'],
-                                   temporal_resolution=temporal_res,
-                                   scale_function=1)
-    noise = noise.astype('double')
-    noise_function = noise[idxs[:, 0], idxs[:, 1], idxs[:, 2], :].T
+    signal_func = sim.convolve_hrf(
+        stimfunction=weights,
+        tr_duration=data_dict['trDuration'],
+        temporal_resolution=1 / data_dict['trDuration'],
+        scale_function=1)
+    roi_noise = noise.astype('double')[vx, vy, vz, :].T
     sf_scaled = sim.compute_signal_change(
-        signal_function=signal_func, noise_function=noise_function,
+        signal_function=signal_func, noise_function=roi_noise,
         noise_dict=data_dict['noise_dict'],
         magnitude=[scale_percentage], method='PSC')
     return sim.apply_signal(sf_scaled, ROI)
@@ -150,23 +148,23 @@ def generate_data(outputDir, user_settings):
 
     # randomized A/B block design
     total_time = int(data_dict['numTRs'] * data_dict['trDuration'])
-    onsets_A, onsets_B = [], []
-    curr_time = data_dict['burn_in']
-    while curr_time < (total_time - data_dict['event_duration']):
-        if np.random.randint(0, 2) == 1:
-            onsets_A.append(curr_time)
-        else:
-            onsets_B.append(curr_time)
-        curr_time += data_dict['event_duration'] + data_dict['isi']
+    slot = data_dict['event_duration'] + data_dict['isi']
+    slot_starts = np.arange(data_dict['burn_in'],
+                            total_time - data_dict['event_duration'],
+                            slot)
+    coin = np.random.randint(0, 2, size=len(slot_starts))
+    onsets_A = [float(t) for t, c in zip(slot_starts, coin) if c == 1]
+    onsets_B = [float(t) for t, c in zip(slot_starts, coin) if c == 0]
 
-    temporal_res = 1 / data_dict['trDuration']
-    event_durations = [data_dict['event_duration']]
-    stimfunc_A = sim.generate_stimfunction(
-        onsets=onsets_A, event_durations=event_durations,
-        total_time=total_time, temporal_resolution=temporal_res)
-    stimfunc_B = sim.generate_stimfunction(
-        onsets=onsets_B, event_durations=event_durations,
-        total_time=total_time, temporal_resolution=temporal_res)
+    def _stimfunction(onsets):
+        return sim.generate_stimfunction(
+            onsets=onsets,
+            event_durations=[data_dict['event_duration']],
+            total_time=total_time,
+            temporal_resolution=1 / data_dict['trDuration'])
+
+    stimfunc_A = _stimfunction(onsets_A)
+    stimfunc_B = _stimfunction(onsets_B)
     np.save(os.path.join(outputDir, 'labels.npy'),
             stimfunc_A + (stimfunc_B * 2))
 
@@ -187,9 +185,10 @@ def generate_data(outputDir, user_settings):
     signal = signal_A + signal_B
 
     logger.info('Generating TRs in real time')
+    composed = noise + signal
     for idx in range(data_dict['numTRs']):
         start = time.time()
-        brain = noise[:, :, :, idx] + signal[:, :, :, idx]
+        brain = composed[..., idx]
         brain_int32 = np.nan_to_num(brain).astype(np.int32)
         if data_dict['save_dicom']:
             from .dicom_minimal import write_dicom
