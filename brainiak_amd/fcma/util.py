@@ -4,23 +4,21 @@
 on GPU tensors the matmul runs on the MFMA path through rocBLAS.
 """
 
-import math
-
 import numpy as np
 import torch
-from scipy.stats import zscore
 
 __all__ = ["compute_correlation"]
 
 
-def _normalize_for_correlation(data, axis, return_nans=False):
-    """Z-score (ddof=0) along ``axis`` then scale by 1/sqrt(n) so that
-    correlation becomes a plain dot product."""
-    shape = data.shape
-    data = zscore(data, axis=axis, ddof=0)
-    if not return_nans:
-        data = np.nan_to_num(data)
-    return data / math.sqrt(shape[axis])
+def _unitize_rows(m, return_nans=False):
+    """Center each row, scale to unit norm x 1/sqrt(n): after this,
+    correlation is a plain dot product.  Zero-variance rows become 0
+    (or NaN when return_nans=True)."""
+    centered = m - m.mean(axis=1, keepdims=True)
+    sd = centered.std(axis=1, keepdims=True)
+    with np.errstate(invalid="ignore", divide="ignore"):
+        unit = centered / (sd * np.sqrt(m.shape[1]))
+    return unit if return_nans else np.nan_to_num(unit)
 
 
 def compute_correlation(matrix1, matrix2, return_nans=False):
@@ -31,12 +29,10 @@ def compute_correlation(matrix1, matrix2, return_nans=False):
     """
     matrix1 = np.asarray(matrix1, dtype=np.float32)
     matrix2 = np.asarray(matrix2, dtype=np.float32)
-    r1, d1 = matrix1.shape
-    r2, d2 = matrix2.shape
-    if d1 != d2:
+    if matrix1.shape[1] != matrix2.shape[1]:
         raise ValueError('Dimension discrepancy')
-    m1 = _normalize_for_correlation(matrix1, 1, return_nans=return_nans)
-    m2 = _normalize_for_correlation(matrix2, 1, return_nans=return_nans)
-    t1 = torch.from_numpy(np.ascontiguousarray(m1))
-    t2 = torch.from_numpy(np.ascontiguousarray(m2))
+    t1 = torch.from_numpy(
+        np.ascontiguousarray(_unitize_rows(matrix1, return_nans)))
+    t2 = torch.from_numpy(
+        np.ascontiguousarray(_unitize_rows(matrix2, return_nans)))
     return (t1 @ t2.T).numpy().astype(np.float32, copy=False)
