@@ -19,6 +19,7 @@ import logging
 import numpy as np
 import torch
 
+from ..utils.timing import stage_timer
 from .core import normalize_correlation_
 
 logger = logging.getLogger(__name__)
@@ -162,7 +163,10 @@ class Classifier:
         self.num_features_ = v1 * v2
         self.num_samples_ = len(X1)
 
-        data = self._generate_training_data(X1, X2, num_training_samples)
+        with stage_timer("classifier training data", logger,
+                         sync_device=self.device):
+            data = self._generate_training_data(X1, X2,
+                                                num_training_samples)
 
         if num_training_samples is not None:
             split = num_training_samples
@@ -181,9 +185,11 @@ class Classifier:
             raise AssertionError(
                 'the number of features does not match the model')
         self.test_raw_data_ = X
-        corr = self._prepare_correlation_data(X1, X2)
-        normalized = self._normalize(corr, len(X1))
-        self.test_data_ = self._prepare_test_data(normalized)
+        with stage_timer("classifier test data", logger,
+                         sync_device=self.device):
+            corr = self._prepare_correlation_data(X1, X2)
+            normalized = self._normalize(corr, len(X1))
+            self.test_data_ = self._prepare_test_data(normalized)
 
     def _prepare_test_data(self, corr_data):
         n_test = corr_data.shape[0]

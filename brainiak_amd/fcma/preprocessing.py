@@ -23,6 +23,7 @@ import numpy as np
 
 from ..image import mask_images, multimask_images
 from ..parallel import DistContext
+from ..utils.timing import stage_timer
 
 logger = logging.getLogger(__name__)
 
@@ -116,7 +117,8 @@ def prepare_fcma_data(images, conditions, mask1, mask2=None,
     ctx = comm if isinstance(comm, DistContext) else DistContext()
     results = ([], [], [])            # raw_data1, raw_data2, labels
     if ctx.is_root:
-        logger.info('start to apply masks and separate epochs')
+        timer = stage_timer("mask + epoch separation", logger)
+        timer.__enter__()
         if mask2 is None:
             per_mask = [list(mask_images(images, mask1, np.float32))]
         else:
@@ -134,6 +136,7 @@ def prepare_fcma_data(images, conditions, mask1, mask2=None,
         results = (epoched[0][0],
                    epoched[1][0] if 1 in epoched else [],
                    epoched[0][1])
+        timer.__exit__(None, None, None)
 
     if ctx.is_distributed:
         results = ctx.broadcast_object(results if ctx.is_root else None)

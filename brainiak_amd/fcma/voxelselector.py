@@ -23,6 +23,7 @@ import numpy as np
 import torch
 
 from ..parallel import DistContext
+from ..utils.timing import stage_timer
 from .core import CorrelationPipeline
 from .svm import cross_validate_voxels
 
@@ -119,16 +120,24 @@ class VoxelSelector:
                 start += count
             scores = None
             if use_gpu_svm:
-                scores = self._overlapped_cv(pipeline, chunks, C, tol)
+                with stage_timer("voxel selection (duo sweep + "
+                                 "overlapped CV)", logger,
+                                 sync_device=pipeline.device):
+                    scores = self._overlapped_cv(pipeline, chunks, C,
+                                                 tol)
             if scores is None:
-                kernels = pipeline.pipelined_kernel_matrices(chunks)
-                if use_gpu_svm:
-                    scores = cross_validate_voxels(kernels, self.labels,
-                                                   self.num_folds, C=C,
-                                                   tol=tol)
-                else:
-                    scores = self._sklearn_cv(clf,
-                                              kernels.cpu().numpy())
+                with stage_timer("correlation/Gram pipeline", logger,
+                                 sync_device=pipeline.device):
+                    kernels = pipeline.pipelined_kernel_matrices(chunks)
+                with stage_timer("cross validation", logger,
+                                 sync_device=pipeline.device):
+                    if use_gpu_svm:
+                        scores = cross_validate_voxels(
+                            kernels, self.labels, self.num_folds, C=C,
+                            tol=tol)
+                    else:
+                        scores = self._sklearn_cv(
+                            clf, kernels.cpu().numpy())
             results.extend((my.start + i, float(scores[i]))
                            for i in range(my.stop - my.start))
         else:

@@ -73,3 +73,23 @@ def test_prepare_mvpa_data(seeded_rng):
     assert labels.tolist() == [0.0, 0.0, 1.0, 1.0] * 2
     # within-subject z-scoring across epochs
     assert np.allclose(processed[:, :4].mean(axis=1), 0, atol=1e-5)
+
+
+def test_stage_timer_logs_and_profiles(tmp_path, caplog, monkeypatch):
+    """stage_timer logs a duration line and (with
+    BRAINIAK_TORCH_PROFILE) drops a chrome trace."""
+    import logging
+
+    from brainiak_amd.utils.timing import stage_timer
+    log = logging.getLogger("timing_test")
+    with caplog.at_level(logging.INFO, logger="timing_test"):
+        with stage_timer("unit stage", log):
+            pass
+    assert any("unit stage took" in r.message for r in caplog.records)
+
+    monkeypatch.setenv("BRAINIAK_TORCH_PROFILE", str(tmp_path))
+    with stage_timer("profiled stage", log):
+        import torch
+        torch.ones(4) @ torch.ones(4)
+    traces = list(tmp_path.glob("profiled_stage_*.json"))
+    assert len(traces) == 1 and traces[0].stat().st_size > 0
