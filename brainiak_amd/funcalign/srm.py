@@ -279,11 +279,18 @@ class SRM(_SRMBase):
 
             # ---- E-step (every rank runs the K×K math redundantly) ----
             rho0 = float((1.0 / rho2).sum())
-            chol_sigma_s = torch.linalg.cholesky(sigma_s)
-            inv_sigma_s = torch.cholesky_inverse(chol_sigma_s)
-            sigma_s_rhos = inv_sigma_s + eye * rho0
-            chol_sigma_s_rhos = torch.linalg.cholesky(sigma_s_rhos)
-            inv_sigma_s_rhos = torch.cholesky_inverse(chol_sigma_s_rhos)
+            # the K x K inversions run on the HOST in fp64: at K~50 a
+            # hipSOLVER cholesky costs ~1 ms of launch overhead per
+            # call (4 calls/iteration) vs ~50 us in numpy, and the
+            # matrices are 10 KB round trips
+            sig_h = sigma_s.double().cpu().numpy()
+            inv_sig_h = np.linalg.inv(sig_h)
+            inv_rhos_h = np.linalg.inv(
+                inv_sig_h + np.eye(K) * rho0)
+            inv_sigma_s = torch.as_tensor(inv_sig_h, dtype=dtype,
+                                          device=device)
+            inv_sigma_s_rhos = torch.as_tensor(inv_rhos_h, dtype=dtype,
+                                               device=device)
 
             # ---- local accumulation: sum_i W_i^T X_i / rho_i^2 ----
             wt_invpsi_x = torch.zeros((K, samples), dtype=dtype,
@@ -319,11 +326,15 @@ class SRM(_SRMBase):
             a_list = [x[s][:, :samples] @ shared_response.T
                       for s in local_idx]                    # [V_i, K]
             w_list = _polar_orthogonal_many(a_list, perturb=0.001)
-            for s, a, wi in zip(local_idx, a_list, w_list):
-                w[s] = wi
-                r = trace_xtx[s]
-                r += -2 * float((wi * a).sum())
-                r += trace_sigma_s
+            # ONE host transfer for all subjects' <W, A> traces (a
+            # per-subject .item() costs a stream sync each)
+            if a_list:
+                dots = torch.stack([(wi * a).sum() for a, wi
+                                    in zip(a_list, w_list)])
+                dots_h = dots.double().cpu().numpy()
+            for j, s in enumerate(local_idx):
+                w[s] = w_list[j]
+                r = trace_xtx[s] - 2 * dots_h[j] + trace_sigma_s
                 rho2_new[s] = r / (samples * voxels[s])
             rho2 = ctx.all_reduce(rho2_new, op="sum")
 
