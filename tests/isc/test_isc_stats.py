@@ -270,3 +270,45 @@ def test_phase_randomize_preserves_spectrum(seeded_rng):
     new_power = np.abs(np.fft.rfft(shuffled, axis=0)) ** 2
     assert np.allclose(orig_power, new_power, rtol=1e-8)
     assert not np.allclose(shuffled, data)
+
+
+def test_bootstrap_pairwise_excludes_self_pairs():
+    """A bootstrap draw that repeats a subject creates pairs of that
+    subject with itself; those rows must not contribute (NaN-excluded
+    from the summary statistic)."""
+    rng = np.random.RandomState(0)
+    n_subj, V = 6, 4
+    iscs = rng.rand(n_subj * (n_subj - 1) // 2, V) * 0.5
+    observed, ci, p, dist = bootstrap_isc(
+        iscs, pairwise=True, n_bootstraps=25, random_state=3)
+    assert dist.shape == (25, V)
+    # every bootstrap summary stays finite despite the NaN-marked
+    # duplicate pairs
+    assert np.all(np.isfinite(dist))
+    # Hall-Wilson: the distribution straddles the observed statistic
+    assert (dist < observed).any() and (dist > observed).any()
+
+
+def test_permutation_two_sample_exact_small_groups():
+    """Two-sample exact test: permutation space 4! = 24 <= n_perm."""
+    rng = np.random.RandomState(1)
+    iscs = np.vstack([rng.rand(2, 3) + 0.5, rng.rand(2, 3) - 0.5])
+    groups = [0, 0, 1, 1]
+    obs, p, dist = permutation_isc(iscs, group_assignment=groups,
+                                   n_permutations=100, random_state=0)
+    assert dist.shape[0] == 24
+    # deterministic regardless of seed in the exact regime
+    obs2, p2, dist2 = permutation_isc(iscs, group_assignment=groups,
+                                      n_permutations=100,
+                                      random_state=77)
+    assert np.array_equal(dist, dist2) and np.array_equal(p, p2)
+
+
+def test_squareform_isfc_roundtrip_with_stack():
+    rng = np.random.RandomState(2)
+    mats = rng.rand(3, 5, 5)
+    mats = (mats + np.transpose(mats, (0, 2, 1))) / 2
+    cond, diag = squareform_isfc(mats)
+    assert cond.shape == (3, 10) and diag.shape == (3, 5)
+    back = squareform_isfc(cond, diag)
+    assert np.allclose(back, mats)
