@@ -108,6 +108,17 @@ def main():
         z_fp8=args.fp8 if args.fp8 else None)
     del raw
 
+    # overlapped CV: per-chunk SMO launches ride a side stream while
+    # the duo sweep continues (same scheme as VoxelSelector.run)
+    cv_plan = cv_stream = None
+    if device.type == "cuda" and not args.no_cv             and pipeline._hip_path and pipeline._raw_split             and not os.environ.get("BRAINIAK_NO_DUO")             and not os.environ.get("BRAINIAK_NO_CV_OVERLAP"):
+        from brainiak_amd.fcma.core import _shrink_
+        from brainiak_amd.fcma.svm import FoldPlan, svm_cv_device
+        cv_plan = FoldPlan(labels, args.num_folds, device)
+        cv_stream = torch.cuda.Stream(
+            device=device,
+            priority=int(os.environ.get("BRAINIAK_CV_PRIO", "0")))
+
     # this rank's voxel range rotates so steps touch different voxels
     vps = min(args.voxels_per_step or V, V)
     chunk = min(args.chunk, vps)
@@ -129,6 +140,22 @@ def main():
             count = min(count, V - start)
             chunks.append((start, count))
             done += count
+        if cv_plan is not None:
+            cv_tol = 1e-2 if args.fp8 else 1e-3
+            accs = []
+
+            def consume(g, start, count):
+                ev = torch.cuda.Event()
+                ev.record(torch.cuda.current_stream(device))
+                with torch.cuda.stream(cv_stream):
+                    cv_stream.wait_event(ev)
+                    _shrink_(g)
+                    accs.append(svm_cv_device(g, cv_plan, 1.0, cv_tol))
+                    g.record_stream(cv_stream)
+
+            pipeline._duo_pipeline(chunks, consumer=consume)
+            torch.cuda.current_stream(device).wait_stream(cv_stream)
+            return float(torch.cat(accs).mean().cpu())
         kernels = pipeline.pipelined_kernel_matrices(chunks)
         if not args.no_cv:
             if device.type == "cuda":
