@@ -37,13 +37,10 @@ __global__ __launch_bounds__(64) void k_svm_cv(
     const int n = n_train[f];
     const int m = n_test[f];
 
-    // fp16 Q throughout (0.05 % rel on the z-scored Gram entries):
-    // at VPL=1 the tile drops 16.9 -> 8.4 KB, doubling resident QPs
-    // per CU — the kernel is a serial SMO chain per wave, so
-    // throughput scales with residency (measured 4.4 -> ~2 ms for a
-    // whole-brain CV pass).  The diagonal/eta arithmetic and all
-    // alpha/gradient state stay fp32.
-    using QT = _Float16;
+    // fp32 Q for the one-variable-per-lane case; fp16 Q (0.05 % rel)
+    // for n <= 128 keeps the tile at 33 KB — under the 64 KB dynamic
+    // LDS launch limit and 4 QPs resident per CU
+    using QT = typename std::conditional<VPL == 1, float, _Float16>::type;
     extern __shared__ char smem_raw[];
     QT (*Q)[MAXN + 1] = (QT (*)[MAXN + 1])smem_raw;
     float* ys = (float*)(smem_raw + sizeof(QT) * MAXN * (MAXN + 1));
@@ -184,7 +181,7 @@ extern "C" void launch_svm_cv(const float* kernels, const float* y,
                               float tol, int max_iter, ll max_n,
                               hipStream_t stream) {
     if (max_n <= 64) {
-        size_t smem = (size_t)64 * 65 * 2 + 2 * 64 * 4;
+        size_t smem = (size_t)64 * 65 * 4 + 2 * 64 * 4;
         hipLaunchKernelGGL(k_svm_cv<1>, dim3((unsigned)(C * F)), dim3(64),
                            smem, stream, kernels, y, train_idx, test_idx,
                            n_train, n_test, correct, C, E, F, Creg, tol,
