@@ -112,7 +112,6 @@ def main():
     # the duo sweep continues (same scheme as VoxelSelector.run)
     cv_plan = cv_stream = None
     if device.type == "cuda" and not args.no_cv             and pipeline._hip_path and pipeline._raw_split             and not os.environ.get("BRAINIAK_NO_DUO")             and not os.environ.get("BRAINIAK_NO_CV_OVERLAP"):
-        from brainiak_amd.fcma.core import _shrink_
         from brainiak_amd.fcma.svm import FoldPlan, svm_cv_device
         cv_plan = FoldPlan(labels, args.num_folds, device)
         cv_stream = torch.cuda.Stream(
@@ -149,7 +148,6 @@ def main():
                 ev.record(torch.cuda.current_stream(device))
                 with torch.cuda.stream(cv_stream):
                     cv_stream.wait_event(ev)
-                    _shrink_(g)
                     accs.append(svm_cv_device(g, cv_plan, 1.0, cv_tol))
                     g.record_stream(cv_stream)
 

@@ -345,8 +345,9 @@ class CorrelationPipeline:
         chunk order, stream-ordered on the current stream) and the
         grams are NOT accumulated — the pipeline returns None.  This
         is how the voxel selector overlaps the per-chunk SVM CV with
-        the remaining duo sweep.  shrink is left to the consumer in
-        that mode."""
+        the remaining duo sweep.  Grams arrive already shrunk when
+        shrink=True (the partial-sum + shrink of chunk i-2 rides the
+        duo grid of launch i as a third block population)."""
         ext = ops.load_extension()
         E = self.num_epochs
         Epad = ((E + 63) // 64) * 64
@@ -366,7 +367,18 @@ class CorrelationPipeline:
         nsplit = min(nsplit, (VB + 127) // 128)
 
         grams = []
+
+        def emit(g, start_v, count_v):
+            if consumer is not None:
+                consumer(g, start_v, count_v)
+            else:
+                grams.append(g)
+
+        def slice_epochs(g):
+            return g if Epad == E else g[:, :E, :E].contiguous()
+
         prev = None                     # (z, count)
+        pending = []                    # [(gp, count, start)] FIFO
         for i, (start, count) in enumerate(chunks):
             z_i = self._zbuf[i % 2]
             if prev is None:
@@ -377,27 +389,37 @@ class CorrelationPipeline:
                 gp = torch.empty((nsplit, cp, Epad, Epad),
                                  dtype=torch.float32,
                                  device=self.device)
+                gout = None
+                extra = {}
+                if pending:
+                    # chunk i-2's partials are complete: their
+                    # reduction + magnitude shrink join this launch's
+                    # grid as a third block population
+                    pgp, pcnt, pstart = pending.pop(0)
+                    gout = torch.empty((pcnt, Epad, Epad),
+                                       dtype=torch.float32,
+                                       device=self.device)
+                    extra = dict(Gsum_part=pgp, Gsum_out=gout,
+                                 shrink=shrink)
                 ext.fcma_corr_gram_duo(self.data, self.data2, start,
                                        count, P, z_i.narrow(0, 0, count),
-                                       Zprev=zp, Gpart=gp)
-                g = gp.sum(0)
-                if Epad != E:
-                    g = g[:, :E, :E].contiguous()
-                if consumer is not None:
-                    consumer(g, *chunks[i - 1])
-                else:
-                    grams.append(g)
+                                       Zprev=zp, Gpart=gp, **extra)
+                if gout is not None:
+                    emit(slice_epochs(gout), pstart, pcnt)
+                pending.append((gp, cp, chunks[i - 1][0]))
             prev = (z_i.narrow(0, 0, count), count)
-        # trailing gram for the last chunk
+        # tail: at most one chunk's partials left, plus the final
+        # chunk's gram (no following launch to ride)
+        for pgp, pcnt, pstart in pending:
+            g = pgp.sum(0)
+            if shrink:
+                _shrink_(g)
+            emit(slice_epochs(g), pstart, pcnt)
         zp, cp = prev
         g = ops.fcma_gram_bf16(zp, norm_P=P)
-        if Epad != E:
-            g = g[:, :E, :E].contiguous()
-        if consumer is not None:
-            consumer(g, *chunks[-1])
-            return None
-        grams.append(g)
-        gram = torch.cat(grams, dim=0)
         if shrink:
-            _shrink_(gram)
-        return gram
+            _shrink_(g)
+        emit(slice_epochs(g), chunks[-1][0], cp)
+        if consumer is not None:
+            return None
+        return torch.cat(grams, dim=0)

@@ -173,7 +173,6 @@ class VoxelSelector:
                 or os.environ.get("BRAINIAK_NO_DUO")
                 or os.environ.get("BRAINIAK_NO_CV_OVERLAP")):
             return None
-        from .core import _shrink_
         from .svm import FoldPlan, svm_cv_device
         try:
             plan = FoldPlan(self.labels, self.num_folds,
@@ -192,7 +191,6 @@ class VoxelSelector:
             ev.record(torch.cuda.current_stream(dev))
             with torch.cuda.stream(cv_stream):
                 cv_stream.wait_event(ev)
-                _shrink_(g)
                 accs.append(svm_cv_device(g, plan, C, tol))
                 # g was allocated on the main stream; keep the
                 # allocator from recycling it under the CV reads

@@ -28,7 +28,8 @@ void launch_fcma_gram_bf16_norm(const void*, float*, ll, ll, ll, ll,
                                 int, void*);
 void launch_fcma_corr_gram_duo(const void*, const void*, void*, ll, ll,
                                ll, ll, ll, int, const void*, float*,
-                               ll, ll, ll, ll, void*);
+                               ll, ll, ll, ll, const float*, float*,
+                               ll, ll, int, void*);
 void launch_fp8_cvt_probe_sw(const float*, void*, ll, void*);
 void launch_fcma_corr_norm_z8(const void*, const void*, void*, ll, ll,
                               ll, ll, ll, int, void*);
@@ -212,7 +213,10 @@ void fcma_corr_gram_duo(torch::Tensor A, torch::Tensor B,
                         int64_t start, int64_t count, int64_t P,
                         torch::Tensor Zout,
                         c10::optional<torch::Tensor> Zprev,
-                        c10::optional<torch::Tensor> Gpart) {
+                        c10::optional<torch::Tensor> Gpart,
+                        c10::optional<torch::Tensor> Gsum_part,
+                        c10::optional<torch::Tensor> Gsum_out,
+                        bool shrink) {
     // one grid = raw-corr blocks for [start, start+count) + Gram
     // (+in-register normalize) blocks for the PREVIOUS chunk's Z
     check_3d(A, torch::kBFloat16, "A");
@@ -248,10 +252,36 @@ void fcma_corr_gram_duo(torch::Tensor A, torch::Tensor B,
         zprev_ptr = Zp.data_ptr();
         g_ptr = G.data_ptr<float>();
     }
+    const float* gps_ptr = nullptr;
+    float* gso_ptr = nullptr;
+    ll Cs = 0, nsplit_sum = 1;
+    if (Gsum_out.has_value()) {
+        TORCH_CHECK(Gsum_part.has_value(), "Gsum_out needs Gsum_part");
+        auto& Gp = Gsum_part.value();
+        auto& Go = Gsum_out.value();
+        TORCH_CHECK(Gp.is_cuda() && Gp.is_contiguous() && Gp.dim() == 4
+                    && Gp.scalar_type() == torch::kFloat32,
+                    "bad Gsum_part");
+        TORCH_CHECK(Go.is_cuda() && Go.is_contiguous() && Go.dim() == 3
+                    && Go.scalar_type() == torch::kFloat32,
+                    "bad Gsum_out");
+        Cs = Go.size(0);
+        nsplit_sum = Gp.size(0);
+        TORCH_CHECK(Gp.size(2) == Go.size(1)
+                    && Gp.size(3) == Go.size(2)
+                    && (Eg == 0 || Go.size(1) == Eg),
+                    "Gsum shapes mismatch");
+        // gsum_body strides by the partials' voxel-row count
+        TORCH_CHECK(Gp.size(1) == Cs,
+                    "Gsum_part rows must equal Gsum_out rows");
+        gps_ptr = Gp.data_ptr<float>();
+        gso_ptr = Go.data_ptr<float>();
+    }
     launch_fcma_corr_gram_duo(At.data_ptr(), B.data_ptr(),
                               Zout.data_ptr(), E, L, VB, count, Eout,
                               (int)P, zprev_ptr, g_ptr, Cg, Eg, Vg,
-                              nsplit, cur_stream());
+                              nsplit, gps_ptr, gso_ptr, Cs, nsplit_sum,
+                              shrink ? 1 : 0, cur_stream());
 }
 
 torch::Tensor debug_fp8_cvt(torch::Tensor x, bool sw) {
@@ -538,12 +568,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fcma_gram_fp8", &fcma_gram_fp8,
           "per-voxel Gram from fp8(e4m3) Z [C,E,V]");
     m.def("fcma_corr_gram_duo", &fcma_corr_gram_duo,
-          "one-grid raw-corr(chunk i) + gram(chunk i-1) co-residency",
+          "one-grid raw-corr(chunk i) + gram(chunk i-1) + partial-sum/"
+          "shrink(chunk i-2) co-residency",
           pybind11::arg("A"), pybind11::arg("B"), pybind11::arg("start"),
           pybind11::arg("count"), pybind11::arg("P"),
           pybind11::arg("Zout"),
           pybind11::arg("Zprev") = pybind11::none(),
-          pybind11::arg("Gpart") = pybind11::none());
+          pybind11::arg("Gpart") = pybind11::none(),
+          pybind11::arg("Gsum_part") = pybind11::none(),
+          pybind11::arg("Gsum_out") = pybind11::none(),
+          pybind11::arg("shrink") = true);
     m.def("fcma_gram_bf16", &fcma_gram_bf16,
           "per-voxel Gram from bf16 Z [C,E,V]; norm_P>0 applies "
           "Fisher-z + z-score to raw correlations in-tile",

@@ -1370,6 +1370,38 @@ __global__ __launch_bounds__(256) void k_gram_bf16(
 }
 
 // ===========================================================================
+// gsum_body: one block per selected voxel c — sums the nsplit
+// partial Gram slabs of a FINISHED chunk into the final [E, E]
+// matrix and applies the reference's magnitude shrink in the same
+// pass.  Folding this into the duo grid removes the per-chunk torch
+// reduce + shrink launches (~3.3 ms serial per whole-brain step) and
+// hides the 143 MB partial re-read under the latency-bound corr/gram
+// waves.
+__device__ __forceinline__ void gsum_body(
+    ll b, const float* __restrict__ Gp, float* __restrict__ Gout,
+    ll Cs, ll EE, ll nsplit, int do_shrink) {
+    const ll c = b;
+    if (c >= Cs) return;
+    const float* base = Gp + c * EE;
+    const ll cstride = Cs * EE;
+    float scale = 1.0f;
+    if (do_shrink) {
+        float lead = 0.0f;
+        for (ll s = 0; s < nsplit; ++s)
+            lead += base[s * cstride];
+        lead = fmaxf(fabsf(lead), 1.0f);
+        float digits = floorf(log10f(lead)) + 1.0f;
+        if (digits > 2.0f)
+            scale = powf(10.0f, 2.0f - digits);
+    }
+    for (ll el = threadIdx.x; el < EE; el += 256) {
+        float acc = 0.0f;
+        for (ll s = 0; s < nsplit; ++s)
+            acc += base[s * cstride + el];
+        Gout[c * EE + el] = acc * scale;
+    }
+}
+
 // k_corr_gram_duo: ONE grid carrying BOTH the raw-correlation blocks of
 // chunk i and the Gram(+normalize) blocks of chunk i-1, proportionally
 // interleaved over blockIdx.  HIP streams do not co-schedule these two
@@ -1383,13 +1415,26 @@ __global__ __launch_bounds__(256) void k_corr_gram_duo(
     bf16_t* __restrict__ zOut, ll E, ll VB, ll C, ll zstride,
     const bf16_t* __restrict__ Zprev, float* __restrict__ G,
     ll Cg, ll Eg, ll Vg, ll nsplit,
-    ll nCorr, ll nGram) {
-    const ll b = blockIdx.x;
-    const ll total = nCorr + nGram;
-    // proportional interleave: block b is a gram block iff the
-    // cumulative gram quota advances at b
-    const ll g_before = (b * nGram) / total;
-    const ll g_at = ((b + 1) * nGram) / total;
+    const float* __restrict__ GpSum, float* __restrict__ GSumOut,
+    ll Cs, ll nsplitSum, int do_shrink,
+    ll nCorr, ll nGram, ll nSum) {
+    ll b = blockIdx.x;
+    const ll total = nCorr + nGram + nSum;
+    // proportional interleave over THREE populations: first peel the
+    // gsum quota (chunk i-2 partial reduction), then split the rest
+    // between gram (chunk i-1) and corr (chunk i) blocks
+    const ll s_before = (b * nSum) / total;
+    const ll s_at = ((b + 1) * nSum) / total;
+    if (s_at != s_before) {
+        if (GSumOut != nullptr)
+            gsum_body(s_before, GpSum, GSumOut, Cs, Eg * Eg,
+                      nsplitSum, do_shrink);
+        return;
+    }
+    b -= s_before;
+    const ll rest = nCorr + nGram;
+    const ll g_before = (b * nGram) / rest;
+    const ll g_at = ((b + 1) * nGram) / rest;
     if (g_at != g_before) {
         if (G != nullptr)
             gram_bf16_body<NP, DZ>(g_before, Zprev, G, Cg, Eg, Vg,
@@ -2188,11 +2233,13 @@ extern "C" void launch_fcma_corr_gram_duo(
     const void* At, const void* B, void* zOut, ll E, ll L, ll VB, ll C,
     ll zstride, int P,
     const void* Zprev, float* G, ll Cg, ll Eg, ll Vg, ll nsplit,
-    hipStream_t stream) {
+    const float* GpSum, float* GSumOut, ll Cs, ll nsplitSum,
+    int do_shrink, hipStream_t stream) {
     ll nCorr = duo_corr_blocks(C, E, P, VB);
     ll nGram = (G != nullptr) ? fcma_duo_gram_blocks(Cg, Eg, nsplit)
                               : 0;
-    ll grid = nCorr + nGram;
+    ll nSum = (GSumOut != nullptr) ? Cs : 0;
+    ll grid = nCorr + nGram + nSum;
     const bool dz = (Eg != 64);
     const bool ct64 = duo_ct() == 64;
     #define DUO_ONE(TP, TL, DZV, CTV)                                    \
@@ -2201,7 +2248,8 @@ extern "C" void launch_fcma_corr_gram_duo(
                            (const bf16_t*)At, (const bf16_t*)B,          \
                            (bf16_t*)zOut, E, VB, C, zstride,             \
                            (const bf16_t*)Zprev, G, Cg, Eg, Vg,          \
-                           nsplit, nCorr, nGram)
+                           nsplit, GpSum, GSumOut, Cs, nsplitSum,        \
+                           do_shrink, nCorr, nGram, nSum)
     #define DUO_CASE(TP, TL)                                             \
         do {                                                             \
             if (dz)         DUO_ONE(TP, TL, true, 128);                  \
