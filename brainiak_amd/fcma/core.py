@@ -391,7 +391,14 @@ class CorrelationPipeline:
                                  device=self.device)
                 gout = None
                 extra = {}
-                if pending:
+                if pending and os.environ.get("BRAINIAK_NO_GSUM"):
+                    # A/B fallback: torch partial-sum + shrink
+                    pgp, pcnt, pstart = pending.pop(0)
+                    g = pgp.sum(0)
+                    if shrink:
+                        _shrink_(g)
+                    emit(slice_epochs(g), pstart, pcnt)
+                elif pending:
                     # chunk i-2's partials are complete: their
                     # reduction + magnitude shrink join this launch's
                     # grid as a third block population
