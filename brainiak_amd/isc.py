@@ -43,6 +43,8 @@ from .utils.utils import (
 
 logger = logging.getLogger(__name__)
 
+MAX_RANDOM_SEED = 2 ** 32 - 1
+
 __all__ = [
     "bootstrap_isc",
     "compute_summary_statistic",

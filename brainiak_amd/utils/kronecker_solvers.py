@@ -18,6 +18,7 @@ import torch
 
 __all__ = [
     "kron_mult",
+    "masked_triangular_solve",
     "solve_lower_triangular_kron",
     "solve_lower_triangular_masked_kron",
     "solve_upper_triangular_kron",
@@ -75,6 +76,26 @@ def solve_upper_triangular_kron(L: List[torch.Tensor],
                                 X: torch.Tensor) -> torch.Tensor:
     """Solve (⊗ L_i)^T z = X (upper-triangular transposes)."""
     return _kron_matmul_solve([m.T for m in L], X, upper=True)
+
+
+def masked_triangular_solve(L: torch.Tensor, y: torch.Tensor,
+                            mask: torch.Tensor, lower: bool = True,
+                            adjoint: bool = False) -> torch.Tensor:
+    """Solve L x = y over the mask's valid rows/columns only
+    (ref ``tf_masked_triangular_solve``, kronecker_solvers.py:150):
+    the solve runs on the mask-selected submatrix and x is scattered
+    back with zeros at masked-out rows."""
+    idx = torch.nonzero(mask.reshape(-1), as_tuple=True)[0]
+    sub = L[idx][:, idx]
+    if adjoint:
+        sub = sub.T
+        lower = not lower
+    squeeze = y.dim() == 1
+    y2 = y.reshape(-1, 1) if squeeze else y
+    xs = torch.linalg.solve_triangular(sub, y2[idx], upper=not lower)
+    x = torch.zeros_like(y2)
+    x[idx] = xs
+    return x.reshape(y.shape)
 
 
 def _masked_dense(L: List[torch.Tensor], mask: torch.Tensor):
