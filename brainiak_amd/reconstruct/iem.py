@@ -276,6 +276,9 @@ class InvertedEncoding2D:
         """Estimate W from training data; C defaults to
         circular-stimulus channel activations built from y."""
         X = np.asarray(X)
+        if np.ndim(X) != 2:
+            raise ValueError("expected a 2-D [observations, voxels] data "
+                             "matrix")
         if self.channels is None:
             raise ValueError("no channel basis defined - call one of the "
                              "define_basis_functions_* methods first")

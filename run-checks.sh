@@ -14,6 +14,7 @@ for p in sorted(pathlib.Path('brainiak_amd').rglob('*.py')):
         print('SYNTAX', p, e); bad += 1
 sys.exit(1 if bad else 0)
 PY
+python scripts/check_api_parity.py
 for f in brainiak_amd/ops/hip/*.hip; do
     echo "hipcc -c $f"
     hipcc --offload-arch=gfx950 -O3 -std=c++17 -fsyntax-only "$f"
