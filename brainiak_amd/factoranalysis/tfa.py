@@ -572,6 +572,24 @@ class TFA:
 
     # -- fitting ---------------------------------------------------------------
 
+    def _prepare_fit(self, X, R, template_prior):
+        """Shared fit preamble (dims, offsets, bounds, subsample
+        scaling, prior init) — also used by HTFA's batched local
+        fit."""
+        self.n_dim = R.shape[1]
+        self.cov_vec_size = np.sum(np.arange(self.n_dim) + 1)
+        self.map_offset = self.get_map_offset()
+        self.bounds = self.get_bounds(R)
+        n_voxel, n_tr = X.shape
+        self.sample_scaling = 0.5 * float(
+            min(self.max_num_voxel, n_voxel)
+            * min(self.max_num_tr, n_tr)) / float(n_voxel * n_tr)
+        if template_prior is None:
+            self.init_prior(R)
+        else:
+            self.local_prior = template_prior[0:self.map_offset[2]]
+        return self
+
     def _fit_tfa(self, data, R, template_prior=None):
         if template_prior is None:
             template_centers = None
@@ -645,18 +663,7 @@ class TFA:
             raise ValueError(
                 "only 'rr' and 'ols' are accepted as weight_method!")
 
-        self.n_dim = R.shape[1]
-        self.cov_vec_size = np.sum(np.arange(self.n_dim) + 1)
-        self.map_offset = self.get_map_offset()
-        self.bounds = self.get_bounds(R)
-        n_voxel, n_tr = X.shape
-        self.sample_scaling = 0.5 * float(
-            min(self.max_num_voxel, n_voxel)
-            * min(self.max_num_tr, n_tr)) / float(n_voxel * n_tr)
-        if template_prior is None:
-            self.init_prior(R)
-        else:
-            self.local_prior = template_prior[0:self.map_offset[2]]
+        self._prepare_fit(X, R, template_prior)
         self._fit_tfa(X, R, template_prior)
         if template_prior is None:
             centers = self.get_centers(self.local_posterior_)

@@ -393,3 +393,48 @@ def test_mnrsa_gpu_runs(cuda, seeded_rng):
     m.fit(X=Y, y=design)
     assert m.U_.shape == (n_c, n_c)
     assert np.isfinite(m.U_).all()
+
+
+@pytest.mark.gpu
+def test_htfa_batched_local_matches_sequential(cuda, seeded_rng):
+    """HTFA's batched-across-subjects LM == the sequential per-subject
+    path on recovery quality (same per-subject math; fp reduction
+    order differs, so compare the recovered template, not bits)."""
+    import os
+
+    from brainiak_amd.factoranalysis.htfa import HTFA
+    rng = seeded_rng
+    K, V, T, S = 4, 3000, 40, 3
+    centers_true = rng.rand(K, 3) * 30
+    X, R = [], []
+    for _ in range(S):
+        coords = rng.rand(V, 3) * 30
+        d2 = ((coords[:, None, :] - centers_true[None, :, :]) ** 2
+              ).sum(-1)
+        F = np.exp(-d2 / 25.0)
+        W = rng.randn(K, T)
+        X.append(F @ W + 0.05 * rng.randn(V, T))
+        R.append(coords)
+
+    def fit(seq):
+        if seq:
+            os.environ["BRAINIAK_HTFA_SEQ"] = "1"
+        try:
+            h = HTFA(K=K, n_subj=S, max_global_iter=2,
+                     max_local_iter=2, device="cuda")
+            h.fit([x.copy() for x in X], [r.copy() for r in R])
+            return h.get_centers(h.global_posterior_)
+        finally:
+            os.environ.pop("BRAINIAK_HTFA_SEQ", None)
+
+    cb = fit(seq=False)
+    cs = fit(seq=True)
+    from scipy.spatial.distance import cdist
+    from scipy.optimize import linear_sum_assignment
+    cost = cdist(cb, cs)
+    rr, cc = linear_sum_assignment(cost)
+    assert cost[rr, cc].max() < 2.0, cost[rr, cc]
+    # and both track the planted centers
+    cost_t = cdist(cb, centers_true)
+    rr, cc = linear_sum_assignment(cost_t)
+    assert np.median(cost_t[rr, cc]) < 3.0, cost_t[rr, cc]
