@@ -134,19 +134,49 @@ class TFA:
         The clustering only SEEDS the NLSS (centers are refined every
         iteration), so it runs single-init on a <=8k-voxel subsample —
         full-brain 10-restart Lloyd was 32 % of an entire HTFA fit
-        (profiles/htfa_cprofile.txt) for identical end fits."""
-        from sklearn.cluster import KMeans
+        (profiles/htfa_cprofile.txt) for identical end fits.  On GPU
+        the Lloyd loop itself runs in torch (sklearn's CPU kmeans was
+        still ~40 % of a 100k-voxel HTFA global iteration)."""
         pts = R
-        if R.shape[0] > 20000:
+        if R.shape[0] > 8000:
             sel = np.random.RandomState(100).choice(
-                R.shape[0], 20000, replace=False)
+                R.shape[0], 8000, replace=False)
             pts = R[sel]
-        kmeans = KMeans(init='k-means++', n_clusters=self.K, n_init=1,
-                        max_iter=50, random_state=100)
-        kmeans.fit(pts)
-        centers = kmeans.cluster_centers_
+        if self._use_gpu():
+            centers = self._kmeans_torch(pts, self.K, seed=100)
+        else:
+            from sklearn.cluster import KMeans
+            kmeans = KMeans(init='k-means++', n_clusters=self.K,
+                            n_init=1, max_iter=50, random_state=100)
+            kmeans.fit(pts)
+            centers = kmeans.cluster_centers_
         widths = self._get_max_sigma(R) * np.ones((self.K, 1))
         return centers, widths
+
+    @staticmethod
+    def _kmeans_torch(pts, K, seed=100, iters=25):
+        """Seeding-quality Lloyd on device: k-means++ init (numpy,
+        deterministic) then fixed-iteration assignments in torch."""
+        rng = np.random.RandomState(seed)
+        P = np.asarray(pts, dtype=np.float64)
+        # k-means++ seeding
+        centers = [P[rng.randint(len(P))]]
+        d2 = ((P - centers[0]) ** 2).sum(1)
+        for _ in range(K - 1):
+            probs = d2 / d2.sum()
+            centers.append(P[rng.choice(len(P), p=probs)])
+            d2 = np.minimum(d2, ((P - centers[-1]) ** 2).sum(1))
+        C = torch.as_tensor(np.asarray(centers), dtype=torch.float32,
+                            device="cuda")
+        X = torch.as_tensor(P, dtype=torch.float32, device="cuda")
+        for _ in range(iters):
+            dist = torch.cdist(X, C)
+            assign = dist.argmin(dim=1)
+            one_hot = torch.zeros((X.shape[0], K), device="cuda")
+            one_hot.scatter_(1, assign[:, None], 1.0)
+            counts = one_hot.sum(0).clamp_min(1.0)
+            C = (one_hot.T @ X) / counts[:, None]
+        return C.double().cpu().numpy()
 
     def get_template(self, R):
         """Template prior (centers | widths | centers cov | widths var)."""
