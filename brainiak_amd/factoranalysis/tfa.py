@@ -169,9 +169,12 @@ class TFA:
         C = torch.as_tensor(np.asarray(centers), dtype=torch.float32,
                             device="cuda")
         X = torch.as_tensor(P, dtype=torch.float32, device="cuda")
+        x_sq = (X * X).sum(1, keepdim=True)
         for _ in range(iters):
-            dist = torch.cdist(X, C)
-            assign = dist.argmin(dim=1)
+            # expanded-norm distances: torch.cdist measured 5 ms/call
+            # at [8k, 3] x [K, 3] vs ~50 us for the GEMM form
+            d2 = x_sq + (C * C).sum(1)[None, :] - 2.0 * (X @ C.T)
+            assign = d2.argmin(dim=1)
             one_hot = torch.zeros((X.shape[0], K), device="cuda")
             one_hot.scatter_(1, assign[:, None], 1.0)
             counts = one_hot.sum(0).clamp_min(1.0)
@@ -609,7 +612,15 @@ class TFA:
         self.n_dim = R.shape[1]
         self.cov_vec_size = np.sum(np.arange(self.n_dim) + 1)
         self.map_offset = self.get_map_offset()
-        self.bounds = self.get_bounds(R)
+        # bounds depend only on R; HTFA re-enters here every global
+        # iteration with the same coordinate matrix (np.std over the
+        # full [V, 3] array per subject per iteration otherwise)
+        cache = getattr(self, "_bounds_cache", None)
+        if cache is not None and cache[0] is R:
+            self.bounds = cache[1]
+        else:
+            self.bounds = self.get_bounds(R)
+            self._bounds_cache = (R, self.bounds)
         n_voxel, n_tr = X.shape
         self.sample_scaling = 0.5 * float(
             min(self.max_num_voxel, n_voxel)
