@@ -7,6 +7,7 @@ One step = fitting ``--regions`` independent EventSegment models
 (ragged lengths grouped by the batched forward-backward)."""
 
 import argparse
+import os
 import sys
 from pathlib import Path
 
@@ -45,10 +46,16 @@ def main():
 
     dev = str(device) if device.type == "cuda" else "cpu"
 
+    batched = not os.environ.get("BRAINIAK_EVENTSEG_SEQ")
+
     def step(i):
-        for d in data:
+        if batched:
             EventSegment(args.events, n_iter=10,
-                         device=dev).fit(d.copy())
+                         device=dev).fit_regions(data)
+        else:
+            for d in data:
+                EventSegment(args.events, n_iter=10,
+                             device=dev).fit(d.copy())
 
     elapsed = timed_steps(step, args.steps, args.warmup, world, device)
     fits_per_sec = regions * world * args.steps / elapsed

@@ -334,6 +334,113 @@ class EventSegment:
             logger.debug("Fitting step %d, LL=%f", step, best_ll)
         return self
 
+    # -- batched independent fits (MI355X addition) ------------------------
+
+    def _logprob_obs_batch_perregion(self, data, mean_pat, var):
+        """Like _logprob_obs_batch but with PER-REGION patterns:
+        data [B, V, T], mean_pat [B, V, K] -> [B, T, K]."""
+        B, V, T = data.shape
+        K = mean_pat.shape[2]
+        v = torch.as_tensor(
+            np.broadcast_to(np.asarray(var, float), (K,)).copy(),
+            dtype=torch.float64, device=data.device)
+        dz = _zscore_rows(data, dim=1)
+        mz = _zscore_rows(mean_pat, dim=1)
+        d2 = (dz * dz).sum(dim=1)                      # [B, T]
+        m2 = (mz * mz).sum(dim=1)                      # [B, K]
+        cross = torch.bmm(dz.transpose(1, 2), mz)      # [B, T, K]
+        sq = d2[:, :, None] + m2[:, None, :] - 2.0 * cross
+        lp = -0.5 * (V * torch.log(2 * torch.pi * v)[None, None, :]
+                     + sq / v[None, None, :])
+        return lp / V
+
+    def fit_regions(self, datasets):
+        """Fit INDEPENDENT segmentations for many regions in one
+        batched EM (MI355X addition — no reference counterpart).
+
+        The per-region model is exactly ``fit`` on that region alone
+        (same annealing schedule, same per-region early stop when the
+        log-likelihood decreases); regions of equal (T, V) shape share
+        each batched observation/forward-backward/M-step contraction,
+        which is what makes many small searchlight-sized regions
+        GPU-viable (one [B, T, K] recursion instead of B kernel-launch
+        -bound fits).  split_merge is not supported here.
+
+        Parameters
+        ----------
+        datasets : list of [T_r, V_r] arrays (TRs by voxels).
+
+        Returns
+        -------
+        list of fitted EventSegment instances (segments_, event_pat_,
+        ll_, event_var_ populated), one per region, in input order.
+        """
+        if self.split_merge:
+            raise ValueError("fit_regions does not support split_merge")
+        dev = torch.device(self.device)
+        K = self.n_events
+        models = [EventSegment(self.n_events, step_var=self.step_var,
+                               n_iter=self.n_iter, device=self.device)
+                  for _ in datasets]
+        # group regions by (T, V) so each group is one dense batch
+        groups = {}
+        for i, d in enumerate(datasets):
+            d2 = _as_valid_2d(d)
+            groups.setdefault(d2.shape, []).append(i)
+        for (T, V), idxs in groups.items():
+            B = len(idxs)
+            X = torch.stack([
+                _zscore_rows(torch.as_tensor(
+                    np.asarray(datasets[i], dtype=np.float64).T,
+                    device=dev), dim=1)
+                for i in idxs])                        # [B, V, T]
+            log_gamma = torch.zeros((B, T, K), dtype=torch.float64,
+                                    device=dev)
+            best_ll = np.full(B, -np.inf)
+            ll_hist = [[] for _ in range(B)]
+            active = np.ones(B, dtype=bool)
+            frozen_gamma = log_gamma.clone()
+            frozen_pat = torch.zeros((B, V, K), dtype=torch.float64,
+                                     device=dev)
+            frozen_var = np.zeros(B)
+            for step in range(1, self.n_iter + 1):
+                if not active.any():
+                    break
+                iteration_var = self.step_var(step)
+                g = torch.exp(log_gamma)
+                g = g / g.sum(dim=1, keepdim=True).clamp_min(1e-300)
+                mean_pat = torch.bmm(X, g)             # [B, V, K]
+                lp = self._logprob_obs_batch_perregion(
+                    X, mean_pat, iteration_var)
+                lg_new, ll_t = self._forward_backward_batch(lp)
+                ll = ll_t.cpu().numpy()
+                improved = active & (ll >= best_ll)
+                stopped = active & ~improved
+                active = improved.copy()
+                if improved.any():
+                    m = torch.as_tensor(improved, device=dev)
+                    log_gamma = torch.where(m[:, None, None], lg_new,
+                                            log_gamma)
+                    frozen_gamma = torch.where(m[:, None, None],
+                                               lg_new, frozen_gamma)
+                    frozen_pat = torch.where(m[:, None, None],
+                                             mean_pat, frozen_pat)
+                    frozen_var[improved] = iteration_var
+                    best_ll[improved] = ll[improved]
+                    for b in np.nonzero(improved)[0]:
+                        ll_hist[b].append(ll[b])
+                del stopped
+            seg = torch.exp(frozen_gamma).cpu().numpy()
+            pat = frozen_pat.cpu().numpy()
+            for j, i in enumerate(idxs):
+                mdl = models[i]
+                mdl.classes_ = np.arange(K)
+                mdl.segments_ = [seg[j]]
+                mdl.event_pat_ = pat[j]
+                mdl.event_var_ = frozen_var[j]
+                mdl.ll_ = np.asarray(ll_hist[j])[:, None]
+        return models
+
     # -- inference ---------------------------------------------------------
 
     def set_event_patterns(self, event_pat):

@@ -200,3 +200,45 @@ def test_logprob_normalization_constant(seeded_rng):
     # doubling the variance raises logprob of far points
     lp2 = es._logprob_obs(d, pat, 4.0)
     assert np.all(np.isfinite(lp1)) and np.all(np.isfinite(lp2))
+
+
+def test_fit_regions_matches_individual_fits(seeded_rng):
+    """Batched independent-region fit == per-region fit (same EM,
+    shared recursions; deterministic given data)."""
+    rng = seeded_rng
+    K, T, V = 4, 60, 30
+    regions = []
+    for _ in range(5):
+        bounds = np.sort(rng.choice(np.arange(1, T), K - 1,
+                                    replace=False))
+        means = rng.randn(K, V)
+        seg = np.zeros((T, V))
+        prev = 0
+        for e, b in enumerate(list(bounds) + [T]):
+            seg[prev:b] = means[e]
+            prev = b
+        regions.append(seg + 0.3 * rng.randn(T, V))
+
+    batched = EventSegment(K, n_iter=25).fit_regions(regions)
+    for d, mb in zip(regions, batched):
+        ms = EventSegment(K, n_iter=25).fit(d.copy())
+        assert np.allclose(mb.segments_[0], ms.segments_[0],
+                           atol=1e-6)
+        assert np.allclose(mb.event_pat_, ms.event_pat_, atol=1e-6)
+        assert np.isclose(mb.ll_[-1, 0], ms.ll_[-1].mean(), atol=1e-8)
+        # boundaries usable downstream
+        assert mb.segments_[0].shape == (T, K)
+
+
+def test_fit_regions_ragged_shapes(seeded_rng):
+    """Different (T, V) regions batch by shape group and come back in
+    input order."""
+    rng = seeded_rng
+    shapes = [(40, 20), (60, 25), (40, 20)]
+    regions = [rng.randn(t, v) for t, v in shapes]
+    models = EventSegment(3, n_iter=5).fit_regions(regions)
+    for (t, v), m in zip(shapes, models):
+        assert m.segments_[0].shape == (t, 3)
+
+    with pytest.raises(ValueError):
+        EventSegment(3, split_merge=True).fit_regions(regions)
