@@ -438,3 +438,21 @@ def test_htfa_batched_local_matches_sequential(cuda, seeded_rng):
     # recovery quality at 2x2 iterations is covered by the TFA
     # torch-LM recovery test, not this parity check)
     assert np.all(cb > -5) and np.all(cb < 35)
+
+
+@pytest.mark.gpu
+def test_eventseg_fit_regions_gpu_parity(cuda, seeded_rng):
+    """Batched region fit on GPU == per-region CPU fits."""
+    from brainiak_amd.eventseg.event import EventSegment
+    rng = seeded_rng
+    K, T, V = 4, 50, 24
+    regions = []
+    for _ in range(6):
+        means = rng.randn(K, V)
+        seg = np.repeat(means, [12, 13, 12, 13], axis=0)
+        regions.append(seg + 0.3 * rng.randn(T, V))
+    gpu_models = EventSegment(K, n_iter=15,
+                              device="cuda").fit_regions(regions)
+    for d, mg in zip(regions, gpu_models):
+        mc = EventSegment(K, n_iter=15, device="cpu").fit(d.copy())
+        assert np.allclose(mg.segments_[0], mc.segments_[0], atol=1e-5)
