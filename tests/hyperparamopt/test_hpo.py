@@ -59,3 +59,42 @@ def test_fmin_validates_dist():
     import pytest
     with pytest.raises(ValueError):
         fmin(lambda d: 0.0, {'x': {'dist': 42}}, 1, [])
+
+
+def test_gmm_matches_reference_oracles(seeded_rng):
+    """The reference's gmm_1d_distribution oracles: density ordering,
+    hard bounds, weight equivalence, scalar==array evaluation."""
+    x = np.array([1., 1., 2., 3., 1.])
+    d = gmm_1d_distribution(x, min_limit=0., max_limit=4.)
+    assert d(1.1) > d(3.5)
+    assert d(2.0) > d(3.0)
+    assert d(-1.0) == 0 and d(9.0) == 0
+    samples = d.get_samples(n=25)
+    assert np.all(samples < 4.) and np.all(samples > 0.)
+
+    # duplicate points == integer weights
+    x_dup = np.array([1., 1., 2., 3., 1., 3.])
+    d_dup = gmm_1d_distribution(x_dup)
+    d_w = gmm_1d_distribution(np.array([1., 2., 3.]),
+                              weights=np.array([3., 1., 2.]))
+    y = d_w(np.array([1.1, 2.0]))
+    assert d_w(1.1) == y[0]
+    assert abs(d_dup(1.1) - d_w(1.1)) < 1e-5
+    assert abs(d_dup(2.0) - d_w(2.0)) < 1e-5
+
+
+def test_fmin_continues_trials(seeded_rng):
+    """fmin resumes from an existing trials list (the reference's
+    continuation contract) and improves or keeps the best loss."""
+    def f(args):
+        return args['x'] ** 2
+
+    space = {'x': {'dist': st.uniform(loc=-10., scale=20),
+                   'lo': -10., 'hi': 10.}}
+    trials = []
+    best1 = fmin(loss_fn=f, space=space, max_evals=30, trials=trials)
+    n1 = len(trials)
+    best2 = fmin(loss_fn=f, space=space, max_evals=30, trials=trials)
+    assert len(trials) == n1 + 30
+    assert best2['loss'] <= best1['loss']
+    assert abs(best2['x']) < 2.5
