@@ -81,7 +81,7 @@ def test_from_masked_images_count_mismatch(seeded_rng):
         MaskedMultiSubjectData.from_masked_images(iter(imgs), 3)
 
 
-def test_single_condition_spec_labels():
+def test_single_condition_spec_partial_epochs():
     from brainiak_amd.image import SingleConditionSpec
     spec = np.zeros((2, 4, 10), dtype=bool)
     spec[0, 0, 0:3] = True
