@@ -121,3 +121,31 @@ def test_shard_slices_cover_and_balance():
         assert covered == list(range(n))
         sizes = [s.stop - s.start for s in sl]
         assert max(sizes) - min(sizes) <= 1
+
+
+def _ws4_roundtrip(ctx, out_dir):
+    import numpy as np
+
+    assert ctx.world_size == 4
+    # all-reduce across 4 ranks
+    v = ctx.all_reduce(np.array([float(ctx.rank + 1)]), op="sum")
+    assert float(v[0]) == 10.0
+    # shard covers the range exactly once
+    sl = ctx.shard(103)
+    spans = ctx.all_gather_object((sl.start, sl.stop))
+    covered = sorted(spans)
+    assert covered[0][0] == 0 and covered[-1][1] == 103
+    for (a, b), (c, d) in zip(covered, covered[1:]):
+        assert b == c
+    # broadcast from root
+    obj = ctx.broadcast_object({"k": 7} if ctx.is_root else None)
+    assert obj == {"k": 7}
+    if ctx.is_root:
+        (out_dir / "ok").write_text("1")
+
+
+def test_collectives_world4(tmp_path):
+    """4-rank gloo plumbing (the 8-GPU driver bench shape, scaled to
+    what CPU CI can run)."""
+    spawn_ranks(_ws4_roundtrip, world_size=4, args=(tmp_path,))
+    assert (tmp_path / "ok").exists()
