@@ -45,8 +45,13 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
     const ll J0 = ti * IT;
     if (ti >= tiles) return;
 
-    __shared__ float mt[IT][IT + IPAD];       // M[J0.., I0..] tile
-    __shared__ float zt[IT][IT + IPAD];       // atanh tile for mirror
+    // bf16 LDS tiles: fp32 tiles (33 KB) capped the kernel at 4
+    // blocks/CU = 16 waves; the kernel is memory-LATENCY bound (PMC:
+    // 85 % wait at 16 waves), so doubling residency buys ~the same
+    // factor.  z is rounded to bf16 once before the mirror
+    // accumulation (~0.4 % rel — the Fisher sums stay fp32).
+    __shared__ bf16_t mt[IT][IT + IPAD];      // M[J0.., I0..] tile
+    __shared__ bf16_t zt[IT][IT + IPAD];      // atanh tile for mirror
     const int tx = threadIdx.x & 63;
     const int ty = threadIdx.x >> 6;          // 4 rows per pass
     const bool diag = (I0 == J0);
@@ -64,7 +69,8 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
         for (int r = ty; r < IT; r += 4) {
             ll row = J0 + r, col = I0 + tx;
             mt[r][tx] = (row < V && col < V)
-                        ? (float)Mb[row * V + col] : 0.0f;
+                        ? (bf16_t)(float)Mb[row * V + col]
+                        : (bf16_t)0.0f;
         }
         __syncthreads();
         #pragma unroll
@@ -73,17 +79,17 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
             float z = 0.0f;
             if (row < V && col < V) {
                 float a = (float)Mb[row * V + col];
-                float sym = 0.5f * (a + mt[tx][r]);
+                float sym = 0.5f * (a + (float)mt[tx][r]);
                 z = atanh_clamped(sym);
                 zsum[q] += z;
             }
-            zt[r][tx] = z;
+            zt[r][tx] = (bf16_t)z;
         }
         if (!diag) {
             __syncthreads();
             #pragma unroll
             for (int r = ty, q = 0; r < IT; r += 4, ++q)
-                zsum_m[q] += zt[tx][r];
+                zsum_m[q] += (float)zt[tx][r];
         }
         __syncthreads();   // tiles reused next bm
     }
