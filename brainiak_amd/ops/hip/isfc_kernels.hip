@@ -45,13 +45,13 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
     const ll J0 = ti * IT;
     if (ti >= tiles) return;
 
-    // bf16 LDS tiles: fp32 tiles (33 KB) capped the kernel at 4
-    // blocks/CU = 16 waves; the kernel is memory-LATENCY bound (PMC:
-    // 85 % wait at 16 waves), so doubling residency buys ~the same
-    // factor.  z is rounded to bf16 once before the mirror
-    // accumulation (~0.4 % rel — the Fisher sums stay fp32).
-    __shared__ bf16_t mt[IT][IT + IPAD];      // M[J0.., I0..] tile
-    __shared__ bf16_t zt[IT][IT + IPAD];      // atanh tile for mirror
+    // ONE fp32 tile, two lives: the mirror tile is pre-read into
+    // registers (16 values/thread), then the same LDS backs the z
+    // tile for the transpose accumulation.  Two fp32 tiles (33 KB)
+    // capped the kernel at 4 blocks/CU = 16 waves and PMC showed
+    // 85 % memory-wait — halving LDS doubles residency with NO
+    // precision change (a bf16-tile variant hit 1e-4-level errors).
+    __shared__ float mt[IT][IT + IPAD];
     const int tx = threadIdx.x & 63;
     const int ty = threadIdx.x >> 6;          // 4 rows per pass
     const bool diag = (I0 == J0);
@@ -69,9 +69,15 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
         for (int r = ty; r < IT; r += 4) {
             ll row = J0 + r, col = I0 + tx;
             mt[r][tx] = (row < V && col < V)
-                        ? (bf16_t)(float)Mb[row * V + col]
-                        : (bf16_t)0.0f;
+                        ? (float)Mb[row * V + col] : 0.0f;
         }
+        __syncthreads();
+        // pre-read this thread's transposed mirror values so the
+        // tile can be reused for z below
+        float mirr[IT / 4];
+        #pragma unroll
+        for (int r = ty, q = 0; r < IT; r += 4, ++q)
+            mirr[q] = mt[tx][r];
         __syncthreads();
         #pragma unroll
         for (int r = ty, q = 0; r < IT; r += 4, ++q) {
@@ -79,19 +85,19 @@ __global__ __launch_bounds__(256) void k_isfc_accum(
             float z = 0.0f;
             if (row < V && col < V) {
                 float a = (float)Mb[row * V + col];
-                float sym = 0.5f * (a + (float)mt[tx][r]);
+                float sym = 0.5f * (a + mirr[q]);
                 z = atanh_clamped(sym);
                 zsum[q] += z;
             }
-            zt[r][tx] = (bf16_t)z;
+            mt[r][tx] = z;                    // second life: z tile
         }
         if (!diag) {
             __syncthreads();
             #pragma unroll
             for (int r = ty, q = 0; r < IT; r += 4, ++q)
-                zsum_m[q] += (float)zt[tx][r];
+                zsum_m[q] += mt[tx][r];
         }
-        __syncthreads();   // tiles reused next bm
+        __syncthreads();   // tile reused next bm
     }
 
     #pragma unroll
