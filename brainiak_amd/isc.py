@@ -28,6 +28,7 @@ Citations as in the reference: [Hasson2004], [Simony2016], [Chen2016],
 
 import logging
 import math
+import os
 from itertools import combinations, permutations
 
 import numpy as np
@@ -284,7 +285,21 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
         from . import ops as _ops
         use_hip = dev.type == 'cuda' and _ops.require_hip()
         acc = torch.zeros((V, V), dtype=torch.float32, device=dev)
-        if use_hip:
+        if use_hip and precision == 'bf16' and local \
+                and not os.environ.get("BRAINIAK_NO_ISFC_FUSED"):
+            # fully fused path: the correlation GEMM tile pairs,
+            # symmetrize, atanh and accumulation all happen in ONE
+            # kernel — the per-subject [V, V] matrix (10 GB at 50k
+            # voxels) never touches HBM.  bf16 mode only (the MFMA
+            # operands are bf16; fp32 keeps the exact GEMM pipeline).
+            Zs = torch.stack([nd.T.contiguous().to(torch.bfloat16)
+                              for nd in normed_local])
+            Zm = torch.stack([
+                _norm((total - d) / (n_total - 1)).T.contiguous()
+                .to(torch.bfloat16) for d in local])
+            _ops.isfc_fused_(acc, Zs, Zm)
+            del Zs, Zm
+        elif use_hip:
             # subjects stream through the fused sym+atanh+accumulate
             # kernel in stacks of up to 4: the [V, V] accumulator's
             # read-modify-write happens once per stack instead of once

@@ -156,6 +156,14 @@ def fcma_fused_gram(data: torch.Tensor, data2: torch.Tensor, start: int,
 # SRM Procrustes
 # ---------------------------------------------------------------------------
 
+def isfc_fused_(acc: torch.Tensor, Zs: torch.Tensor,
+                Zm: torch.Tensor) -> torch.Tensor:
+    """acc += per-subject atanh(sym(Zs_b @ Zm_b^T)) with the [V, V]
+    correlation matrices never materialized (bf16 MFMA tile pairs,
+    fp32 accumulation).  Zs/Zm: bf16 [B, V, T] unit rows."""
+    return _ext().isfc_fused_(acc, Zs, Zm)
+
+
 def svm_cv(kernels: torch.Tensor, y: torch.Tensor, train_idx, test_idx,
            n_train, n_test, C: float = 1.0, tol: float = 1e-3,
            max_iter: int = 10000, max_n: int = -1) -> torch.Tensor:

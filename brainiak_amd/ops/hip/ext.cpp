@@ -45,6 +45,8 @@ ll fcma_supported_L(ll);
 int fcma_corr_norm_smem(ll, int);
 int fcma_fused_gram_supported(ll, int, ll);
 void launch_isfc_accum(float*, const void*, ll, ll, int, void*);
+void launch_isfc_fused(float*, const void*, const void*, ll, ll, ll,
+                       void*);
 void launch_stencil3d(const float*, const float*, float*, ll, int, int,
                       int, int, void*);
 void launch_fcma_fused_corr_gram(const void*, const void*, float*, ll,
@@ -524,6 +526,25 @@ torch::Tensor stencil3d(torch::Tensor x, torch::Tensor w) {
     return out;
 }
 
+torch::Tensor isfc_fused_(torch::Tensor acc, torch::Tensor Zs,
+                          torch::Tensor Zm) {
+    TORCH_CHECK(acc.is_cuda() && acc.is_contiguous() && acc.dim() == 2
+                && acc.scalar_type() == torch::kFloat32
+                && acc.size(0) == acc.size(1), "acc must be fp32 [V,V]");
+    for (auto* Z : {&Zs, &Zm}) {
+        TORCH_CHECK(Z->is_cuda() && Z->is_contiguous()
+                    && Z->dim() == 3
+                    && Z->scalar_type() == torch::kBFloat16,
+                    "Z stacks must be bf16 [B,V,T]");
+    }
+    TORCH_CHECK(Zs.sizes() == Zm.sizes(), "Zs/Zm shape mismatch");
+    TORCH_CHECK(Zs.size(1) == acc.size(0), "V mismatch");
+    launch_isfc_fused(acc.data_ptr<float>(), Zs.data_ptr(),
+                      Zm.data_ptr(), Zs.size(1), Zs.size(2),
+                      Zs.size(0), cur_stream());
+    return acc;
+}
+
 torch::Tensor isfc_accum_(torch::Tensor acc, torch::Tensor M) {
     // M: [V, V] or a [B, V, V] subject stack (the batch amortizes the
     // acc read-modify-write across B matrices in one pass)
@@ -548,6 +569,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "direct [K,K,K] valid conv over [B,X,Y,Z] (searchlight ball)");
     m.def("isfc_accum_", &isfc_accum_,
           "acc += atanh(clamp((M + M^T)/2)) fused in one pass");
+    m.def("isfc_fused_", &isfc_fused_,
+          "acc += atanh(sym(Zs_b Zm_b^T)) per subject, M never "
+          "materialized (bf16 MFMA tile pairs)");
     m.def("svm_cv", &svm_cv,
           "batched precomputed-kernel SVC cross-validation");
     m.def("fcma_normalize_", &fcma_normalize_,

@@ -572,3 +572,53 @@ def test_duo_gsum_fuzz_shapes(ops):
         assert g_duo.shape == g_str.shape == (V, E, E), (E, L, V, P)
         assert torch.allclose(g_duo.float(), g_str.float(),
                               atol=2e-3, rtol=1e-3), (E, L, V, P)
+
+
+def test_isfc_fused_matches_gemm_accum(ops):
+    """Fused tile-pair ISFC kernel == bf16 GEMM + k_isfc_accum on the
+    same inputs (both bf16 operands, fp32 accumulation)."""
+    import numpy as np
+
+    from brainiak_amd import ops as _ops
+    torch.manual_seed(0)
+    dev = "cuda"
+    B, V, T = 3, 333, 50          # odd V exercises the tail guards
+    Zs = torch.randn(B, V, T, device=dev)
+    Zs = Zs / Zs.norm(dim=2, keepdim=True)
+    Zm = torch.randn(B, V, T, device=dev)
+    Zm = Zm / Zm.norm(dim=2, keepdim=True)
+    Zs16, Zm16 = Zs.to(torch.bfloat16), Zm.to(torch.bfloat16)
+
+    acc_f = torch.zeros(V, V, device=dev)
+    _ops.isfc_fused_(acc_f, Zs16.contiguous(), Zm16.contiguous())
+
+    acc_r = torch.zeros(V, V, device=dev)
+    for b in range(B):
+        m = (Zs16[b] @ Zm16[b].T).float()
+        m = (m + m.T) / 2
+        acc_r += torch.atanh(m.clamp(-1 + 1e-7, 1 - 1e-7))
+    assert torch.allclose(acc_f, acc_r, atol=5e-3, rtol=1e-3), \
+        (acc_f - acc_r).abs().max()
+
+
+def test_isfc_distributed_fused_matches_gemm_path(ops):
+    """isfc_distributed bf16: fused kernel vs the GEMM+accum pipeline."""
+    import os
+
+    import numpy as np
+
+    from brainiak_amd.isc import isfc_distributed
+    from brainiak_amd.parallel import DistContext
+    rng = np.random.RandomState(3)
+    data = [rng.randn(40, 300).astype(np.float32) for _ in range(5)]
+    ctx = DistContext(device="cuda")
+    fused = isfc_distributed(data, ctx, summary_statistic='mean',
+                             precision='bf16', return_tensor=True)
+    os.environ["BRAINIAK_NO_ISFC_FUSED"] = "1"
+    try:
+        ref = isfc_distributed(data, ctx, summary_statistic='mean',
+                               precision='bf16', return_tensor=True)
+    finally:
+        del os.environ["BRAINIAK_NO_ISFC_FUSED"]
+    assert torch.allclose(fused, ref, atol=2e-3), \
+        (fused - ref).abs().max()
