@@ -286,12 +286,16 @@ def isfc_distributed(local_data, comm, summary_statistic=None,
         use_hip = dev.type == 'cuda' and _ops.require_hip()
         acc = torch.zeros((V, V), dtype=torch.float32, device=dev)
         if use_hip and precision == 'bf16' and local \
-                and not os.environ.get("BRAINIAK_NO_ISFC_FUSED"):
-            # fully fused path: the correlation GEMM tile pairs,
-            # symmetrize, atanh and accumulation all happen in ONE
-            # kernel — the per-subject [V, V] matrix (10 GB at 50k
-            # voxels) never touches HBM.  bf16 mode only (the MFMA
-            # operands are bf16; fp32 keeps the exact GEMM pipeline).
+                and os.environ.get("BRAINIAK_ISFC_FUSED"):
+            # fully fused path (OPT-IN): correlation tile pairs,
+            # symmetrize, atanh and accumulation in ONE kernel — the
+            # per-subject [V, V] matrix never touches HBM.  Measured
+            # 270 vs 259 ms against the GEMM+accum pipeline at 50k x
+            # 32: the hand MFMA schedule gives back more than the M
+            # round trip saves (hipBLASLt's GEMM is ~3x our MFMA
+            # utilization), so the GEMM path stays the default —
+            # profiles/NEXT.md lists the tiling work that would flip
+            # it.
             Zs = torch.stack([nd.T.contiguous().to(torch.bfloat16)
                               for nd in normed_local])
             Zm = torch.stack([

@@ -612,13 +612,13 @@ def test_isfc_distributed_fused_matches_gemm_path(ops):
     rng = np.random.RandomState(3)
     data = [rng.randn(40, 300).astype(np.float32) for _ in range(5)]
     ctx = DistContext(device="cuda")
-    fused = isfc_distributed(data, ctx, summary_statistic='mean',
-                             precision='bf16', return_tensor=True)
-    os.environ["BRAINIAK_NO_ISFC_FUSED"] = "1"
+    os.environ["BRAINIAK_ISFC_FUSED"] = "1"
     try:
-        ref = isfc_distributed(data, ctx, summary_statistic='mean',
-                               precision='bf16', return_tensor=True)
+        fused = isfc_distributed(data, ctx, summary_statistic='mean',
+                                 precision='bf16', return_tensor=True)
     finally:
-        del os.environ["BRAINIAK_NO_ISFC_FUSED"]
+        del os.environ["BRAINIAK_ISFC_FUSED"]
+    ref = isfc_distributed(data, ctx, summary_statistic='mean',
+                           precision='bf16', return_tensor=True)
     assert torch.allclose(fused, ref, atol=2e-3), \
         (fused - ref).abs().max()
