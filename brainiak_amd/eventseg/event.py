@@ -441,6 +441,52 @@ class EventSegment:
                 mdl.ll_ = np.asarray(ll_hist[j])[:, None]
         return models
 
+    def find_events_regions(self, models, datasets, var=None):
+        """Batched inference counterpart of ``fit_regions``
+        (MI355X addition): segment region r's dataset with region r's
+        fitted model, all regions riding one [B, T, K] forward-
+        backward per shape group.
+
+        Parameters
+        ----------
+        models : list of fitted EventSegment (e.g. from fit_regions).
+        datasets : list of [T_r, V_r] arrays, same length.
+        var : optional shared variance override.
+
+        Returns
+        -------
+        (segments, lls): list of [T_r, K] soft segmentations and an
+        [n_regions] array of log-likelihoods, in input order.
+        """
+        if len(models) != len(datasets):
+            raise ValueError("models and datasets must pair up")
+        dev = torch.device(self.device)
+        K = self.n_events
+        groups = {}
+        for i, d in enumerate(datasets):
+            d2 = _as_valid_2d(d)
+            groups.setdefault(d2.shape, []).append(i)
+        segments = [None] * len(datasets)
+        lls = np.empty(len(datasets))
+        for (T, V), idxs in groups.items():
+            X = torch.stack([
+                torch.as_tensor(np.asarray(datasets[i],
+                                           dtype=np.float64).T,
+                                device=dev) for i in idxs])
+            pats = torch.stack([
+                torch.as_tensor(np.asarray(models[i].event_pat_,
+                                           dtype=np.float64),
+                                device=dev) for i in idxs])
+            vs = var if var is not None else models[idxs[0]].event_var_
+            lp = self._logprob_obs_batch_perregion(X, pats, vs)
+            lg, ll = self._forward_backward_batch(lp)
+            seg = torch.exp(lg).cpu().numpy()
+            ll_h = ll.cpu().numpy()
+            for j, i in enumerate(idxs):
+                segments[i] = seg[j]
+                lls[i] = ll_h[j]
+        return segments, lls
+
     # -- inference ---------------------------------------------------------
 
     def set_event_patterns(self, event_pat):

@@ -242,3 +242,20 @@ def test_fit_regions_ragged_shapes(seeded_rng):
 
     with pytest.raises(ValueError):
         EventSegment(3, split_merge=True).fit_regions(regions)
+
+
+def test_find_events_regions_matches_individual(seeded_rng):
+    """Batched inference == per-region find_events."""
+    rng = seeded_rng
+    K, T, V = 3, 50, 20
+    regions = [rng.randn(T, V) for _ in range(4)]
+    models = EventSegment(K, n_iter=10).fit_regions(regions)
+    test_sets = [rng.randn(T, V) for _ in range(4)]
+    es = EventSegment(K, n_iter=10)
+    segs, lls = es.find_events_regions(models, test_sets)
+    for m, d, seg, ll in zip(models, test_sets, segs, lls):
+        ref_seg, ref_ll = m.find_events(d)
+        assert np.allclose(seg, ref_seg, atol=1e-9)
+        assert np.isclose(ll, ref_ll, atol=1e-7)
+    with pytest.raises(ValueError):
+        es.find_events_regions(models, test_sets[:2])
