@@ -138,6 +138,26 @@ class _BatchedLM:
             J = torch.cat([J, Jp], dim=1)
         return r, J
 
+    def eval_costs(self, thetas):
+        """soft_l1 costs for a [N, n_par] candidate stack whose rows
+        cycle through the subjects in repeat-interleaved order
+        (subject s occupies rows s*k..s*k+k-1)."""
+        N = thetas.shape[0]
+        k = N // self.S
+        rep = lambda t: torch.repeat_interleave(t, k, dim=0)  # noqa: E731
+        saved = (self.co, self.X, self.W, self.sigma, self.scale,
+                 self.tc, self.tw, self.reci, self.S)
+        try:
+            self.co, self.X, self.W = rep(self.co), rep(self.X),                 rep(self.W)
+            self.sigma = torch.repeat_interleave(self.sigma, k)
+            self.scale = torch.repeat_interleave(self.scale, k)
+            self.S = N
+            r, _ = self.residual_jac(thetas, want_jac=False)
+            return _soft_l1_cost_batch(r).double().cpu().numpy()
+        finally:
+            (self.co, self.X, self.W, self.sigma, self.scale,
+             self.tc, self.tw, self.reci, self.S) = saved
+
     def solve(self):
         """Run the batched LM; returns theta [S, n_par] fp64 numpy."""
         S = self.S
@@ -156,34 +176,52 @@ class _BatchedLM:
             A_h = A.double().cpu().numpy()
             g_h = g.double().cpu().numpy()
             dA_h = np.clip(np.einsum('sii->si', A_h), 1e-12, None)
+            # evaluate the whole per-subject damping ladder
+            # (lam, 4*lam, ..., 4^5*lam) in ONE stacked host solve and
+            # ONE batched candidate-cost evaluation: the sequential
+            # try loop cost up to 6 stream syncs per LM iteration.
+            NT = 6
+            ladder = lam[:, None] * (4.0 ** np.arange(NT))[None, :]
+            damp = (A_h[:, None]
+                    + ladder[:, :, None, None]
+                    * np.eye(self.n_par)[None, None]
+                    * dA_h[:, None, :, None])
+            try:
+                delta_h = np.linalg.solve(
+                    damp.reshape(S * NT, self.n_par, self.n_par),
+                    -np.repeat(g_h, NT, axis=0))
+            except np.linalg.LinAlgError:
+                delta_h = np.stack([
+                    np.linalg.lstsq(d, -r, rcond=None)[0]
+                    for d, r in zip(
+                        damp.reshape(S * NT, self.n_par, self.n_par),
+                        np.repeat(g_h, NT, axis=0))])
+            delta = torch.as_tensor(delta_h.reshape(S, NT, self.n_par),
+                                    dtype=torch.float32,
+                                    device=self.theta.device)
+            cand = (self.theta[:, None] + delta).clamp(
+                self.lb[:, None], self.ub[:, None])   # [S, NT, n_par]
+            costs = self.eval_costs(cand.reshape(S * NT, self.n_par))                 .reshape(S, NT)
             improved = np.zeros(S, dtype=bool)
-            for _try in range(6):
-                need = alive & ~improved
-                if not need.any():
-                    break
-                delta_h = np.zeros((S, self.n_par))
-                for s in np.nonzero(need)[0]:
-                    damp = A_h[s] + lam[s] * np.diag(dA_h[s])
-                    try:
-                        delta_h[s] = np.linalg.solve(damp, -g_h[s])
-                    except np.linalg.LinAlgError:
-                        lam[s] *= 10
-                        need[s] = False
-                delta = torch.as_tensor(delta_h, dtype=torch.float32,
-                                        device=self.theta.device)
-                cand = (self.theta + delta).clamp(self.lb, self.ub)
-                rc_, _ = self.residual_jac(cand, want_jac=False)
-                c2 = _soft_l1_cost_batch(rc_).double().cpu().numpy()
-                accept = need & (c2 < cost)
-                if accept.any():
-                    m = torch.as_tensor(accept, device=cand.device)
-                    self.theta = torch.where(m[:, None], cand,
-                                             self.theta)
-                    cost[accept] = c2[accept]
-                    lam[accept] = np.maximum(lam[accept] / 3, 1e-8)
-                    improved |= accept
-                reject = need & ~accept
-                lam[reject] *= 4
+            first_ok = np.full(S, -1)
+            for t in range(NT):                       # host-side argfirst
+                hit = alive & (first_ok < 0) & (costs[:, t] < cost)
+                first_ok[hit] = t
+            accept = first_ok >= 0
+            if accept.any():
+                sel = torch.as_tensor(np.maximum(first_ok, 0),
+                                      device=cand.device)
+                chosen = cand[torch.arange(S, device=cand.device), sel]
+                m = torch.as_tensor(accept, device=cand.device)
+                self.theta = torch.where(m[:, None], chosen,
+                                         self.theta)
+                t_acc = first_ok[accept]
+                cost[accept] = costs[accept, t_acc]
+                lam[accept] = np.maximum(
+                    ladder[accept, t_acc] / 3, 1e-8)
+                improved |= accept
+            reject = alive & ~accept
+            lam[reject] *= 4.0 ** NT
             alive &= improved
             if alive.any():
                 r, J = self.residual_jac(self.theta)
