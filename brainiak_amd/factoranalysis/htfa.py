@@ -189,7 +189,8 @@ class _BatchedLM:
             try:
                 delta_h = np.linalg.solve(
                     damp.reshape(S * NT, self.n_par, self.n_par),
-                    -np.repeat(g_h, NT, axis=0))
+                    -np.repeat(g_h, NT, axis=0)[:, :, None]
+                )[:, :, 0]
             except np.linalg.LinAlgError:
                 delta_h = np.stack([
                     np.linalg.lstsq(d, -r, rcond=None)[0]
