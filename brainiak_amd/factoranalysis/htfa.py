@@ -138,25 +138,27 @@ class _BatchedLM:
             J = torch.cat([J, Jp], dim=1)
         return r, J
 
-    def eval_costs(self, thetas):
-        """soft_l1 costs for a [N, n_par] candidate stack whose rows
-        cycle through the subjects in repeat-interleaved order
-        (subject s occupies rows s*k..s*k+k-1)."""
+    def eval_costs_subset(self, subj_idx, thetas):
+        """soft_l1 costs for a [R*k, n_par] candidate stack covering
+        subjects ``subj_idx`` (k candidates each, repeat-interleaved
+        order) — one residual batch, one host sync."""
         N = thetas.shape[0]
-        k = N // self.S
-        rep = lambda t: torch.repeat_interleave(t, k, dim=0)  # noqa: E731
+        k = N // len(subj_idx)
+        idx = torch.as_tensor(np.repeat(subj_idx, k),
+                              device=self.theta.device)
         saved = (self.co, self.X, self.W, self.sigma, self.scale,
-                 self.tc, self.tw, self.reci, self.S)
+                 self.S)
         try:
-            self.co, self.X, self.W = rep(self.co), rep(self.X),                 rep(self.W)
-            self.sigma = torch.repeat_interleave(self.sigma, k)
-            self.scale = torch.repeat_interleave(self.scale, k)
+            self.co, self.X, self.W = (self.co[idx], self.X[idx],
+                                       self.W[idx])
+            self.sigma = self.sigma[idx]
+            self.scale = self.scale[idx]
             self.S = N
             r, _ = self.residual_jac(thetas, want_jac=False)
             return _soft_l1_cost_batch(r).double().cpu().numpy()
         finally:
             (self.co, self.X, self.W, self.sigma, self.scale,
-             self.tc, self.tw, self.reci, self.S) = saved
+             self.S) = saved
 
     def solve(self):
         """Run the batched LM; returns theta [S, n_par] fp64 numpy."""
@@ -176,53 +178,84 @@ class _BatchedLM:
             A_h = A.double().cpu().numpy()
             g_h = g.double().cpu().numpy()
             dA_h = np.clip(np.einsum('sii->si', A_h), 1e-12, None)
-            # evaluate the whole per-subject damping ladder
-            # (lam, 4*lam, ..., 4^5*lam) in ONE stacked host solve and
-            # ONE batched candidate-cost evaluation: the sequential
-            # try loop cost up to 6 stream syncs per LM iteration.
+            # hybrid damping: try the current lambda for every
+            # subject first (the typical accept path — one solve, one
+            # cost sync, exactly like the sequential loop's best
+            # case); ONLY subjects that reject get the remaining
+            # ladder (4*lam .. 4^5*lam) as one stacked solve + one
+            # batched cost evaluation, instead of up to 5 more
+            # sync rounds.
             NT = 6
-            ladder = lam[:, None] * (4.0 ** np.arange(NT))[None, :]
-            damp = (A_h[:, None]
-                    + ladder[:, :, None, None]
-                    * np.eye(self.n_par)[None, None]
-                    * dA_h[:, None, :, None])
-            try:
-                delta_h = np.linalg.solve(
-                    damp.reshape(S * NT, self.n_par, self.n_par),
-                    -np.repeat(g_h, NT, axis=0)[:, :, None]
-                )[:, :, 0]
-            except np.linalg.LinAlgError:
-                delta_h = np.stack([
-                    np.linalg.lstsq(d, -r, rcond=None)[0]
-                    for d, r in zip(
-                        damp.reshape(S * NT, self.n_par, self.n_par),
-                        np.repeat(g_h, NT, axis=0))])
-            delta = torch.as_tensor(delta_h.reshape(S, NT, self.n_par),
-                                    dtype=torch.float32,
-                                    device=self.theta.device)
-            cand = (self.theta[:, None] + delta).clamp(
-                self.lb[:, None], self.ub[:, None])   # [S, NT, n_par]
-            costs = self.eval_costs(cand.reshape(S * NT, self.n_par))                 .reshape(S, NT)
-            improved = np.zeros(S, dtype=bool)
-            first_ok = np.full(S, -1)
-            for t in range(NT):                       # host-side argfirst
-                hit = alive & (first_ok < 0) & (costs[:, t] < cost)
-                first_ok[hit] = t
-            accept = first_ok >= 0
-            if accept.any():
-                sel = torch.as_tensor(np.maximum(first_ok, 0),
-                                      device=cand.device)
-                chosen = cand[torch.arange(S, device=cand.device), sel]
-                m = torch.as_tensor(accept, device=cand.device)
-                self.theta = torch.where(m[:, None], chosen,
+
+            def ladder_deltas(As, gs, lams):
+                n = self.n_par
+                dAs = np.einsum('sii->si', As).clip(1e-12, None)
+                damp = As + lams[:, None, None] \
+                    * (np.eye(n)[None] * dAs[:, :, None])
+                try:
+                    return np.linalg.solve(
+                        damp, -gs[:, :, None])[:, :, 0]
+                except np.linalg.LinAlgError:
+                    return np.stack([
+                        np.linalg.lstsq(d, -r, rcond=None)[0]
+                        for d, r in zip(damp, gs)])
+
+            d0 = ladder_deltas(A_h, g_h, lam)
+            cand0 = (self.theta + torch.as_tensor(
+                d0, dtype=torch.float32,
+                device=self.theta.device)).clamp(self.lb, self.ub)
+            rc0, _ = self.residual_jac(cand0, want_jac=False)
+            c0 = _soft_l1_cost_batch(rc0).double().cpu().numpy()
+            improved = alive & (c0 < cost)
+            if improved.any():
+                m = torch.as_tensor(improved, device=cand0.device)
+                self.theta = torch.where(m[:, None], cand0,
                                          self.theta)
-                t_acc = first_ok[accept]
-                cost[accept] = costs[accept, t_acc]
-                lam[accept] = np.maximum(
-                    ladder[accept, t_acc] / 3, 1e-8)
-                improved |= accept
-            reject = alive & ~accept
-            lam[reject] *= 4.0 ** NT
+                cost[improved] = c0[improved]
+                lam[improved] = np.maximum(lam[improved] / 3, 1e-8)
+            rej = np.nonzero(alive & ~improved)[0]
+            if len(rej):
+                R = len(rej)
+                lads = lam[rej, None] * (4.0 ** np.arange(1, NT))
+                dR = ladder_deltas(
+                    np.repeat(A_h[rej], NT - 1, axis=0),
+                    np.repeat(g_h[rej], NT - 1, axis=0),
+                    lads.reshape(-1))
+                dR = torch.as_tensor(
+                    dR.reshape(R, NT - 1, self.n_par),
+                    dtype=torch.float32, device=self.theta.device)
+                rej_t = torch.as_tensor(rej,
+                                        device=self.theta.device)
+                candR = (self.theta[rej_t][:, None] + dR).clamp(
+                    self.lb[rej_t][:, None], self.ub[rej_t][:, None])
+                costsR = self.eval_costs_subset(
+                    rej,
+                    candR.reshape(R * (NT - 1), self.n_par)
+                ).reshape(R, NT - 1)
+                first_ok = np.full(R, -1)
+                for t in range(NT - 1):
+                    hit = (first_ok < 0) & (costsR[:, t] < cost[rej])
+                    first_ok[hit] = t
+                acc = first_ok >= 0
+                if acc.any():
+                    sel = torch.as_tensor(np.maximum(first_ok, 0),
+                                          device=candR.device)
+                    chosen = candR[
+                        torch.arange(R, device=candR.device), sel]
+                    mfull = np.zeros(S, dtype=bool)
+                    mfull[rej[acc]] = True
+                    mt = torch.as_tensor(mfull, device=candR.device)
+                    full = torch.zeros_like(self.theta)
+                    full[rej_t] = chosen
+                    self.theta = torch.where(mt[:, None], full,
+                                             self.theta)
+                    t_acc = first_ok[acc]
+                    cost[rej[acc]] = costsR[acc, t_acc]
+                    lam[rej[acc]] = np.maximum(
+                        lads[acc, t_acc] / 3, 1e-8)
+                    improved[rej[acc]] = True
+                still = rej[~acc]
+                lam[still] *= 4.0 ** NT
             alive &= improved
             if alive.any():
                 r, J = self.residual_jac(self.theta)
