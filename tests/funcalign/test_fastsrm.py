@@ -258,3 +258,17 @@ def test_fastsrm_validation_errors(seeded_rng):
     bad = [X[0], X[1][:10]]
     with pytest.raises(ValueError):
         FastSRM(n_components=3).fit(bad)
+
+
+def test_fastsrm_low_ram_matches_in_memory(tmp_path, seeded_rng):
+    """low_ram=True (disk-backed reduced data) produces the same fit
+    as the in-memory path."""
+    data, bases, S = _lownoise_data(seeded_rng, voxels=30, trs=(20, 20),
+                                    k=3)
+    mem = FastSRM(n_components=3, n_iter=8, seed=0,
+                  aggregate="mean").fit(data)
+    disk = FastSRM(n_components=3, n_iter=8, seed=0, low_ram=True,
+                   temp_dir=str(tmp_path), aggregate="mean").fit(data)
+    sm = np.asarray(mem.transform(data))
+    sd = np.asarray(disk.transform(data))
+    assert np.allclose(sm, sd, atol=1e-8)
