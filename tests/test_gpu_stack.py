@@ -456,3 +456,32 @@ def test_eventseg_fit_regions_gpu_parity(cuda, seeded_rng):
     for d, mg in zip(regions, gpu_models):
         mc = EventSegment(K, n_iter=15, device="cpu").fit(d.copy())
         assert np.allclose(mg.segments_[0], mc.segments_[0], atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_stage_timer_gpu_sync(cuda):
+    """stage_timer with sync_device brackets the region with device
+    synchronization: the logged time covers queued GPU work."""
+    import logging
+    import re
+    import time
+
+    from brainiak_amd.utils.timing import stage_timer
+    log = logging.getLogger("timing_gpu_test")
+    records = []
+    handler = logging.Handler()
+    handler.emit = lambda r: records.append(r.getMessage())
+    log.addHandler(handler)
+    log.setLevel(logging.INFO)
+    try:
+        a = torch.randn(4096, 4096, device="cuda")
+        with stage_timer("gpu stage", log, sync_device="cuda"):
+            for _ in range(30):
+                a = a @ a * 1e-3
+        msg = [m for m in records if "gpu stage took" in m][0]
+        logged = float(re.search(r"took ([0-9.]+) s", msg).group(1))
+        # a no-sync timer would read near zero; synced must cover the
+        # 30 chained 4096^3 GEMMs (>= a few ms)
+        assert logged > 1e-3
+    finally:
+        log.removeHandler(handler)
