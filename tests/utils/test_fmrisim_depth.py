@@ -250,3 +250,36 @@ def test_1d_rf_tuning_curves():
     for t, feat in enumerate(trial_list):
         best_vox = np.argmax(resp[:, t])
         assert abs(tuning[best_vox] - feat) <= 6
+
+
+def test_compute_signal_change_all_methods():
+    """Every method's scaling rule against its closed form
+    (ref fmrisim.py:3072 semantics)."""
+    rng = np.random.RandomState(0)
+    T, V = 60, 2
+    sf = np.abs(rng.randn(T, V)) + 0.1
+    noise = rng.randn(T, V) * 3 + 100.0
+    nd = {'sfnr': 50.0}
+    sfn = sf / np.max(np.abs(sf))
+    for method, expect in [
+        ('SFNR', lambda s, n, m: s * (n.mean() / nd['sfnr'] * m)),
+        ('CNR_Amp/Noise-SD', lambda s, n, m: s * (m * n.std())),
+        ('PSC', lambda s, n, m: s * (n.mean() / 100 * m)),
+        ('CNR_Signal-SD/Noise-SD',
+         lambda s, n, m: s * ((m / np.max(np.abs(s))) * n.std()
+                              / s.std())),
+    ]:
+        out = sim.compute_signal_change(sf.copy(), noise, nd,
+                                        [1.5, 1.5], method=method)
+        for v in range(V):
+            want = expect(sfn[:, v], noise[:, v], 1.5)
+            assert np.allclose(out[:, v], want), method
+    # dB variants scale with 10^(mag/20)
+    out_db = sim.compute_signal_change(sf.copy(), noise, nd, [20.0],
+                                       method='CNR_Amp2/Noise-Var_dB')
+    out_db0 = sim.compute_signal_change(sf.copy(), noise, nd, [0.0],
+                                        method='CNR_Amp2/Noise-Var_dB')
+    assert np.allclose(out_db, out_db0 * 10.0)
+    with pytest.raises(ValueError):
+        sim.compute_signal_change(sf, noise, nd, [1.0],
+                                  method='nonsense')
