@@ -345,3 +345,32 @@ def test_device_batched_matches_host_batched(seeded_rng):
                 if not np.isclose(float(a[i, j, k] or 0),
                                   float(b[i, j, k] or 0))]
         assert not mism, mism[:5]
+
+
+def _count_mask_fn(d, m, r, b):
+    return int(np.count_nonzero(m))
+
+
+def test_min_active_voxels_proportion_gates_centers(seeded_rng):
+    """Centers whose searchlight has too few in-mask voxels are left
+    None when min_active_voxels_proportion is set."""
+    dims = (7, 7, 7)
+    data = [seeded_rng.rand(*dims, 4).astype(np.float32)]
+    mask = np.zeros(dims, dtype=bool)
+    mask[3, 3, 3] = True          # isolated center: sparse light
+    mask[1, 1:6, 1:6] = True      # dense plane of centers
+    sl_all = Searchlight(sl_rad=1, max_blk_edge=5)
+    sl_all.distribute(data, mask)
+    sl_all.broadcast(None)
+    out_all = sl_all.run_searchlight(_count_mask_fn)
+    assert out_all[3, 3, 3] == 1          # only itself in the light
+
+    sl_gated = Searchlight(sl_rad=1, max_blk_edge=5,
+                           min_active_voxels_proportion=0.2)
+    sl_gated.distribute(data, mask)
+    sl_gated.broadcast(None)
+    out = sl_gated.run_searchlight(_count_mask_fn)
+    # the isolated center falls below 20 % active and is skipped
+    assert out[3, 3, 3] is None
+    # dense-plane centers stay
+    assert out[1, 3, 3] is not None
