@@ -251,3 +251,21 @@ def test_array_correlation_matches_corrcoef(seeded_rng):
     assert np.allclose(r1, r)
     with pytest.raises(ValueError):
         array_correlation(x, y[:10])
+
+
+def test_p_from_null_sides_and_validation(seeded_rng):
+    """left/right/two-sided relationships and the invalid-side error
+    (ref tests/utils/test_utils.py:184-197)."""
+    null = np.arange(100, dtype=float)
+    p_r = utils.p_from_null(80.0, null, side='right')
+    p_l = utils.p_from_null(80.0, null, side='left')
+    p_t = utils.p_from_null(80.0, null, side='two-sided')
+    # Phipson-Smyth smoothing: (count + 1) / (n + 1)
+    assert np.isclose(p_r, (19 + 1 + 1) / 101)   # >= 80: 80..99 + obs
+    assert p_l > p_r
+    assert p_t <= 2 * min(p_l, p_r) + 1e-12
+    with pytest.raises(ValueError):
+        utils.p_from_null(1.0, null, side='wrong')
+    # exact mode skips the +1 smoothing
+    p_exact = utils.p_from_null(80.0, null, side='right', exact=True)
+    assert p_exact < p_r
