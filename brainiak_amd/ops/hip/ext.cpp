@@ -157,8 +157,8 @@ torch::Tensor fcma_corr_norm_z(torch::Tensor A, torch::Tensor B,
     }
     if (fp8) {
         TORCH_CHECK(At_ptr != nullptr,
-                    "fp8 Z output needs the dot3s variant "
-                    "(BRAINIAK_CORR_VARIANT=3, the default)");
+                    "fp8 Z output needs the dot3s corr kernel "
+                    "(BRAINIAK_CORR_KERNEL=dot3s, the default)");
         TORCH_CHECK(fcma_corr_norm_z8_supported(E, (int)P, L),
                     "fp8 Z output needs P in {2,4} and even L");
         launch_fcma_corr_norm_z8(At_ptr, B.data_ptr(), Z.data_ptr(),
